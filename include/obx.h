@@ -268,10 +268,13 @@ int obx_gpu_free_blocks(obx_gpu_ctx *ctx, int handle);
  * compacted row ids (selection vector of ObVectorStore::fill_output_rows /
  * get_row_ids, ob_block_batched_row_store.cpp:107). Outputs stay on device;
  * obx_gpu_fetch_* copies them back for parity tests. */
-int obx_gpu_filter(obx_gpu_ctx *ctx, int handle, const obx_filter_desc *filter);
+int obx_gpu_filter(obx_gpu_ctx *ctx, int handle, const obx_filter_desc *filter,
+                   int want_row_ids);
 int obx_gpu_fetch_bitmap(obx_gpu_ctx *ctx, int handle, uint8_t *out, int64_t cap);
 int obx_gpu_fetch_row_ids(obx_gpu_ctx *ctx, int handle, int32_t *out,
                           int64_t cap, uint64_t *n_out);
+int obx_gpu_fetch_blk_counts(obx_gpu_ctx *ctx, int handle, uint32_t *out,
+                             int64_t cap);
 
 /* get_rows equivalent: decode projected columns for all rows into device
  * VEC_FIXED arrays; fetch for parity. */
@@ -289,6 +292,9 @@ int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
 /* Timing helpers: milliseconds of device time of the last
  * filter/scan_filter_agg call (HIP events on the context stream). */
 double obx_gpu_last_kernel_ms(obx_gpu_ctx *ctx);
+uint64_t obx_gpu_total_rows(obx_gpu_ctx *ctx, int handle);
+uint64_t obx_gpu_total_bytes(obx_gpu_ctx *ctx, int handle);
+uint64_t obx_gpu_last_survivors(obx_gpu_ctx *ctx, int handle);
 
 #ifdef __cplusplus
 }

@@ -187,25 +187,36 @@ static void fill_cols(int config, uint64_t seed, uint64_t row0, uint32_t n,
   }
 }
 
+typedef struct blk_loc {
+  int64_t size;
+  int32_t worker;
+  int64_t arena_off;
+} blk_loc;
+
 typedef struct gen_job {
   int config;
   uint64_t seed, row_count, row_id_base;
   uint32_t rows_per_block;
   uint64_t n_blocks;
-  int64_t slot_bytes;
-  uint8_t *slab;           /* n_blocks * slot_bytes */
-  int64_t *sizes;          /* per-block encoded size */
+  int64_t slot_bytes;      /* per-block worst-case bound */
+  blk_loc *locs;           /* per-block location in its worker's arena */
+  uint8_t **arenas;        /* per-worker arena (realloc-grown) */
+  int64_t *arena_len;
   const gen_layout *L;
   _Atomic uint64_t *next;
   _Atomic int *err;
+  int worker_id;
 } gen_job;
 
 static void *gen_worker(void *arg) {
   gen_job *J = (gen_job *)arg;
   const gen_layout *L = J->L;
+  const int w = J->worker_id;
   uint8_t *cbuf[8];
   for (int c = 0; c < L->n_cols; c++)
     cbuf[c] = (uint8_t *)malloc((size_t)J->rows_per_block * L->cols[c].len);
+  int64_t cap = 0, len = 0;
+  uint8_t *arena = NULL;
   for (;;) {
     uint64_t b = atomic_fetch_add(J->next, 1);
     if (b >= J->n_blocks) break;
@@ -214,13 +225,22 @@ static void *gen_worker(void *arg) {
                                 ? J->row_count - b * J->rows_per_block
                                 : J->rows_per_block);
     fill_cols(J->config, J->seed, row0, n, cbuf, L);
+    if (len + J->slot_bytes > cap) {
+      cap = cap ? cap * 2 : (J->slot_bytes * 1024);
+      arena = (uint8_t *)realloc(arena, (size_t)cap);
+      if (!arena) { atomic_store(J->err, OBX_INTERNAL_ERROR); break; }
+    }
     int64_t sz = obx_encode_block(L->cols, L->n_cols,
                                   (const uint8_t *const *)cbuf, NULL, n,
-                                  L->enc, J->slab + b * J->slot_bytes,
-                                  J->slot_bytes);
+                                  L->enc, arena + len, J->slot_bytes);
     if (sz < 0) { atomic_store(J->err, (int)sz); break; }
-    J->sizes[b] = sz;
+    J->locs[b].size = sz;
+    J->locs[b].worker = w;
+    J->locs[b].arena_off = len;
+    len += sz;
   }
+  J->arenas[w] = arena;
+  J->arena_len[w] = len;
   for (int c = 0; c < L->n_cols; c++) free(cbuf[c]);
   return NULL;
 }
@@ -238,42 +258,67 @@ int64_t obx_gen_lineitem(int config, uint64_t row_count, uint64_t seed,
   if (rpb < 16) rpb = 16;
   if (rpb > 65535) rpb = 65535;
   uint64_t n_blocks = (row_count + rpb - 1) / rpb;
-  /* generous slot bound: worst case all-raw datums + headers */
-  int64_t slot = 512;
-  for (int c = 0; c < L.n_cols; c++) slot += (int64_t)rpb * (L.cols[c].len + 1);
-  gen_job J;
-  memset(&J, 0, sizeof(J));
-  J.config = config; J.seed = seed; J.row_count = row_count;
-  J.row_id_base = row_id_base;
-  J.rows_per_block = rpb; J.n_blocks = n_blocks; J.slot_bytes = slot;
-  J.slab = (uint8_t *)malloc((size_t)(n_blocks * slot));
-  J.sizes = (int64_t *)calloc(n_blocks, sizeof(int64_t));
-  J.L = &L;
-  _Atomic uint64_t next = 0;
-  _Atomic int err = 0;
-  J.next = &next; J.err = &err;
-  if (!J.slab || !J.sizes) { free(J.slab); free(J.sizes); return OBX_INTERNAL_ERROR; }
+  /* per-block worst-case bound (encoded blocks are ~target size; forced
+     encodings keep this tight) */
+  int64_t slot = 2 * (int64_t)target_block_bytes + 4096;
 
   int nthreads = (int)sysconf(_SC_NPROCESSORS_ONLN);
+  const char *envt = getenv("OBX_GEN_THREADS");
+  if (envt && atoi(envt) > 0) nthreads = atoi(envt);
   if (nthreads < 1) nthreads = 1;
   if ((uint64_t)nthreads > n_blocks) nthreads = (int)n_blocks;
   if (nthreads > 256) nthreads = 256;
+
+  blk_loc *locs = (blk_loc *)calloc(n_blocks, sizeof(blk_loc));
+  uint8_t **arenas = (uint8_t **)calloc((size_t)nthreads, sizeof(uint8_t *));
+  int64_t *arena_len = (int64_t *)calloc((size_t)nthreads, sizeof(int64_t));
+  gen_job *jobs = (gen_job *)calloc((size_t)nthreads, sizeof(gen_job));
+  if (!locs || !arenas || !arena_len || !jobs) {
+    free(locs); free(arenas); free(arena_len); free(jobs);
+    return OBX_INTERNAL_ERROR;
+  }
+  _Atomic uint64_t next = 0;
+  _Atomic int err = 0;
+  for (int t = 0; t < nthreads; t++) {
+    jobs[t].config = config; jobs[t].seed = seed;
+    jobs[t].row_count = row_count; jobs[t].row_id_base = row_id_base;
+    jobs[t].rows_per_block = rpb; jobs[t].n_blocks = n_blocks;
+    jobs[t].slot_bytes = slot; jobs[t].locs = locs;
+    jobs[t].arenas = arenas; jobs[t].arena_len = arena_len;
+    jobs[t].L = &L; jobs[t].next = &next; jobs[t].err = &err;
+    jobs[t].worker_id = t;
+  }
   pthread_t tids[256];
-  for (int t = 1; t < nthreads; t++) pthread_create(&tids[t], NULL, gen_worker, &J);
-  gen_worker(&J);
+  for (int t = 1; t < nthreads; t++)
+    pthread_create(&tids[t], NULL, gen_worker, &jobs[t]);
+  gen_worker(&jobs[0]);
   for (int t = 1; t < nthreads; t++) pthread_join(tids[t], NULL);
-  if (atomic_load(&err)) { free(J.slab); free(J.sizes); return atomic_load(&err); }
+  if (atomic_load(&err)) {
+    for (int t = 0; t < nthreads; t++) free(arenas[t]);
+    free(locs); free(arenas); free(arena_len); free(jobs);
+    return atomic_load(&err);
+  }
 
   uint64_t *offs = (uint64_t *)malloc(sizeof(uint64_t) * (n_blocks + 1));
   int64_t total = 0;
-  for (uint64_t b = 0; b < n_blocks; b++) { offs[b] = (uint64_t)total; total += J.sizes[b]; }
+  for (uint64_t b = 0; b < n_blocks; b++) {
+    offs[b] = (uint64_t)total;
+    total += locs[b].size;
+  }
   offs[n_blocks] = (uint64_t)total;
   uint8_t *data = (uint8_t *)malloc((size_t)total + 16);
-  if (!offs || !data) { free(J.slab); free(J.sizes); free(offs); free(data); return OBX_INTERNAL_ERROR; }
+  if (!offs || !data) {
+    for (int t = 0; t < nthreads; t++) free(arenas[t]);
+    free(locs); free(arenas); free(arena_len); free(jobs);
+    free(offs); free(data);
+    return OBX_INTERNAL_ERROR;
+  }
   for (uint64_t b = 0; b < n_blocks; b++)
-    memcpy(data + offs[b], J.slab + b * slot, (size_t)J.sizes[b]);
+    memcpy(data + offs[b], arenas[locs[b].worker] + locs[b].arena_off,
+           (size_t)locs[b].size);
   memset(data + total, 0, 16); /* slack for 9-byte bitstream reads */
-  free(J.slab); free(J.sizes);
+  for (int t = 0; t < nthreads; t++) free(arenas[t]);
+  free(locs); free(arenas); free(arena_len); free(jobs);
   *out_data = data;
   *out_offsets = offs;
   if (out_cols) memcpy(out_cols, L.cols, sizeof(obx_col_schema) * L.n_cols);
