@@ -72,9 +72,14 @@ typedef struct dev_block {
   uint32_t row_start_lo;   /* global row index of row 0 of this block */
   uint32_t row_start_hi;
   uint32_t row_count;
-  uint32_t pad;
+  uint32_t block_len;      /* block byte length (for LDS staging) */
+  uint64_t block_byte;     /* block start offset in the staged buffer
+                              (16-B aligned when the LDS kernels are used) */
   dev_col cols[OBX_DEV_MAX_COLS];
 } dev_block;
+
+/* max block size eligible for LDS staging (kernel stages the whole block) */
+#define OBX_LDS_STAGE_BYTES (20 * 1024)
 
 #if defined(__HIPCC__) || defined(__HIP_DEVICE_COMPILE__)
 __host__ __device__

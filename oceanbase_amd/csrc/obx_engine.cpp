@@ -27,7 +27,14 @@
 extern "C" __global__ void k_scan_filter_agg(
     const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
     const blk_leaf *, const dev_plan_hdr, gslot *, unsigned long long *);
+extern "C" __global__ void k_scan_filter_agg_lds(
+    const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
+    const blk_leaf *, const dev_plan_hdr, gslot *, unsigned long long *);
 extern "C" __global__ void k_filter(
+    const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
+    const blk_leaf *, const dev_plan_hdr, uint64_t *, int32_t *, uint32_t *,
+    unsigned long long *);
+extern "C" __global__ void k_filter_lds(
     const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
     const blk_leaf *, const dev_plan_hdr, uint64_t *, int32_t *, uint32_t *,
     unsigned long long *);
@@ -69,6 +76,7 @@ struct obx_handle {
   uint32_t *d_blk_counts = nullptr;
   uint8_t *d_decode_out[OBX_DEV_MAX_COLS] = {};
   uint64_t last_survivors = 0;
+  bool lds_ok = false;   /* all blocks 16-B aligned and <= LDS stage size */
   bool in_use = false;
 };
 
@@ -116,7 +124,8 @@ extern "C" int obx_gpu_close(obx_gpu_ctx *ctx) {
  * cited in oracle/obx_codec.c) ------------------------------------------- */
 static int parse_block(const obx_col_schema *cols, uint16_t n_cols,
                        const uint8_t *block, uint64_t dev_off,
-                       uint64_t row_start, dev_block *out) {
+                       uint64_t block_len, uint64_t row_start,
+                       dev_block *out) {
   const obx_micro_header *h = (const obx_micro_header *)block;
   if (h->magic != OBX_MICRO_BLOCK_MAGIC) return OBX_INVALID_ARGUMENT;
   if (h->column_count != n_cols) return OBX_INVALID_ARGUMENT;
@@ -124,6 +133,8 @@ static int parse_block(const obx_col_schema *cols, uint16_t n_cols,
   out->row_start_lo = (uint32_t)row_start;
   out->row_start_hi = (uint32_t)(row_start >> 32);
   out->row_count = h->row_count;
+  out->block_len = (uint32_t)block_len;
+  out->block_byte = dev_off;
   const uint64_t meta_base =
       dev_off + h->header_size + (uint64_t)n_cols * sizeof(obx_col_header);
   const obx_col_header *chp =
@@ -252,13 +263,18 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
 
   std::vector<dev_block> blocks(bs->n_blocks);
   uint64_t row_start = 0;
+  bool lds_ok = true;
   for (uint32_t b = 0; b < bs->n_blocks; b++) {
+    uint64_t blen = bs->block_offsets[b + 1] - bs->block_offsets[b];
     int rc = parse_block(bs->cols, bs->n_cols, bs->data + bs->block_offsets[b],
-                         bs->block_offsets[b], row_start, &blocks[b]);
+                         bs->block_offsets[b], blen, row_start, &blocks[b]);
     if (rc != OBX_SUCCESS) return rc;
+    if ((bs->block_offsets[b] & 15) || blen + 24 > OBX_LDS_STAGE_BYTES)
+      lds_ok = false;
     row_start += blocks[b].row_count;
   }
   h.total_rows = row_start;
+  h.lds_ok = lds_ok;
 
   HIP_TRY(hipMalloc(&h.d_buf, h.total_bytes + 64));
   HIP_TRY(hipMemset(h.d_buf + h.total_bytes, 0, 64));
@@ -393,11 +409,18 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
   }
   HIP_TRY(hipMemsetAsync(h.d_counters, 0, 16 * 8, ctx->stream));
   HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
-  hipLaunchKernelGGL(k_filter, dim3(grid_for(h.n_blocks)), dim3(256), 0,
-                     ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
-                     h.d_pleaves, h.d_bleaves, ph, h.d_bitmap,
-                     want_row_ids ? h.d_row_ids : nullptr,
-                     want_row_ids ? h.d_blk_counts : nullptr, h.d_counters);
+  if (h.lds_ok)
+    hipLaunchKernelGGL(k_filter_lds, dim3(grid_for(h.n_blocks)), dim3(256), 0,
+                       ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
+                       h.d_pleaves, h.d_bleaves, ph, h.d_bitmap,
+                       want_row_ids ? h.d_row_ids : nullptr,
+                       want_row_ids ? h.d_blk_counts : nullptr, h.d_counters);
+  else
+    hipLaunchKernelGGL(k_filter, dim3(grid_for(h.n_blocks)), dim3(256), 0,
+                       ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
+                       h.d_pleaves, h.d_bleaves, ph, h.d_bitmap,
+                       want_row_ids ? h.d_row_ids : nullptr,
+                       want_row_ids ? h.d_blk_counts : nullptr, h.d_counters);
   HIP_TRY(hipEventRecord(ctx->ev_stop, ctx->stream));
   HIP_TRY(hipStreamSynchronize(ctx->stream));
   float ms = 0;
@@ -511,9 +534,16 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
   HIP_TRY(hipMemsetAsync(h.d_counters, 0, 16 * 8, ctx->stream));
 
   HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
-  hipLaunchKernelGGL(k_scan_filter_agg, dim3(grid_for(h.n_blocks)), dim3(256),
-                     0, ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
-                     h.d_pleaves, h.d_bleaves, ph, h.d_gtable, h.d_counters);
+  if (h.lds_ok)
+    hipLaunchKernelGGL(k_scan_filter_agg_lds, dim3(grid_for(h.n_blocks)),
+                       dim3(256), 0, ctx->stream, h.d_buf, h.d_blocks,
+                       h.n_blocks, h.d_pleaves, h.d_bleaves, ph, h.d_gtable,
+                       h.d_counters);
+  else
+    hipLaunchKernelGGL(k_scan_filter_agg, dim3(grid_for(h.n_blocks)),
+                       dim3(256), 0, ctx->stream, h.d_buf, h.d_blocks,
+                       h.n_blocks, h.d_pleaves, h.d_bleaves, ph, h.d_gtable,
+                       h.d_counters);
   HIP_TRY(hipEventRecord(ctx->ev_stop, ctx->stream));
   HIP_TRY(hipStreamSynchronize(ctx->stream));
   float ms = 0;

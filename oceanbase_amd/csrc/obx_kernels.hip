@@ -4,11 +4,11 @@
  *
  * Design (MI355X-first, not a translation of the reference's AVX loops):
  *  - one 256-thread workgroup processes whole microblocks from a grid-stride
- *    loop (the reference batches 256 rows on CPU, ob_parameter_seed.ipp:415;
- *    the GPU engine honors that granule only at the host boundary)
- *  - every bit/byte read is an aligned-u64 + funnel-shift load relative to
- *    the staged buffer base (coalesced across lanes for the row-strided
- *    packed streams; the encoded column pages are the algorithmic bytes)
+ *    loop; each block's encoded bytes are STAGED THROUGH LDS with coalesced
+ *    dwordx4 copies (HBM sees exactly the encoded column pages, read once,
+ *    fully coalesced), then all bit-granular decode reads hit LDS
+ *    (~16 KB block << 160 KB LDS/CU). Blocks larger than
+ *    OBX_LDS_STAGE_BYTES take a generic global-memory path.
  *  - white filters arrive pre-lowered per block (obx_dev.h): dict-domain
  *    64-bit ref masks or packed-domain unsigned ranges, so the inner loop is
  *    unpack + mask/range test + __ballot (cf. the reference's AVX512
@@ -32,16 +32,29 @@
 #define WG 256
 #define WAVES (WG / 64)
 
-/* ---------------- bit reads (buffer base is 256-B aligned) -------------- */
-__device__ __forceinline__ uint64_t bit_read(const uint8_t *__restrict__ buf,
-                                             uint64_t bitpos, uint32_t k) {
-  const uint64_t *w = (const uint64_t *)buf;
+/* view of a block's bytes: either the global buffer (bias 0) or the LDS
+ * staging copy (bias = block_byte * 8). All dev_col offsets are absolute;
+ * reads subtract the bias. */
+struct blk_view {
+  const uint8_t *base;
+  uint64_t bit_bias;
+};
+
+/* ---------------- bit reads (base is >=16-B aligned) -------------------- */
+__device__ __forceinline__ uint64_t bit_read_at(const uint8_t *__restrict__ b,
+                                                uint64_t bitpos, uint32_t k) {
+  const uint64_t *w = (const uint64_t *)b;
   uint64_t widx = bitpos >> 6;
   uint32_t sh = (uint32_t)(bitpos & 63);
   uint64_t v = w[widx] >> sh;
   if (sh + k > 64) v |= w[widx + 1] << (64 - sh);
   if (k < 64) v &= (((uint64_t)1 << k) - 1);
   return v;
+}
+
+__device__ __forceinline__ uint64_t bit_read(const blk_view &bv,
+                                             uint64_t bitpos, uint32_t k) {
+  return bit_read_at(bv.base, bitpos - bv.bit_bias, k);
 }
 
 __device__ __forceinline__ int64_t sext(uint64_t v, uint32_t bytes) {
@@ -62,66 +75,66 @@ __device__ __forceinline__ uint64_t shflxor64(uint64_t v, int mask) {
 }
 
 /* unpack the column's packed stream entry for row r (refs/diffs/values) */
-__device__ __forceinline__ uint64_t col_packed(const uint8_t *buf,
+__device__ __forceinline__ uint64_t col_packed(const blk_view &bv,
                                                const dev_col &c, uint32_t r) {
   uint32_t W = (c.flags & OBX_DF_BITPACK) ? c.width : (uint32_t)c.width * 8;
-  return bit_read(buf, c.data_bit + (uint64_t)r * W, W);
+  return bit_read(bv, c.data_bit + (uint64_t)r * W, W);
 }
 
-__device__ __forceinline__ bool col_is_null_ext(const uint8_t *buf,
+__device__ __forceinline__ bool col_is_null_ext(const blk_view &bv,
                                                 const dev_col &c, uint32_t r) {
   if (!(c.flags & OBX_DF_HAS_EXT)) return false;
-  return bit_read(buf, c.ext_bit + (uint64_t)r * c.ext_width, c.ext_width) != 0;
+  return bit_read(bv, c.ext_bit + (uint64_t)r * c.ext_width, c.ext_width) != 0;
 }
 
 /* RLE: ref of row r by binary search over run starts
  * (ob_rle_decoder.cpp:18-31) */
-__device__ __forceinline__ uint64_t rle_ref(const uint8_t *buf,
+__device__ __forceinline__ uint64_t rle_ref(const blk_view &bv,
                                             const dev_col &c, uint32_t r) {
   uint64_t lo = 0, hi = c.runs;
   while (lo < hi) {
     uint64_t mid = (lo + hi) >> 1;
-    uint64_t s = bit_read(buf, (c.aux_byte + mid * c.rib) * 8,
+    uint64_t s = bit_read(bv, (c.aux_byte + mid * c.rib) * 8,
                           (uint32_t)c.rib * 8);
     if (s <= r) lo = mid + 1; else hi = mid;
   }
   uint64_t run = lo - 1;
-  return bit_read(buf,
+  return bit_read(bv,
                   (c.aux_byte + (uint64_t)c.runs * c.rib + run * c.rfb) * 8,
                   (uint32_t)c.rfb * 8);
 }
 
-__device__ __forceinline__ int64_t dict_entry(const uint8_t *buf,
+__device__ __forceinline__ int64_t dict_entry(const blk_view &bv,
                                               const dev_col &c, uint64_t ref) {
-  uint64_t v = bit_read(buf, (c.dict_byte + ref * c.entry_len) * 8,
+  uint64_t v = bit_read(bv, (c.dict_byte + ref * c.entry_len) * 8,
                         (uint32_t)c.entry_len * 8);
   if (c.flags & OBX_DF_STRING) return (int64_t)v; /* raw LE bytes */
   return (c.flags & OBX_DF_SIGNED) ? sext(v, c.tss) : sext(v, c.entry_len);
 }
 
 /* full value decode: int64 (sign-extended), is_null out */
-__device__ int64_t col_value(const uint8_t *buf, const dev_col &c, uint32_t r,
-                             bool &null_out) {
+__device__ __forceinline__ int64_t col_value(const blk_view &bv,
+                                             const dev_col &c, uint32_t r,
+                                             bool &null_out) {
   null_out = false;
   switch (c.enc) {
     case OBX_D_RAW: {
-      if (col_is_null_ext(buf, c, r)) { null_out = true; return 0; }
-      uint64_t v = col_packed(buf, c, r);
+      if (col_is_null_ext(bv, c, r)) { null_out = true; return 0; }
+      uint64_t v = col_packed(bv, c, r);
       if (c.flags & OBX_DF_BITPACK) return (int64_t)v; /* zero-extended */
       if (c.flags & OBX_DF_STRING) return (int64_t)v;  /* raw LE bytes */
       return (c.flags & OBX_DF_SIGNED) ? sext(v, c.tss)
-                                       : sext(v, c.width); /* decimal: width
-                                           bytes == datum len */
+                                       : sext(v, c.width);
     }
     case OBX_D_DICT: {
-      uint64_t ref = col_packed(buf, c, r);
+      uint64_t ref = col_packed(bv, c, r);
       if (ref >= c.count) { null_out = true; return 0; }
-      return dict_entry(buf, c, ref);
+      return dict_entry(bv, c, ref);
     }
     case OBX_D_RLE: {
-      uint64_t ref = rle_ref(buf, c, r);
+      uint64_t ref = rle_ref(bv, c, r);
       if (ref >= c.count) { null_out = true; return 0; }
-      return dict_entry(buf, c, ref);
+      return dict_entry(bv, c, ref);
     }
     case OBX_D_CONST: {
       if (c.runs == 0) {
@@ -131,19 +144,19 @@ __device__ int64_t col_value(const uint8_t *buf, const dev_col &c, uint32_t r,
       uint64_t ref = c.rfb; /* const_ref */
       for (uint32_t i = 0; i < c.runs; i++) {
         uint64_t rid = bit_read(
-            buf, (c.aux_byte + c.runs + (uint64_t)i * c.rib) * 8,
+            bv, (c.aux_byte + c.runs + (uint64_t)i * c.rib) * 8,
             (uint32_t)c.rib * 8);
         if (rid == r) {
-          ref = bit_read(buf, (c.aux_byte + i) * 8, 8);
+          ref = bit_read(bv, (c.aux_byte + i) * 8, 8);
           break;
         }
       }
       if (ref >= c.count) { null_out = true; return 0; }
-      return dict_entry(buf, c, ref);
+      return dict_entry(bv, c, ref);
     }
     case OBX_D_INTDIFF: {
-      if (col_is_null_ext(buf, c, r)) { null_out = true; return 0; }
-      uint64_t diff = col_packed(buf, c, r);
+      if (col_is_null_ext(bv, c, r)) { null_out = true; return 0; }
+      uint64_t diff = col_packed(bv, c, r);
       return (int64_t)((uint64_t)c.base + diff);
     }
   }
@@ -173,7 +186,7 @@ __device__ __forceinline__ bool leaf_value_match(const dev_leaf &lf, int64_t v,
   return false;
 }
 
-__device__ __forceinline__ bool leaf_match(const uint8_t *buf,
+__device__ __forceinline__ bool leaf_match(const blk_view &bv,
                                            const dev_block &blk,
                                            const dev_leaf &plf,
                                            const blk_leaf &blf, uint32_t r) {
@@ -182,27 +195,27 @@ __device__ __forceinline__ bool leaf_match(const uint8_t *buf,
     case OBX_LEAF_NONE: return false;
     case OBX_LEAF_ALL:
       if (c.enc == OBX_D_RAW || c.enc == OBX_D_INTDIFF)
-        return !col_is_null_ext(buf, c, r);
+        return !col_is_null_ext(bv, c, r);
       return true;
     case OBX_LEAF_REF_MASK: {
-      uint64_t ref = (c.enc == OBX_D_RLE) ? rle_ref(buf, c, r)
-                                          : col_packed(buf, c, r);
+      uint64_t ref = (c.enc == OBX_D_RLE) ? rle_ref(bv, c, r)
+                                          : col_packed(bv, c, r);
       return (blf.mask >> ref) & 1; /* host guarantees count<=63 here */
     }
     case OBX_LEAF_RANGE: {
-      if (col_is_null_ext(buf, c, r)) return false;
-      uint64_t v = col_packed(buf, c, r);
+      if (col_is_null_ext(bv, c, r)) return false;
+      uint64_t v = col_packed(bv, c, r);
       bool in = (v - blf.lo) <= (blf.hi - blf.lo);
       return in != (bool)blf.invert;
     }
     case OBX_LEAF_NULL: {
       bool isn;
-      (void)col_value(buf, c, r, isn);
+      (void)col_value(bv, c, r, isn);
       return isn != (bool)blf.invert;
     }
     default: {
       bool isn;
-      int64_t v = col_value(buf, c, r, isn);
+      int64_t v = col_value(bv, c, r, isn);
       return leaf_value_match(plf, v, isn);
     }
   }
@@ -270,8 +283,8 @@ __device__ __forceinline__ void lds_acc_i128(unsigned long long *cell,
   if (hi) atomicAdd(&cell[1], hi);
 }
 
-/* CAS-based signed min/max on an LDS u64 holding an int64 */
-__device__ __forceinline__ void lds_minmax(unsigned long long *slot,
+/* CAS-based signed min/max on a u64 holding an int64 (LDS or global) */
+__device__ __forceinline__ void cas_minmax(unsigned long long *slot,
                                            int64_t v, bool is_min) {
   unsigned long long cur = *slot;
   for (;;) {
@@ -299,24 +312,25 @@ __device__ __forceinline__ void g_acc_i128(unsigned long long *limbs,
   }
 }
 
-__device__ __forceinline__ void g_minmax(unsigned long long *slot, int64_t v,
-                                         bool is_min) {
-  unsigned long long cur = *slot;
-  for (;;) {
-    int64_t c = (int64_t)cur;
-    if (is_min ? (v >= c) : (v <= c)) return;
-    unsigned long long prev = atomicCAS(slot, cur, (unsigned long long)v);
-    if (prev == cur) return;
-    cur = prev;
-  }
+/* stage one block's bytes into LDS (coalesced dwordx4; block_byte is
+ * 16-B aligned, container blocks are 16-B aligned with zero padding) */
+__device__ __forceinline__ void stage_block(const uint8_t *__restrict__ buf,
+                                            const dev_block &blk,
+                                            uint8_t *lds_blk) {
+  const uint32_t n16 = (blk.block_len + 8 + 15) >> 4;
+  const uint4 *src = (const uint4 *)(buf + blk.block_byte);
+  uint4 *dst = (uint4 *)lds_blk;
+  for (uint32_t i = threadIdx.x; i < n16; i += WG) dst[i] = src[i];
 }
 
 /* ---------------- fused scan->filter->aggregate kernel ------------------ */
-extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
+template <bool STAGE>
+__device__ void scan_filter_agg_body(
     const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
     uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
-    const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
-    gslot *__restrict__ gtable, unsigned long long *__restrict__ counters) {
+    const blk_leaf *__restrict__ bleaves, const dev_plan_hdr &ph,
+    gslot *__restrict__ gtable, unsigned long long *__restrict__ counters,
+    uint8_t *lds_blk) {
   __shared__ lds_table tab;
   __shared__ dev_block cur;
   __shared__ unsigned long long wg_passed;
@@ -350,6 +364,14 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
       for (uint32_t i = tid; i < words; i += WG) dst[i] = src[i];
     }
     __syncthreads();
+    if (STAGE) {
+      stage_block(buf, cur, lds_blk);
+      __syncthreads();
+    }
+    blk_view bv;
+    bv.base = STAGE ? lds_blk : buf;
+    bv.bit_bias = STAGE ? cur.block_byte * 8 : 0;
+
     const uint32_t rows = cur.row_count;
     const uint32_t iters = (rows + WG - 1) / WG;
     const blk_leaf *bl = bleaves + (uint64_t)b * ph.n_leaves;
@@ -358,7 +380,7 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
       uint32_t r = it * WG + tid;
       bool pass = r < rows;
       for (uint32_t i = 0; i < ph.n_leaves && pass; i++)
-        pass = leaf_match(buf, cur, plan_leaves[i], bl[i], r);
+        pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
 
       uint64_t pass_mask = __ballot(pass);
       if (lane == 0 && pass_mask)
@@ -372,7 +394,7 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
       if (pass) {
         for (uint32_t i = 0; i < ph.n_need; i++) {
           bool isn;
-          vals[i] = col_value(buf, cur.cols[ph.need_cols[i]], r, isn);
+          vals[i] = col_value(bv, cur.cols[ph.need_cols[i]], r, isn);
           nullbits |= (isn ? 1u : 0u) << i;
         }
         uint32_t koff = 0;
@@ -398,18 +420,18 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
           bool nb = ag.ib != 0xFF && ((nullbits >> ag.ib) & 1);
           bool nc = ag.ic != 0xFF && ((nullbits >> ag.ic) & 1);
           switch (ag.kind) {
-            case 0: /* COUNT */
+            case 0:
               pv[a] = i128_from_i64((ag.ia != 0xFF && na) ? 0 : 1);
               break;
-            case 1: /* SUM */
+            case 1:
               pv[a] = i128_from_i64(na ? 0 : vals[ag.ia]);
               break;
-            case 4: /* SUM_PROD2 */
+            case 4:
               pv[a] = (na || nb)
                           ? i128_from_i64(0)
                           : i128_mul_i64(vals[ag.ia], ag.one_b - vals[ag.ib]);
               break;
-            case 5: /* SUM_PROD3 */
+            case 5:
               pv[a] = (na || nb || nc)
                           ? i128_from_i64(0)
                           : i128_mul_pos_i64(
@@ -417,7 +439,7 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
                                              ag.one_b - vals[ag.ib]),
                                 ag.one_c + vals[ag.ic]);
               break;
-            case 6: /* SUM_MUL */
+            case 6:
               pv[a] = (na || nb)
                           ? i128_from_i64(0)
                           : i128_mul_i64(vals[ag.ia], vals[ag.ib]);
@@ -431,7 +453,6 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
       }
 
       if (any_minmax) {
-        /* correct fallback: per-lane LDS updates */
         if (pass) {
           int s = lds_slot(&tab, key);
           if (s < 0) { atomicAdd(&counters[1], 1ull); continue; }
@@ -440,7 +461,7 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
             uint8_t k = ph.aggs[a].kind;
             if (k == 2 || k == 3) {
               if (!pv[a].hi) {
-                lds_minmax(&tab.cell[s][a][0], (int64_t)pv[a].lo, k == 2);
+                cas_minmax(&tab.cell[s][a][0], (int64_t)pv[a].lo, k == 2);
                 tab.cell[s][a][1] = 1;
               }
             } else {
@@ -449,7 +470,6 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
           }
         }
       } else {
-        /* wave-level cell clustering + 64-wide reductions */
         uint64_t active = pass_mask;
         while (active) {
           int leader = __ffsll((unsigned long long)active) - 1;
@@ -494,8 +514,8 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
       uint8_t kind = ph.aggs[a].kind;
       if (kind == 2 || kind == 3) {
         if (tab.cell[s][a][1]) {
-          g_minmax(&gtable[idx].cells[a][0], (int64_t)tab.cell[s][a][0],
-                   kind == 2);
+          cas_minmax(&gtable[idx].cells[a][0], (int64_t)tab.cell[s][a][0],
+                     kind == 2);
           gtable[idx].cells[a][1] = 1;
         }
       } else {
@@ -506,22 +526,45 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
   }
 }
 
-/* ---------------- filter-only kernel (bitmap + selection vectors) ------- */
-extern "C" __global__ __launch_bounds__(WG, 2) void k_filter(
+extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
     const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
     uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
     const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
+    gslot *__restrict__ gtable, unsigned long long *__restrict__ counters) {
+  scan_filter_agg_body<false>(buf, blocks, n_blocks, plan_leaves, bleaves, ph,
+                              gtable, counters, nullptr);
+}
+
+extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg_lds(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
+    const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
+    gslot *__restrict__ gtable, unsigned long long *__restrict__ counters) {
+  __shared__ uint8_t lds_blk[OBX_LDS_STAGE_BYTES + 32];
+  scan_filter_agg_body<true>(buf, blocks, n_blocks, plan_leaves, bleaves, ph,
+                             gtable, counters, lds_blk);
+}
+
+/* ---------------- filter-only kernel (bitmap + selection vectors) ------- */
+template <bool STAGE>
+__device__ void filter_body(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
+    const blk_leaf *__restrict__ bleaves, const dev_plan_hdr &ph,
     uint64_t *__restrict__ bitmap, int32_t *__restrict__ row_ids,
     uint32_t *__restrict__ blk_counts,
-    unsigned long long *__restrict__ counters) {
+    unsigned long long *__restrict__ counters, uint8_t *lds_blk) {
   __shared__ dev_block cur;
   __shared__ uint32_t wv_cnt[WAVES];
   __shared__ uint32_t wv_scan[WAVES];
   __shared__ uint32_t blk_written;
+  __shared__ unsigned long long wg_passed;
 
   const uint32_t tid = threadIdx.x;
   const uint32_t lane = tid & 63;
   const uint32_t wv = tid >> 6;
+  if (tid == 0) wg_passed = 0;
+  __syncthreads();
 
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
     {
@@ -532,6 +575,14 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_filter(
       if (tid == 0) blk_written = 0;
     }
     __syncthreads();
+    if (STAGE) {
+      stage_block(buf, cur, lds_blk);
+      __syncthreads();
+    }
+    blk_view bv;
+    bv.base = STAGE ? lds_blk : buf;
+    bv.bit_bias = STAGE ? cur.block_byte * 8 : 0;
+
     const uint32_t rows = cur.row_count;
     const uint64_t row_start = dev_block_row_start(&cur);
     const uint32_t iters = (rows + WG - 1) / WG;
@@ -541,7 +592,7 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_filter(
       uint32_t r = it * WG + tid;
       bool pass = r < rows;
       for (uint32_t i = 0; i < ph.n_leaves && pass; i++)
-        pass = leaf_match(buf, cur, plan_leaves[i], bl[i], r);
+        pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
       uint64_t m = __ballot(pass);
 
       if (bitmap && lane == 0 && m) {
@@ -574,18 +625,42 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_filter(
         }
         __syncthreads();
       } else if (lane == 0 && m) {
-        atomicAdd(&counters[0], (unsigned long long)__popcll(m));
+        atomicAdd(&wg_passed, (unsigned long long)__popcll(m));
       }
     }
     if (row_ids) {
       __syncthreads();
       if (tid == 0) {
         blk_counts[b] = blk_written;
-        atomicAdd(&counters[0], (unsigned long long)blk_written);
+        atomicAdd(&wg_passed, (unsigned long long)blk_written);
       }
     }
     __syncthreads();
   }
+  if (tid == 0 && wg_passed) atomicAdd(&counters[0], wg_passed);
+}
+
+extern "C" __global__ __launch_bounds__(WG, 2) void k_filter(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
+    const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
+    uint64_t *__restrict__ bitmap, int32_t *__restrict__ row_ids,
+    uint32_t *__restrict__ blk_counts,
+    unsigned long long *__restrict__ counters) {
+  filter_body<false>(buf, blocks, n_blocks, plan_leaves, bleaves, ph, bitmap,
+                     row_ids, blk_counts, counters, nullptr);
+}
+
+extern "C" __global__ __launch_bounds__(WG, 2) void k_filter_lds(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
+    const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
+    uint64_t *__restrict__ bitmap, int32_t *__restrict__ row_ids,
+    uint32_t *__restrict__ blk_counts,
+    unsigned long long *__restrict__ counters) {
+  __shared__ uint8_t lds_blk[OBX_LDS_STAGE_BYTES + 32];
+  filter_body<true>(buf, blocks, n_blocks, plan_leaves, bleaves, ph, bitmap,
+                    row_ids, blk_counts, counters, lds_blk);
 }
 
 /* ---------------- decode kernel (get_rows equivalent, parity) ----------- */
@@ -603,11 +678,12 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_decode(
       for (uint32_t i = tid; i < words; i += WG) dst[i] = src[i];
     }
     __syncthreads();
+    blk_view bv; bv.base = buf; bv.bit_bias = 0;
     const uint32_t rows = cur.row_count;
     const uint64_t row_start = dev_block_row_start(&cur);
     for (uint32_t r = tid; r < rows; r += WG) {
       bool isn;
-      int64_t v = col_value(buf, cur.cols[col], r, isn);
+      int64_t v = col_value(bv, cur.cols[col], r, isn);
       uint64_t uv = isn ? 0 : (uint64_t)v;
       uint8_t *dst = out + (row_start + r) * datum_len;
       for (uint32_t i = 0; i < datum_len; i++)
@@ -626,7 +702,6 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_decode(
 __device__ __forceinline__ void lower_range(int64_t base, uint64_t dmax,
                                             const dev_leaf &lf,
                                             blk_leaf &out) {
-  /* unsigned packed domain: v in [0, dmax]; value = base + v */
   __int128 D = (__int128)lf.vlo - base;
   __int128 D2 = (__int128)lf.vhi - base;
   out.invert = 0;
@@ -684,24 +759,24 @@ extern "C" __global__ void k_lower_leaves(
   uint32_t i = (uint32_t)(idx % n_leaves);
   const dev_leaf &lf = pl[i];
   const dev_col &c = blocks[b].cols[lf.col];
+  blk_view bv; bv.base = buf; bv.bit_bias = 0;
   blk_leaf o;
   o.mask = 0; o.lo = 0; o.hi = 0; o.mode = OBX_LEAF_VALUE; o.invert = 0;
 
   const bool is_dict = (c.enc == OBX_D_DICT || c.enc == OBX_D_RLE ||
                         (c.enc == OBX_D_CONST && c.runs > 0));
   if (is_dict && c.count < 64) {
-    if (lf.op == 8) {            /* NU: only the null ref passes */
+    if (lf.op == 8) {
       o.mask = 1ull << c.count;
-    } else if (lf.op == 9) {     /* NN */
+    } else if (lf.op == 9) {
       o.mask = (c.count == 0) ? 0 : ((1ull << c.count) - 1);
     } else {
       for (uint32_t e = 0; e < c.count; e++) {
-        int64_t v = dict_entry(buf, c, e);
+        int64_t v = dict_entry(bv, c, e);
         if (leaf_value_match(lf, v, false)) o.mask |= 1ull << e;
       }
     }
     o.mode = (o.mask == 0) ? OBX_LEAF_NONE : OBX_LEAF_REF_MASK;
-    /* keep REF_MASK even for mask==0 on CONST (mode NONE is fine) */
   } else if (c.enc == OBX_D_RAW && (c.flags & OBX_DF_BITPACK) && lf.op <= 6) {
     uint64_t dmax = (c.width >= 64) ? ~0ull : ((1ull << c.width) - 1);
     lower_range(0, dmax, lf, o);
@@ -718,12 +793,11 @@ extern "C" __global__ void k_lower_leaves(
       o.mode = (lf.op == 8) ? OBX_LEAF_NONE : OBX_LEAF_ALL;
     }
   } else if (c.enc == OBX_D_CONST && c.runs == 0) {
-    /* whole-block constant: evaluate once */
     bool isn = (c.count == 0);
     bool m = leaf_value_match(lf, c.base, isn);
     o.mode = m ? OBX_LEAF_ALL : OBX_LEAF_NONE;
   } else {
-    o.mode = OBX_LEAF_VALUE; /* generic fallback (large dicts, IN on raw) */
+    o.mode = OBX_LEAF_VALUE;
   }
   out[idx] = o;
 }

@@ -303,7 +303,9 @@ int64_t obx_gen_lineitem(int config, uint64_t row_count, uint64_t seed,
   int64_t total = 0;
   for (uint64_t b = 0; b < n_blocks; b++) {
     offs[b] = (uint64_t)total;
-    total += locs[b].size;
+    /* 16-B align each block in the container (the GPU engine stages blocks
+       to LDS with dwordx4 copies; macro-block writers align similarly) */
+    total = (total + locs[b].size + 15) & ~(int64_t)15;
   }
   offs[n_blocks] = (uint64_t)total;
   uint8_t *data = (uint8_t *)malloc((size_t)total + 16);
@@ -313,9 +315,12 @@ int64_t obx_gen_lineitem(int config, uint64_t row_count, uint64_t seed,
     free(offs); free(data);
     return OBX_INTERNAL_ERROR;
   }
-  for (uint64_t b = 0; b < n_blocks; b++)
+  for (uint64_t b = 0; b < n_blocks; b++) {
     memcpy(data + offs[b], arenas[locs[b].worker] + locs[b].arena_off,
            (size_t)locs[b].size);
+    int64_t pad = (int64_t)offs[b + 1] - (int64_t)offs[b] - locs[b].size;
+    if (pad > 0) memset(data + offs[b] + locs[b].size, 0, (size_t)pad);
+  }
   memset(data + total, 0, 16); /* slack for 9-byte bitstream reads */
   for (int t = 0; t < nthreads; t++) free(arenas[t]);
   free(locs); free(arenas); free(arena_len); free(jobs);
