@@ -51,15 +51,18 @@ def build_descs(workload):
     if workload == "q1":
         filt = abi.make_filter(
             [dict(col=6, op=abi.OP_LE, lo=oracle.date_days(1998, 9, 2))])
-        agg = abi.make_agg([4, 5], [
+        aggs = [
             dict(kind=abi.AGG_COUNT),
             dict(kind=abi.AGG_SUM, col_a=0),
             dict(kind=abi.AGG_SUM, col_a=1),
             dict(kind=abi.AGG_SUM_PROD2, col_a=1, col_b=2),
             dict(kind=abi.AGG_SUM_PROD3, col_a=1, col_b=2, col_c=3),
             dict(kind=abi.AGG_SUM, col_a=2),
-        ])
-        return filt, agg, 6
+        ]
+        n = int(os.environ.get("OBX_Q1_AGGS", "6"))  # diagnostics knob
+        aggs = aggs[:n]
+        agg = abi.make_agg([4, 5], aggs)
+        return filt, agg, len(aggs)
     if workload == "q6":
         d94, d95 = oracle.date_days(1994, 1, 1), oracle.date_days(1995, 1, 1)
         filt = abi.make_filter([

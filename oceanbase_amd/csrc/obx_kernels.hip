@@ -267,9 +267,13 @@ __device__ __forceinline__ uint32_t key_hash(uint64_t k) {
 __device__ __forceinline__ int lds_slot(lds_table *t, uint64_t key) {
   uint32_t idx = key_hash(key);
   for (int probe = 0; probe < OBX_LTABLE_SLOTS; probe++) {
-    unsigned long long cur = atomicCAS(&t->key[idx], OBX_KEY_EMPTY,
-                                       (unsigned long long)key);
-    if (cur == OBX_KEY_EMPTY || cur == key) return (int)idx;
+    unsigned long long k = t->key[idx];      /* fast path: plain read */
+    if (k == key) return (int)idx;
+    if (k == OBX_KEY_EMPTY) {
+      unsigned long long cur = atomicCAS(&t->key[idx], OBX_KEY_EMPTY,
+                                         (unsigned long long)key);
+      if (cur == OBX_KEY_EMPTY || cur == key) return (int)idx;
+    }
     idx = (idx + 1) & (OBX_LTABLE_SLOTS - 1);
   }
   return -1;
@@ -332,7 +336,6 @@ __device__ void scan_filter_agg_body(
     gslot *__restrict__ gtable, unsigned long long *__restrict__ counters,
     uint8_t *lds_blk) {
   __shared__ lds_table tab;
-  __shared__ dev_block cur;
   __shared__ unsigned long long wg_passed;
 
   const uint32_t tid = threadIdx.x;
@@ -357,14 +360,10 @@ __device__ void scan_filter_agg_body(
   __syncthreads();
 
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
-    {
-      const uint32_t words = sizeof(dev_block) / 4;
-      const uint32_t *src = (const uint32_t *)&blocks[b];
-      uint32_t *dst = (uint32_t *)&cur;
-      for (uint32_t i = tid; i < words; i += WG) dst[i] = src[i];
-    }
-    __syncthreads();
+    /* descriptor fields are read via uniform (scalar-cached) loads */
+    const dev_block &cur = blocks[b];
     if (STAGE) {
+      __syncthreads(); /* previous block's LDS reads must drain */
       stage_block(buf, cur, lds_blk);
       __syncthreads();
     }
@@ -452,49 +451,29 @@ __device__ void scan_filter_agg_body(
         }
       }
 
-      if (any_minmax) {
-        if (pass) {
-          int s = lds_slot(&tab, key);
-          if (s < 0) { atomicAdd(&counters[1], 1ull); continue; }
-          atomicAdd(&tab.count[s], 1ull);
-          for (uint32_t a = 0; a < ph.n_aggs; a++) {
-            uint8_t k = ph.aggs[a].kind;
-            if (k == 2 || k == 3) {
-              if (!pv[a].hi) {
-                cas_minmax(&tab.cell[s][a][0], (int64_t)pv[a].lo, k == 2);
-                tab.cell[s][a][1] = 1;
-              }
-            } else {
-              lds_acc_i128(tab.cell[s][a], pv[a]);
+      /* per-lane LDS atomic accumulation (profiled faster than wave-
+         clustered shuffle reductions: ds_bpermute chains are latency-bound
+         at ~4 clusters/wave; same-address LDS atomic adds serialize only
+         within a lane group and carry adds are rare) */
+      if (pass) {
+        int s = lds_slot(&tab, key);
+        if (s < 0) { atomicAdd(&counters[1], 1ull); continue; }
+        atomicAdd(&tab.count[s], 1ull);
+        for (uint32_t a = 0; a < ph.n_aggs; a++) {
+          uint8_t k = ph.aggs[a].kind;
+          if (any_minmax && (k == 2 || k == 3)) {
+            if (!pv[a].hi) {
+              cas_minmax(&tab.cell[s][a][0], (int64_t)pv[a].lo, k == 2);
+              tab.cell[s][a][1] = 1;
             }
+          } else {
+            lds_acc_i128(tab.cell[s][a], pv[a]);
           }
-        }
-      } else {
-        uint64_t active = pass_mask;
-        while (active) {
-          int leader = __ffsll((unsigned long long)active) - 1;
-          uint64_t lkey = shfl64(key, leader);
-          uint64_t same = __ballot(pass && key == lkey) & active;
-          bool member = (same >> lane) & 1;
-          int slot = -1;
-          if ((int)lane == leader) slot = lds_slot(&tab, lkey);
-          for (uint32_t a = 0; a < ph.n_aggs; a++) {
-            i128v contrib = member ? pv[a] : i128_from_i64(0);
-            i128v tot = wave_sum_i128(contrib);
-            if ((int)lane == leader) {
-              if (slot >= 0) lds_acc_i128(tab.cell[slot][a], tot);
-              else atomicAdd(&counters[1], 1ull);
-            }
-          }
-          if ((int)lane == leader && slot >= 0)
-            atomicAdd(&tab.count[slot],
-                      (unsigned long long)__popcll(same));
-          active &= ~same;
         }
       }
     }
-    __syncthreads();
   }
+  __syncthreads();
 
   /* flush LDS table to the global table */
   if (tid == 0 && wg_passed) atomicAdd(&counters[0], wg_passed);
@@ -554,7 +533,6 @@ __device__ void filter_body(
     uint64_t *__restrict__ bitmap, int32_t *__restrict__ row_ids,
     uint32_t *__restrict__ blk_counts,
     unsigned long long *__restrict__ counters, uint8_t *lds_blk) {
-  __shared__ dev_block cur;
   __shared__ uint32_t wv_cnt[WAVES];
   __shared__ uint32_t wv_scan[WAVES];
   __shared__ uint32_t blk_written;
@@ -567,18 +545,13 @@ __device__ void filter_body(
   __syncthreads();
 
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
-    {
-      const uint32_t words = sizeof(dev_block) / 4;
-      const uint32_t *src = (const uint32_t *)&blocks[b];
-      uint32_t *dst = (uint32_t *)&cur;
-      for (uint32_t i = tid; i < words; i += WG) dst[i] = src[i];
-      if (tid == 0) blk_written = 0;
+    const dev_block &cur = blocks[b];
+    if (tid == 0) blk_written = 0;
+    if (STAGE) {
+      __syncthreads();
+      stage_block(buf, cur, lds_blk);
     }
     __syncthreads();
-    if (STAGE) {
-      stage_block(buf, cur, lds_blk);
-      __syncthreads();
-    }
     blk_view bv;
     bv.base = STAGE ? lds_blk : buf;
     bv.bit_bias = STAGE ? cur.block_byte * 8 : 0;
@@ -668,16 +641,9 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_decode(
     const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
     uint32_t n_blocks, uint32_t col, uint32_t datum_len,
     uint8_t *__restrict__ out, uint8_t *__restrict__ out_null) {
-  __shared__ dev_block cur;
   const uint32_t tid = threadIdx.x;
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
-    {
-      const uint32_t words = sizeof(dev_block) / 4;
-      const uint32_t *src = (const uint32_t *)&blocks[b];
-      uint32_t *dst = (uint32_t *)&cur;
-      for (uint32_t i = tid; i < words; i += WG) dst[i] = src[i];
-    }
-    __syncthreads();
+    const dev_block &cur = blocks[b];
     blk_view bv; bv.base = buf; bv.bit_bias = 0;
     const uint32_t rows = cur.row_count;
     const uint64_t row_start = dev_block_row_start(&cur);
@@ -690,7 +656,6 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_decode(
         dst[i] = (uint8_t)(uv >> (i * 8));
       if (out_null) out_null[row_start + r] = isn ? 1 : 0;
     }
-    __syncthreads();
   }
 }
 
