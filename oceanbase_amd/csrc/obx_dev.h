@@ -79,7 +79,7 @@ typedef struct dev_block {
 } dev_block;
 
 /* max block size eligible for LDS staging (kernel stages the whole block) */
-#define OBX_LDS_STAGE_BYTES (20 * 1024)
+#define OBX_LDS_STAGE_BYTES 17408
 
 #if defined(__HIPCC__) || defined(__HIP_DEVICE_COMPILE__)
 __host__ __device__

@@ -269,7 +269,8 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
     int rc = parse_block(bs->cols, bs->n_cols, bs->data + bs->block_offsets[b],
                          bs->block_offsets[b], blen, row_start, &blocks[b]);
     if (rc != OBX_SUCCESS) return rc;
-    if ((bs->block_offsets[b] & 15) || blen + 24 > OBX_LDS_STAGE_BYTES)
+    if ((bs->block_offsets[b] & 15) || blen + 24 > OBX_LDS_STAGE_BYTES ||
+        blocks[b].row_count > 4096 /* OBX_MAX_BLOCK_ROWS */)
       lds_ok = false;
     row_start += blocks[b].row_count;
   }

@@ -112,8 +112,184 @@ __device__ __forceinline__ int64_t dict_entry(const blk_view &bv,
   return (c.flags & OBX_DF_SIGNED) ? sext(v, c.tss) : sext(v, c.entry_len);
 }
 
-/* full value decode: int64 (sign-extended), is_null out */
-__device__ __forceinline__ int64_t col_value(const blk_view &bv,
+/* ---------------- per-block register contexts ---------------------------
+ * Built ONCE per block from the (uniform but vector-loaded) descriptors so
+ * the row loops touch only registers; without this the compiler re-issues
+ * global byte loads of descriptor fields per row (SMEM scalarization is
+ * blocked by the kernel's global stores). */
+struct col_ctx {
+  uint64_t data_bit;   /* packed stream start (absolute bits) */
+  uint64_t aux;        /* DICT: dict payload bit pos; INTDIFF/CONST: base */
+  uint64_t ext_bit;    /* ext-bit stream (HAS_EXT) */
+  uint32_t count;      /* DICT: count (null ref >= count); else 0 */
+  uint16_t W;          /* packed entry width in bits */
+  uint8_t entry_bits;  /* DICT: dict entry width in bits */
+  uint8_t sext_sh;     /* value: shift for sign extension (0 = none) */
+  uint8_t ent_sh;      /* DICT entry sign-extension shift */
+  uint8_t kind;        /* 0 raw, 1 dict, 2 intdiff, 3 const, 4 slow(RLE/exc) */
+  uint8_t has_ext;
+  uint8_t ext_w;
+};
+
+__device__ __forceinline__ col_ctx make_col_ctx(const dev_col &c) {
+  col_ctx x;
+  x.data_bit = c.data_bit;
+  x.aux = 0; x.ext_bit = c.ext_bit; x.count = 0;
+  x.W = (c.flags & OBX_DF_BITPACK) ? c.width : (uint16_t)(c.width * 8);
+  x.entry_bits = 0; x.sext_sh = 0; x.ent_sh = 0;
+  x.has_ext = (c.flags & OBX_DF_HAS_EXT) ? 1 : 0;
+  x.ext_w = c.ext_width;
+  switch (c.enc) {
+    case OBX_D_RAW:
+      x.kind = 0;
+      if (!(c.flags & OBX_DF_BITPACK) && !(c.flags & OBX_DF_STRING))
+        x.sext_sh = (uint8_t)(64 - 8 * ((c.flags & OBX_DF_SIGNED) ? c.tss
+                                                                  : c.width));
+      if (x.sext_sh == 0 || x.sext_sh >= 64) x.sext_sh = 0;
+      break;
+    case OBX_D_DICT:
+      x.kind = 1;
+      x.count = c.count;
+      x.aux = c.dict_byte * 8;
+      x.entry_bits = (uint8_t)(c.entry_len * 8);
+      if (!(c.flags & OBX_DF_STRING)) {
+        uint32_t sb = (c.flags & OBX_DF_SIGNED) ? c.tss : c.entry_len;
+        x.ent_sh = (uint8_t)(sb >= 8 ? 0 : 64 - 8 * sb);
+      }
+      break;
+    case OBX_D_INTDIFF:
+      x.kind = 2;
+      x.aux = (uint64_t)c.base;
+      break;
+    case OBX_D_CONST:
+      if (c.runs == 0) {
+        x.kind = 3;
+        x.aux = (uint64_t)c.base;
+        x.count = c.count; /* 0 = null const */
+      } else {
+        x.kind = 4;
+      }
+      break;
+    default:
+      x.kind = 4; /* RLE and exotic: slow path */
+      break;
+  }
+  return x;
+}
+
+__device__ __forceinline__ int64_t ctx_value(const blk_view &bv,
+                                             const col_ctx &x, uint32_t r,
+                                             bool &null_out) {
+  null_out = false;
+  switch (x.kind) {
+    case 0: { /* RAW */
+      if (x.has_ext &&
+          bit_read(bv, x.ext_bit + (uint64_t)r * x.ext_w, x.ext_w)) {
+        null_out = true; return 0;
+      }
+      uint64_t v = bit_read(bv, x.data_bit + (uint64_t)r * x.W, x.W);
+      return x.sext_sh ? (((int64_t)(v << x.sext_sh)) >> x.sext_sh)
+                       : (int64_t)v;
+    }
+    case 1: { /* DICT */
+      uint64_t ref = bit_read(bv, x.data_bit + (uint64_t)r * x.W, x.W);
+      if (ref >= x.count) { null_out = true; return 0; }
+      uint64_t v = bit_read(bv, x.aux + ref * x.entry_bits, x.entry_bits);
+      return x.ent_sh ? (((int64_t)(v << x.ent_sh)) >> x.ent_sh) : (int64_t)v;
+    }
+    case 2: { /* INTDIFF */
+      if (x.has_ext &&
+          bit_read(bv, x.ext_bit + (uint64_t)r * x.ext_w, x.ext_w)) {
+        null_out = true; return 0;
+      }
+      return (int64_t)(x.aux + bit_read(bv, x.data_bit + (uint64_t)r * x.W,
+                                        x.W));
+    }
+    case 3: { /* CONST, no exceptions */
+      if (x.count == 0) { null_out = true; return 0; }
+      return (int64_t)x.aux;
+    }
+    default:
+      return 0; /* caller uses the slow path for kind 4 */
+  }
+}
+
+/* per-block filter-leaf context (registers) */
+struct leaf_ctx {
+  col_ctx cc;              /* decode context of the filter column */
+  uint64_t mask, lo, hi;   /* REF_MASK / packed RANGE */
+  int64_t vlo, vhi;        /* VALUE-mode operands */
+  uint8_t mode, invert, op, slow;
+};
+
+__device__ __forceinline__ leaf_ctx make_leaf_ctx(const dev_block &blk,
+                                                  const dev_leaf &plf,
+                                                  const blk_leaf &blf) {
+  const dev_col &c = blk.cols[plf.col];
+  leaf_ctx x;
+  x.cc = make_col_ctx(c);
+  x.mask = blf.mask; x.lo = blf.lo; x.hi = blf.hi;
+  x.vlo = plf.vlo; x.vhi = plf.vhi;
+  x.mode = blf.mode; x.invert = blf.invert; x.op = plf.op;
+  /* slow fallback: RLE/exception decode, or IN lists (operand array stays
+     in the global leaf) */
+  x.slow = (x.cc.kind == 4 ||
+            (blf.mode == OBX_LEAF_VALUE && plf.op == 7)) ? 1 : 0;
+  return x;
+}
+
+__device__ __forceinline__ bool leaf_ctx_match(const blk_view &bv,
+                                               const leaf_ctx &x,
+                                               uint32_t r) {
+  switch (x.mode) {
+    case OBX_LEAF_NONE: return false;
+    case OBX_LEAF_ALL:
+      if (x.cc.has_ext &&
+          bit_read(bv, x.cc.ext_bit + (uint64_t)r * x.cc.ext_w, x.cc.ext_w))
+        return false;
+      return true;
+    case OBX_LEAF_REF_MASK: {
+      uint64_t ref = bit_read(bv, x.cc.data_bit + (uint64_t)r * x.cc.W,
+                              x.cc.W);
+      return (x.mask >> ref) & 1;
+    }
+    case OBX_LEAF_RANGE: {
+      if (x.cc.has_ext &&
+          bit_read(bv, x.cc.ext_bit + (uint64_t)r * x.cc.ext_w, x.cc.ext_w))
+        return false;
+      uint64_t v = bit_read(bv, x.cc.data_bit + (uint64_t)r * x.cc.W, x.cc.W);
+      bool in = (v - x.lo) <= (x.hi - x.lo);
+      return in != (bool)x.invert;
+    }
+    case OBX_LEAF_NULL: {
+      bool isn;
+      (void)ctx_value(bv, x.cc, r, isn);
+      return isn != (bool)x.invert;
+    }
+    default: { /* VALUE (op != IN) */
+      bool isn;
+      int64_t v = ctx_value(bv, x.cc, r, isn);
+      if (x.op == 8) return isn;
+      if (x.op == 9) return !isn;
+      if (isn) return false;
+      switch (x.op) {
+        case 0: return v == x.vlo;
+        case 1: return v <= x.vlo;
+        case 2: return v < x.vlo;
+        case 3: return v >= x.vlo;
+        case 4: return v > x.vlo;
+        case 5: return v != x.vlo;
+        case 6: return v >= x.vlo && v <= x.vhi;
+      }
+      return false;
+    }
+  }
+}
+
+/* full value decode: int64 (sign-extended), is_null out.
+ * __noinline__: slow/cold generic path — keeps register pressure off the
+ * hot ctx_value loops. */
+__device__ __noinline__ int64_t col_value(const blk_view &bv,
                                              const dev_col &c, uint32_t r,
                                              bool &null_out) {
   null_out = false;
@@ -186,7 +362,7 @@ __device__ __forceinline__ bool leaf_value_match(const dev_leaf &lf, int64_t v,
   return false;
 }
 
-__device__ __forceinline__ bool leaf_match(const blk_view &bv,
+__device__ __noinline__ bool leaf_match(const blk_view &bv,
                                            const dev_block &blk,
                                            const dev_leaf &plf,
                                            const blk_leaf &blf, uint32_t r) {
@@ -316,18 +492,35 @@ __device__ __forceinline__ void g_acc_i128(unsigned long long *limbs,
   }
 }
 
-/* stage one block's bytes into LDS (coalesced dwordx4; block_byte is
- * 16-B aligned, container blocks are 16-B aligned with zero padding) */
-__device__ __forceinline__ void stage_block(const uint8_t *__restrict__ buf,
+/* issue one block's bytes as async LDS-DMA (global_load_lds_dwordx4;
+ * block_byte is 16-B aligned, container blocks are 16-B aligned with zero
+ * padding). Completion: the issuing thread's s_waitcnt vmcnt(0), then a
+ * workgroup barrier. */
+__device__ __forceinline__ void stage_issue(const uint8_t *__restrict__ buf,
                                             const dev_block &blk,
                                             uint8_t *lds_blk) {
   const uint32_t n16 = (blk.block_len + 8 + 15) >> 4;
-  const uint4 *src = (const uint4 *)(buf + blk.block_byte);
-  uint4 *dst = (uint4 *)lds_blk;
-  for (uint32_t i = threadIdx.x; i < n16; i += WG) dst[i] = src[i];
+  const uint8_t *src = buf + blk.block_byte;
+  for (uint32_t i = threadIdx.x; i < n16; i += WG) {
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) uint32_t *)(src +
+                                                             (size_t)i * 16),
+        (__attribute__((address_space(3))) uint32_t *)(lds_blk +
+                                                       (size_t)i * 16),
+        16, 0, 0);
+  }
+}
+
+__device__ __forceinline__ void stage_wait() {
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
 }
 
 /* ---------------- fused scan->filter->aggregate kernel ------------------ */
+#define OBX_FAST_LEAVES 4
+#define OBX_FAST_NEED 8
+#define OBX_MAX_BLOCK_ROWS 4096
+
 template <bool STAGE>
 __device__ void scan_filter_agg_body(
     const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
@@ -336,14 +529,14 @@ __device__ void scan_filter_agg_body(
     gslot *__restrict__ gtable, unsigned long long *__restrict__ counters,
     uint8_t *lds_blk) {
   __shared__ lds_table tab;
+  __shared__ uint64_t pass_bm[OBX_MAX_BLOCK_ROWS / 64];
+  __shared__ uint8_t row_slot[OBX_MAX_BLOCK_ROWS];
   __shared__ unsigned long long wg_passed;
 
   const uint32_t tid = threadIdx.x;
   const uint32_t lane = tid & 63;
+  const uint32_t wv = tid >> 6;
 
-  bool any_minmax = false;
-  for (uint32_t a = 0; a < ph.n_aggs; a++)
-    if (ph.aggs[a].kind == 2 || ph.aggs[a].kind == 3) any_minmax = true;
 
   for (uint32_t s = tid; s < OBX_LTABLE_SLOTS; s += WG) {
     tab.key[s] = OBX_KEY_EMPTY;
@@ -359,119 +552,194 @@ __device__ void scan_filter_agg_body(
   if (tid == 0) wg_passed = 0;
   __syncthreads();
 
+  uint32_t par = 0;
+  if (STAGE && blockIdx.x < n_blocks)
+    stage_issue(buf, blocks[blockIdx.x], lds_blk);
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
-    /* descriptor fields are read via uniform (scalar-cached) loads */
     const dev_block &cur = blocks[b];
     if (STAGE) {
-      __syncthreads(); /* previous block's LDS reads must drain */
-      stage_block(buf, cur, lds_blk);
-      __syncthreads();
+      stage_wait();
+      uint32_t b2 = b + gridDim.x;
+      if (b2 < n_blocks)
+        stage_issue(buf, blocks[b2],
+                    lds_blk + (par ^ 1) * OBX_LDS_STAGE_BYTES);
     }
     blk_view bv;
-    bv.base = STAGE ? lds_blk : buf;
+    bv.base = STAGE ? lds_blk + par * OBX_LDS_STAGE_BYTES : buf;
     bv.bit_bias = STAGE ? cur.block_byte * 8 : 0;
+    par ^= 1;
 
-    const uint32_t rows = cur.row_count;
-    const uint32_t iters = (rows + WG - 1) / WG;
+    const uint32_t all_rows = cur.row_count;
     const blk_leaf *bl = bleaves + (uint64_t)b * ph.n_leaves;
+    for (uint32_t w0 = 0; w0 < all_rows; w0 += OBX_MAX_BLOCK_ROWS) {
+    const uint32_t rows = (all_rows - w0 < OBX_MAX_BLOCK_ROWS)
+                              ? all_rows - w0 : OBX_MAX_BLOCK_ROWS;
+    const uint32_t iters = (rows + WG - 1) / WG;
 
-    for (uint32_t it = 0; it < iters; it++) {
-      uint32_t r = it * WG + tid;
-      bool pass = r < rows;
-      for (uint32_t i = 0; i < ph.n_leaves && pass; i++)
-        pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
-
-      uint64_t pass_mask = __ballot(pass);
-      if (lane == 0 && pass_mask)
-        atomicAdd(&wg_passed, (unsigned long long)__popcll(pass_mask));
-      if (ph.n_aggs == 0 || !pass_mask) continue;
-
-      /* decode needed values once per surviving row */
-      int64_t vals[OBX_DEV_MAX_NEED];
-      uint32_t nullbits = 0;
-      uint64_t key = 0;
-      if (pass) {
-        for (uint32_t i = 0; i < ph.n_need; i++) {
-          bool isn;
-          vals[i] = col_value(bv, cur.cols[ph.need_cols[i]], r, isn);
-          nullbits |= (isn ? 1u : 0u) << i;
-        }
-        uint32_t koff = 0;
-        for (uint32_t gi = 0; gi < ph.n_group_cols; gi++) {
-          uint32_t vi = ph.group_idx[gi];
-          uint32_t kl = ph.group_len[gi];
-          uint64_t kb = (nullbits >> vi) & 1
-                            ? 0x00ffffffffffffffull
-                            : ((uint64_t)vals[vi] &
-                               ((kl >= 8) ? ~0ull
-                                          : (((uint64_t)1 << (kl * 8)) - 1)));
-          key |= kb << (koff * 8);
-          koff += kl;
+    /* ---- phase 1: filter into the LDS pass bitmap ---- */
+    {
+      leaf_ctx lcs[OBX_FAST_LEAVES];
+      uint32_t slow_set = 0;
+#pragma unroll
+      for (uint32_t i = 0; i < OBX_FAST_LEAVES; i++) {
+        if (i < ph.n_leaves) {
+          lcs[i] = make_leaf_ctx(cur, plan_leaves[i], bl[i]);
+          if (lcs[i].slow) slow_set |= 1u << i;
         }
       }
-
-      /* per-row aggregate contributions */
-      i128v pv[OBX_DEV_MAX_AGGS];
-      if (pass) {
-        for (uint32_t a = 0; a < ph.n_aggs; a++) {
-          const dev_agg &ag = ph.aggs[a];
-          bool na = ag.ia != 0xFF && ((nullbits >> ag.ia) & 1);
-          bool nb = ag.ib != 0xFF && ((nullbits >> ag.ib) & 1);
-          bool nc = ag.ic != 0xFF && ((nullbits >> ag.ic) & 1);
-          switch (ag.kind) {
-            case 0:
-              pv[a] = i128_from_i64((ag.ia != 0xFF && na) ? 0 : 1);
-              break;
-            case 1:
-              pv[a] = i128_from_i64(na ? 0 : vals[ag.ia]);
-              break;
-            case 4:
-              pv[a] = (na || nb)
-                          ? i128_from_i64(0)
-                          : i128_mul_i64(vals[ag.ia], ag.one_b - vals[ag.ib]);
-              break;
-            case 5:
-              pv[a] = (na || nb || nc)
-                          ? i128_from_i64(0)
-                          : i128_mul_pos_i64(
-                                i128_mul_i64(vals[ag.ia],
-                                             ag.one_b - vals[ag.ib]),
-                                ag.one_c + vals[ag.ic]);
-              break;
-            case 6:
-              pv[a] = (na || nb)
-                          ? i128_from_i64(0)
-                          : i128_mul_i64(vals[ag.ia], vals[ag.ib]);
-              break;
-            default: /* MIN/MAX: lo = value, hi = null flag */
-              pv[a].lo = (uint64_t)(na ? 0 : vals[ag.ia]);
-              pv[a].hi = na ? 1 : 0;
-              break;
+      for (uint32_t it = 0; it < iters; it++) {
+        uint32_t rr = it * WG + tid;
+        uint32_t r = w0 + rr;
+        bool pass = rr < rows;
+#pragma unroll
+        for (uint32_t i = 0; i < OBX_FAST_LEAVES; i++) {
+          if (i < ph.n_leaves && pass) {
+            if (slow_set & (1u << i))
+              pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
+            else
+              pass = leaf_ctx_match(bv, lcs[i], r);
           }
         }
+        for (uint32_t i = OBX_FAST_LEAVES; i < ph.n_leaves && pass; i++)
+          pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
+        uint64_t m = __ballot(pass);
+        if (lane == 0) {
+          pass_bm[it * WAVES + wv] = m;
+          if (m) atomicAdd(&wg_passed, (unsigned long long)__popcll(m));
+        }
       }
+    }
+    __syncthreads();
 
-      /* per-lane LDS atomic accumulation (profiled faster than wave-
-         clustered shuffle reductions: ds_bpermute chains are latency-bound
-         at ~4 clusters/wave; same-address LDS atomic adds serialize only
-         within a lane group and carry adds are rare) */
-      if (pass) {
-        int s = lds_slot(&tab, key);
-        if (s < 0) { atomicAdd(&counters[1], 1ull); continue; }
-        atomicAdd(&tab.count[s], 1ull);
-        for (uint32_t a = 0; a < ph.n_aggs; a++) {
-          uint8_t k = ph.aggs[a].kind;
-          if (any_minmax && (k == 2 || k == 3)) {
-            if (!pv[a].hi) {
-              cas_minmax(&tab.cell[s][a][0], (int64_t)pv[a].lo, k == 2);
+    /* ---- phase 2: group rows to LDS-table slots (row_slot map) ---- */
+    if (ph.n_aggs) {
+      col_ctx g0, g1;
+      const dev_col *gd0 = nullptr, *gd1 = nullptr;
+      if (ph.n_group_cols > 0) {
+        gd0 = &cur.cols[ph.need_cols[ph.group_idx[0]]];
+        g0 = make_col_ctx(*gd0);
+      }
+      if (ph.n_group_cols > 1) {
+        gd1 = &cur.cols[ph.need_cols[ph.group_idx[1]]];
+        g1 = make_col_ctx(*gd1);
+      }
+      const uint32_t kl0 = ph.group_len[0], kl1 = ph.group_len[1];
+      for (uint32_t it = 0; it < iters; it++) {
+        uint32_t rr = it * WG + tid;
+        uint32_t r = w0 + rr;
+        uint64_t m = pass_bm[it * WAVES + wv];
+        bool pass = (m >> lane) & 1;
+        if (!m) { if (rr < rows) row_slot[rr] = 255; continue; }
+        uint8_t slot8 = 255;
+        if (pass) {
+          uint64_t key = 0;
+          if (ph.n_group_cols > 0) {
+            bool isn;
+            int64_t kv = (g0.kind == 4) ? col_value(bv, *gd0, r, isn)
+                                        : ctx_value(bv, g0, r, isn);
+            key = isn ? 0x00ffffffffffffffull
+                      : ((uint64_t)kv &
+                         ((kl0 >= 8) ? ~0ull
+                                     : (((uint64_t)1 << (kl0 * 8)) - 1)));
+            if (ph.n_group_cols > 1) {
+              int64_t kv1 = (g1.kind == 4) ? col_value(bv, *gd1, r, isn)
+                                           : ctx_value(bv, g1, r, isn);
+              uint64_t kb = isn ? 0x00ffffffffffffffull
+                                : ((uint64_t)kv1 &
+                                   ((kl1 >= 8)
+                                        ? ~0ull
+                                        : (((uint64_t)1 << (kl1 * 8)) - 1)));
+              key |= kb << (kl0 * 8);
+            }
+          }
+          int s = lds_slot(&tab, key);
+          if (s < 0) {
+            atomicAdd(&counters[1], 1ull);
+          } else {
+            slot8 = (uint8_t)s;
+            atomicAdd(&tab.count[s], 1ull);
+          }
+        }
+        if (rr < rows) row_slot[rr] = slot8;
+      }
+      __syncthreads();
+
+      /* ---- phase 3: one pass per aggregate (ctxs live one at a time) -- */
+      for (uint32_t a = 0; a < ph.n_aggs; a++) {
+        const dev_agg ag = ph.aggs[a];
+        col_ctx ca, cb, cc;
+        const dev_col *da = nullptr, *db = nullptr, *dc2 = nullptr;
+        if (ag.ia != 0xFF) {
+          da = &cur.cols[ph.need_cols[ag.ia]];
+          ca = make_col_ctx(*da);
+        }
+        if (ag.ib != 0xFF) {
+          db = &cur.cols[ph.need_cols[ag.ib]];
+          cb = make_col_ctx(*db);
+        }
+        if (ag.ic != 0xFF) {
+          dc2 = &cur.cols[ph.need_cols[ag.ic]];
+          cc = make_col_ctx(*dc2);
+        }
+        for (uint32_t it = 0; it < iters; it++) {
+          uint32_t rr = it * WG + tid;
+          uint32_t r = w0 + rr;
+          uint64_t m = pass_bm[it * WAVES + wv];
+          if (!m) continue;
+          bool pass = (m >> lane) & 1;
+          uint8_t slot8 = pass && rr < rows ? row_slot[rr] : 255;
+          if (slot8 == 255) continue;
+          int s = slot8;
+          bool na = false, nb = false, nc = false;
+          int64_t va = 0, vb = 0, vc = 0;
+          if (ag.ia != 0xFF)
+            va = (ca.kind == 4) ? col_value(bv, *da, r, na)
+                                : ctx_value(bv, ca, r, na);
+          if (ag.ib != 0xFF)
+            vb = (cb.kind == 4) ? col_value(bv, *db, r, nb)
+                                : ctx_value(bv, cb, r, nb);
+          if (ag.ic != 0xFF)
+            vc = (cc.kind == 4) ? col_value(bv, *dc2, r, nc)
+                                : ctx_value(bv, cc, r, nc);
+          i128v pv;
+          switch (ag.kind) {
+            case 0:
+              pv = i128_from_i64((ag.ia != 0xFF && na) ? 0 : 1);
+              break;
+            case 1:
+              pv = i128_from_i64(na ? 0 : va);
+              break;
+            case 4:
+              pv = (na || nb) ? i128_from_i64(0)
+                              : i128_mul_i64(va, ag.one_b - vb);
+              break;
+            case 5:
+              pv = (na || nb || nc)
+                       ? i128_from_i64(0)
+                       : i128_mul_pos_i64(i128_mul_i64(va, ag.one_b - vb),
+                                          ag.one_c + vc);
+              break;
+            case 6:
+              pv = (na || nb) ? i128_from_i64(0) : i128_mul_i64(va, vb);
+              break;
+            default: /* MIN/MAX */
+              pv.lo = (uint64_t)(na ? 0 : va);
+              pv.hi = na ? 1 : 0;
+              break;
+          }
+          if (ag.kind == 2 || ag.kind == 3) {
+            if (!pv.hi) {
+              cas_minmax(&tab.cell[s][a][0], (int64_t)pv.lo, ag.kind == 2);
               tab.cell[s][a][1] = 1;
             }
           } else {
-            lds_acc_i128(tab.cell[s][a], pv[a]);
+            lds_acc_i128(tab.cell[s][a], pv);
           }
         }
       }
     }
+    __syncthreads(); /* pass_bm reuse across windows */
+    } /* window loop */
   }
   __syncthreads();
 
@@ -519,7 +787,7 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg_lds(
     uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
     const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
     gslot *__restrict__ gtable, unsigned long long *__restrict__ counters) {
-  __shared__ uint8_t lds_blk[OBX_LDS_STAGE_BYTES + 32];
+  __shared__ uint8_t lds_blk[2 * OBX_LDS_STAGE_BYTES + 32];
   scan_filter_agg_body<true>(buf, blocks, n_blocks, plan_leaves, bleaves, ph,
                              gtable, counters, lds_blk);
 }
@@ -544,27 +812,54 @@ __device__ void filter_body(
   if (tid == 0) wg_passed = 0;
   __syncthreads();
 
+  uint32_t par = 0;
+  if (STAGE && blockIdx.x < n_blocks)
+    stage_issue(buf, blocks[blockIdx.x], lds_blk);
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
     const dev_block &cur = blocks[b];
     if (tid == 0) blk_written = 0;
     if (STAGE) {
-      __syncthreads();
-      stage_block(buf, cur, lds_blk);
+      stage_wait();
+      uint32_t b2 = b + gridDim.x;
+      if (b2 < n_blocks)
+        stage_issue(buf, blocks[b2],
+                    lds_blk + (par ^ 1) * OBX_LDS_STAGE_BYTES);
+    } else {
+      __syncthreads(); /* blk_written reset */
     }
-    __syncthreads();
     blk_view bv;
-    bv.base = STAGE ? lds_blk : buf;
+    bv.base = STAGE ? lds_blk + par * OBX_LDS_STAGE_BYTES : buf;
     bv.bit_bias = STAGE ? cur.block_byte * 8 : 0;
+    par ^= 1;
 
     const uint32_t rows = cur.row_count;
     const uint64_t row_start = dev_block_row_start(&cur);
     const uint32_t iters = (rows + WG - 1) / WG;
     const blk_leaf *bl = bleaves + (uint64_t)b * ph.n_leaves;
 
+    leaf_ctx lcs[OBX_FAST_LEAVES];
+    uint32_t slow_set = 0;
+#pragma unroll
+    for (uint32_t i = 0; i < OBX_FAST_LEAVES; i++) {
+      if (i < ph.n_leaves) {
+        lcs[i] = make_leaf_ctx(cur, plan_leaves[i], bl[i]);
+        if (lcs[i].slow) slow_set |= 1u << i;
+      }
+    }
+
     for (uint32_t it = 0; it < iters; it++) {
       uint32_t r = it * WG + tid;
       bool pass = r < rows;
-      for (uint32_t i = 0; i < ph.n_leaves && pass; i++)
+#pragma unroll
+      for (uint32_t i = 0; i < OBX_FAST_LEAVES; i++) {
+        if (i < ph.n_leaves && pass) {
+          if (slow_set & (1u << i))
+            pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
+          else
+            pass = leaf_ctx_match(bv, lcs[i], r);
+        }
+      }
+      for (uint32_t i = OBX_FAST_LEAVES; i < ph.n_leaves && pass; i++)
         pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
       uint64_t m = __ballot(pass);
 
@@ -631,7 +926,7 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_filter_lds(
     uint64_t *__restrict__ bitmap, int32_t *__restrict__ row_ids,
     uint32_t *__restrict__ blk_counts,
     unsigned long long *__restrict__ counters) {
-  __shared__ uint8_t lds_blk[OBX_LDS_STAGE_BYTES + 32];
+  __shared__ uint8_t lds_blk[2 * OBX_LDS_STAGE_BYTES + 32];
   filter_body<true>(buf, blocks, n_blocks, plan_leaves, bleaves, ph, bitmap,
                     row_ids, blk_counts, counters, lds_blk);
 }
