@@ -23,7 +23,7 @@
 #define OBX_DEV_MAX_LEAVES 8
 #define OBX_DEV_MAX_AGGS 8
 #define OBX_GTABLE_SLOTS 256   /* global group table (open addressing) */
-#define OBX_LTABLE_SLOTS 32    /* per-workgroup LDS group table */
+#define OBX_LTABLE_SLOTS 16    /* per-workgroup LDS group table */
 
 /* column encodings (== ObColumnHeader::Type subset) */
 enum {
