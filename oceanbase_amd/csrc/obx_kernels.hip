@@ -31,6 +31,9 @@
 
 #define WG 256
 #define WAVES (WG / 64)
+#ifndef OBX_PIPELINE
+#define OBX_PIPELINE 0   /* 0 = single-buffer staging (higher occupancy; measured faster), 1 = double-buffered prefetch */
+#endif
 
 /* view of a block's bytes: either the global buffer (bias 0) or the LDS
  * staging copy (bias = block_byte * 8). All dev_col offsets are absolute;
@@ -432,7 +435,9 @@ __device__ __forceinline__ i128v wave_sum_i128(i128v v) {
  * Cells are 4-way lane-striped (stripe = lane & 3): same-slot atomic adds
  * from one wave serialize only within a 16-lane stripe group, cutting the
  * measured LDS same-address conflict cycles ~4x. Stripes merge at flush. */
-#define OBX_STRIPES 4
+#ifndef OBX_STRIPES
+#define OBX_STRIPES 2
+#endif
 struct lds_table {
   unsigned long long key[OBX_LTABLE_SLOTS];
   unsigned long long count[OBX_LTABLE_SLOTS][OBX_STRIPES];
@@ -523,7 +528,7 @@ __device__ __forceinline__ void stage_wait() {
 /* ---------------- fused scan->filter->aggregate kernel ------------------ */
 #define OBX_FAST_LEAVES 4
 #define OBX_FAST_NEED 8
-#define OBX_MAX_BLOCK_ROWS 4096
+#define OBX_MAX_BLOCK_ROWS 2048
 
 template <bool STAGE>
 __device__ void scan_filter_agg_body(
@@ -559,19 +564,26 @@ __device__ void scan_filter_agg_body(
   __syncthreads();
 
   uint32_t par = 0;
-  if (STAGE && blockIdx.x < n_blocks)
+  if (STAGE && OBX_PIPELINE && blockIdx.x < n_blocks)
     stage_issue(buf, blocks[blockIdx.x], lds_blk);
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
     const dev_block &cur = blocks[b];
     if (STAGE) {
+#if OBX_PIPELINE
       stage_wait();
       uint32_t b2 = b + gridDim.x;
       if (b2 < n_blocks)
         stage_issue(buf, blocks[b2],
                     lds_blk + (par ^ 1) * OBX_LDS_STAGE_BYTES);
+#else
+      __syncthreads(); /* drain previous block's reads */
+      stage_issue(buf, cur, lds_blk);
+      stage_wait();
+#endif
     }
     blk_view bv;
-    bv.base = STAGE ? lds_blk + par * OBX_LDS_STAGE_BYTES : buf;
+    bv.base = STAGE ? lds_blk + (OBX_PIPELINE ? par * OBX_LDS_STAGE_BYTES : 0)
+                    : buf;
     bv.bit_bias = STAGE ? cur.block_byte * 8 : 0;
     par ^= 1;
 
@@ -876,7 +888,7 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg_lds(
     uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
     const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
     gslot *__restrict__ gtable, unsigned long long *__restrict__ counters) {
-  __shared__ uint8_t lds_blk[2 * OBX_LDS_STAGE_BYTES + 32];
+  __shared__ uint8_t lds_blk[(1 + OBX_PIPELINE) * OBX_LDS_STAGE_BYTES + 32];
   scan_filter_agg_body<true>(buf, blocks, n_blocks, plan_leaves, bleaves, ph,
                              gtable, counters, lds_blk);
 }
@@ -902,22 +914,29 @@ __device__ void filter_body(
   __syncthreads();
 
   uint32_t par = 0;
-  if (STAGE && blockIdx.x < n_blocks)
+  if (STAGE && OBX_PIPELINE && blockIdx.x < n_blocks)
     stage_issue(buf, blocks[blockIdx.x], lds_blk);
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
     const dev_block &cur = blocks[b];
     if (tid == 0) blk_written = 0;
     if (STAGE) {
+#if OBX_PIPELINE
       stage_wait();
       uint32_t b2 = b + gridDim.x;
       if (b2 < n_blocks)
         stage_issue(buf, blocks[b2],
                     lds_blk + (par ^ 1) * OBX_LDS_STAGE_BYTES);
+#else
+      __syncthreads();
+      stage_issue(buf, cur, lds_blk);
+      stage_wait();
+#endif
     } else {
       __syncthreads(); /* blk_written reset */
     }
     blk_view bv;
-    bv.base = STAGE ? lds_blk + par * OBX_LDS_STAGE_BYTES : buf;
+    bv.base = STAGE ? lds_blk + (OBX_PIPELINE ? par * OBX_LDS_STAGE_BYTES : 0)
+                    : buf;
     bv.bit_bias = STAGE ? cur.block_byte * 8 : 0;
     par ^= 1;
 
@@ -1015,7 +1034,7 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_filter_lds(
     uint64_t *__restrict__ bitmap, int32_t *__restrict__ row_ids,
     uint32_t *__restrict__ blk_counts,
     unsigned long long *__restrict__ counters) {
-  __shared__ uint8_t lds_blk[2 * OBX_LDS_STAGE_BYTES + 32];
+  __shared__ uint8_t lds_blk[(1 + OBX_PIPELINE) * OBX_LDS_STAGE_BYTES + 32];
   filter_body<true>(buf, blocks, n_blocks, plan_leaves, bleaves, ph, bitmap,
                     row_ids, blk_counts, counters, lds_blk);
 }
