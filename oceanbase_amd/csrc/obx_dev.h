@@ -128,16 +128,37 @@ typedef struct dev_agg {
 } dev_agg;
 
 #define OBX_DEV_MAX_NEED 12
+
+/* aggregate-pass grouping: aggs whose inputs share <=3 distinct columns run
+ * in one row pass (engine builds these in prep_query) */
+typedef struct dev_pass_agg {
+  uint8_t kind;        /* obx_agg_kind */
+  uint8_t agg_idx;     /* output cell index */
+  uint8_t sa, sb, sc;  /* selectors into the pass's decoded values (0..2) */
+  uint8_t pad[3];
+  int64_t one_b, one_c;
+} dev_pass_agg;
+
+typedef struct dev_pass {
+  uint8_t n_aggs;
+  uint8_t n_cols;      /* distinct decoded columns (<=3) */
+  uint8_t cols[3];     /* need-slot index per decoded value */
+  uint8_t pad[3];
+  dev_pass_agg aggs[OBX_DEV_MAX_AGGS];
+} dev_pass;
+
 /* plan header, passed to kernels by value */
 typedef struct dev_plan_hdr {
   uint32_t n_leaves;
   uint32_t n_aggs;
   uint32_t n_group_cols;
   uint32_t n_need;              /* decoded columns per surviving row */
+  uint32_t n_passes;
   uint16_t need_cols[OBX_DEV_MAX_NEED]; /* column index per val slot */
   uint8_t group_idx[2];         /* val-slot index of group cols */
   uint8_t group_len[2];         /* datum byte lens of group cols */
   dev_agg aggs[OBX_DEV_MAX_AGGS];
+  dev_pass passes[OBX_DEV_MAX_AGGS];
 } dev_plan_hdr;
 
 /* global group table slot: key + count + n_aggs 256-bit cells.

@@ -365,6 +365,52 @@ static int prep_query(obx_gpu_ctx *ctx, obx_handle &h,
       }
     }
   }
+  /* group aggregates into row passes: aggs whose inputs fit in <=3 shared
+     decoded columns run together (COUNT(*) needs no pass — filled from the
+     group counts at flush) */
+  if (agg) {
+    bool used[OBX_DEV_MAX_AGGS] = {};
+    ph.n_passes = 0;
+    for (int a = 0; a < agg->n_aggs; a++) {
+      if (used[a]) continue;
+      const dev_agg *da = &ph.aggs[a];
+      if (da->kind == OBX_AGG_COUNT && da->ia == 0xFF) { used[a] = true; continue; }
+      dev_pass *pp = &ph.passes[ph.n_passes++];
+      memset(pp, 0, sizeof(*pp));
+      auto col_sel = [&](uint8_t need_idx) -> int {
+        if (need_idx == 0xFF) return -1;
+        for (int k = 0; k < pp->n_cols; k++)
+          if (pp->cols[k] == need_idx) return k;
+        if (pp->n_cols >= 3) return -2; /* no room */
+        pp->cols[pp->n_cols] = need_idx;
+        return pp->n_cols++;
+      };
+      auto try_add = [&](int idx) -> bool {
+        const dev_agg *d = &ph.aggs[idx];
+        uint8_t save_n = pp->n_cols;
+        int sa = col_sel(d->ia), sb = col_sel(d->ib), sc = col_sel(d->ic);
+        if (sa == -2 || sb == -2 || sc == -2) { pp->n_cols = save_n; return false; }
+        dev_pass_agg *pa = &pp->aggs[pp->n_aggs++];
+        pa->kind = d->kind;
+        pa->agg_idx = (uint8_t)idx;
+        pa->sa = (uint8_t)(sa < 0 ? 0xFF : sa);
+        pa->sb = (uint8_t)(sb < 0 ? 0xFF : sb);
+        pa->sc = (uint8_t)(sc < 0 ? 0xFF : sc);
+        pa->one_b = d->one_b;
+        pa->one_c = d->one_c;
+        return true;
+      };
+      try_add(a);
+      used[a] = true;
+      for (int b2 = a + 1; b2 < agg->n_aggs; b2++) {
+        if (used[b2]) continue;
+        const dev_agg *d2 = &ph.aggs[b2];
+        if (d2->kind == OBX_AGG_COUNT && d2->ia == 0xFF) { used[b2] = true; continue; }
+        if (try_add(b2)) used[b2] = true;
+      }
+    }
+  }
+
   /* upload plan leaves + lower per-block tests on device */
   HIP_TRY(hipMemcpyAsync(h.d_pleaves, pl, sizeof(pl), hipMemcpyHostToDevice,
                          ctx->stream));
