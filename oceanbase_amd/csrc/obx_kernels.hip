@@ -260,7 +260,8 @@ __device__ __forceinline__ bool leaf_ctx_match(const blk_view &bv,
       if (x.cc.has_ext &&
           bit_read(bv, x.cc.ext_bit + (uint64_t)r * x.cc.ext_w, x.cc.ext_w))
         return false;
-      uint64_t v = bit_read(bv, x.cc.data_bit + (uint64_t)r * x.cc.W, x.cc.W);
+      uint64_t v = bit_read(bv, x.cc.data_bit + (uint64_t)r * x.cc.W, x.cc.W)
+                   ^ x.mask; /* signed domains are order-mapped by xor bias */
       bool in = (v - x.lo) <= (x.hi - x.lo);
       return in != (bool)x.invert;
     }
@@ -383,7 +384,7 @@ __device__ __noinline__ bool leaf_match(const blk_view &bv,
     }
     case OBX_LEAF_RANGE: {
       if (col_is_null_ext(bv, c, r)) return false;
-      uint64_t v = col_packed(bv, c, r);
+      uint64_t v = col_packed(bv, c, r) ^ blf.mask;
       bool in = (v - blf.lo) <= (blf.hi - blf.lo);
       return in != (bool)blf.invert;
     }
@@ -525,6 +526,35 @@ __device__ __forceinline__ void stage_wait() {
   __syncthreads();
 }
 
+/* full group-key build (cold path: slow group columns or first occurrence
+ * of a dict-ref cell). __noinline__ keeps its register use off the hot
+ * loops. NULL group values set flag bits in the top key byte. */
+__device__ __noinline__ uint64_t build_group_key(
+    const blk_view &bv, uint32_t n_group_cols, const col_ctx &g0,
+    const col_ctx &g1, const dev_col *gd0, const dev_col *gd1, uint32_t kl0,
+    uint32_t kl1, uint32_t r) {
+  uint64_t key = 0;
+  if (n_group_cols > 0) {
+    bool isn;
+    int64_t kv = (g0.kind == 4) ? col_value(bv, *gd0, r, isn)
+                                : ctx_value(bv, g0, r, isn);
+    if (isn) key |= 1ull << 56;
+    else key |= (uint64_t)kv &
+                ((kl0 >= 8) ? ~0ull : (((uint64_t)1 << (kl0 * 8)) - 1));
+    if (n_group_cols > 1) {
+      bool isn1;
+      int64_t kv1 = (g1.kind == 4) ? col_value(bv, *gd1, r, isn1)
+                                   : ctx_value(bv, g1, r, isn1);
+      uint64_t kb = 0;
+      if (isn1) key |= 1ull << 57;
+      else kb = (uint64_t)kv1 &
+                ((kl1 >= 8) ? ~0ull : (((uint64_t)1 << (kl1 * 8)) - 1));
+      key |= kb << (kl0 * 8);
+    }
+  }
+  return key;
+}
+
 /* ---------------- fused scan->filter->aggregate kernel ------------------ */
 #define OBX_FAST_LEAVES 4
 #define OBX_FAST_NEED 8
@@ -540,6 +570,7 @@ __device__ void scan_filter_agg_body(
   __shared__ lds_table tab;
   __shared__ uint64_t pass_bm[OBX_MAX_BLOCK_ROWS / 64];
   __shared__ uint8_t row_slot[OBX_MAX_BLOCK_ROWS];
+  __shared__ uint8_t cell_slot[64];
   __shared__ unsigned long long wg_passed;
 
   const uint32_t tid = threadIdx.x;
@@ -629,7 +660,13 @@ __device__ void scan_filter_agg_body(
     }
     __syncthreads();
 
-    /* ---- phase 2: group rows to LDS-table slots (row_slot map) ---- */
+    /* ---- phase 2: group rows to LDS-table slots (row_slot map).
+       Fast path (the reference's storage group-by pushdown design,
+       ob_pushdown_aggregate_vec.h:445 / read_reference): when every group
+       column is dict-encoded and the ref product is small, rows map to
+       cells by packed REFS alone; cell -> slot resolves lazily, once per
+       block, via the dict values. NULL group values set flag bits in the
+       top key byte (data bytes stay zero). ---- */
     if (ph.n_aggs) {
       col_ctx g0, g1;
       const dev_col *gd0 = nullptr, *gd1 = nullptr;
@@ -642,43 +679,73 @@ __device__ void scan_filter_agg_body(
         g1 = make_col_ctx(*gd1);
       }
       const uint32_t kl0 = ph.group_len[0], kl1 = ph.group_len[1];
-      for (uint32_t it = 0; it < iters; it++) {
-        uint32_t rr = it * WG + tid;
-        uint32_t r = w0 + rr;
-        uint64_t m = pass_bm[it * WAVES + wv];
-        bool pass = (m >> lane) & 1;
-        if (!m) { if (rr < rows) row_slot[rr] = 255; continue; }
-        uint8_t slot8 = 255;
-        if (pass) {
-          uint64_t key = 0;
-          if (ph.n_group_cols > 0) {
-            bool isn;
-            int64_t kv = (g0.kind == 4) ? col_value(bv, *gd0, r, isn)
-                                        : ctx_value(bv, g0, r, isn);
-            key = isn ? 0x00ffffffffffffffull
-                      : ((uint64_t)kv &
-                         ((kl0 >= 8) ? ~0ull
-                                     : (((uint64_t)1 << (kl0 * 8)) - 1)));
+      const uint32_t dim0 = (ph.n_group_cols > 0) ? g0.count + 1 : 1;
+      const uint32_t dim1 = (ph.n_group_cols > 1) ? g1.count + 1 : 1;
+      const bool ref_fast =
+          ph.n_group_cols > 0 && g0.kind == 1 &&
+          (ph.n_group_cols < 2 || g1.kind == 1) && dim0 * dim1 <= 64;
+
+      if (ref_fast) {
+        for (uint32_t i = tid; i < 64; i += WG) cell_slot[i] = 254;
+        __syncthreads();
+        for (uint32_t it = 0; it < iters; it++) {
+          uint32_t rr = it * WG + tid;
+          uint32_t r = w0 + rr;
+          uint64_t m = pass_bm[it * WAVES + wv];
+          bool pass = (m >> lane) & 1;
+          if (!m) { if (rr < rows) row_slot[rr] = 255; continue; }
+          uint8_t slot8 = 255;
+          if (pass) {
+            uint32_t ref0 = (uint32_t)bit_read(
+                bv, g0.data_bit + (uint64_t)r * g0.W, g0.W);
+            if (ref0 > g0.count) ref0 = g0.count; /* nope -> null bucket */
+            uint32_t cell = ref0;
             if (ph.n_group_cols > 1) {
-              int64_t kv1 = (g1.kind == 4) ? col_value(bv, *gd1, r, isn)
-                                           : ctx_value(bv, g1, r, isn);
-              uint64_t kb = isn ? 0x00ffffffffffffffull
-                                : ((uint64_t)kv1 &
-                                   ((kl1 >= 8)
-                                        ? ~0ull
-                                        : (((uint64_t)1 << (kl1 * 8)) - 1)));
-              key |= kb << (kl0 * 8);
+              uint32_t ref1 = (uint32_t)bit_read(
+                  bv, g1.data_bit + (uint64_t)r * g1.W, g1.W);
+              if (ref1 > g1.count) ref1 = g1.count;
+              cell += ref1 * dim0;
+            }
+            uint8_t cs = cell_slot[cell];
+            if (cs == 254) {
+              int s = lds_slot(&tab, build_group_key(bv, ph.n_group_cols,
+                                                      g0, g1, gd0, gd1, kl0,
+                                                      kl1, r));
+              if (s < 0) {
+                atomicAdd(&counters[1], 1ull);
+                cs = 255;
+              } else {
+                cs = (uint8_t)s;
+              }
+              cell_slot[cell] = cs;
+            }
+            slot8 = cs;
+            if (slot8 != 255)
+              atomicAdd(&tab.count[slot8][lane & (OBX_STRIPES - 1)], 1ull);
+          }
+          if (rr < rows) row_slot[rr] = slot8;
+        }
+      } else {
+        for (uint32_t it = 0; it < iters; it++) {
+          uint32_t rr = it * WG + tid;
+          uint32_t r = w0 + rr;
+          uint64_t m = pass_bm[it * WAVES + wv];
+          bool pass = (m >> lane) & 1;
+          if (!m) { if (rr < rows) row_slot[rr] = 255; continue; }
+          uint8_t slot8 = 255;
+          if (pass) {
+            int s = lds_slot(&tab, build_group_key(bv, ph.n_group_cols, g0,
+                                                   g1, gd0, gd1, kl0, kl1,
+                                                   r));
+            if (s < 0) {
+              atomicAdd(&counters[1], 1ull);
+            } else {
+              slot8 = (uint8_t)s;
+              atomicAdd(&tab.count[s][lane & (OBX_STRIPES - 1)], 1ull);
             }
           }
-          int s = lds_slot(&tab, key);
-          if (s < 0) {
-            atomicAdd(&counters[1], 1ull);
-          } else {
-            slot8 = (uint8_t)s;
-            atomicAdd(&tab.count[s][lane & (OBX_STRIPES - 1)], 1ull);
-          }
+          if (rr < rows) row_slot[rr] = slot8;
         }
-        if (rr < rows) row_slot[rr] = slot8;
       }
       __syncthreads();
 
@@ -1067,13 +1134,13 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_decode(
  * domain (the reference evaluates dict-domain filters once per dict entry,
  * ob_dict_decoder.cpp:810-886,1481-1561; and maps raw compares to its
  * fast SIMD paths, ob_raw_decoder.cpp:707-790). */
-__device__ __forceinline__ void lower_range(int64_t base, uint64_t dmax,
-                                            const dev_leaf &lf,
+__device__ __forceinline__ void lower_range(__int128 D, __int128 D2,
+                                            uint8_t op, uint64_t dmax,
                                             blk_leaf &out) {
-  __int128 D = (__int128)lf.vlo - base;
-  __int128 D2 = (__int128)lf.vhi - base;
+  /* unsigned packed domain: v in [0, dmax]; D/D2 are the operands already
+     mapped into that domain (may lie outside [0, dmax]) */
   out.invert = 0;
-  switch (lf.op) {
+  switch (op) {
     case 0: /* EQ */
       if (D < 0 || D > (__int128)dmax) { out.mode = OBX_LEAF_NONE; return; }
       out.mode = OBX_LEAF_RANGE; out.lo = (uint64_t)D; out.hi = (uint64_t)D;
@@ -1147,11 +1214,35 @@ extern "C" __global__ void k_lower_leaves(
     o.mode = (o.mask == 0) ? OBX_LEAF_NONE : OBX_LEAF_REF_MASK;
   } else if (c.enc == OBX_D_RAW && (c.flags & OBX_DF_BITPACK) && lf.op <= 6) {
     uint64_t dmax = (c.width >= 64) ? ~0ull : ((1ull << c.width) - 1);
-    lower_range(0, dmax, lf, o);
+    lower_range((__int128)lf.vlo, (__int128)lf.vhi, lf.op, dmax, o);
+  } else if (c.enc == OBX_D_RAW && !(c.flags & OBX_DF_BITPACK) &&
+             !(c.flags & OBX_DF_STRING) && lf.op <= 6 && lf.op != 7 /* IN */) {
+    /* fixed-width RAW numeric: k-byte domain. Signed iff the sign bit lives
+       inside the stored bytes (INT class with width==tss, or DECIMAL);
+       order-map signed domains to unsigned with an xor bias. */
+    uint32_t kbits = (uint32_t)c.width * 8;
+    uint64_t dmax = (kbits >= 64) ? ~0ull : ((1ull << kbits) - 1);
+    bool signed_dom = ((c.flags & OBX_DF_SIGNED) && c.width >= c.tss) ||
+                      (!(c.flags & OBX_DF_SIGNED) && c.width == 8);
+    if (signed_dom) {
+      uint64_t bias = 1ull << (kbits - 1);
+      __int128 smin = -(__int128)bias, smax = (__int128)bias - 1;
+      auto biasmap = [&](int64_t x) -> __int128 {
+        __int128 v = x;
+        if (v < smin) return (__int128)-1;          /* below domain */
+        if (v > smax) return (__int128)dmax + 1;    /* above domain */
+        return v + (__int128)bias;                  /* in [0, dmax] */
+      };
+      lower_range(biasmap(lf.vlo), biasmap(lf.vhi), lf.op, dmax, o);
+      if (o.mode == OBX_LEAF_RANGE) o.mask = bias;
+    } else {
+      lower_range((__int128)lf.vlo, (__int128)lf.vhi, lf.op, dmax, o);
+    }
   } else if (c.enc == OBX_D_INTDIFF && lf.op <= 6) {
     uint32_t kbits = (c.flags & OBX_DF_BITPACK) ? c.width : c.width * 8;
     uint64_t dmax = (kbits >= 64) ? ~0ull : ((1ull << kbits) - 1);
-    lower_range(c.base, dmax, lf, o);
+    lower_range((__int128)lf.vlo - c.base, (__int128)lf.vhi - c.base, lf.op,
+                dmax, o);
   } else if ((c.enc == OBX_D_RAW || c.enc == OBX_D_INTDIFF) &&
              (lf.op == 8 || lf.op == 9)) {
     if (c.flags & OBX_DF_HAS_EXT) {

@@ -129,15 +129,19 @@ typedef struct work_ctx {
   int rc;
 } work_ctx;
 
+/* key bytes = concatenated datums (zeros for NULL) + one null-flags byte
+ * (bit per group col) appended at key[key_len] — NULL forms its own group
+ * without colliding with value 0 (mirrors the GPU key layout). */
 static grp *find_group(work_ctx *w, const uint8_t *key, uint8_t key_len) {
   for (uint32_t i = 0; i < w->n_groups; i++) {
     grp *g = &w->groups[i];
-    if (g->key_len == key_len && memcmp(g->key, key, key_len) == 0) return g;
+    if (g->key_len == key_len && memcmp(g->key, key, (size_t)key_len + 1) == 0)
+      return g;
   }
   if (w->n_groups >= OBX_MAX_GROUPS) return NULL;
   grp *g = &w->groups[w->n_groups++];
   memset(g, 0, sizeof(*g));
-  memcpy(g->key, key, key_len);
+  memcpy(g->key, key, (size_t)key_len + 1);
   g->key_len = key_len;
   g->used = 1;
   return g;
@@ -211,14 +215,20 @@ static void *worker(void *arg) {
         obx__col_dec_row(decbuf + dsz * i, h, &bs->cols[need[i]], r, &vals[i],
                          &nulls[i]);
       }
-      /* group key */
-      uint8_t key[OBX_MAX_KEY_BYTES]; int klen = 0;
+      /* group key (+ null-flags byte at key[klen]) */
+      uint8_t key[OBX_MAX_KEY_BYTES]; int klen = 0; uint8_t nflags = 0;
       for (int i = 0; i < agg->n_group_cols; i++) {
         uint16_t c = agg->group_cols[i];
         int di = dec_of_col[c];
-        memcpy(key + klen, &vals[di], bs->cols[c].len);
+        if (nulls[di]) {
+          memset(key + klen, 0, bs->cols[c].len);
+          nflags |= (uint8_t)(1u << i);
+        } else {
+          memcpy(key + klen, &vals[di], bs->cols[c].len);
+        }
         klen += bs->cols[c].len;
       }
+      key[klen] = nflags;
       grp *g = find_group(w, key, (uint8_t)klen);
       if (!g) { w->rc = OBX_BUF_NOT_ENOUGH; free(decbuf); return NULL; }
       g->row_count++;
@@ -347,7 +357,8 @@ int obx_cpu_scan_filter_agg(const obx_blockset *bs,
     while (j > 0) {
       grp *p = &merged.groups[j - 1];
       int c = memcmp(p->key, tmp.key,
-                     p->key_len < tmp.key_len ? p->key_len : tmp.key_len);
+                     (size_t)(p->key_len < tmp.key_len ? p->key_len
+                                                       : tmp.key_len) + 1);
       if (c > 0 || (c == 0 && p->key_len > tmp.key_len)) {
         merged.groups[j] = *p; j--;
       } else break;
