@@ -20,6 +20,10 @@
 #include <algorithm>
 
 #include "obx_dev.h"
+
+#ifndef OBX_WG_HOST
+#define OBX_WG_HOST 256  /* must match the kernels' WG */
+#endif
 #include "../../include/obx.h"
 #include "../../oracle/obx_format.h"
 
@@ -422,8 +426,8 @@ static int prep_query(obx_gpu_ctx *ctx, obx_handle &h,
       h.bleaves_cap = needed;
     }
     uint32_t total = (uint32_t)needed;
-    uint32_t grid = (total + 255) / 256;
-    hipLaunchKernelGGL(k_lower_leaves, dim3(grid), dim3(256), 0, ctx->stream,
+    uint32_t grid = (total + OBX_WG_HOST - 1) / OBX_WG_HOST;
+    hipLaunchKernelGGL(k_lower_leaves, dim3(grid), dim3(OBX_WG_HOST), 0, ctx->stream,
                        h.d_buf, h.d_blocks, h.n_blocks, h.d_pleaves, nl,
                        h.d_bleaves);
   }
@@ -457,13 +461,13 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
   HIP_TRY(hipMemsetAsync(h.d_counters, 0, 16 * 8, ctx->stream));
   HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
   if (h.lds_ok)
-    hipLaunchKernelGGL(k_filter_lds, dim3(grid_for(h.n_blocks)), dim3(256), 0,
+    hipLaunchKernelGGL(k_filter_lds, dim3(grid_for(h.n_blocks)), dim3(OBX_WG_HOST), 0,
                        ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
                        h.d_pleaves, h.d_bleaves, ph, h.d_bitmap,
                        want_row_ids ? h.d_row_ids : nullptr,
                        want_row_ids ? h.d_blk_counts : nullptr, h.d_counters);
   else
-    hipLaunchKernelGGL(k_filter, dim3(grid_for(h.n_blocks)), dim3(256), 0,
+    hipLaunchKernelGGL(k_filter, dim3(grid_for(h.n_blocks)), dim3(OBX_WG_HOST), 0,
                        ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
                        h.d_pleaves, h.d_bleaves, ph, h.d_bitmap,
                        want_row_ids ? h.d_row_ids : nullptr,
@@ -530,7 +534,7 @@ extern "C" int obx_gpu_decode(obx_gpu_ctx *ctx, int handle,
     if (c >= h.n_cols) return OBX_INVALID_ARGUMENT;
     if (!h.d_decode_out[c])
       HIP_TRY(hipMalloc(&h.d_decode_out[c], h.total_rows * h.cols[c].len));
-    hipLaunchKernelGGL(k_decode, dim3(grid_for(h.n_blocks)), dim3(256), 0,
+    hipLaunchKernelGGL(k_decode, dim3(grid_for(h.n_blocks)), dim3(OBX_WG_HOST), 0,
                        ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
                        (uint32_t)c, (uint32_t)h.cols[c].len,
                        h.d_decode_out[c], (uint8_t *)nullptr);
@@ -583,12 +587,12 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
   HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
   if (h.lds_ok)
     hipLaunchKernelGGL(k_scan_filter_agg_lds, dim3(grid_for(h.n_blocks)),
-                       dim3(256), 0, ctx->stream, h.d_buf, h.d_blocks,
+                       dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf, h.d_blocks,
                        h.n_blocks, h.d_pleaves, h.d_bleaves, ph, h.d_gtable,
                        h.d_counters);
   else
     hipLaunchKernelGGL(k_scan_filter_agg, dim3(grid_for(h.n_blocks)),
-                       dim3(256), 0, ctx->stream, h.d_buf, h.d_blocks,
+                       dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf, h.d_blocks,
                        h.n_blocks, h.d_pleaves, h.d_bleaves, ph, h.d_gtable,
                        h.d_counters);
   HIP_TRY(hipEventRecord(ctx->ev_stop, ctx->stream));

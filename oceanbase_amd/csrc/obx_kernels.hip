@@ -556,7 +556,11 @@ __device__ __noinline__ uint64_t build_group_key(
 }
 
 /* ---------------- fused scan->filter->aggregate kernel ------------------ */
+#ifdef OBX_FAST_LEAVES_OVR
+#define OBX_FAST_LEAVES OBX_FAST_LEAVES_OVR
+#else
 #define OBX_FAST_LEAVES 4
+#endif
 #define OBX_FAST_NEED 8
 #define OBX_MAX_BLOCK_ROWS 2048
 
