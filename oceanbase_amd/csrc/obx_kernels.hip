@@ -40,7 +40,8 @@
  * reads subtract the bias. */
 struct blk_view {
   const uint8_t *base;
-  uint64_t bit_bias;
+  uint64_t bit_bias;   /* slow paths: absolute offsets minus this bias */
+  uint64_t rbase_bit;  /* ctx fast paths: block-relative offsets plus this */
 };
 
 /* ---------------- bit reads (base is >=16-B aligned) -------------------- */
@@ -121,10 +122,14 @@ __device__ __forceinline__ int64_t dict_entry(const blk_view &bv,
  * global byte loads of descriptor fields per row (SMEM scalarization is
  * blocked by the kernel's global stores). */
 struct col_ctx {
-  uint64_t data_bit;   /* packed stream start (absolute bits) */
-  uint64_t aux;        /* DICT: dict payload bit pos; INTDIFF/CONST: base */
-  uint64_t ext_bit;    /* ext-bit stream (HAS_EXT) */
+  /* offsets are BLOCK-RELATIVE bits (blocks are <= 17 KB, so 32 bits
+     suffice) — halves the uniform (SGPR) footprint of a context; the
+     102-SGPR budget was spilling to scratch inside the row loops */
+  uint32_t data_bit;   /* packed stream start (block-relative bits) */
+  uint32_t aux;        /* DICT: dict payload bit pos (block-relative) */
+  uint32_t ext_bit;    /* ext-bit stream (HAS_EXT, block-relative) */
   uint32_t count;      /* DICT: count (null ref >= count); else 0 */
+  int64_t base;        /* INTDIFF/CONST value */
   uint16_t W;          /* packed entry width in bits */
   uint8_t entry_bits;  /* DICT: dict entry width in bits */
   uint8_t sext_sh;     /* value: shift for sign extension (0 = none) */
@@ -134,10 +139,13 @@ struct col_ctx {
   uint8_t ext_w;
 };
 
-__device__ __forceinline__ col_ctx make_col_ctx(const dev_col &c) {
+__device__ __forceinline__ col_ctx make_col_ctx(const dev_col &c,
+                                                uint64_t blk_bit) {
   col_ctx x;
-  x.data_bit = c.data_bit;
-  x.aux = 0; x.ext_bit = c.ext_bit; x.count = 0;
+  x.data_bit = (uint32_t)(c.data_bit - blk_bit);
+  x.aux = 0; x.base = 0;
+  x.ext_bit = (uint32_t)(c.ext_bit - blk_bit);
+  x.count = 0;
   x.W = (c.flags & OBX_DF_BITPACK) ? c.width : (uint16_t)(c.width * 8);
   x.entry_bits = 0; x.sext_sh = 0; x.ent_sh = 0;
   x.has_ext = (c.flags & OBX_DF_HAS_EXT) ? 1 : 0;
@@ -153,7 +161,7 @@ __device__ __forceinline__ col_ctx make_col_ctx(const dev_col &c) {
     case OBX_D_DICT:
       x.kind = 1;
       x.count = c.count;
-      x.aux = c.dict_byte * 8;
+      x.aux = (uint32_t)(c.dict_byte * 8 - blk_bit);
       x.entry_bits = (uint8_t)(c.entry_len * 8);
       if (!(c.flags & OBX_DF_STRING)) {
         uint32_t sb = (c.flags & OBX_DF_SIGNED) ? c.tss : c.entry_len;
@@ -162,12 +170,12 @@ __device__ __forceinline__ col_ctx make_col_ctx(const dev_col &c) {
       break;
     case OBX_D_INTDIFF:
       x.kind = 2;
-      x.aux = (uint64_t)c.base;
+      x.base = c.base;
       break;
     case OBX_D_CONST:
       if (c.runs == 0) {
         x.kind = 3;
-        x.aux = (uint64_t)c.base;
+        x.base = c.base;
         x.count = c.count; /* 0 = null const */
       } else {
         x.kind = 4;
@@ -186,31 +194,40 @@ __device__ __forceinline__ int64_t ctx_value(const blk_view &bv,
   null_out = false;
   switch (x.kind) {
     case 0: { /* RAW */
-      if (x.has_ext &&
-          bit_read(bv, x.ext_bit + (uint64_t)r * x.ext_w, x.ext_w)) {
+      if (x.has_ext && bit_read_at(bv.base,
+                                   bv.rbase_bit + x.ext_bit +
+                                       (uint64_t)r * x.ext_w, x.ext_w)) {
         null_out = true; return 0;
       }
-      uint64_t v = bit_read(bv, x.data_bit + (uint64_t)r * x.W, x.W);
+      uint64_t v = bit_read_at(bv.base,
+                               bv.rbase_bit + x.data_bit + (uint64_t)r * x.W,
+                               x.W);
       return x.sext_sh ? (((int64_t)(v << x.sext_sh)) >> x.sext_sh)
                        : (int64_t)v;
     }
     case 1: { /* DICT */
-      uint64_t ref = bit_read(bv, x.data_bit + (uint64_t)r * x.W, x.W);
+      uint64_t ref = bit_read_at(
+          bv.base, bv.rbase_bit + x.data_bit + (uint64_t)r * x.W, x.W);
       if (ref >= x.count) { null_out = true; return 0; }
-      uint64_t v = bit_read(bv, x.aux + ref * x.entry_bits, x.entry_bits);
+      uint64_t v = bit_read_at(bv.base,
+                               bv.rbase_bit + x.aux + ref * x.entry_bits,
+                               x.entry_bits);
       return x.ent_sh ? (((int64_t)(v << x.ent_sh)) >> x.ent_sh) : (int64_t)v;
     }
     case 2: { /* INTDIFF */
-      if (x.has_ext &&
-          bit_read(bv, x.ext_bit + (uint64_t)r * x.ext_w, x.ext_w)) {
+      if (x.has_ext && bit_read_at(bv.base,
+                                   bv.rbase_bit + x.ext_bit +
+                                       (uint64_t)r * x.ext_w, x.ext_w)) {
         null_out = true; return 0;
       }
-      return (int64_t)(x.aux + bit_read(bv, x.data_bit + (uint64_t)r * x.W,
-                                        x.W));
+      return (int64_t)((uint64_t)x.base +
+                       bit_read_at(bv.base,
+                                   bv.rbase_bit + x.data_bit +
+                                       (uint64_t)r * x.W, x.W));
     }
     case 3: { /* CONST, no exceptions */
       if (x.count == 0) { null_out = true; return 0; }
-      return (int64_t)x.aux;
+      return x.base;
     }
     default:
       return 0; /* caller uses the slow path for kind 4 */
@@ -220,19 +237,18 @@ __device__ __forceinline__ int64_t ctx_value(const blk_view &bv,
 /* per-block filter-leaf context (registers) */
 struct leaf_ctx {
   col_ctx cc;              /* decode context of the filter column */
-  uint64_t mask, lo, hi;   /* REF_MASK / packed RANGE */
-  int64_t vlo, vhi;        /* VALUE-mode operands */
+  uint64_t mask, lo, hi;   /* REF_MASK / packed RANGE (mask = xor bias) */
   uint8_t mode, invert, op, slow;
 };
 
 __device__ __forceinline__ leaf_ctx make_leaf_ctx(const dev_block &blk,
                                                   const dev_leaf &plf,
-                                                  const blk_leaf &blf) {
+                                                  const blk_leaf &blf,
+                                                  uint64_t blk_bit) {
   const dev_col &c = blk.cols[plf.col];
   leaf_ctx x;
-  x.cc = make_col_ctx(c);
+  x.cc = make_col_ctx(c, blk_bit);
   x.mask = blf.mask; x.lo = blf.lo; x.hi = blf.hi;
-  x.vlo = plf.vlo; x.vhi = plf.vhi;
   x.mode = blf.mode; x.invert = blf.invert; x.op = plf.op;
   /* slow fallback: RLE/exception decode, or IN lists (operand array stays
      in the global leaf) */
@@ -243,24 +259,29 @@ __device__ __forceinline__ leaf_ctx make_leaf_ctx(const dev_block &blk,
 
 __device__ __forceinline__ bool leaf_ctx_match(const blk_view &bv,
                                                const leaf_ctx &x,
+                                               const dev_leaf &plf,
                                                uint32_t r) {
   switch (x.mode) {
     case OBX_LEAF_NONE: return false;
     case OBX_LEAF_ALL:
       if (x.cc.has_ext &&
-          bit_read(bv, x.cc.ext_bit + (uint64_t)r * x.cc.ext_w, x.cc.ext_w))
+          bit_read_at(bv.base, bv.rbase_bit + x.cc.ext_bit +
+                                   (uint64_t)r * x.cc.ext_w, x.cc.ext_w))
         return false;
       return true;
     case OBX_LEAF_REF_MASK: {
-      uint64_t ref = bit_read(bv, x.cc.data_bit + (uint64_t)r * x.cc.W,
-                              x.cc.W);
+      uint64_t ref = bit_read_at(
+          bv.base, bv.rbase_bit + x.cc.data_bit + (uint64_t)r * x.cc.W,
+          x.cc.W);
       return (x.mask >> ref) & 1;
     }
     case OBX_LEAF_RANGE: {
       if (x.cc.has_ext &&
-          bit_read(bv, x.cc.ext_bit + (uint64_t)r * x.cc.ext_w, x.cc.ext_w))
+          bit_read_at(bv.base, bv.rbase_bit + x.cc.ext_bit +
+                                   (uint64_t)r * x.cc.ext_w, x.cc.ext_w))
         return false;
-      uint64_t v = bit_read(bv, x.cc.data_bit + (uint64_t)r * x.cc.W, x.cc.W)
+      uint64_t v = bit_read_at(bv.base, bv.rbase_bit + x.cc.data_bit +
+                                            (uint64_t)r * x.cc.W, x.cc.W)
                    ^ x.mask; /* signed domains are order-mapped by xor bias */
       bool in = (v - x.lo) <= (x.hi - x.lo);
       return in != (bool)x.invert;
@@ -270,20 +291,20 @@ __device__ __forceinline__ bool leaf_ctx_match(const blk_view &bv,
       (void)ctx_value(bv, x.cc, r, isn);
       return isn != (bool)x.invert;
     }
-    default: { /* VALUE (op != IN) */
+    default: { /* VALUE (op != IN); operands read from the plan leaf */
       bool isn;
       int64_t v = ctx_value(bv, x.cc, r, isn);
       if (x.op == 8) return isn;
       if (x.op == 9) return !isn;
       if (isn) return false;
       switch (x.op) {
-        case 0: return v == x.vlo;
-        case 1: return v <= x.vlo;
-        case 2: return v < x.vlo;
-        case 3: return v >= x.vlo;
-        case 4: return v > x.vlo;
-        case 5: return v != x.vlo;
-        case 6: return v >= x.vlo && v <= x.vhi;
+        case 0: return v == plf.vlo;
+        case 1: return v <= plf.vlo;
+        case 2: return v < plf.vlo;
+        case 3: return v >= plf.vlo;
+        case 4: return v > plf.vlo;
+        case 5: return v != plf.vlo;
+        case 6: return v >= plf.vlo && v <= plf.vhi;
       }
       return false;
     }
@@ -616,10 +637,12 @@ __device__ void scan_filter_agg_body(
       stage_wait();
 #endif
     }
+    const uint64_t blk_bit = cur.block_byte * 8;
     blk_view bv;
     bv.base = STAGE ? lds_blk + (OBX_PIPELINE ? par * OBX_LDS_STAGE_BYTES : 0)
                     : buf;
-    bv.bit_bias = STAGE ? cur.block_byte * 8 : 0;
+    bv.bit_bias = STAGE ? blk_bit : 0;
+    bv.rbase_bit = STAGE ? 0 : blk_bit;
     par ^= 1;
 
     const uint32_t all_rows = cur.row_count;
@@ -636,7 +659,7 @@ __device__ void scan_filter_agg_body(
 #pragma unroll
       for (uint32_t i = 0; i < OBX_FAST_LEAVES; i++) {
         if (i < ph.n_leaves) {
-          lcs[i] = make_leaf_ctx(cur, plan_leaves[i], bl[i]);
+          lcs[i] = make_leaf_ctx(cur, plan_leaves[i], bl[i], blk_bit);
           if (lcs[i].slow) slow_set |= 1u << i;
         }
       }
@@ -650,7 +673,7 @@ __device__ void scan_filter_agg_body(
             if (slow_set & (1u << i))
               pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
             else
-              pass = leaf_ctx_match(bv, lcs[i], r);
+              pass = leaf_ctx_match(bv, lcs[i], plan_leaves[i], r);
           }
         }
         for (uint32_t i = OBX_FAST_LEAVES; i < ph.n_leaves && pass; i++)
@@ -676,11 +699,11 @@ __device__ void scan_filter_agg_body(
       const dev_col *gd0 = nullptr, *gd1 = nullptr;
       if (ph.n_group_cols > 0) {
         gd0 = &cur.cols[ph.need_cols[ph.group_idx[0]]];
-        g0 = make_col_ctx(*gd0);
+        g0 = make_col_ctx(*gd0, blk_bit);
       }
       if (ph.n_group_cols > 1) {
         gd1 = &cur.cols[ph.need_cols[ph.group_idx[1]]];
-        g1 = make_col_ctx(*gd1);
+        g1 = make_col_ctx(*gd1, blk_bit);
       }
       const uint32_t kl0 = ph.group_len[0], kl1 = ph.group_len[1];
       const uint32_t dim0 = (ph.n_group_cols > 0) ? g0.count + 1 : 1;
@@ -700,13 +723,15 @@ __device__ void scan_filter_agg_body(
           if (!m) { if (rr < rows) row_slot[rr] = 255; continue; }
           uint8_t slot8 = 255;
           if (pass) {
-            uint32_t ref0 = (uint32_t)bit_read(
-                bv, g0.data_bit + (uint64_t)r * g0.W, g0.W);
+            uint32_t ref0 = (uint32_t)bit_read_at(
+                bv.base, bv.rbase_bit + g0.data_bit + (uint64_t)r * g0.W,
+                g0.W);
             if (ref0 > g0.count) ref0 = g0.count; /* nope -> null bucket */
             uint32_t cell = ref0;
             if (ph.n_group_cols > 1) {
-              uint32_t ref1 = (uint32_t)bit_read(
-                  bv, g1.data_bit + (uint64_t)r * g1.W, g1.W);
+              uint32_t ref1 = (uint32_t)bit_read_at(
+                  bv.base, bv.rbase_bit + g1.data_bit + (uint64_t)r * g1.W,
+                  g1.W);
               if (ref1 > g1.count) ref1 = g1.count;
               cell += ref1 * dim0;
             }
@@ -769,16 +794,16 @@ __device__ void scan_filter_agg_body(
         const dev_col *da = nullptr, *db = nullptr, *dc2 = nullptr;
         if (ag.ia != 0xFF) {
           da = &cur.cols[ph.need_cols[ag.ia]];
-          ca = make_col_ctx(*da);
+          ca = make_col_ctx(*da, blk_bit);
         }
         if (ag.ib != 0xFF) {
           db = &cur.cols[ph.need_cols[ag.ib]];
-          cb = make_col_ctx(*db);
+          cb = make_col_ctx(*db, blk_bit);
         }
         uint16_t ic = fuse_p3 ? ag2.ic : ag.ic;
         if (ic != 0xFF && (ag.kind == 5 || fuse_p3)) {
           dc2 = &cur.cols[ph.need_cols[ic]];
-          cc = make_col_ctx(*dc2);
+          cc = make_col_ctx(*dc2, blk_bit);
         }
         const uint32_t stripe = lane & (OBX_STRIPES - 1);
 
@@ -1005,10 +1030,12 @@ __device__ void filter_body(
     } else {
       __syncthreads(); /* blk_written reset */
     }
+    const uint64_t blk_bit = cur.block_byte * 8;
     blk_view bv;
     bv.base = STAGE ? lds_blk + (OBX_PIPELINE ? par * OBX_LDS_STAGE_BYTES : 0)
                     : buf;
-    bv.bit_bias = STAGE ? cur.block_byte * 8 : 0;
+    bv.bit_bias = STAGE ? blk_bit : 0;
+    bv.rbase_bit = STAGE ? 0 : blk_bit;
     par ^= 1;
 
     const uint32_t rows = cur.row_count;
@@ -1021,7 +1048,7 @@ __device__ void filter_body(
 #pragma unroll
     for (uint32_t i = 0; i < OBX_FAST_LEAVES; i++) {
       if (i < ph.n_leaves) {
-        lcs[i] = make_leaf_ctx(cur, plan_leaves[i], bl[i]);
+        lcs[i] = make_leaf_ctx(cur, plan_leaves[i], bl[i], blk_bit);
         if (lcs[i].slow) slow_set |= 1u << i;
       }
     }
@@ -1035,7 +1062,7 @@ __device__ void filter_body(
           if (slow_set & (1u << i))
             pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
           else
-            pass = leaf_ctx_match(bv, lcs[i], r);
+            pass = leaf_ctx_match(bv, lcs[i], plan_leaves[i], r);
         }
       }
       for (uint32_t i = OBX_FAST_LEAVES; i < ph.n_leaves && pass; i++)
@@ -1118,7 +1145,7 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_decode(
   const uint32_t tid = threadIdx.x;
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
     const dev_block &cur = blocks[b];
-    blk_view bv; bv.base = buf; bv.bit_bias = 0;
+    blk_view bv; bv.base = buf; bv.bit_bias = 0; bv.rbase_bit = 0;
     const uint32_t rows = cur.row_count;
     const uint64_t row_start = dev_block_row_start(&cur);
     for (uint32_t r = tid; r < rows; r += WG) {
@@ -1198,7 +1225,7 @@ extern "C" __global__ void k_lower_leaves(
   uint32_t i = (uint32_t)(idx % n_leaves);
   const dev_leaf &lf = pl[i];
   const dev_col &c = blocks[b].cols[lf.col];
-  blk_view bv; bv.base = buf; bv.bit_bias = 0;
+  blk_view bv; bv.base = buf; bv.bit_bias = 0; bv.rbase_bit = 0;
   blk_leaf o;
   o.mask = 0; o.lo = 0; o.hi = 0; o.mode = OBX_LEAF_VALUE; o.invert = 0;
 
