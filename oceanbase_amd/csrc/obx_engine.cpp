@@ -452,6 +452,7 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
   uint64_t bm_words = (h.total_rows + 63) / 64 + 1;
   if (!h.d_bitmap) HIP_TRY(hipMalloc(&h.d_bitmap, bm_words * 8));
   HIP_TRY(hipMemsetAsync(h.d_bitmap, 0, bm_words * 8, ctx->stream));
+  const bool no_bitmap = getenv("OBX_NO_BITMAP") != nullptr; /* perf A/B */
   if (want_row_ids) {
     if (!h.d_row_ids)
       HIP_TRY(hipMalloc(&h.d_row_ids, h.total_rows * sizeof(int32_t)));
@@ -463,7 +464,8 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
   if (h.lds_ok)
     hipLaunchKernelGGL(k_filter_lds, dim3(grid_for(h.n_blocks)), dim3(OBX_WG_HOST), 0,
                        ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
-                       h.d_pleaves, h.d_bleaves, ph, h.d_bitmap,
+                       h.d_pleaves, h.d_bleaves, ph,
+                       no_bitmap ? nullptr : h.d_bitmap,
                        want_row_ids ? h.d_row_ids : nullptr,
                        want_row_ids ? h.d_blk_counts : nullptr, h.d_counters);
   else
