@@ -112,6 +112,9 @@ def main():
     ap.add_argument("--block-bytes", type=int, default=16384)
     ap.add_argument("--seed", type=int, default=42)
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--trace", action="store_true",
+                    help="per-step device-time table (prep/main kernel) — "
+                         "the per-operator monitor rebuild (SURVEY §5)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -172,13 +175,21 @@ def main():
         torch.cuda.synchronize() if torch.cuda.is_available() else None
 
     kernel_ms = []
+    prep_ms = []
     for _ in range(args.warmup):
         one_step()
     barrier_sync()
     t_start = time.time()
     for _ in range(args.steps):
+        t_step = time.time()
         one_step()
         kernel_ms.append(eng.last_kernel_ms())
+        prep_ms.append(eng.last_prep_ms())
+        if args.trace and rank == 0:
+            print(f"# step {len(kernel_ms)}: prep={prep_ms[-1]:.3f} ms "
+                  f"main_kernel={kernel_ms[-1]:.3f} ms "
+                  f"step_wall={(time.time()-t_step)*1e3:.3f} ms",
+                  file=sys.stderr)
     barrier_sync()
     elapsed = time.time() - t_start
 
@@ -257,7 +268,9 @@ def main():
                         block_bytes=args.block_bytes,
                         gen_seconds=round(gen_s, 2),
                         h2d_seconds=round(load_s, 2),
-                        kernel_ms_avg=round(avg_kms, 4)),
+                        kernel_ms_avg=round(avg_kms, 4),
+                        prep_ms_avg=round(sum(prep_ms) / max(len(prep_ms), 1),
+                                          4)),
             roofline=roofline,
             cpu_baseline=cpu_baseline,
         )

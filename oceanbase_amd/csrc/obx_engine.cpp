@@ -88,7 +88,9 @@ struct obx_gpu_ctx {
   int device = 0;
   hipStream_t stream = nullptr;
   hipEvent_t ev_start = nullptr, ev_stop = nullptr;
+  hipEvent_t ev_p0 = nullptr, ev_p1 = nullptr;
   double last_ms = 0.0;
+  double last_prep_ms = 0.0;  /* plan upload + per-block filter lowering */
   std::vector<obx_handle> handles;
 };
 
@@ -102,6 +104,8 @@ extern "C" int obx_gpu_open(int device, obx_gpu_ctx **out) {
   HIP_TRY(hipStreamCreate(&ctx->stream));
   HIP_TRY(hipEventCreate(&ctx->ev_start));
   HIP_TRY(hipEventCreate(&ctx->ev_stop));
+  HIP_TRY(hipEventCreate(&ctx->ev_p0));
+  HIP_TRY(hipEventCreate(&ctx->ev_p1));
   *out = ctx;
   return OBX_SUCCESS;
 }
@@ -416,6 +420,7 @@ static int prep_query(obx_gpu_ctx *ctx, obx_handle &h,
   }
 
   /* upload plan leaves + lower per-block tests on device */
+  HIP_TRY(hipEventRecord(ctx->ev_p0, ctx->stream));
   HIP_TRY(hipMemcpyAsync(h.d_pleaves, pl, sizeof(pl), hipMemcpyHostToDevice,
                          ctx->stream));
   if (nl > 0) {
@@ -431,6 +436,7 @@ static int prep_query(obx_gpu_ctx *ctx, obx_handle &h,
                        h.d_buf, h.d_blocks, h.n_blocks, h.d_pleaves, nl,
                        h.d_bleaves);
   }
+  HIP_TRY(hipEventRecord(ctx->ev_p1, ctx->stream));
   return OBX_SUCCESS;
 }
 
@@ -479,6 +485,8 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
   float ms = 0;
   HIP_TRY(hipEventElapsedTime(&ms, ctx->ev_start, ctx->ev_stop));
   ctx->last_ms = ms;
+  HIP_TRY(hipEventElapsedTime(&ms, ctx->ev_p0, ctx->ev_p1));
+  ctx->last_prep_ms = ms;
   unsigned long long cnt[2];
   HIP_TRY(hipMemcpy(cnt, h.d_counters, 16, hipMemcpyDeviceToHost));
   h.last_survivors = cnt[0];
@@ -602,6 +610,8 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
   float ms = 0;
   HIP_TRY(hipEventElapsedTime(&ms, ctx->ev_start, ctx->ev_stop));
   ctx->last_ms = ms;
+  HIP_TRY(hipEventElapsedTime(&ms, ctx->ev_p0, ctx->ev_p1));
+  ctx->last_prep_ms = ms;
 
   std::vector<gslot> gt(OBX_GTABLE_SLOTS);
   unsigned long long cnt[2];
@@ -654,6 +664,13 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
 
 extern "C" double obx_gpu_last_kernel_ms(obx_gpu_ctx *ctx) {
   return ctx ? ctx->last_ms : -1.0;
+}
+
+/* per-query monitoring (the rebuild of the reference's per-operator
+ * sql_plan_monitor / NG_TRACE stats): device times of the
+ * last query's prep (plan upload + k_lower_leaves) and main kernel. */
+extern "C" double obx_gpu_last_prep_ms(obx_gpu_ctx *ctx) {
+  return ctx ? ctx->last_prep_ms : -1.0;
 }
 
 extern "C" uint64_t obx_gpu_total_rows(obx_gpu_ctx *ctx, int handle) {

@@ -58,6 +58,8 @@ def _load():
                                             C.POINTER(abi.AggResult)]
     lib.obx_gpu_last_kernel_ms.restype = C.c_double
     lib.obx_gpu_last_kernel_ms.argtypes = [C.c_void_p]
+    lib.obx_gpu_last_prep_ms.restype = C.c_double
+    lib.obx_gpu_last_prep_ms.argtypes = [C.c_void_p]
     for f in ("obx_gpu_total_rows", "obx_gpu_total_bytes",
               "obx_gpu_last_survivors"):
         getattr(lib, f).restype = C.c_uint64
@@ -176,6 +178,11 @@ class GpuEngine:
 
     def last_kernel_ms(self):
         return float(self._lib.obx_gpu_last_kernel_ms(self._ctx))
+
+    def last_prep_ms(self):
+        """Device time of the last query's prep (plan upload + per-block
+        filter lowering) — per-operator monitoring, SURVEY §5."""
+        return float(self._lib.obx_gpu_last_prep_ms(self._ctx))
 
     def total_bytes(self, handle):
         return int(self._lib.obx_gpu_total_bytes(self._ctx, handle))
