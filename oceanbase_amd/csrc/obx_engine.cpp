@@ -45,6 +45,12 @@ extern "C" __global__ void k_filter_lds(
 extern "C" __global__ void k_decode(
     const uint8_t *, const dev_block *, uint32_t, uint32_t, uint32_t,
     uint8_t *, uint8_t *);
+extern "C" __global__ void k_group_pass(
+    const uint8_t *, const dev_block *, uint32_t, const dev_plan_hdr,
+    const uint64_t *, uint8_t *, gslot *, unsigned long long *);
+extern "C" __global__ void k_agg_pass(
+    const uint8_t *, const dev_block *, uint32_t, const dev_plan_hdr,
+    uint32_t, const uint64_t *, const uint8_t *, gslot *);
 extern "C" __global__ void k_lower_leaves(
     const uint8_t *, const dev_block *, uint32_t, const dev_leaf *, uint32_t,
     blk_leaf *);
@@ -78,6 +84,7 @@ struct obx_handle {
   uint64_t *d_bitmap = nullptr;
   int32_t *d_row_ids = nullptr;
   uint32_t *d_blk_counts = nullptr;
+  uint8_t *d_row_slot = nullptr;
   uint8_t *d_decode_out[OBX_DEV_MAX_COLS] = {};
   uint64_t last_survivors = 0;
   bool lds_ok = false;   /* all blocks 16-B aligned and <= LDS stage size */
@@ -118,6 +125,7 @@ extern "C" int obx_gpu_close(obx_gpu_ctx *ctx) {
     (void)hipFree(h.d_buf); (void)hipFree(h.d_blocks); (void)hipFree(h.d_bleaves);
     (void)hipFree(h.d_pleaves); (void)hipFree(h.d_gtable); (void)hipFree(h.d_counters);
     (void)hipFree(h.d_bitmap); (void)hipFree(h.d_row_ids); (void)hipFree(h.d_blk_counts);
+    (void)hipFree(h.d_row_slot);
     for (auto *p : h.d_decode_out) (void)hipFree(p);
   }
   (void)hipStreamDestroy(ctx->stream);
@@ -308,6 +316,7 @@ extern "C" int obx_gpu_free_blocks(obx_gpu_ctx *ctx, int handle) {
   (void)hipFree(h.d_buf); (void)hipFree(h.d_blocks); (void)hipFree(h.d_bleaves);
   (void)hipFree(h.d_pleaves); (void)hipFree(h.d_gtable); (void)hipFree(h.d_counters);
   (void)hipFree(h.d_bitmap); (void)hipFree(h.d_row_ids); (void)hipFree(h.d_blk_counts);
+  (void)hipFree(h.d_row_slot);
   for (auto *&p : h.d_decode_out) { (void)hipFree(p); p = nullptr; }
   h = obx_handle();
   return OBX_SUCCESS;
@@ -594,17 +603,59 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
                          hipMemcpyHostToDevice, ctx->stream));
   HIP_TRY(hipMemsetAsync(h.d_counters, 0, 16 * 8, ctx->stream));
 
-  HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
-  if (h.lds_ok)
-    hipLaunchKernelGGL(k_scan_filter_agg_lds, dim3(grid_for(h.n_blocks)),
+  /* Default: the fused single-kernel path (measured fastest in round 1).
+     OBX_PIPELINE_AGG=1 switches to the filter->group->agg-pass kernel DAG
+     (cleaner structure, currently slower — round-2 exploration; both paths
+     are parity-tested). */
+  const bool use_pipeline = getenv("OBX_PIPELINE_AGG") != nullptr;
+  if (use_pipeline) {
+    uint64_t bm_words = (h.total_rows + 63) / 64 + 1;
+    if (!h.d_bitmap) HIP_TRY(hipMalloc(&h.d_bitmap, bm_words * 8));
+    HIP_TRY(hipMemsetAsync(h.d_bitmap, 0, bm_words * 8, ctx->stream));
+    if (!h.d_row_slot) HIP_TRY(hipMalloc(&h.d_row_slot, h.total_rows + 1));
+
+    HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
+    if (h.lds_ok)
+      hipLaunchKernelGGL(k_filter_lds, dim3(grid_for(h.n_blocks)),
+                         dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf,
+                         h.d_blocks, h.n_blocks, h.d_pleaves, h.d_bleaves, ph,
+                         h.d_bitmap, (int32_t *)nullptr, (uint32_t *)nullptr,
+                         h.d_counters);
+    else
+      hipLaunchKernelGGL(k_filter, dim3(grid_for(h.n_blocks)),
+                         dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf,
+                         h.d_blocks, h.n_blocks, h.d_pleaves, h.d_bleaves, ph,
+                         h.d_bitmap, (int32_t *)nullptr, (uint32_t *)nullptr,
+                         h.d_counters);
+    hipLaunchKernelGGL(k_group_pass, dim3(grid_for(h.n_blocks)),
                        dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf, h.d_blocks,
-                       h.n_blocks, h.d_pleaves, h.d_bleaves, ph, h.d_gtable,
+                       h.n_blocks, ph, h.d_bitmap, h.d_row_slot, h.d_gtable,
                        h.d_counters);
-  else
-    hipLaunchKernelGGL(k_scan_filter_agg, dim3(grid_for(h.n_blocks)),
-                       dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf, h.d_blocks,
-                       h.n_blocks, h.d_pleaves, h.d_bleaves, ph, h.d_gtable,
-                       h.d_counters);
+    for (uint32_t a = 0; a < ph.n_aggs;) {
+      const dev_agg &ag = ph.aggs[a];
+      if (ag.kind == OBX_AGG_COUNT && ag.ia == 0xFF) { a++; continue; }
+      bool fuse = (ag.kind == OBX_AGG_SUM_PROD2 && a + 1 < ph.n_aggs &&
+                   ph.aggs[a + 1].kind == OBX_AGG_SUM_PROD3 &&
+                   ph.aggs[a + 1].ia == ag.ia && ph.aggs[a + 1].ib == ag.ib);
+      hipLaunchKernelGGL(k_agg_pass, dim3(grid_for(h.n_blocks)),
+                         dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf,
+                         h.d_blocks, h.n_blocks, ph, a, h.d_bitmap,
+                         h.d_row_slot, h.d_gtable);
+      a += fuse ? 2 : 1;
+    }
+  } else {
+    HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
+    if (h.lds_ok)
+      hipLaunchKernelGGL(k_scan_filter_agg_lds, dim3(grid_for(h.n_blocks)),
+                         dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf,
+                         h.d_blocks, h.n_blocks, h.d_pleaves, h.d_bleaves, ph,
+                         h.d_gtable, h.d_counters);
+    else
+      hipLaunchKernelGGL(k_scan_filter_agg, dim3(grid_for(h.n_blocks)),
+                         dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf,
+                         h.d_blocks, h.n_blocks, h.d_pleaves, h.d_bleaves, ph,
+                         h.d_gtable, h.d_counters);
+  }
   HIP_TRY(hipEventRecord(ctx->ev_stop, ctx->stream));
   HIP_TRY(hipStreamSynchronize(ctx->stream));
   float ms = 0;
@@ -647,7 +698,11 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
     uint32_t na = agg ? agg->n_aggs : 0;
     for (uint32_t a = 0; a < na; a++) {
       uint8_t kind = agg->aggs[a].kind;
-      if (kind == OBX_AGG_MIN || kind == OBX_AGG_MAX) {
+      if (kind == OBX_AGG_COUNT && agg->aggs[a].col_a == UINT16_MAX) {
+        /* COUNT(*) = the group's row count (group pass) */
+        g->cells[a].limb[0] = live[i]->count;
+        g->cells[a].limb[1] = g->cells[a].limb[2] = g->cells[a].limb[3] = 0;
+      } else if (kind == OBX_AGG_MIN || kind == OBX_AGG_MAX) {
         /* sign-extend the int64 min/max into the 256-bit cell */
         int64_t v = (int64_t)live[i]->cells[a][0];
         g->cells[a].limb[0] = (uint64_t)v;

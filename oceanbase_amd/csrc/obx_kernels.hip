@@ -1291,3 +1291,281 @@ extern "C" __global__ void k_lower_leaves(
   }
   out[idx] = o;
 }
+
+/* ======================================================================
+ * Pass-pipeline aggregation (round-1 final design).
+ *
+ * The one-kernel fused design pays SGPR-spill scratch traffic and staging
+ * barriers (profiles/r01_q1_sf100.md). The query instead runs as a short
+ * kernel DAG — each pass small and register-lean, reading only its own
+ * column streams directly from HBM (consecutive rows of one column are
+ * contiguous -> coalesced; dict payloads stay hot in L1/L2):
+ *
+ *   k_filter(_lds)  -> survivor bitmap (1 bit/row)   [existing kernel]
+ *   k_group_pass    -> row_slot[] (u8/row), global group keys + counts
+ *   k_agg_pass x P  -> one aggregate (or a fused PROD2+PROD3 pair) per
+ *                      pass; per-slot LDS cells, one global flush per WG
+ *
+ * Kernel boundaries are the inter-pass synchronization (~1.5 us each).
+ * Slots are indices into the global group table (gslot); row_slot stores
+ * them as u8 (255 = filtered out or table overflow; overflow counted).
+ * ====================================================================== */
+
+/* find-or-insert a key in the global group table; returns slot or -1 */
+__device__ __forceinline__ int g_slot(gslot *gtable, uint64_t key) {
+  uint32_t idx = (uint32_t)((key * 0x9E3779B97F4A7C15ull) >> 56) &
+                 (OBX_GTABLE_SLOTS - 1);
+  for (int probe = 0; probe < OBX_GTABLE_SLOTS; probe++) {
+    unsigned long long k = __hip_atomic_load(
+        &gtable[idx].key, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    if (k == key) return (int)idx;
+    if (k == OBX_KEY_EMPTY) {
+      unsigned long long cur = atomicCAS(&gtable[idx].key, OBX_KEY_EMPTY,
+                                         (unsigned long long)key);
+      if (cur == OBX_KEY_EMPTY || cur == key) return (int)idx;
+    }
+    idx = (idx + 1) & (OBX_GTABLE_SLOTS - 1);
+  }
+  return -1;
+}
+
+/* this row's bit from the global survivor bitmap */
+__device__ __forceinline__ bool pass_bit(const uint64_t *__restrict__ bitmap,
+                                         uint64_t grow) {
+  return (bitmap[grow >> 6] >> (grow & 63)) & 1;
+}
+
+/* ---- group pass: map each surviving row to a global-table slot ---- */
+extern "C" __global__ __launch_bounds__(WG, 2) void k_group_pass(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, const dev_plan_hdr ph,
+    const uint64_t *__restrict__ bitmap, uint8_t *__restrict__ row_slot,
+    gslot *__restrict__ gtable, unsigned long long *__restrict__ counters) {
+  __shared__ uint8_t cell_slot[64];
+  __shared__ unsigned long long cnt[OBX_GTABLE_SLOTS][4]; /* lane-striped */
+  const uint32_t tid = threadIdx.x;
+  const uint32_t lane = tid & 63;
+
+  for (uint32_t i = tid; i < OBX_GTABLE_SLOTS; i += WG) {
+    cnt[i][0] = 0; cnt[i][1] = 0; cnt[i][2] = 0; cnt[i][3] = 0;
+  }
+  __syncthreads();
+
+  for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
+    const dev_block &cur = blocks[b];
+    const uint64_t blk_bit = cur.block_byte * 8;
+    blk_view bv;
+    bv.base = buf; bv.bit_bias = 0; bv.rbase_bit = blk_bit;
+    const uint32_t rows = cur.row_count;
+    const uint64_t row_start = dev_block_row_start(&cur);
+
+    col_ctx g0, g1;
+    const dev_col *gd0 = nullptr, *gd1 = nullptr;
+    if (ph.n_group_cols > 0) {
+      gd0 = &cur.cols[ph.need_cols[ph.group_idx[0]]];
+      g0 = make_col_ctx(*gd0, blk_bit);
+    }
+    if (ph.n_group_cols > 1) {
+      gd1 = &cur.cols[ph.need_cols[ph.group_idx[1]]];
+      g1 = make_col_ctx(*gd1, blk_bit);
+    }
+    const uint32_t kl0 = ph.group_len[0], kl1 = ph.group_len[1];
+    const uint32_t dim0 = (ph.n_group_cols > 0) ? g0.count + 1 : 1;
+    const uint32_t dim1 = (ph.n_group_cols > 1) ? g1.count + 1 : 1;
+    const bool ref_fast = ph.n_group_cols > 0 && g0.kind == 1 &&
+                          (ph.n_group_cols < 2 || g1.kind == 1) &&
+                          dim0 * dim1 <= 64;
+    if (ref_fast) {
+      __syncthreads(); /* previous block's cell reads drain */
+      for (uint32_t i = tid; i < 64; i += WG) cell_slot[i] = 254;
+      __syncthreads();
+    }
+
+    const uint32_t iters = (rows + WG - 1) / WG;
+    for (uint32_t it = 0; it < iters; it++) {
+      uint32_t r = it * WG + tid;
+      bool in = r < rows;
+      bool pass = in && pass_bit(bitmap, row_start + r);
+      uint8_t slot8 = 255;
+      if (pass) {
+        if (ref_fast) {
+          uint32_t ref0 = (uint32_t)bit_read_at(
+              bv.base, bv.rbase_bit + g0.data_bit + (uint64_t)r * g0.W,
+              g0.W);
+          if (ref0 > g0.count) ref0 = g0.count;
+          uint32_t cell = ref0;
+          if (ph.n_group_cols > 1) {
+            uint32_t ref1 = (uint32_t)bit_read_at(
+                bv.base, bv.rbase_bit + g1.data_bit + (uint64_t)r * g1.W,
+                g1.W);
+            if (ref1 > g1.count) ref1 = g1.count;
+            cell += ref1 * dim0;
+          }
+          uint8_t cs = cell_slot[cell];
+          if (cs == 254) {
+            int s = g_slot(gtable, build_group_key(bv, ph.n_group_cols, g0,
+                                                   g1, gd0, gd1, kl0, kl1,
+                                                   r));
+            cs = (s < 0 || s >= 255) ? 255 : (uint8_t)s;
+            if (cs == 255) atomicAdd(&counters[1], 1ull);
+            cell_slot[cell] = cs;
+          }
+          slot8 = cs;
+        } else {
+          int s = g_slot(gtable, build_group_key(bv, ph.n_group_cols, g0, g1,
+                                                 gd0, gd1, kl0, kl1, r));
+          if (s < 0 || s >= 255) {
+            atomicAdd(&counters[1], 1ull);
+          } else {
+            slot8 = (uint8_t)s;
+          }
+        }
+        if (slot8 != 255) atomicAdd(&cnt[slot8][lane & 3], 1ull);
+      }
+      if (in) row_slot[row_start + r] = slot8;
+    }
+  }
+  __syncthreads();
+  for (uint32_t s = tid; s < OBX_GTABLE_SLOTS; s += WG) {
+    unsigned long long c = cnt[s][0] + cnt[s][1] + cnt[s][2] + cnt[s][3];
+    if (c) atomicAdd(&gtable[s].count, c);
+  }
+}
+
+/* ---- aggregate pass: one aggregate (or a fused PROD2+PROD3 pair) ---- */
+extern "C" __global__ __launch_bounds__(WG, 2) void k_agg_pass(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, const dev_plan_hdr ph, uint32_t a,
+    const uint64_t *__restrict__ bitmap, const uint8_t *__restrict__ row_slot,
+    gslot *__restrict__ gtable) {
+  /* per-slot LDS cells for this pass's aggregate(s): [slot][stripe][2] */
+  __shared__ unsigned long long cells[2][OBX_GTABLE_SLOTS][4][2];
+  const uint32_t tid = threadIdx.x;
+  const uint32_t lane = tid & 63;
+  const uint32_t stripe = lane & 3;
+
+  const dev_agg ag = ph.aggs[a];
+  const bool fuse_p3 =
+      (ag.kind == 4 && a + 1 < ph.n_aggs && ph.aggs[a + 1].kind == 5 &&
+       ph.aggs[a + 1].ia == ag.ia && ph.aggs[a + 1].ib == ag.ib);
+  const dev_agg ag2 = fuse_p3 ? ph.aggs[a + 1] : ag;
+  const bool is_minmax = (ag.kind == 2 || ag.kind == 3);
+
+  for (uint32_t s = tid; s < OBX_GTABLE_SLOTS; s += WG) {
+    for (uint32_t f = 0; f < 2; f++) {
+      for (uint32_t st = 0; st < 4; st++) {
+        cells[f][s][st][0] = (ag.kind == 2) ? (unsigned long long)INT64_MAX
+                             : (ag.kind == 3) ? (unsigned long long)INT64_MIN
+                                              : 0ull;
+        cells[f][s][st][1] = 0;
+      }
+    }
+  }
+  __syncthreads();
+
+  unsigned long long *c0 = &cells[0][0][0][0];
+  unsigned long long *c1 = &cells[1][0][0][0];
+
+  for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
+    const dev_block &cur = blocks[b];
+    const uint64_t blk_bit = cur.block_byte * 8;
+    blk_view bv;
+    bv.base = buf; bv.bit_bias = 0; bv.rbase_bit = blk_bit;
+    const uint32_t rows = cur.row_count;
+    const uint64_t row_start = dev_block_row_start(&cur);
+
+    col_ctx ca, cb, cc;
+    const dev_col *da = nullptr, *db = nullptr, *dc2 = nullptr;
+    if (ag.ia != 0xFF) {
+      da = &cur.cols[ph.need_cols[ag.ia]];
+      ca = make_col_ctx(*da, blk_bit);
+    }
+    if (ag.ib != 0xFF) {
+      db = &cur.cols[ph.need_cols[ag.ib]];
+      cb = make_col_ctx(*db, blk_bit);
+    }
+    uint16_t icx = fuse_p3 ? ag2.ic : ag.ic;
+    if (icx != 0xFF && (ag.kind == 5 || fuse_p3)) {
+      dc2 = &cur.cols[ph.need_cols[icx]];
+      cc = make_col_ctx(*dc2, blk_bit);
+    }
+
+    const uint32_t iters = (rows + WG - 1) / WG;
+    for (uint32_t it = 0; it < iters; it++) {
+      uint32_t r = it * WG + tid;
+      if (r >= rows) continue;
+      uint8_t slot8 = row_slot[row_start + r];
+      if (slot8 == 255) continue;
+      uint32_t idx = ((uint32_t)slot8 * 4 + stripe) * 2;
+      bool na = false, nb = false, nc = false;
+      int64_t va = 0, vb = 0, vc = 0;
+      if (ag.ia != 0xFF)
+        va = (ca.kind == 4) ? col_value(bv, *da, r, na)
+                            : ctx_value(bv, ca, r, na);
+      if (ag.ib != 0xFF)
+        vb = (cb.kind == 4) ? col_value(bv, *db, r, nb)
+                            : ctx_value(bv, cb, r, nb);
+      if (dc2)
+        vc = (cc.kind == 4) ? col_value(bv, *dc2, r, nc)
+                            : ctx_value(bv, cc, r, nc);
+      switch (ag.kind) {
+        case 0: /* COUNT(col) */
+          if (!na) atomicAdd(&c0[idx], 1ull);
+          break;
+        case 1: /* SUM */
+          if (!na) lds_acc_i128(&c0[idx], i128_from_i64(va));
+          break;
+        case 2: case 3: /* MIN/MAX */
+          if (!na) {
+            cas_minmax(&c0[idx], va, ag.kind == 2);
+            c0[idx + 1] = 1;
+          }
+          break;
+        case 4: /* SUM_PROD2 (+ fused PROD3 mate) */
+          if (!na && !nb) {
+            i128v p2 = i128_mul_i64(va, ag.one_b - vb);
+            lds_acc_i128(&c0[idx], p2);
+            if (fuse_p3 && !nc)
+              lds_acc_i128(&c1[idx], i128_mul_pos_i64(p2, ag2.one_c + vc));
+          }
+          break;
+        case 5: /* SUM_PROD3 standalone */
+          if (!na && !nb && !nc)
+            lds_acc_i128(&c0[idx],
+                         i128_mul_pos_i64(i128_mul_i64(va, ag.one_b - vb),
+                                          ag.one_c + vc));
+          break;
+        case 6: /* SUM_MUL */
+          if (!na && !nb) lds_acc_i128(&c0[idx], i128_mul_i64(va, vb));
+          break;
+        default: break;
+      }
+    }
+  }
+  __syncthreads();
+
+  /* flush: merge the two lane stripes, one global accumulate per slot */
+  for (uint32_t s = tid; s < OBX_GTABLE_SLOTS; s += WG) {
+    for (uint32_t f = 0; f < (fuse_p3 ? 2u : 1u); f++) {
+      uint32_t aa = a + f;
+      if (is_minmax) {
+        for (uint32_t st = 0; st < 4; st++) {
+          if (cells[f][s][st][1]) {
+            cas_minmax(&gtable[s].cells[aa][0],
+                       (int64_t)cells[f][s][st][0], ag.kind == 2);
+            gtable[s].cells[aa][1] = 1;
+          }
+        }
+      } else {
+        uint64_t lo = 0, hi = 0;
+        for (uint32_t st = 0; st < 4; st++) {
+          uint64_t l = cells[f][s][st][0];
+          uint64_t nlo = lo + l;
+          hi += cells[f][s][st][1] + (nlo < l);
+          lo = nlo;
+        }
+        if (lo | hi) g_acc_i128(gtable[s].cells[aa], lo, hi);
+      }
+    }
+  }
+}
