@@ -1,0 +1,54 @@
+"""C-ABI loadability (CPU-only contract check).
+
+The product library must load and export every entry point include/obx.h
+declares, without a GPU; compute calls must fail loudly (OBX_NO_GPU), never
+fall back to CPU.
+"""
+import ctypes as C
+import os
+import re
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _declared_functions():
+    hdr = open(os.path.join(REPO, "include", "obx.h")).read()
+    names = re.findall(r"^(?:int|int64_t|double|uint64_t)\s+(obx_\w+)\(",
+                       hdr, re.M)
+    assert len(names) >= 15
+    return names
+
+
+def test_engine_exports_every_declared_symbol():
+    so = os.path.join(REPO, "oceanbase_amd", "libobx.so")
+    if not os.path.exists(so):
+        import subprocess
+        subprocess.run(["bash", os.path.join(REPO, "oceanbase_amd", "csrc",
+                                             "build.sh")], check=True)
+    lib = C.CDLL(so)
+    gpu_syms = [n for n in _declared_functions() if n.startswith("obx_gpu")]
+    assert gpu_syms
+    for n in gpu_syms:
+        assert getattr(lib, n, None) is not None, f"missing export: {n}"
+
+
+def test_oracle_exports_cpu_symbols():
+    from oceanbase_amd import oracle  # builds liboracle.so if needed
+    lib = C.CDLL(os.path.join(REPO, "oracle", "liboracle.so"))
+    for n in _declared_functions():
+        if n.startswith("obx_gpu"):
+            continue
+        assert getattr(lib, n, None) is not None, f"missing export: {n}"
+    assert oracle is not None
+
+
+def test_product_refuses_without_gpu():
+    """On a GPU-less host, the product path raises — no CPU fallback."""
+    import torch
+    if torch.cuda.is_available():
+        pytest.skip("GPU present: refusal path not exercised here")
+    from oceanbase_amd.engine import GpuEngine, EngineUnavailable
+    with pytest.raises(EngineUnavailable):
+        GpuEngine(0)
