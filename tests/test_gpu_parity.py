@@ -285,3 +285,43 @@ def test_filter_ops_on_date_intdiff(eng):
         res = oracle.scan_filter_agg(li.bs, fd, None)
         assert survivors == res.rows_passed, op
     eng.free(h)
+
+
+def test_q1_parity_2m_rows(eng):
+    """Larger-scale parity: full oracle comparison at 2M rows plus the
+    size-independent invariants (sum of group counts == rows passed)."""
+    li, h = _load_lineitem(eng, 4, 2_000_000)
+    filt, agg = _q1_descs()
+    res_gpu = eng.scan_filter_agg(h, filt, agg)
+    res_cpu = oracle.scan_filter_agg(li.bs, filt, agg)
+    assert abi.result_rows(res_gpu, 6) == abi.result_rows(res_cpu, 6)
+    rows = abi.result_rows(res_gpu, 6)
+    assert sum(rc for _, rc, _ in rows) == res_gpu.rows_passed
+    assert all(cells[0] == rc for _, rc, cells in rows)  # COUNT(*) == rows
+    # sum_disc_price <= sum_charge (tax >= 0), qty>0 -> sums positive
+    for _, rc, cells in rows:
+        assert 0 < cells[3] <= cells[4]
+        assert cells[1] > 0 and cells[2] > 0
+    eng.free(h)
+
+
+def test_wide_rows_per_block(eng):
+    """Blocks near the row_count cap (>> the 2048-row kernel window)."""
+    rng = np.random.default_rng(17)
+    rows = 60000
+    vals = rng.integers(0, 1 << 40, rows, dtype=np.int64)
+    grp = rng.choice(np.frombuffer(b"PQ", dtype=np.uint8), rows)
+    schema = oracle.make_schema([(abi.T_INT, 0, 19, 8), (abi.T_CHAR, 0, 0, 1)])
+    block = oracle.encode_block(schema, [vals.view(np.uint8), grp],
+                                [abi.ENC_RAW, abi.ENC_DICT])
+    bs = _manual_blockset(schema, [block])
+    filt = abi.make_filter([dict(col=0, op=abi.OP_GE, lo=1 << 39)])
+    agg = abi.make_agg([1], [dict(kind=abi.AGG_SUM, col_a=0),
+                             dict(kind=abi.AGG_MIN, col_a=0),
+                             dict(kind=abi.AGG_MAX, col_a=0),
+                             dict(kind=abi.AGG_COUNT)])
+    res_cpu = oracle.scan_filter_agg(bs, filt, agg)
+    h = eng.load(bs)
+    res_gpu = eng.scan_filter_agg(h, filt, agg)
+    assert abi.result_rows(res_gpu, 4) == abi.result_rows(res_cpu, 4)
+    eng.free(h)
