@@ -624,6 +624,16 @@ __device__ void scan_filter_agg_body(
     stage_issue(buf, blocks[blockIdx.x], lds_blk);
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
     const dev_block &cur = blocks[b];
+    /* constant-result short-circuit: an AND leaf lowered to NONE means no
+       row of this block can pass — skip the block before staging (cf. the
+       reference's filter constant results / skip-index pruning) */
+    if (!OBX_PIPELINE || !STAGE) {
+      bool block_none = false;
+      for (uint32_t i = 0; i < ph.n_leaves; i++)
+        if (bleaves[(uint64_t)b * ph.n_leaves + i].mode == OBX_LEAF_NONE)
+          block_none = true;
+      if (block_none) continue;
+    }
     if (STAGE) {
 #if OBX_PIPELINE
       stage_wait();
@@ -652,35 +662,47 @@ __device__ void scan_filter_agg_body(
                               ? all_rows - w0 : OBX_MAX_BLOCK_ROWS;
     const uint32_t iters = (rows + WG - 1) / WG;
 
-    /* ---- phase 1: filter into the LDS pass bitmap ---- */
+    /* ---- phase 1: filter into the LDS pass bitmap, one leaf at a time
+       (the reference's per-leaf bitmap AND, ob_pushdown_filter.cpp:1559;
+       a single live leaf context keeps uniform state inside the SGPR
+       budget — the fused multi-leaf loop spilled). Blocks whose lowered
+       leaves are NONE are skipped before staging (constant-result
+       short-circuit; the ALL case skips the leaf's row pass). ---- */
     {
-      leaf_ctx lcs[OBX_FAST_LEAVES];
-      uint32_t slow_set = 0;
-#pragma unroll
-      for (uint32_t i = 0; i < OBX_FAST_LEAVES; i++) {
-        if (i < ph.n_leaves) {
-          lcs[i] = make_leaf_ctx(cur, plan_leaves[i], bl[i], blk_bit);
-          if (lcs[i].slow) slow_set |= 1u << i;
+      bool first = true;
+      for (uint32_t i = 0; i < ph.n_leaves; i++) {
+        const blk_leaf lfb = bl[i];
+        if (lfb.mode == OBX_LEAF_ALL &&
+            !(cur.cols[plan_leaves[i].col].flags & OBX_DF_HAS_EXT))
+          continue; /* every row passes this leaf */
+        leaf_ctx lc = make_leaf_ctx(cur, plan_leaves[i], bl[i], blk_bit);
+        const bool slow = lc.slow;
+        for (uint32_t it = 0; it < iters; it++) {
+          uint64_t prev = first ? ~0ull : pass_bm[it * WAVES + wv];
+          if (!prev) { if (first && lane == 0) pass_bm[it * WAVES + wv] = 0;
+                       continue; }
+          uint32_t rr = it * WG + tid;
+          uint32_t r = w0 + rr;
+          bool pass = rr < rows && ((prev >> lane) & 1);
+          if (pass)
+            pass = slow ? leaf_match(bv, cur, plan_leaves[i], bl[i], r)
+                        : leaf_ctx_match(bv, lc, plan_leaves[i], r);
+          uint64_t m = __ballot(pass);
+          if (lane == 0) pass_bm[it * WAVES + wv] = m;
+        }
+        first = false;
+      }
+      if (first) { /* no live leaves: all in-range rows pass */
+        for (uint32_t it = 0; it < iters; it++) {
+          uint32_t rr = it * WG + tid;
+          uint64_t m = __ballot(rr < rows);
+          if (lane == 0) pass_bm[it * WAVES + wv] = m;
         }
       }
+      /* rows-passed count (lane 0 of each wave over its own words) */
       for (uint32_t it = 0; it < iters; it++) {
-        uint32_t rr = it * WG + tid;
-        uint32_t r = w0 + rr;
-        bool pass = rr < rows;
-#pragma unroll
-        for (uint32_t i = 0; i < OBX_FAST_LEAVES; i++) {
-          if (i < ph.n_leaves && pass) {
-            if (slow_set & (1u << i))
-              pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
-            else
-              pass = leaf_ctx_match(bv, lcs[i], plan_leaves[i], r);
-          }
-        }
-        for (uint32_t i = OBX_FAST_LEAVES; i < ph.n_leaves && pass; i++)
-          pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
-        uint64_t m = __ballot(pass);
         if (lane == 0) {
-          pass_bm[it * WAVES + wv] = m;
+          uint64_t m = pass_bm[it * WAVES + wv];
           if (m) atomicAdd(&wg_passed, (unsigned long long)__popcll(m));
         }
       }
@@ -1002,6 +1024,7 @@ __device__ void filter_body(
   __shared__ uint32_t wv_scan[WAVES];
   __shared__ uint32_t blk_written;
   __shared__ unsigned long long wg_passed;
+  __shared__ uint64_t pass_bm[OBX_MAX_BLOCK_ROWS / 64];
 
   const uint32_t tid = threadIdx.x;
   const uint32_t lane = tid & 63;
@@ -1014,6 +1037,17 @@ __device__ void filter_body(
     stage_issue(buf, blocks[blockIdx.x], lds_blk);
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
     const dev_block &cur = blocks[b];
+    if (!OBX_PIPELINE || !STAGE) {
+      /* constant-result short-circuit (see scan_filter_agg_body) */
+      bool block_none = false;
+      for (uint32_t i = 0; i < ph.n_leaves; i++)
+        if (bleaves[(uint64_t)b * ph.n_leaves + i].mode == OBX_LEAF_NONE)
+          block_none = true;
+      if (block_none) {
+        if (row_ids && tid == 0) blk_counts[b] = 0;
+        continue; /* bitmap is pre-zeroed */
+      }
+    }
     if (tid == 0) blk_written = 0;
     if (STAGE) {
 #if OBX_PIPELINE
@@ -1038,39 +1072,54 @@ __device__ void filter_body(
     bv.rbase_bit = STAGE ? 0 : blk_bit;
     par ^= 1;
 
-    const uint32_t rows = cur.row_count;
+    const uint32_t all_rows = cur.row_count;
     const uint64_t row_start = dev_block_row_start(&cur);
-    const uint32_t iters = (rows + WG - 1) / WG;
     const blk_leaf *bl = bleaves + (uint64_t)b * ph.n_leaves;
+    for (uint32_t w0 = 0; w0 < all_rows; w0 += OBX_MAX_BLOCK_ROWS) {
+    const uint32_t rows = (all_rows - w0 < OBX_MAX_BLOCK_ROWS)
+                              ? all_rows - w0 : OBX_MAX_BLOCK_ROWS;
+    const uint32_t iters = (rows + WG - 1) / WG;
 
-    leaf_ctx lcs[OBX_FAST_LEAVES];
-    uint32_t slow_set = 0;
-#pragma unroll
-    for (uint32_t i = 0; i < OBX_FAST_LEAVES; i++) {
-      if (i < ph.n_leaves) {
-        lcs[i] = make_leaf_ctx(cur, plan_leaves[i], bl[i], blk_bit);
-        if (lcs[i].slow) slow_set |= 1u << i;
+    /* leaf-by-leaf fission into the pass bitmap (one live context) */
+    {
+      bool first = true;
+      for (uint32_t i = 0; i < ph.n_leaves; i++) {
+        const blk_leaf lfb = bl[i];
+        if (lfb.mode == OBX_LEAF_ALL &&
+            !(cur.cols[plan_leaves[i].col].flags & OBX_DF_HAS_EXT))
+          continue;
+        leaf_ctx lc = make_leaf_ctx(cur, plan_leaves[i], bl[i], blk_bit);
+        const bool slow = lc.slow;
+        for (uint32_t it = 0; it < iters; it++) {
+          uint64_t prev = first ? ~0ull : pass_bm[it * WAVES + wv];
+          if (!prev) { if (first && lane == 0) pass_bm[it * WAVES + wv] = 0;
+                       continue; }
+          uint32_t rr = it * WG + tid;
+          uint32_t r = w0 + rr;
+          bool pass = rr < rows && ((prev >> lane) & 1);
+          if (pass)
+            pass = slow ? leaf_match(bv, cur, plan_leaves[i], bl[i], r)
+                        : leaf_ctx_match(bv, lc, plan_leaves[i], r);
+          uint64_t m = __ballot(pass);
+          if (lane == 0) pass_bm[it * WAVES + wv] = m;
+        }
+        first = false;
+      }
+      if (first) {
+        for (uint32_t it = 0; it < iters; it++) {
+          uint32_t rr = it * WG + tid;
+          uint64_t m = __ballot(rr < rows);
+          if (lane == 0) pass_bm[it * WAVES + wv] = m;
+        }
       }
     }
 
+    /* output: global bitmap words, pass counts, selection vectors */
     for (uint32_t it = 0; it < iters; it++) {
-      uint32_t r = it * WG + tid;
-      bool pass = r < rows;
-#pragma unroll
-      for (uint32_t i = 0; i < OBX_FAST_LEAVES; i++) {
-        if (i < ph.n_leaves && pass) {
-          if (slow_set & (1u << i))
-            pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
-          else
-            pass = leaf_ctx_match(bv, lcs[i], plan_leaves[i], r);
-        }
-      }
-      for (uint32_t i = OBX_FAST_LEAVES; i < ph.n_leaves && pass; i++)
-        pass = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
-      uint64_t m = __ballot(pass);
-
+      uint64_t m = pass_bm[it * WAVES + wv];
       if (bitmap && lane == 0 && m) {
-        uint64_t gbit = row_start + (uint64_t)it * WG + (uint64_t)wv * 64;
+        uint64_t gbit = row_start + w0 + (uint64_t)it * WG +
+                        (uint64_t)wv * 64;
         uint64_t widx = gbit >> 6;
         uint32_t sh = (uint32_t)(gbit & 63);
         atomicOr((unsigned long long *)&bitmap[widx],
@@ -1079,7 +1128,6 @@ __device__ void filter_body(
           atomicOr((unsigned long long *)&bitmap[widx + 1],
                    (unsigned long long)(m >> (64 - sh)));
       }
-
       if (row_ids) {
         if (lane == 0) wv_cnt[wv] = (uint32_t)__popcll(m);
         __syncthreads();
@@ -1092,6 +1140,8 @@ __device__ void filter_body(
           blk_written = acc;
         }
         __syncthreads();
+        uint32_t r = w0 + it * WG + tid;
+        bool pass = (m >> lane) & 1;
         if (pass) {
           uint32_t below = __builtin_amdgcn_mbcnt_lo((uint32_t)m, 0);
           below = __builtin_amdgcn_mbcnt_hi((uint32_t)(m >> 32), below);
@@ -1102,6 +1152,8 @@ __device__ void filter_body(
         atomicAdd(&wg_passed, (unsigned long long)__popcll(m));
       }
     }
+    __syncthreads(); /* pass_bm reuse across windows */
+    } /* window loop */
     if (row_ids) {
       __syncthreads();
       if (tid == 0) {
