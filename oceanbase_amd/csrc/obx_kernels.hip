@@ -314,7 +314,7 @@ __device__ __forceinline__ bool leaf_ctx_match(const blk_view &bv,
 /* full value decode: int64 (sign-extended), is_null out.
  * __noinline__: slow/cold generic path — keeps register pressure off the
  * hot ctx_value loops. */
-__device__ __noinline__ int64_t col_value(const blk_view &bv,
+__device__ __noinline__ int64_t col_value(const blk_view bv,
                                              const dev_col &c, uint32_t r,
                                              bool &null_out) {
   null_out = false;
@@ -387,7 +387,7 @@ __device__ __forceinline__ bool leaf_value_match(const dev_leaf &lf, int64_t v,
   return false;
 }
 
-__device__ __noinline__ bool leaf_match(const blk_view &bv,
+__device__ __noinline__ bool leaf_match(const blk_view bv,
                                            const dev_block &blk,
                                            const dev_leaf &plf,
                                            const blk_leaf &blf, uint32_t r) {
@@ -551,21 +551,18 @@ __device__ __forceinline__ void stage_wait() {
  * of a dict-ref cell). __noinline__ keeps its register use off the hot
  * loops. NULL group values set flag bits in the top key byte. */
 __device__ __noinline__ uint64_t build_group_key(
-    const blk_view &bv, uint32_t n_group_cols, const col_ctx &g0,
-    const col_ctx &g1, const dev_col *gd0, const dev_col *gd1, uint32_t kl0,
-    uint32_t kl1, uint32_t r) {
+    const blk_view bv, uint32_t n_group_cols, const dev_col *gd0,
+    const dev_col *gd1, uint32_t kl0, uint32_t kl1, uint32_t r) {
   uint64_t key = 0;
   if (n_group_cols > 0) {
     bool isn;
-    int64_t kv = (g0.kind == 4) ? col_value(bv, *gd0, r, isn)
-                                : ctx_value(bv, g0, r, isn);
+    int64_t kv = col_value(bv, *gd0, r, isn);
     if (isn) key |= 1ull << 56;
     else key |= (uint64_t)kv &
                 ((kl0 >= 8) ? ~0ull : (((uint64_t)1 << (kl0 * 8)) - 1));
     if (n_group_cols > 1) {
       bool isn1;
-      int64_t kv1 = (g1.kind == 4) ? col_value(bv, *gd1, r, isn1)
-                                   : ctx_value(bv, g1, r, isn1);
+      int64_t kv1 = col_value(bv, *gd1, r, isn1);
       uint64_t kb = 0;
       if (isn1) key |= 1ull << 57;
       else kb = (uint64_t)kv1 &
@@ -760,8 +757,8 @@ __device__ void scan_filter_agg_body(
             uint8_t cs = cell_slot[cell];
             if (cs == 254) {
               int s = lds_slot(&tab, build_group_key(bv, ph.n_group_cols,
-                                                      g0, g1, gd0, gd1, kl0,
-                                                      kl1, r));
+                                                      gd0, gd1, kl0, kl1,
+                                                      r));
               if (s < 0) {
                 atomicAdd(&counters[1], 1ull);
                 cs = 255;
@@ -785,9 +782,8 @@ __device__ void scan_filter_agg_body(
           if (!m) { if (rr < rows) row_slot[rr] = 255; continue; }
           uint8_t slot8 = 255;
           if (pass) {
-            int s = lds_slot(&tab, build_group_key(bv, ph.n_group_cols, g0,
-                                                   g1, gd0, gd1, kl0, kl1,
-                                                   r));
+            int s = lds_slot(&tab, build_group_key(bv, ph.n_group_cols,
+                                                   gd0, gd1, kl0, kl1, r));
             if (s < 0) {
               atomicAdd(&counters[1], 1ull);
             } else {
@@ -1455,17 +1451,16 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_group_pass(
           }
           uint8_t cs = cell_slot[cell];
           if (cs == 254) {
-            int s = g_slot(gtable, build_group_key(bv, ph.n_group_cols, g0,
-                                                   g1, gd0, gd1, kl0, kl1,
-                                                   r));
+            int s = g_slot(gtable, build_group_key(bv, ph.n_group_cols,
+                                                   gd0, gd1, kl0, kl1, r));
             cs = (s < 0 || s >= 255) ? 255 : (uint8_t)s;
             if (cs == 255) atomicAdd(&counters[1], 1ull);
             cell_slot[cell] = cs;
           }
           slot8 = cs;
         } else {
-          int s = g_slot(gtable, build_group_key(bv, ph.n_group_cols, g0, g1,
-                                                 gd0, gd1, kl0, kl1, r));
+          int s = g_slot(gtable, build_group_key(bv, ph.n_group_cols, gd0,
+                                                 gd1, kl0, kl1, r));
           if (s < 0 || s >= 255) {
             atomicAdd(&counters[1], 1ull);
           } else {
