@@ -105,7 +105,9 @@ typedef struct dev_leaf {
   uint16_t col;
   uint8_t op;           /* obx_white_op for VALUE mode */
   uint8_t n_in;
-  uint8_t pad[4];
+  uint8_t char_len;     /* >0: char column — order-map values before compare
+                           (byte-lexicographic, like the oracle's char_key) */
+  uint8_t pad[3];
   int64_t in_list[8];   /* VALUE-mode IN */
 } dev_leaf;
 

@@ -341,6 +341,19 @@ static int prep_query(obx_gpu_ctx *ctx, obx_handle &h,
     pl[i].vhi = lf->hi;
     pl[i].n_in = lf->n_in;
     for (int j = 0; j < lf->n_in && j < 8; j++) pl[i].in_list[j] = lf->in_list[j];
+    /* char columns compare byte-lexicographically: order-map the operands
+       exactly like the oracle's char_key (low len LE bytes -> BE int) */
+    if (obx_store_class(h.cols[lf->col].obj_type) == OBX_SC_STRING) {
+      uint32_t len = h.cols[lf->col].len;
+      auto ck = [len](int64_t x) -> int64_t {
+        return (int64_t)(__builtin_bswap64((uint64_t)x) >> (8 * (8 - len)));
+      };
+      pl[i].char_len = (uint8_t)len;
+      pl[i].vlo = ck(lf->lo);
+      pl[i].vhi = ck(lf->hi);
+      for (int j = 0; j < lf->n_in && j < 8; j++)
+        pl[i].in_list[j] = ck(lf->in_list[j]);
+    }
   }
   /* resolve needed value slots */
   int slot_of_col[64];

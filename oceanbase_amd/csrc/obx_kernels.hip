@@ -67,6 +67,12 @@ __device__ __forceinline__ int64_t sext(uint64_t v, uint32_t bytes) {
   return ((int64_t)(v << sh)) >> sh;
 }
 
+/* byte-lexicographic order mapping for char values (== the oracle's
+ * char_key: low `len` LE bytes -> big-endian integer) */
+__device__ __forceinline__ int64_t dev_char_key(int64_t raw_le, uint32_t len) {
+  return (int64_t)(__builtin_bswap64((uint64_t)raw_le) >> (8 * (8 - len)));
+}
+
 __device__ __forceinline__ uint64_t shfl64(uint64_t v, int lane) {
   uint32_t lo = __shfl((int)(uint32_t)v, lane, 64);
   uint32_t hi = __shfl((int)(uint32_t)(v >> 32), lane, 64);
@@ -297,6 +303,7 @@ __device__ __forceinline__ bool leaf_ctx_match(const blk_view &bv,
       if (x.op == 8) return isn;
       if (x.op == 9) return !isn;
       if (isn) return false;
+      if (plf.char_len) v = dev_char_key(v, plf.char_len);
       switch (x.op) {
         case 0: return v == plf.vlo;
         case 1: return v <= plf.vlo;
@@ -370,6 +377,7 @@ __device__ __forceinline__ bool leaf_value_match(const dev_leaf &lf, int64_t v,
   if (lf.op == 8) return isn;
   if (lf.op == 9) return !isn;
   if (isn) return false;
+  if (lf.char_len) v = dev_char_key(v, lf.char_len);
   switch (lf.op) {
     case 0: return v == lf.vlo;
     case 1: return v <= lf.vlo;

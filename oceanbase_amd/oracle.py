@@ -95,9 +95,11 @@ def encode_block(schema, col_arrays, encodings, null_bitmaps=None):
             nulls[i] = (None if nb is None
                         else np.ascontiguousarray(nb).ctypes.data_as(C.c_void_p).value)
     encs = (C.c_uint8 * n_cols)(*encodings)
-    cap = 1024
+    cap = 4096
     for i in range(n_cols):
-        cap += rows * (schema[i].len + 2) + 1024
+        # worst case: forced RLE/DICT on high-cardinality data stores the
+        # runs/dict alongside full-width entries
+        cap += rows * (3 * schema[i].len + 10) + 4096
     out = np.zeros(cap, dtype=np.uint8)
     sz = _lib.obx_encode_block(schema, n_cols, ptrs, nulls, rows, encs,
                                out.ctypes.data_as(C.c_void_p), cap)
