@@ -145,10 +145,22 @@ typedef struct obx_filter_leaf {
   int64_t in_list[OBX_MAX_IN_LIST];
 } obx_filter_leaf;
 
+/* Filter combine program: the POD restatement of the executor tree's
+ * AND/OR combine (ObPushdownFilterExecutor::execute,
+ * ob_pushdown_filter.cpp:1559-1632). Postfix tokens over leaf bitmaps:
+ *   0..n_leaves-1 : push leaf i's result
+ *   OBX_TOK_AND   : pop two, push AND
+ *   OBX_TOK_OR    : pop two, push OR
+ * n_prog == 0 means AND of all leaves (the common conjunctive shape). */
+#define OBX_TOK_AND 128
+#define OBX_TOK_OR 129
+#define OBX_MAX_PROG 15
+
 typedef struct obx_filter_desc {
   uint16_t n_leaves;              /* 0 = no filter (all rows pass) */
-  obx_filter_leaf leaves[8];      /* AND-combined (Q1/Q6 shape);
-                                     OR/tree composition is a later round */
+  uint8_t n_prog;                 /* postfix program length (0 = AND-all) */
+  uint8_t prog[OBX_MAX_PROG];
+  obx_filter_leaf leaves[8];
 } obx_filter_desc;
 
 /* ---- aggregate descriptor ----------------------------------------------- */

@@ -38,8 +38,12 @@ class FilterLeaf(C.Structure):
                 ("in_list", C.c_int64 * 8)]
 
 
+TOK_AND, TOK_OR = 128, 129
+
+
 class FilterDesc(C.Structure):
-    _fields_ = [("n_leaves", C.c_uint16), ("leaves", FilterLeaf * 8)]
+    _fields_ = [("n_leaves", C.c_uint16), ("n_prog", C.c_uint8),
+                ("prog", C.c_uint8 * 15), ("leaves", FilterLeaf * 8)]
 
 
 class AggExpr(C.Structure):
@@ -75,10 +79,16 @@ class AggResult(C.Structure):
                 ("rows_scanned", C.c_uint64), ("rows_passed", C.c_uint64)]
 
 
-def make_filter(leaves):
-    """leaves: list of dicts {col, op, lo, hi?, in_list?}."""
+def make_filter(leaves, prog=None):
+    """leaves: list of dicts {col, op, lo, hi?, in_list?}; prog: optional
+    postfix combine program (ints: leaf index, TOK_AND, TOK_OR); None = AND
+    of all leaves."""
     f = FilterDesc()
     f.n_leaves = len(leaves)
+    if prog:
+        f.n_prog = len(prog)
+        for i, t in enumerate(prog):
+            f.prog[i] = t
     for i, lf in enumerate(leaves):
         f.leaves[i].col = lf["col"]
         f.leaves[i].op = lf["op"]

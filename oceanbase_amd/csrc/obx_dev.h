@@ -156,6 +156,8 @@ typedef struct dev_plan_hdr {
   uint32_t n_group_cols;
   uint32_t n_need;              /* decoded columns per surviving row */
   uint32_t n_passes;
+  uint8_t n_prog;               /* filter combine program (0 = AND-all) */
+  uint8_t prog[15];             /* postfix: leaf idx | 128=AND | 129=OR */
   uint16_t need_cols[OBX_DEV_MAX_NEED]; /* column index per val slot */
   uint8_t group_idx[2];         /* val-slot index of group cols */
   uint8_t group_len[2];         /* datum byte lens of group cols */

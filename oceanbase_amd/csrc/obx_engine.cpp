@@ -42,6 +42,23 @@ extern "C" __global__ void k_filter_lds(
     const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
     const blk_leaf *, const dev_plan_hdr, uint64_t *, int32_t *, uint32_t *,
     unsigned long long *);
+/* _prog variants: compiled with the postfix AND/OR combine-program path
+   (reference: ObPushdownFilterExecutor AND/OR trees); separate entry points
+   so the AND-only fast path keeps its lean register budget. */
+extern "C" __global__ void k_scan_filter_agg_prog(
+    const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
+    const blk_leaf *, const dev_plan_hdr, gslot *, unsigned long long *);
+extern "C" __global__ void k_scan_filter_agg_prog_lds(
+    const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
+    const blk_leaf *, const dev_plan_hdr, gslot *, unsigned long long *);
+extern "C" __global__ void k_filter_prog(
+    const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
+    const blk_leaf *, const dev_plan_hdr, uint64_t *, int32_t *, uint32_t *,
+    unsigned long long *);
+extern "C" __global__ void k_filter_prog_lds(
+    const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
+    const blk_leaf *, const dev_plan_hdr, uint64_t *, int32_t *, uint32_t *,
+    unsigned long long *);
 extern "C" __global__ void k_decode(
     const uint8_t *, const dev_block *, uint32_t, uint32_t, uint32_t,
     uint8_t *, uint8_t *);
@@ -332,6 +349,20 @@ static int prep_query(obx_gpu_ctx *ctx, obx_handle &h,
   uint16_t nl = filter ? filter->n_leaves : 0;
   if (nl > OBX_DEV_MAX_LEAVES) return OBX_INVALID_ARGUMENT;
   ph.n_leaves = nl;
+  if (filter && filter->n_prog) {
+    /* validate the combine program (operand range, stack discipline) */
+    int sp = 0;
+    for (int p = 0; p < filter->n_prog; p++) {
+      uint8_t t = filter->prog[p];
+      if (t < nl) sp++;
+      else if ((t == OBX_TOK_AND || t == OBX_TOK_OR) && sp >= 2) sp--;
+      else return OBX_INVALID_ARGUMENT;
+      if (sp > 8) return OBX_INVALID_ARGUMENT;
+    }
+    if (sp != 1) return OBX_INVALID_ARGUMENT;
+    ph.n_prog = filter->n_prog;
+    memcpy(ph.prog, filter->prog, filter->n_prog);
+  }
   for (uint16_t i = 0; i < nl; i++) {
     const obx_filter_leaf *lf = &filter->leaves[i];
     if (lf->col >= h.n_cols) return OBX_INVALID_ARGUMENT;
@@ -489,19 +520,16 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
   }
   HIP_TRY(hipMemsetAsync(h.d_counters, 0, 16 * 8, ctx->stream));
   HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
-  if (h.lds_ok)
-    hipLaunchKernelGGL(k_filter_lds, dim3(grid_for(h.n_blocks)), dim3(OBX_WG_HOST), 0,
+  {
+    auto kfn = h.lds_ok ? (ph.n_prog ? k_filter_prog_lds : k_filter_lds)
+                        : (ph.n_prog ? k_filter_prog : k_filter);
+    hipLaunchKernelGGL(kfn, dim3(grid_for(h.n_blocks)), dim3(OBX_WG_HOST), 0,
                        ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
                        h.d_pleaves, h.d_bleaves, ph,
-                       no_bitmap ? nullptr : h.d_bitmap,
+                       (h.lds_ok && no_bitmap) ? nullptr : h.d_bitmap,
                        want_row_ids ? h.d_row_ids : nullptr,
                        want_row_ids ? h.d_blk_counts : nullptr, h.d_counters);
-  else
-    hipLaunchKernelGGL(k_filter, dim3(grid_for(h.n_blocks)), dim3(OBX_WG_HOST), 0,
-                       ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
-                       h.d_pleaves, h.d_bleaves, ph, h.d_bitmap,
-                       want_row_ids ? h.d_row_ids : nullptr,
-                       want_row_ids ? h.d_blk_counts : nullptr, h.d_counters);
+  }
   HIP_TRY(hipEventRecord(ctx->ev_stop, ctx->stream));
   HIP_TRY(hipStreamSynchronize(ctx->stream));
   float ms = 0;
@@ -628,18 +656,15 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
     if (!h.d_row_slot) HIP_TRY(hipMalloc(&h.d_row_slot, h.total_rows + 1));
 
     HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
-    if (h.lds_ok)
-      hipLaunchKernelGGL(k_filter_lds, dim3(grid_for(h.n_blocks)),
-                         dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf,
-                         h.d_blocks, h.n_blocks, h.d_pleaves, h.d_bleaves, ph,
-                         h.d_bitmap, (int32_t *)nullptr, (uint32_t *)nullptr,
+    {
+      auto kfn = h.lds_ok ? (ph.n_prog ? k_filter_prog_lds : k_filter_lds)
+                          : (ph.n_prog ? k_filter_prog : k_filter);
+      hipLaunchKernelGGL(kfn, dim3(grid_for(h.n_blocks)), dim3(OBX_WG_HOST), 0,
+                         ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
+                         h.d_pleaves, h.d_bleaves, ph, h.d_bitmap,
+                         (int32_t *)nullptr, (uint32_t *)nullptr,
                          h.d_counters);
-    else
-      hipLaunchKernelGGL(k_filter, dim3(grid_for(h.n_blocks)),
-                         dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf,
-                         h.d_blocks, h.n_blocks, h.d_pleaves, h.d_bleaves, ph,
-                         h.d_bitmap, (int32_t *)nullptr, (uint32_t *)nullptr,
-                         h.d_counters);
+    }
     hipLaunchKernelGGL(k_group_pass, dim3(grid_for(h.n_blocks)),
                        dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf, h.d_blocks,
                        h.n_blocks, ph, h.d_bitmap, h.d_row_slot, h.d_gtable,
@@ -658,16 +683,16 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
     }
   } else {
     HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
-    if (h.lds_ok)
-      hipLaunchKernelGGL(k_scan_filter_agg_lds, dim3(grid_for(h.n_blocks)),
-                         dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf,
-                         h.d_blocks, h.n_blocks, h.d_pleaves, h.d_bleaves, ph,
-                         h.d_gtable, h.d_counters);
-    else
-      hipLaunchKernelGGL(k_scan_filter_agg, dim3(grid_for(h.n_blocks)),
-                         dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf,
-                         h.d_blocks, h.n_blocks, h.d_pleaves, h.d_bleaves, ph,
-                         h.d_gtable, h.d_counters);
+    {
+      auto kfn = h.lds_ok
+                     ? (ph.n_prog ? k_scan_filter_agg_prog_lds
+                                  : k_scan_filter_agg_lds)
+                     : (ph.n_prog ? k_scan_filter_agg_prog : k_scan_filter_agg);
+      hipLaunchKernelGGL(kfn, dim3(grid_for(h.n_blocks)), dim3(OBX_WG_HOST), 0,
+                         ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
+                         h.d_pleaves, h.d_bleaves, ph, h.d_gtable,
+                         h.d_counters);
+    }
   }
   HIP_TRY(hipEventRecord(ctx->ev_stop, ctx->stream));
   HIP_TRY(hipStreamSynchronize(ctx->stream));

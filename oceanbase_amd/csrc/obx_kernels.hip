@@ -590,7 +590,115 @@ __device__ __noinline__ uint64_t build_group_key(
 #define OBX_FAST_NEED 8
 #define OBX_MAX_BLOCK_ROWS 2048
 
-template <bool STAGE>
+/* three-valued fold of the filter combine program over per-block leaf
+ * classes (0 = never passes, 1 = always passes, 2 = row-dependent);
+ * implements the executor tree's constant-result short-circuit
+ * (ob_pushdown_filter.cpp:1559-1632) at block granularity. */
+__device__ __forceinline__ uint8_t leaf_class(const dev_block &cur,
+                                              const dev_leaf &plf,
+                                              const blk_leaf &blf) {
+  if (blf.mode == OBX_LEAF_NONE) return 0;
+  if (blf.mode == OBX_LEAF_ALL &&
+      !(cur.cols[plf.col].flags & OBX_DF_HAS_EXT))
+    return 1;
+  return 2;
+}
+
+__device__ __forceinline__ uint8_t fold_prog3(
+    const dev_plan_hdr &ph, const dev_block &cur,
+    const dev_leaf *__restrict__ pl, const blk_leaf *__restrict__ bl) {
+  if (ph.n_leaves == 0) return 1;
+  if (ph.n_prog == 0) { /* AND of all leaves */
+    uint8_t res = 1;
+    for (uint32_t i = 0; i < ph.n_leaves; i++) {
+      uint8_t c = leaf_class(cur, pl[i], bl[i]);
+      if (c == 0) return 0;
+      if (c == 2) res = 2;
+    }
+    return res;
+  }
+  uint8_t stack[16];
+  int sp = 0;
+  for (uint32_t p = 0; p < ph.n_prog; p++) {
+    uint8_t t = ph.prog[p];
+    if (t < ph.n_leaves) {
+      stack[sp++] = leaf_class(cur, pl[t], bl[t]);
+    } else {
+      uint8_t b2 = stack[--sp], a2 = stack[sp - 1];
+      if (t == 128) /* AND */
+        stack[sp - 1] = (a2 == 0 || b2 == 0) ? 0
+                        : (a2 == 1 && b2 == 1) ? 1 : 2;
+      else /* OR */
+        stack[sp - 1] = (a2 == 1 || b2 == 1) ? 1
+                        : (a2 == 0 && b2 == 0) ? 0 : 2;
+    }
+  }
+  return stack[0];
+}
+
+typedef __attribute__((address_space(3))) uint64_t lds3_u64;
+struct prog_spec { uint8_t n; uint8_t t[15]; };
+
+/* cold path: evaluate a combine-program filter for one row window.
+ * __noinline__ + by-value args keep its registers off the hot AND path
+ * (called once per block window only when the plan has a program). */
+__device__ __forceinline__ void filter_prog_phase(
+    const blk_view bv, const dev_block &cur,
+    const dev_leaf *__restrict__ plan_leaves,
+    const blk_leaf *__restrict__ bl, uint32_t n_leaves, prog_spec ps,
+    uint32_t rows, uint32_t w0, uint64_t blk_bit, lds3_u64 *pass_bm,
+    lds3_u64 *leaf_bm_flat) {
+  const uint32_t tid = threadIdx.x;
+  const uint32_t lane = tid & 63;
+  const uint32_t wv = tid >> 6;
+  const uint32_t iters = (rows + WG - 1) / WG;
+  uint8_t cls[8];
+  for (uint32_t i = 0; i < n_leaves; i++) {
+    cls[i] = leaf_class(cur, plan_leaves[i], bl[i]);
+    if (cls[i] != 2) continue;
+    leaf_ctx lc = make_leaf_ctx(cur, plan_leaves[i], bl[i], blk_bit);
+    const bool slow = lc.slow;
+    for (uint32_t it = 0; it < iters; it++) {
+      uint32_t rr = it * WG + tid;
+      uint32_t r = w0 + rr;
+      bool pass = rr < rows;
+      if (pass)
+        pass = slow ? leaf_match(bv, cur, plan_leaves[i], bl[i], r)
+                    : leaf_ctx_match(bv, lc, plan_leaves[i], r);
+      uint64_t m = __ballot(pass);
+      if (lane == 0)
+        leaf_bm_flat[i * (OBX_MAX_BLOCK_ROWS / 64) + it * WAVES + wv] = m;
+    }
+  }
+  __syncthreads();
+  for (uint32_t it = 0; it < iters; it++) {
+    uint32_t rr = it * WG + tid;
+    uint64_t vm = __ballot(rr < rows);
+    if (lane == 0) {
+      uint64_t stack[8];
+      int sp = 0;
+      for (uint32_t p = 0; p < ps.n; p++) {
+        uint8_t t = ps.t[p];
+        if (t < n_leaves) {
+          stack[sp++] =
+              cls[t] == 2
+                  ? leaf_bm_flat[t * (OBX_MAX_BLOCK_ROWS / 64) +
+                                 it * WAVES + wv]
+                  : cls[t] == 1 ? vm : 0;
+        } else if (t == 128) {
+          sp--; stack[sp - 1] &= stack[sp];
+        } else {
+          sp--; stack[sp - 1] |= stack[sp];
+        }
+      }
+      pass_bm[it * WAVES + wv] = stack[0] & vm;
+    }
+  }
+  __syncthreads();
+}
+
+
+template <bool STAGE, bool PROG>
 __device__ void scan_filter_agg_body(
     const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
     uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
@@ -599,7 +707,12 @@ __device__ void scan_filter_agg_body(
     uint8_t *lds_blk) {
   __shared__ lds_table tab;
   __shared__ uint64_t pass_bm[OBX_MAX_BLOCK_ROWS / 64];
-  __shared__ uint8_t row_slot[OBX_MAX_BLOCK_ROWS];
+  /* leaf_bm (phase 1, combine programs only) and row_slot (phases 2-3) are
+     live in disjoint phases — share the same 2 KB of LDS */
+  __shared__ union {
+    uint64_t leaf_bm[8][OBX_MAX_BLOCK_ROWS / 64];
+    uint8_t row_slot[OBX_MAX_BLOCK_ROWS];
+  } u;
   __shared__ uint8_t cell_slot[64];
   __shared__ unsigned long long wg_passed;
 
@@ -629,15 +742,14 @@ __device__ void scan_filter_agg_body(
     stage_issue(buf, blocks[blockIdx.x], lds_blk);
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
     const dev_block &cur = blocks[b];
-    /* constant-result short-circuit: an AND leaf lowered to NONE means no
-       row of this block can pass — skip the block before staging (cf. the
-       reference's filter constant results / skip-index pruning) */
+    /* constant-result short-circuit: fold the combine program over the
+       lowered leaf classes — a FALSE block is skipped before staging (cf.
+       the reference's filter constant results / skip-index pruning) */
+    uint8_t blk_verdict = 2;
     if (!OBX_PIPELINE || !STAGE) {
-      bool block_none = false;
-      for (uint32_t i = 0; i < ph.n_leaves; i++)
-        if (bleaves[(uint64_t)b * ph.n_leaves + i].mode == OBX_LEAF_NONE)
-          block_none = true;
-      if (block_none) continue;
+      blk_verdict = fold_prog3(ph, cur, plan_leaves,
+                               bleaves + (uint64_t)b * ph.n_leaves);
+      if (blk_verdict == 0) continue;
     }
     if (STAGE) {
 #if OBX_PIPELINE
@@ -668,12 +780,17 @@ __device__ void scan_filter_agg_body(
     const uint32_t iters = (rows + WG - 1) / WG;
 
     /* ---- phase 1: filter into the LDS pass bitmap, one leaf at a time
-       (the reference's per-leaf bitmap AND, ob_pushdown_filter.cpp:1559;
+       (the reference's per-leaf bitmap combine, ob_pushdown_filter.cpp:1559;
        a single live leaf context keeps uniform state inside the SGPR
-       budget — the fused multi-leaf loop spilled). Blocks whose lowered
-       leaves are NONE are skipped before staging (constant-result
-       short-circuit; the ALL case skips the leaf's row pass). ---- */
-    {
+       budget). AND-only plans fold progressively; plans with a combine
+       program evaluate per-leaf bitmaps then a word-wise postfix pass. ---- */
+    if (blk_verdict == 1) {
+      for (uint32_t it = 0; it < iters; it++) {
+        uint32_t rr = it * WG + tid;
+        uint64_t m = __ballot(rr < rows);
+        if (lane == 0) pass_bm[it * WAVES + wv] = m;
+      }
+    } else if constexpr (!PROG) {
       bool first = true;
       for (uint32_t i = 0; i < ph.n_leaves; i++) {
         const blk_leaf lfb = bl[i];
@@ -704,12 +821,19 @@ __device__ void scan_filter_agg_body(
           if (lane == 0) pass_bm[it * WAVES + wv] = m;
         }
       }
-      /* rows-passed count (lane 0 of each wave over its own words) */
-      for (uint32_t it = 0; it < iters; it++) {
-        if (lane == 0) {
-          uint64_t m = pass_bm[it * WAVES + wv];
-          if (m) atomicAdd(&wg_passed, (unsigned long long)__popcll(m));
-        }
+    } else {
+      prog_spec ps;
+      ps.n = ph.n_prog;
+      for (int p = 0; p < 15; p++) ps.t[p] = ph.prog[p];
+      filter_prog_phase(bv, cur, plan_leaves, bl, ph.n_leaves, ps, rows, w0,
+                        blk_bit, (lds3_u64 *)pass_bm,
+                        (lds3_u64 *)&u.leaf_bm[0][0]);
+    }
+    /* rows-passed count (lane 0 of each wave over its own words) */
+    for (uint32_t it = 0; it < iters; it++) {
+      if (lane == 0) {
+        uint64_t m = pass_bm[it * WAVES + wv];
+        if (m) atomicAdd(&wg_passed, (unsigned long long)__popcll(m));
       }
     }
     __syncthreads();
@@ -747,7 +871,7 @@ __device__ void scan_filter_agg_body(
           uint32_t r = w0 + rr;
           uint64_t m = pass_bm[it * WAVES + wv];
           bool pass = (m >> lane) & 1;
-          if (!m) { if (rr < rows) row_slot[rr] = 255; continue; }
+          if (!m) { if (rr < rows) u.row_slot[rr] = 255; continue; }
           uint8_t slot8 = 255;
           if (pass) {
             uint32_t ref0 = (uint32_t)bit_read_at(
@@ -779,7 +903,7 @@ __device__ void scan_filter_agg_body(
             if (slot8 != 255)
               atomicAdd(&tab.count[slot8][lane & (OBX_STRIPES - 1)], 1ull);
           }
-          if (rr < rows) row_slot[rr] = slot8;
+          if (rr < rows) u.row_slot[rr] = slot8;
         }
       } else {
         for (uint32_t it = 0; it < iters; it++) {
@@ -787,7 +911,7 @@ __device__ void scan_filter_agg_body(
           uint32_t r = w0 + rr;
           uint64_t m = pass_bm[it * WAVES + wv];
           bool pass = (m >> lane) & 1;
-          if (!m) { if (rr < rows) row_slot[rr] = 255; continue; }
+          if (!m) { if (rr < rows) u.row_slot[rr] = 255; continue; }
           uint8_t slot8 = 255;
           if (pass) {
             int s = lds_slot(&tab, build_group_key(bv, ph.n_group_cols,
@@ -799,7 +923,7 @@ __device__ void scan_filter_agg_body(
               atomicAdd(&tab.count[s][lane & (OBX_STRIPES - 1)], 1ull);
             }
           }
-          if (rr < rows) row_slot[rr] = slot8;
+          if (rr < rows) u.row_slot[rr] = slot8;
         }
       }
       __syncthreads();
@@ -840,7 +964,7 @@ __device__ void scan_filter_agg_body(
           uint64_t m = pass_bm[it * WAVES + wv];                        \
           if (!m) continue;                                             \
           bool pass = (m >> lane) & 1;                                  \
-          uint8_t slot8 = pass && rr < rows ? row_slot[rr] : 255;       \
+          uint8_t slot8 = pass && rr < rows ? u.row_slot[rr] : 255;       \
           if (slot8 == 255) continue;                                   \
           int s = slot8;                                                \
           (void)r; __VA_ARGS__                                          \
@@ -1001,8 +1125,17 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg(
     uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
     const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
     gslot *__restrict__ gtable, unsigned long long *__restrict__ counters) {
-  scan_filter_agg_body<false>(buf, blocks, n_blocks, plan_leaves, bleaves, ph,
-                              gtable, counters, nullptr);
+  scan_filter_agg_body<false, false>(buf, blocks, n_blocks, plan_leaves,
+                                     bleaves, ph, gtable, counters, nullptr);
+}
+
+extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg_prog(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
+    const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
+    gslot *__restrict__ gtable, unsigned long long *__restrict__ counters) {
+  scan_filter_agg_body<false, true>(buf, blocks, n_blocks, plan_leaves,
+                                    bleaves, ph, gtable, counters, nullptr);
 }
 
 extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg_lds(
@@ -1011,12 +1144,22 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg_lds(
     const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
     gslot *__restrict__ gtable, unsigned long long *__restrict__ counters) {
   __shared__ uint8_t lds_blk[(1 + OBX_PIPELINE) * OBX_LDS_STAGE_BYTES + 32];
-  scan_filter_agg_body<true>(buf, blocks, n_blocks, plan_leaves, bleaves, ph,
-                             gtable, counters, lds_blk);
+  scan_filter_agg_body<true, false>(buf, blocks, n_blocks, plan_leaves,
+                                    bleaves, ph, gtable, counters, lds_blk);
+}
+
+extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_filter_agg_prog_lds(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
+    const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
+    gslot *__restrict__ gtable, unsigned long long *__restrict__ counters) {
+  __shared__ uint8_t lds_blk[(1 + OBX_PIPELINE) * OBX_LDS_STAGE_BYTES + 32];
+  scan_filter_agg_body<true, true>(buf, blocks, n_blocks, plan_leaves,
+                                   bleaves, ph, gtable, counters, lds_blk);
 }
 
 /* ---------------- filter-only kernel (bitmap + selection vectors) ------- */
-template <bool STAGE>
+template <bool STAGE, bool PROG>
 __device__ void filter_body(
     const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
     uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
@@ -1029,6 +1172,7 @@ __device__ void filter_body(
   __shared__ uint32_t blk_written;
   __shared__ unsigned long long wg_passed;
   __shared__ uint64_t pass_bm[OBX_MAX_BLOCK_ROWS / 64];
+  __shared__ uint64_t leaf_bm[8][OBX_MAX_BLOCK_ROWS / 64];
 
   const uint32_t tid = threadIdx.x;
   const uint32_t lane = tid & 63;
@@ -1041,13 +1185,12 @@ __device__ void filter_body(
     stage_issue(buf, blocks[blockIdx.x], lds_blk);
   for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
     const dev_block &cur = blocks[b];
+    uint8_t blk_verdict = 2;
     if (!OBX_PIPELINE || !STAGE) {
       /* constant-result short-circuit (see scan_filter_agg_body) */
-      bool block_none = false;
-      for (uint32_t i = 0; i < ph.n_leaves; i++)
-        if (bleaves[(uint64_t)b * ph.n_leaves + i].mode == OBX_LEAF_NONE)
-          block_none = true;
-      if (block_none) {
+      blk_verdict = fold_prog3(ph, cur, plan_leaves,
+                               bleaves + (uint64_t)b * ph.n_leaves);
+      if (blk_verdict == 0) {
         if (row_ids && tid == 0) blk_counts[b] = 0;
         continue; /* bitmap is pre-zeroed */
       }
@@ -1084,8 +1227,15 @@ __device__ void filter_body(
                               ? all_rows - w0 : OBX_MAX_BLOCK_ROWS;
     const uint32_t iters = (rows + WG - 1) / WG;
 
-    /* leaf-by-leaf fission into the pass bitmap (one live context) */
-    {
+    /* leaf-by-leaf fission into the pass bitmap (one live context);
+       combine programs evaluate per-leaf bitmaps then a word-wise postfix */
+    if (blk_verdict == 1) {
+      for (uint32_t it = 0; it < iters; it++) {
+        uint32_t rr = it * WG + tid;
+        uint64_t m = __ballot(rr < rows);
+        if (lane == 0) pass_bm[it * WAVES + wv] = m;
+      }
+    } else if constexpr (!PROG) {
       bool first = true;
       for (uint32_t i = 0; i < ph.n_leaves; i++) {
         const blk_leaf lfb = bl[i];
@@ -1116,6 +1266,13 @@ __device__ void filter_body(
           if (lane == 0) pass_bm[it * WAVES + wv] = m;
         }
       }
+    } else {
+      prog_spec ps;
+      ps.n = ph.n_prog;
+      for (int p = 0; p < 15; p++) ps.t[p] = ph.prog[p];
+      filter_prog_phase(bv, cur, plan_leaves, bl, ph.n_leaves, ps, rows, w0,
+                        blk_bit, (lds3_u64 *)pass_bm,
+                        (lds3_u64 *)&leaf_bm[0][0]);
     }
 
     /* output: global bitmap words, pass counts, selection vectors */
@@ -1177,8 +1334,19 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_filter(
     uint64_t *__restrict__ bitmap, int32_t *__restrict__ row_ids,
     uint32_t *__restrict__ blk_counts,
     unsigned long long *__restrict__ counters) {
-  filter_body<false>(buf, blocks, n_blocks, plan_leaves, bleaves, ph, bitmap,
-                     row_ids, blk_counts, counters, nullptr);
+  filter_body<false, false>(buf, blocks, n_blocks, plan_leaves, bleaves, ph,
+                            bitmap, row_ids, blk_counts, counters, nullptr);
+}
+
+extern "C" __global__ __launch_bounds__(WG, 2) void k_filter_prog(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
+    const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
+    uint64_t *__restrict__ bitmap, int32_t *__restrict__ row_ids,
+    uint32_t *__restrict__ blk_counts,
+    unsigned long long *__restrict__ counters) {
+  filter_body<false, true>(buf, blocks, n_blocks, plan_leaves, bleaves, ph,
+                           bitmap, row_ids, blk_counts, counters, nullptr);
 }
 
 extern "C" __global__ __launch_bounds__(WG, 2) void k_filter_lds(
@@ -1189,8 +1357,20 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_filter_lds(
     uint32_t *__restrict__ blk_counts,
     unsigned long long *__restrict__ counters) {
   __shared__ uint8_t lds_blk[(1 + OBX_PIPELINE) * OBX_LDS_STAGE_BYTES + 32];
-  filter_body<true>(buf, blocks, n_blocks, plan_leaves, bleaves, ph, bitmap,
-                    row_ids, blk_counts, counters, lds_blk);
+  filter_body<true, false>(buf, blocks, n_blocks, plan_leaves, bleaves, ph,
+                           bitmap, row_ids, blk_counts, counters, lds_blk);
+}
+
+extern "C" __global__ __launch_bounds__(WG, 2) void k_filter_prog_lds(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
+    const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
+    uint64_t *__restrict__ bitmap, int32_t *__restrict__ row_ids,
+    uint32_t *__restrict__ blk_counts,
+    unsigned long long *__restrict__ counters) {
+  __shared__ uint8_t lds_blk[(1 + OBX_PIPELINE) * OBX_LDS_STAGE_BYTES + 32];
+  filter_body<true, true>(buf, blocks, n_blocks, plan_leaves, bleaves, ph,
+                          bitmap, row_ids, blk_counts, counters, lds_blk);
 }
 
 /* ---------------- decode kernel (get_rows equivalent, parity) ----------- */

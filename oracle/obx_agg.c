@@ -42,6 +42,8 @@ int obx__col_dec_row(const void *d, const void *h, const obx_col_schema *cs,
                      uint32_t r, int64_t *out, int *is_null);
 int obx__leaf_match(const obx_filter_leaf *lf, int64_t v, int is_null, int sc,
                     int len);
+int obx__combine_leaves(const obx_filter_desc *f, const int *leaf_res);
+int obx__prog_valid(const obx_filter_desc *f);
 size_t obx__col_dec_size(void);
 
 /* ---- 256-bit two's-complement accumulation ------------------------------ */
@@ -193,9 +195,9 @@ static void *worker(void *arg) {
     w->rows_scanned += rows;
     int64_t vals[32]; int nulls[32];
     for (uint32_t r = 0; r < rows; r++) {
-      /* filter */
+      /* filter (conjunctive fast path or combine program) */
       int pass = 1;
-      if (filter) {
+      if (filter && filter->n_prog == 0) {
         for (int i = 0; i < filter->n_leaves && pass; i++) {
           const obx_filter_leaf *lf = &filter->leaves[i];
           int di = dec_of_col[lf->col];
@@ -206,6 +208,19 @@ static void *worker(void *arg) {
                                  obx_store_class(bs->cols[lf->col].obj_type),
                                  bs->cols[lf->col].len);
         }
+      } else if (filter && filter->n_leaves > 0) {
+        int leaf_res[8];
+        for (int i = 0; i < filter->n_leaves; i++) {
+          const obx_filter_leaf *lf = &filter->leaves[i];
+          int di = dec_of_col[lf->col];
+          int64_t v; int isn;
+          obx__col_dec_row(decbuf + dsz * di, h, &bs->cols[lf->col], r, &v,
+                           &isn);
+          leaf_res[i] = obx__leaf_match(
+              lf, v, isn, obx_store_class(bs->cols[lf->col].obj_type),
+              bs->cols[lf->col].len);
+        }
+        pass = obx__combine_leaves(filter, leaf_res);
       }
       if (!pass) continue;
       w->rows_passed++;
@@ -300,6 +315,7 @@ int obx_cpu_scan_filter_agg(const obx_blockset *bs,
                             const obx_agg_desc *agg, int nthreads,
                             obx_agg_result *out) {
   if (!bs || !out) return OBX_INVALID_ARGUMENT;
+  if (filter && !obx__prog_valid(filter)) return OBX_INVALID_ARGUMENT;
   if (nthreads <= 0) nthreads = (int)sysconf(_SC_NPROCESSORS_ONLN);
   if (nthreads < 1) nthreads = 1;
   if (nthreads > 256) nthreads = 256;

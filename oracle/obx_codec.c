@@ -933,6 +933,50 @@ static inline int leaf_match(const obx_filter_leaf *lf, int64_t v, int is_null,
   }
 }
 
+/* evaluate the filter's combine program over per-leaf booleans
+ * (ObPushdownFilterExecutor::execute AND/OR, ob_pushdown_filter.cpp:1559;
+ * n_prog == 0 is the conjunctive default) */
+static int combine_leaves(const obx_filter_desc *f, const int *leaf_res) {
+  if (f->n_prog == 0) {
+    for (int i = 0; i < f->n_leaves; i++)
+      if (!leaf_res[i]) return 0;
+    return 1;
+  }
+  int stack[16];
+  int sp = 0;
+  for (int p = 0; p < f->n_prog; p++) {
+    uint8_t t = f->prog[p];
+    if (t < f->n_leaves) {
+      stack[sp++] = leaf_res[t];
+    } else if (t == OBX_TOK_AND && sp >= 2) {
+      sp--; stack[sp - 1] = stack[sp - 1] && stack[sp];
+    } else if (t == OBX_TOK_OR && sp >= 2) {
+      sp--; stack[sp - 1] = stack[sp - 1] || stack[sp];
+    } else {
+      return 0; /* malformed program: validated by callers */
+    }
+  }
+  return sp == 1 ? stack[0] : 0;
+}
+
+int obx__combine_leaves(const obx_filter_desc *f, const int *leaf_res) {
+  return combine_leaves(f, leaf_res);
+}
+
+/* program validity: operands in range, final stack depth 1 */
+int obx__prog_valid(const obx_filter_desc *f) {
+  if (f->n_prog == 0) return 1;
+  int sp = 0;
+  for (int p = 0; p < f->n_prog; p++) {
+    uint8_t t = f->prog[p];
+    if (t < f->n_leaves) sp++;
+    else if ((t == OBX_TOK_AND || t == OBX_TOK_OR) && sp >= 2) sp--;
+    else return 0;
+    if (sp > 15) return 0;
+  }
+  return sp == 1;
+}
+
 int obx_cpu_filter_block(const obx_col_schema *cols, uint16_t n_cols,
                          const uint8_t *block, int64_t block_len,
                          const obx_filter_desc *filter,
@@ -961,14 +1005,29 @@ int obx_cpu_filter_block(const obx_col_schema *cols, uint16_t n_cols,
     sc_of[i] = obx_store_class(cols[c].obj_type);
     len_of[i] = cols[c].len;
   }
+  if (filter && !obx__prog_valid(filter)) return OBX_INVALID_ARGUMENT;
   uint32_t pc = 0;
   for (uint32_t r = 0; r < rows; r++) {
     int pass = 1;
-    for (uint16_t i = 0; i < nl && pass; i++) {
-      int64_t v; int isn;
-      if (col_dec_row(&dec[i], h, &cols[filter->leaves[i].col], r, &v, &isn))
-        return OBX_INTERNAL_ERROR;
-      pass = leaf_match(&filter->leaves[i], v, isn, sc_of[i], len_of[i]);
+    if (filter && filter->n_prog == 0) {
+      for (uint16_t i = 0; i < nl && pass; i++) {
+        int64_t v; int isn;
+        if (col_dec_row(&dec[i], h, &cols[filter->leaves[i].col], r, &v,
+                        &isn))
+          return OBX_INTERNAL_ERROR;
+        pass = leaf_match(&filter->leaves[i], v, isn, sc_of[i], len_of[i]);
+      }
+    } else if (filter && nl > 0) {
+      int leaf_res[8];
+      for (uint16_t i = 0; i < nl; i++) {
+        int64_t v; int isn;
+        if (col_dec_row(&dec[i], h, &cols[filter->leaves[i].col], r, &v,
+                        &isn))
+          return OBX_INTERNAL_ERROR;
+        leaf_res[i] = leaf_match(&filter->leaves[i], v, isn, sc_of[i],
+                                 len_of[i]);
+      }
+      pass = combine_leaves(filter, leaf_res);
     }
     if (pass) { result_bits[r >> 3] |= (uint8_t)(1u << (r & 7)); pc++; }
   }
