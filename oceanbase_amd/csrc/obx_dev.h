@@ -17,7 +17,23 @@
 #ifndef OBX_DEV_H_
 #define OBX_DEV_H_
 
+#ifdef __HIPRTC__
+/* hipRTC has no libc headers; these match the HIP device ABI */
+typedef signed char int8_t;
+typedef unsigned char uint8_t;
+typedef short int16_t;
+typedef unsigned short uint16_t;
+typedef int int32_t;
+typedef unsigned int uint32_t;
+typedef long long int64_t;
+typedef unsigned long long uint64_t;
+typedef unsigned long size_t;
+#define INT64_MAX 0x7fffffffffffffffll
+#define INT64_MIN (-INT64_MAX - 1ll)
+#define UINT64_MAX 0xffffffffffffffffull
+#else
 #include <stdint.h>
+#endif
 
 #define OBX_DEV_MAX_COLS 8
 #define OBX_DEV_MAX_LEAVES 8
