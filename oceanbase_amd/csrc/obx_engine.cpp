@@ -106,6 +106,11 @@ struct obx_handle {
   uint64_t last_survivors = 0;
   bool lds_ok = false;   /* all blocks 16-B aligned and <= LDS stage size */
   bool in_use = false;
+  /* per-(block,col) encoding summary captured at load time for the JIT
+     eligibility walk (obx_jit.inc): low nibble = decode class (0 raw,
+     1 dict, 2 intdiff, 3 const, 4 slow), bit4 = string; col_cnt = dict
+     count capped at 255 */
+  std::vector<uint8_t> col_class, col_cnt;
 };
 
 struct obx_gpu_ctx {
@@ -305,6 +310,20 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
     if ((bs->block_offsets[b] & 15) || blen + 24 > OBX_LDS_STAGE_BYTES ||
         blocks[b].row_count > 4096 /* OBX_MAX_BLOCK_ROWS */)
       lds_ok = false;
+    for (uint16_t c = 0; c < bs->n_cols; c++) {
+      const dev_col &dc = blocks[b].cols[c];
+      uint8_t cls;
+      switch (dc.enc) {
+        case OBX_D_RAW: cls = 0; break;
+        case OBX_D_DICT: cls = 1; break;
+        case OBX_D_INTDIFF: cls = 2; break;
+        case OBX_D_CONST: cls = dc.runs == 0 ? 3 : 4; break;
+        default: cls = 4; break;
+      }
+      if (dc.flags & OBX_DF_STRING) cls |= 0x10;
+      h.col_class.push_back(cls);
+      h.col_cnt.push_back((uint8_t)(dc.count > 254 ? 255 : dc.count));
+    }
     row_start += blocks[b].row_count;
   }
   h.total_rows = row_start;
@@ -498,6 +517,9 @@ static uint32_t grid_for(uint32_t n_blocks) {
   return g ? g : 1;
 }
 
+#include "obx_jit.inc"
+
+
 extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
                               const obx_filter_desc *filter,
                               int want_row_ids) {
@@ -682,8 +704,14 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
       a += fuse ? 2 : 1;
     }
   } else {
+    /* plan-specialized JIT kernel when eligible (compiled once per plan
+       signature, before the timed region; obx_jit.inc) */
+    jit_entry *je = jit_prepare(h, ph);
     HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
-    {
+    if (je) {
+      if (jit_launch(je, h, grid_for(h.n_blocks), ctx->stream) != 0)
+        return OBX_INTERNAL_ERROR;
+    } else {
       auto kfn = h.lds_ok
                      ? (ph.n_prog ? k_scan_filter_agg_prog_lds
                                   : k_scan_filter_agg_lds)
