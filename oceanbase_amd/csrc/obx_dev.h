@@ -17,7 +17,7 @@
 #ifndef OBX_DEV_H_
 #define OBX_DEV_H_
 
-#ifdef __HIPRTC__
+#if defined(__HIPRTC__) || defined(OBX_HIPRTC)
 /* hipRTC has no libc headers; these match the HIP device ABI */
 typedef signed char int8_t;
 typedef unsigned char uint8_t;

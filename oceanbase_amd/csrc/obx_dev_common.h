@@ -9,7 +9,7 @@
  */
 #ifndef OBX_DEV_COMMON_H
 #define OBX_DEV_COMMON_H
-#ifndef __HIPRTC__
+#if !defined(__HIPRTC__) && !defined(OBX_HIPRTC)
 #include <hip/hip_runtime.h>
 #endif
 #include "obx_dev.h"
