@@ -304,6 +304,9 @@ int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
 /* Timing helpers: milliseconds of device time of the last
  * filter/scan_filter_agg call (HIP events on the context stream). */
 double obx_gpu_last_kernel_ms(obx_gpu_ctx *ctx);
+/* 1 if the last scan_filter_agg ran the hipRTC plan-specialized kernel
+ * (query codegen), 0 if the precompiled generic kernels ran. */
+int obx_gpu_last_jit(obx_gpu_ctx *ctx);
 uint64_t obx_gpu_total_rows(obx_gpu_ctx *ctx, int handle);
 uint64_t obx_gpu_total_bytes(obx_gpu_ctx *ctx, int handle);
 uint64_t obx_gpu_last_survivors(obx_gpu_ctx *ctx, int handle);

@@ -120,6 +120,7 @@ struct obx_gpu_ctx {
   hipEvent_t ev_p0 = nullptr, ev_p1 = nullptr;
   double last_ms = 0.0;
   double last_prep_ms = 0.0;  /* plan upload + per-block filter lowering */
+  int last_jit = 0;           /* last scan used the specialized kernel */
   std::vector<obx_handle> handles;
 };
 
@@ -405,6 +406,38 @@ static int prep_query(obx_gpu_ctx *ctx, obx_handle &h,
         pl[i].in_list[j] = ck(lf->in_list[j]);
     }
   }
+  /* AND-only plans: merge a lower-bound and an upper-bound leaf on the
+     same column into one OP_BT leaf (the reference's range-node build,
+     ObWhiteFilterExecutor; one pass instead of two in the kernels) */
+  if (filter && !filter->n_prog && nl > 1) {
+    for (uint16_t i = 0; i < nl; i++) {
+      if (pl[i].op != OBX_OP_GE && pl[i].op != OBX_OP_GT) continue;
+      for (uint16_t k = 0; k < nl; k++) {
+        if (k == i || pl[k].col != pl[i].col) continue;
+        if (pl[k].op != OBX_OP_LE && pl[k].op != OBX_OP_LT) continue;
+        int64_t lo = pl[i].vlo, hi = pl[k].vlo;
+        if (pl[i].op == OBX_OP_GT) {
+          if (lo == INT64_MAX) continue;
+          lo++;
+        }
+        if (pl[k].op == OBX_OP_LT) {
+          if (hi == INT64_MIN) continue;
+          hi--;
+        }
+        pl[i].op = OBX_OP_BT;
+        pl[i].vlo = lo;
+        pl[i].vhi = hi;
+        /* drop leaf k */
+        for (uint16_t m = k; m + 1 < nl; m++) pl[m] = pl[m + 1];
+        nl--;
+        ph.n_leaves = nl;
+        if (k < i) i--;
+        i--; /* re-examine the merged leaf (another pair may exist) */
+        break;
+      }
+    }
+  }
+
   /* resolve needed value slots */
   int slot_of_col[64];
   for (int i = 0; i < 64; i++) slot_of_col[i] = -1;
@@ -648,6 +681,7 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
   dev_plan_hdr ph;
   int rc = prep_query(ctx, h, filter, agg, ph);
   if (rc != OBX_SUCCESS) return rc;
+  ctx->last_jit = 0;
 
   /* init global table (min/max cells need INT64_MAX/MIN) */
   std::vector<gslot> init(OBX_GTABLE_SLOTS);
@@ -707,6 +741,7 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
     /* plan-specialized JIT kernel when eligible (compiled once per plan
        signature, before the timed region; obx_jit.inc) */
     jit_entry *je = jit_prepare(h, ph);
+    ctx->last_jit = je != nullptr;
     HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
     if (je) {
       if (jit_launch(je, h, grid_for(h.n_blocks), ctx->stream) != 0)
@@ -792,6 +827,10 @@ extern "C" double obx_gpu_last_kernel_ms(obx_gpu_ctx *ctx) {
  * last query's prep (plan upload + k_lower_leaves) and main kernel. */
 extern "C" double obx_gpu_last_prep_ms(obx_gpu_ctx *ctx) {
   return ctx ? ctx->last_prep_ms : -1.0;
+}
+
+extern "C" int obx_gpu_last_jit(obx_gpu_ctx *ctx) {
+  return ctx ? ctx->last_jit : 0;
 }
 
 extern "C" uint64_t obx_gpu_total_rows(obx_gpu_ctx *ctx, int handle) {

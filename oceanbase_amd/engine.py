@@ -60,6 +60,8 @@ def _load():
     lib.obx_gpu_last_kernel_ms.argtypes = [C.c_void_p]
     lib.obx_gpu_last_prep_ms.restype = C.c_double
     lib.obx_gpu_last_prep_ms.argtypes = [C.c_void_p]
+    lib.obx_gpu_last_jit.restype = C.c_int
+    lib.obx_gpu_last_jit.argtypes = [C.c_void_p]
     for f in ("obx_gpu_total_rows", "obx_gpu_total_bytes",
               "obx_gpu_last_survivors"):
         getattr(lib, f).restype = C.c_uint64
@@ -178,6 +180,10 @@ class GpuEngine:
 
     def last_kernel_ms(self):
         return float(self._lib.obx_gpu_last_kernel_ms(self._ctx))
+
+    def last_jit(self):
+        """True if the last scan ran the hipRTC plan-specialized kernel."""
+        return bool(self._lib.obx_gpu_last_jit(self._ctx))
 
     def last_prep_ms(self):
         """Device time of the last query's prep (plan upload + per-block

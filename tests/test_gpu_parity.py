@@ -325,3 +325,27 @@ def test_wide_rows_per_block(eng):
     res_gpu = eng.scan_filter_agg(h, filt, agg)
     assert abi.result_rows(res_gpu, 4) == abi.result_rows(res_cpu, 4)
     eng.free(h)
+
+
+@pytest.mark.gpu
+def test_jit_engages_on_q1_shape():
+    """The hipRTC plan-specialized kernel must actually run for the
+    flagship Q1 plan shape (guards against silent fallback to the generic
+    kernels, which once hid a broken JIT compile)."""
+    from oceanbase_amd.engine import GpuEngine
+    li = oracle.Lineitem(4, 50_000, seed=11)
+    eng = GpuEngine(0)
+    h = eng.load(li.bs)
+    filt = abi.make_filter(
+        [dict(col=6, op=abi.OP_LE, lo=oracle.date_days(1998, 9, 2))])
+    aggs = [dict(kind=abi.AGG_COUNT), dict(kind=abi.AGG_SUM, col_a=0),
+            dict(kind=abi.AGG_SUM_PROD2, col_a=1, col_b=2),
+            dict(kind=abi.AGG_SUM_PROD3, col_a=1, col_b=2, col_c=3)]
+    agg = abi.make_agg([4, 5], aggs)
+    res = eng.scan_filter_agg(h, filt, agg)
+    assert eng.last_jit(), "JIT kernel did not engage for the Q1 shape"
+    ores = oracle.scan_filter_agg(li.bs, filt, agg)
+    assert res.rows_passed == ores.rows_passed
+    assert sorted(abi.result_rows(res, 4)) == sorted(
+        abi.result_rows(ores, 4))
+    eng.close()
