@@ -67,6 +67,10 @@ enum obx_encoding {
   OBX_ENC_RLE = 2,
   OBX_ENC_CONST = 3,
   OBX_ENC_INTEGER_BASE_DIFF = 4,
+  OBX_ENC_STRING_DIFF = 5,       /* fixed char: common/diff byte runs
+                                    (ob_string_diff_encoder.h) */
+  OBX_ENC_HEX_PACKING = 6,       /* fixed char, <=16 distinct chars: nibble
+                                    packing (ob_hex_string_encoder.h) */
   OBX_ENC_MAX = 10,
   OBX_ENC_AUTO = 255,            /* writer picks (cost-ranked like
                                     ob_encoding_util.h:270-303) */

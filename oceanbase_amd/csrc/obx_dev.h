@@ -48,6 +48,8 @@ enum {
   OBX_D_RLE = 2,
   OBX_D_CONST = 3,
   OBX_D_INTDIFF = 4,
+  OBX_D_SDIFF = 5,   /* STRING_DIFF: common/diff byte runs */
+  OBX_D_HEX = 6,     /* HEX_PACKING: nibble-packed chars */
 };
 
 /* dev_col flags */

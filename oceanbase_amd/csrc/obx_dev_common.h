@@ -307,6 +307,54 @@ __device__ __forceinline__ bool leaf_ctx_match(const blk_view &bv,
 /* full value decode: int64 (sign-extended), is_null out.
  * __noinline__: slow/cold generic path — keeps register pressure off the
  * hot ctx_value loops. */
+/* HEX_PACKING / STRING_DIFF row decode (ObHexStringUnpacker /
+ * ObStringDiffHeader::copy_string). __noinline__ with unrolling disabled:
+ * this is a cold generic path, and its register footprint rolls up into
+ * every kernel that can reach col_value (AMDGPU call-graph budget). */
+__device__ __noinline__ int64_t str_value(const blk_view bv,
+                                          const dev_col &c, uint32_t r) {
+  uint64_t v = 0;
+  const uint64_t row_byte = c.data_bit / 8 + (uint64_t)r * c.width;
+  if (c.enc == OBX_D_HEX) {
+#pragma clang loop unroll(disable)
+    for (uint32_t i = 0; i < c.datum_len; i++) {
+      uint8_t b8 = (uint8_t)bit_read(bv, (row_byte + i / 2) * 8, 8);
+      uint8_t nib = (uint8_t)((b8 >> (((i + 1) % 2) * 4)) & 0xF);
+      uint8_t ch8 = (uint8_t)bit_read(bv, (c.dict_byte + nib) * 8, 8);
+      v |= (uint64_t)ch8 << (8 * i);
+    }
+    return (int64_t)v;
+  }
+  /* STRING_DIFF */
+  const bool hex = c.entry_len > 0;
+  const uint64_t chars_byte = c.dict_byte + c.runs;
+  uint32_t pos = 0, cpos = 0, dpos = 0;
+#pragma clang loop unroll(disable)
+  for (uint32_t i = 0; i < c.runs; i++) {
+    uint8_t d8 = (uint8_t)bit_read(bv, (c.dict_byte + i) * 8, 8);
+    uint32_t cnt = d8 >> 1;
+#pragma clang loop unroll(disable)
+    for (uint32_t j = 0; j < cnt; j++, pos++) {
+      uint8_t ch8;
+      if (d8 & 1) {
+        if (hex) {
+          uint8_t b8 = (uint8_t)bit_read(bv, (row_byte + dpos / 2) * 8, 8);
+          uint8_t nib = (uint8_t)((b8 >> (((dpos + 1) % 2) * 4)) & 0xF);
+          ch8 = (uint8_t)bit_read(bv, (chars_byte + nib) * 8, 8);
+        } else {
+          ch8 = (uint8_t)bit_read(bv, (row_byte + dpos) * 8, 8);
+        }
+        dpos++;
+      } else {
+        ch8 = (uint8_t)bit_read(bv, (c.aux_byte + cpos) * 8, 8);
+        cpos++;
+      }
+      v |= (uint64_t)ch8 << (8 * pos);
+    }
+  }
+  return (int64_t)v;
+}
+
 __device__ __noinline__ int64_t col_value(const blk_view bv,
                                              const dev_col &c, uint32_t r,
                                              bool &null_out) {
@@ -352,6 +400,11 @@ __device__ __noinline__ int64_t col_value(const blk_view bv,
       if (col_is_null_ext(bv, c, r)) { null_out = true; return 0; }
       uint64_t diff = col_packed(bv, c, r);
       return (int64_t)((uint64_t)c.base + diff);
+    }
+    case OBX_D_HEX:
+    case OBX_D_SDIFF: {
+      if (col_is_null_ext(bv, c, r)) { null_out = true; return 0; }
+      return str_value(bv, c, r);
     }
   }
   return 0;

@@ -283,6 +283,43 @@ static int parse_block(const obx_col_schema *cols, uint16_t n_cols,
         else dc->data_bit = (data0 + (ext_bits + 7) / 8) * 8;
         break;
       }
+      case OBX_COL_HEX_PACKING: {
+        const obx_hex_meta *hm = (const obx_hex_meta *)col_host;
+        dc->width = (uint8_t)((hm->string_size + 1) / 2); /* row stride B */
+        dc->count = hm->char_cnt;
+        dc->dict_byte = col_base + sizeof(obx_hex_meta); /* char array */
+        uint64_t data0 = col_base + ch->length;
+        uint64_t ext_bits = has_ext ? (uint64_t)evb * h->row_count : 0;
+        if (has_ext) {
+          dc->flags |= OBX_DF_HAS_EXT;
+          dc->ext_bit = data0 * 8;
+        }
+        dc->data_bit = (data0 + (ext_bits + 7) / 8) * 8;
+        break;
+      }
+      case OBX_COL_STRING_DIFF: {
+        const obx_sdiff_meta *sm = (const obx_sdiff_meta *)col_host;
+        dc->runs = sm->diff_desc_cnt;
+        dc->entry_len = sm->hex_char_cnt;
+        dc->dict_byte = col_base + sizeof(obx_sdiff_meta); /* descs */
+        const uint8_t *descs = col_host + sizeof(obx_sdiff_meta);
+        uint32_t diff_len = 0;
+        for (uint32_t i = 0; i < sm->diff_desc_cnt; i++)
+          if (descs[i] & 1) diff_len += descs[i] >> 1;
+        dc->rib = (uint8_t)diff_len;
+        dc->width = (uint8_t)(sm->hex_char_cnt ? (diff_len + 1) / 2
+                                               : diff_len);
+        dc->aux_byte = dc->dict_byte + sm->diff_desc_cnt +
+                       sm->hex_char_cnt; /* common bytes */
+        uint64_t data0 = col_base + ch->length;
+        uint64_t ext_bits = has_ext ? (uint64_t)evb * h->row_count : 0;
+        if (has_ext) {
+          dc->flags |= OBX_DF_HAS_EXT;
+          dc->ext_bit = data0 * 8;
+        }
+        dc->data_bit = (data0 + (ext_bits + 7) / 8) * 8;
+        break;
+      }
       default:
         return OBX_NOT_SUPPORTED;
     }

@@ -533,6 +533,135 @@ static int encode_column(enc_buf *mb, obx_col_header *ch,
       if (rc) return rc;
       break;
     }
+    case OBX_ENC_HEX_PACKING: {
+      /* ObHexStringEncoder (ob_hex_string_encoder.{h,cpp}): <=16 distinct
+         chars map to nibbles; build_index assigns indices in ascending
+         char order; packing is HIGH nibble first. */
+      if (sc != OBX_SC_STRING) return OBX_NOT_SUPPORTED;
+      uint8_t seen[256]; memset(seen, 0, sizeof(seen));
+      uint32_t nch = 0; int any = 0;
+      for (uint32_t r = 0; r < rows; r++) {
+        if (null_at(nulls, r)) continue;
+        any = 1;
+        for (int i = 0; i < len; i++) {
+          uint8_t cc = data[(size_t)r * len + i];
+          if (!seen[cc]) { seen[cc] = 1; nch++; }
+        }
+      }
+      if (!any || nch > 16) return OBX_NOT_SUPPORTED;
+      uint8_t chars[16]; uint8_t idx_of[256]; uint32_t k = 0;
+      for (int i = 0; i < 256; i++)
+        if (seen[i]) { idx_of[i] = (uint8_t)k; chars[k++] = (uint8_t)i; }
+      ch->type = OBX_COL_HEX_PACKING;
+      ch->attr |= OBX_COL_ATTR_FIX_LENGTH;
+      if (null_cnt) ch->attr |= OBX_COL_ATTR_HAS_EXTEND_VALUE;
+      obx_hex_meta hm = { 0, (uint8_t)nch, (uint16_t)len };
+      int64_t meta_size = (int64_t)sizeof(hm) + nch;
+      if (mb->len + meta_size > mb->cap) return OBX_BUF_NOT_ENOUGH;
+      memcpy(mb->p + mb->len, &hm, sizeof(hm));
+      memcpy(mb->p + mb->len + sizeof(hm), chars, nch);
+      mb->len += meta_size;
+      ch->length = (uint32_t)meta_size;
+      int stride = (len + 1) / 2;
+      uint8_t *scr = (uint8_t *)calloc((size_t)rows * stride + 1, 1);
+      if (!scr) return OBX_INTERNAL_ERROR;
+      for (uint32_t r = 0; r < rows; r++) {
+        if (null_at(nulls, r)) continue;
+        uint8_t *rp = scr + (size_t)r * stride;
+        for (int i = 0; i < len; i++) {
+          uint8_t nib = idx_of[data[(size_t)r * len + i]];
+          rp[i / 2] |= (uint8_t)(nib << (((i + 1) % 2) * 4));
+        }
+      }
+      fix_store_spec s = { block_ext_bit, null_cnt > 0, 0, stride };
+      int rc = store_fix_region(mb, &s, rows, nulls, NULL, NULL, scr, stride);
+      free(scr);
+      if (rc) return rc;
+      break;
+    }
+    case OBX_ENC_STRING_DIFF: {
+      /* ObStringDiffEncoder (ob_string_diff_encoder.{h,cpp}): byte
+         positions that never vary across non-null rows are stored once
+         (common_data); varying runs are described by DiffDesc bytes
+         (bit0 = diff, bits1-7 = run length) and each row stores only its
+         diff bytes, nibble-packed like HEX_PACKING when the diff chars
+         fit 16 (is_hex_packing). */
+      if (sc != OBX_SC_STRING || len < 1) return OBX_NOT_SUPPORTED;
+      int first = -1;
+      for (uint32_t r = 0; r < rows; r++)
+        if (!null_at(nulls, r)) { first = (int)r; break; }
+      if (first < 0) return OBX_NOT_SUPPORTED;
+      uint8_t vary[256]; memset(vary, 0, (size_t)len);
+      for (uint32_t r = 0; r < rows; r++) {
+        if (null_at(nulls, r)) continue;
+        for (int i = 0; i < len; i++)
+          if (data[(size_t)r * len + i] != data[(size_t)first * len + i])
+            vary[i] = 1;
+      }
+      uint8_t descs[256]; int nd = 0; int diff_len = 0;
+      for (int i = 0; i < len;) {
+        int j = i;
+        while (j < len && vary[j] == vary[i] && j - i < 127) j++;
+        descs[nd++] = (uint8_t)((vary[i] & 1) | ((j - i) << 1));
+        if (vary[i]) diff_len += j - i;
+        i = j;
+      }
+      /* hex-packability of the diff bytes only */
+      uint8_t seen[256]; memset(seen, 0, sizeof(seen));
+      uint32_t nch = 0;
+      for (uint32_t r = 0; r < rows && nch <= 16; r++) {
+        if (null_at(nulls, r)) continue;
+        for (int i = 0; i < len; i++) {
+          if (!vary[i]) continue;
+          uint8_t cc = data[(size_t)r * len + i];
+          if (!seen[cc]) { seen[cc] = 1; nch++; }
+        }
+      }
+      int use_hex = diff_len > 0 && nch <= 16 && (diff_len + 1) / 2 < diff_len;
+      uint8_t chars[16]; uint8_t idx_of[256]; uint32_t k = 0;
+      if (use_hex)
+        for (int i = 0; i < 256; i++)
+          if (seen[i]) { idx_of[i] = (uint8_t)k; chars[k++] = (uint8_t)i; }
+      ch->type = OBX_COL_STRING_DIFF;
+      ch->attr |= OBX_COL_ATTR_FIX_LENGTH;
+      if (null_cnt) ch->attr |= OBX_COL_ATTR_HAS_EXTEND_VALUE;
+      obx_sdiff_meta sm2 = { 0, (uint8_t)(use_hex ? nch : 0), (uint16_t)len,
+                             (uint8_t)nd };
+      int64_t meta_size = (int64_t)sizeof(sm2) + nd + (use_hex ? nch : 0) +
+                          (len - diff_len);
+      if (mb->len + meta_size > mb->cap) return OBX_BUF_NOT_ENOUGH;
+      uint8_t *mp = mb->p + mb->len;
+      memcpy(mp, &sm2, sizeof(sm2)); mp += sizeof(sm2);
+      memcpy(mp, descs, (size_t)nd); mp += nd;
+      if (use_hex) { memcpy(mp, chars, nch); mp += nch; }
+      for (int i = 0; i < len; i++)
+        if (!vary[i]) *mp++ = data[(size_t)first * len + i];
+      mb->len += meta_size;
+      ch->length = (uint32_t)meta_size;
+      int stride = use_hex ? (diff_len + 1) / 2 : diff_len;
+      uint8_t *scr = (uint8_t *)calloc((size_t)rows * (stride ? stride : 1) + 1, 1);
+      if (!scr) return OBX_INTERNAL_ERROR;
+      for (uint32_t r = 0; r < rows && stride; r++) {
+        if (null_at(nulls, r)) continue;
+        uint8_t *rp = scr + (size_t)r * stride;
+        int pos = 0;
+        for (int i = 0; i < len; i++) {
+          if (!vary[i]) continue;
+          uint8_t cc = data[(size_t)r * len + i];
+          if (use_hex)
+            rp[pos / 2] |= (uint8_t)(idx_of[cc] << (((pos + 1) % 2) * 4));
+          else
+            rp[pos] = cc;
+          pos++;
+        }
+      }
+      fix_store_spec s = { block_ext_bit, null_cnt > 0, 0, stride };
+      int rc = store_fix_region(mb, &s, rows, nulls, NULL, NULL, scr,
+                                stride ? stride : 1);
+      free(scr);
+      if (rc) return rc;
+      break;
+    }
     default:
       return OBX_NOT_SUPPORTED;
   }
@@ -635,6 +764,12 @@ typedef struct col_dec {
   /* int diff */
   const obx_intdiff_meta *im;
   uint64_t base;
+  /* string transforms (HEX_PACKING / STRING_DIFF) */
+  const obx_hex_meta *hm;
+  const obx_sdiff_meta *sm;
+  const uint8_t *str_chars;  /* hex char array */
+  const uint8_t *str_descs;  /* diff descs */
+  const uint8_t *str_common; /* common bytes */
 } col_dec;
 
 static int col_dec_init(col_dec *d, const obx_micro_header *h,
@@ -699,6 +834,24 @@ static int col_dec_init(col_dec *d, const obx_micro_header *h,
       }
       break;
     }
+    case OBX_COL_HEX_PACKING: {
+      d->hm = (const obx_hex_meta *)d->meta;
+      d->str_chars = d->meta + sizeof(obx_hex_meta);
+      d->data = d->meta + ch->length;
+      int64_t ext_bits = ext ? (int64_t)evb * h->row_count : 0;
+      d->fix_off = (ext_bits + 7) / 8;
+      break;
+    }
+    case OBX_COL_STRING_DIFF: {
+      d->sm = (const obx_sdiff_meta *)d->meta;
+      d->str_descs = d->meta + sizeof(obx_sdiff_meta);
+      d->str_chars = d->str_descs + d->sm->diff_desc_cnt;
+      d->str_common = d->str_chars + d->sm->hex_char_cnt;
+      d->data = d->meta + ch->length;
+      int64_t ext_bits = ext ? (int64_t)evb * h->row_count : 0;
+      d->fix_off = (ext_bits + 7) / 8;
+      break;
+    }
     default:
       return OBX_NOT_SUPPORTED;
   }
@@ -713,6 +866,8 @@ static inline int col_dec_is_null(const col_dec *d, const obx_micro_header *h,
   switch (d->ch->type) {
     case OBX_COL_RAW:
     case OBX_COL_INTEGER_BASE_DIFF:
+    case OBX_COL_HEX_PACKING:
+    case OBX_COL_STRING_DIFF:
       if (!(d->ch->attr & OBX_COL_ATTR_HAS_EXTEND_VALUE)) return 0;
       return obx_bs_get(d->data, (int64_t)r * evb, evb) != 0;
     default:
@@ -823,6 +978,53 @@ static int col_dec_row(const col_dec *d, const obx_micro_header *h,
       uint64_t v = 0;
       memcpy(&v, d->dict_pay + ref * d->dm->data_size, d->dm->data_size);
       if (sc == OBX_SC_INT) v = obx_sign_extend(v, tss, 1);
+      *out = (int64_t)v;
+      return 0;
+    }
+    case OBX_COL_HEX_PACKING: {
+      if (col_dec_is_null(d, h, r)) { *is_null = 1; *out = 0; return 0; }
+      int n = d->hm->string_size;
+      const uint8_t *rp = d->data + d->fix_off + (size_t)r * ((n + 1) / 2);
+      uint64_t v = 0;
+      for (int i = 0; i < n; i++) {
+        uint8_t nib = (uint8_t)((rp[i / 2] >> (((i + 1) % 2) * 4)) & 0xF);
+        v |= (uint64_t)d->str_chars[nib] << (8 * i);
+      }
+      *out = (int64_t)v;
+      return 0;
+    }
+    case OBX_COL_STRING_DIFF: {
+      if (col_dec_is_null(d, h, r)) { *is_null = 1; *out = 0; return 0; }
+      int n = d->sm->string_size;
+      int use_hex = d->sm->hex_char_cnt > 0;
+      /* diff_len from the descs */
+      int diff_len = 0;
+      for (int i = 0; i < d->sm->diff_desc_cnt; i++)
+        if (d->str_descs[i] & 1) diff_len += d->str_descs[i] >> 1;
+      int stride = use_hex ? (diff_len + 1) / 2 : diff_len;
+      const uint8_t *rp = d->data + d->fix_off + (size_t)r * stride;
+      uint64_t v = 0;
+      int pos = 0, cpos = 0, dpos = 0;
+      for (int i = 0; i < d->sm->diff_desc_cnt; i++) {
+        int cnt = d->str_descs[i] >> 1;
+        if (d->str_descs[i] & 1) {
+          for (int j = 0; j < cnt; j++, pos++, dpos++) {
+            uint8_t cc;
+            if (use_hex) {
+              uint8_t nib = (uint8_t)((rp[dpos / 2] >>
+                                       (((dpos + 1) % 2) * 4)) & 0xF);
+              cc = d->str_chars[nib];
+            } else {
+              cc = rp[dpos];
+            }
+            v |= (uint64_t)cc << (8 * pos);
+          }
+        } else {
+          for (int j = 0; j < cnt; j++, pos++, cpos++)
+            v |= (uint64_t)d->str_common[cpos] << (8 * pos);
+        }
+      }
+      (void)n;
       *out = (int64_t)v;
       return 0;
     }

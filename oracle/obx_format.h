@@ -95,6 +95,8 @@ enum {
   OBX_COL_RLE = 2,
   OBX_COL_CONST = 3,
   OBX_COL_INTEGER_BASE_DIFF = 4,
+  OBX_COL_STRING_DIFF = 5,
+  OBX_COL_HEX_PACKING = 6,
   /* 5..9: string encoders, out of scope this round */
 };
 enum {
@@ -146,6 +148,35 @@ typedef struct obx_const_meta {
      count*row_id_byte], then at +offset dict meta.  No-exception path:
      payload at +offset = the const datum bytes (no dict). */
 } obx_const_meta;        /* 6 bytes packed */
+
+/* HEX_PACKING meta (ObHexStringHeader, ob_hex_string_encoder.h:128-139;
+   the reference's offset_/length_ var-data bookkeeping is dropped — our
+   boundary stores fixed char(N<=8) only, so string_size is the fixed N).
+   Followed by char_cnt bytes of hex_char_array (build_index order: the
+   distinct chars ASCENDING; nibble k decodes as hex_char_array[k],
+   ob_hex_string_encoder.cpp:22-34). Row data (fix region, after the usual
+   ext bits): ceil(N/2) bytes per row, HIGH nibble first
+   (ObHexStringPacker::pack, pos even -> shift 4). */
+typedef struct obx_hex_meta {
+  uint8_t version;       /* 0 */
+  uint8_t char_cnt;      /* distinct chars, <= 16 */
+  uint16_t string_size;  /* N (== schema len) */
+} obx_hex_meta;          /* 4 bytes packed */
+
+/* STRING_DIFF meta (ObStringDiffHeader, ob_string_diff_encoder.h:20-97).
+   diff_descs: 1 byte each, bit0 = diff_ (1: positions vary per row),
+   bits1-7 = count_ (run length; reference bitfield diff_:1,count_:7).
+   Followed by: hex_char_array (hex_char_cnt bytes; >0 means the per-row
+   diff bytes are nibble-packed exactly like HEX_PACKING), then
+   common_data (the bytes of all non-diff runs, in position order).
+   Row data (fix region): diff bytes of the row in position order,
+   hex-packed when hex_char_cnt > 0; stride = diff_len or ceil(diff_len/2). */
+typedef struct obx_sdiff_meta {
+  uint8_t version;       /* 0 */
+  uint8_t hex_char_cnt;  /* 0 = plain diff bytes */
+  uint16_t string_size;  /* N */
+  uint8_t diff_desc_cnt;
+} obx_sdiff_meta;        /* 5 bytes packed */
 
 typedef struct obx_intdiff_meta {
   uint8_t version;       /* 0 */

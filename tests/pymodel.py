@@ -12,7 +12,7 @@ All values are returned as Python ints (already sign-extended) or None (NULL).
 """
 import struct
 
-ENC_RAW, ENC_DICT, ENC_RLE, ENC_CONST, ENC_INT_DIFF = range(5)
+ENC_RAW, ENC_DICT, ENC_RLE, ENC_CONST, ENC_INT_DIFF, ENC_SDIFF, ENC_HEX = range(7)
 
 ATTR_FIX = 1
 ATTR_EXT = 2
@@ -220,6 +220,62 @@ class Block:
                     diff = int.from_bytes(d[f0 + r * k:f0 + (r + 1) * k],
                                           "little")
                 out.append(bval + diff)
+        elif t == ENC_SDIFF:
+            # obx_sdiff_meta: version u8, hex_char_cnt u8, string_size u16,
+            # diff_desc_cnt u8; then descs, hex chars, common bytes
+            ver, hex_cnt, ssize, nd = struct.unpack_from("<BBHB", d, base)
+            descs = list(d[base + 5:base + 5 + nd])
+            chars = list(d[base + 5 + nd:base + 5 + nd + hex_cnt])
+            common = d[base + 5 + nd + hex_cnt:base + ch["length"]]
+            diff_len = sum(dd >> 1 for dd in descs if dd & 1)
+            stride = (diff_len + 1) // 2 if hex_cnt else diff_len
+            data0 = base + ch["length"]
+            ext_bits = evb * rows if has_ext else 0
+            fix0 = data0 + (ext_bits + 7) // 8
+            for r in range(rows):
+                if has_ext and bs_get(d[data0:], r * evb, evb):
+                    out.append(None)
+                    continue
+                rp = d[fix0 + r * stride:fix0 + (r + 1) * stride]
+                v = 0
+                pos = cpos = dpos = 0
+                for dd in descs:
+                    cnt = dd >> 1
+                    if dd & 1:
+                        for _ in range(cnt):
+                            if hex_cnt:
+                                nib = (rp[dpos // 2] >>
+                                       (((dpos + 1) % 2) * 4)) & 0xF
+                                cc = chars[nib]
+                            else:
+                                cc = rp[dpos]
+                            v |= cc << (8 * pos)
+                            pos += 1
+                            dpos += 1
+                    else:
+                        for _ in range(cnt):
+                            v |= common[cpos] << (8 * pos)
+                            pos += 1
+                            cpos += 1
+                out.append(v)
+        elif t == ENC_HEX:
+            # obx_hex_meta: version u8, char_cnt u8, string_size u16; chars
+            ver, nch, ssize = struct.unpack_from("<BBH", d, base)
+            chars = list(d[base + 4:base + 4 + nch])
+            stride = (ssize + 1) // 2
+            data0 = base + ch["length"]
+            ext_bits = evb * rows if has_ext else 0
+            fix0 = data0 + (ext_bits + 7) // 8
+            for r in range(rows):
+                if has_ext and bs_get(d[data0:], r * evb, evb):
+                    out.append(None)
+                    continue
+                rp = d[fix0 + r * stride:fix0 + (r + 1) * stride]
+                v = 0
+                for i in range(ssize):
+                    nib = (rp[i // 2] >> (((i + 1) % 2) * 4)) & 0xF
+                    v |= chars[nib] << (8 * i)
+                out.append(v)
         else:
             raise NotImplementedError(t)
         return out

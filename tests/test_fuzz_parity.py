@@ -54,7 +54,10 @@ def _gen_column(rng, tspec, rows, style):
 
 def _encoding_for(rng, tspec, enc, style):
     t = tspec[0]
-    if enc == abi.ENC_INT_DIFF and t in (abi.T_CHAR, abi.T_DECIMAL_INT):
+    if enc == abi.ENC_INT_DIFF and t == abi.T_CHAR:
+        # chars can't int-diff: exercise the string transforms instead
+        return abi.ENC_HEX if style % 2 else abi.ENC_SDIFF
+    if enc == abi.ENC_INT_DIFF and t == abi.T_DECIMAL_INT:
         return abi.ENC_RAW
     if enc == abi.ENC_RLE and t == abi.T_CHAR:
         return abi.ENC_DICT
