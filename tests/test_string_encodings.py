@@ -140,3 +140,40 @@ def test_gpu_parity_filter_and_group(enc):
     # pure filter too (bitmap/row-id path)
     assert eng.filter(h, filt) == res_cpu.rows_passed
     eng.close()
+
+
+@pytest.mark.gpu
+def test_jit_group_spill_past_wg_table():
+    """>16 distinct groups through the JIT path: per-block dict products
+    stay small (eligible) but the union exceeds the 16-slot workgroup
+    table, exercising the cold global-table spill pass."""
+    from oceanbase_amd.engine import GpuEngine
+    from test_gpu_parity import _manual_blockset
+    schema = oracle.make_schema([(abi.T_CHAR, 0, 0, 2), (abi.T_INT, 0, 0, 8)])
+    rng = np.random.default_rng(31)
+    blocks = []
+    all_keys, all_q = [], []
+    for bi in range(8):  # 8 blocks x 4 disjoint keys = 32 groups
+        rows = 600
+        keys = [bytes([65 + bi, 97 + j]) for j in range(4)]
+        pick = rng.integers(0, 4, rows)
+        a = np.frombuffer(b"".join(keys[p] for p in pick), dtype=np.uint8)
+        q = rng.integers(1, 100, rows, dtype=np.int64)
+        all_keys.extend(keys[p] for p in pick)
+        all_q.append(q)
+        blocks.append(oracle.encode_block(
+            schema, [a.copy(), q.view(np.uint8)],
+            [abi.ENC_DICT, abi.ENC_RAW], None))
+    bs = _manual_blockset(schema, blocks)
+    bs.total_rows = 8 * 600
+    eng = GpuEngine(0)
+    h = eng.load(bs)
+    agg = abi.make_agg([0], [dict(kind=abi.AGG_COUNT),
+                             dict(kind=abi.AGG_SUM, col_a=1)])
+    res_gpu = eng.scan_filter_agg(h, None, agg)
+    assert eng.last_jit(), "expected the JIT path for this plan shape"
+    res_cpu = oracle.scan_filter_agg(bs, None, agg)
+    assert res_gpu.n_groups == 32 == res_cpu.n_groups
+    assert sorted(abi.result_rows(res_gpu, 2)) == sorted(
+        abi.result_rows(res_cpu, 2))
+    eng.close()
