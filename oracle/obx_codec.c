@@ -284,6 +284,37 @@ static int encode_column(enc_buf *mb, obx_col_header *ch,
     else if (runs * 8 <= rows) enc = OBX_ENC_RLE;
     else if (nd <= 64 && nd * 4 <= rows) enc = OBX_ENC_DICT;
     else if (sc == OBX_SC_INT && null_cnt < rows) enc = OBX_ENC_INTEGER_BASE_DIFF;
+    else if (sc == OBX_SC_STRING && null_cnt < rows) {
+      /* char columns with wide dictionaries: prefer the string
+         transforms (the reference's cost ranking covers these the same
+         way, ob_encoding_util.h:270-303) — STRING_DIFF when some byte
+         positions are common, else HEX_PACKING when <=16 distinct chars */
+      int vary_any = 0, common_any = 0;
+      uint8_t seen[256]; memset(seen, 0, sizeof(seen));
+      uint32_t nch = 0; int first = -1;
+      for (uint32_t r = 0; r < rows; r++) {
+        if (null_at(nulls, r)) continue;
+        if (first < 0) first = (int)r;
+        for (int i = 0; i < len; i++) {
+          uint8_t cc = data[(size_t)r * len + i];
+          if (!seen[cc]) { seen[cc] = 1; nch++; }
+        }
+      }
+      for (int i = 0; first >= 0 && i < len; i++) {
+        int v = 0;
+        for (uint32_t r = 0; r < rows; r++) {
+          if (null_at(nulls, r)) continue;
+          if (data[(size_t)r * len + i] != data[(size_t)first * len + i]) {
+            v = 1; break;
+          }
+        }
+        if (v) vary_any = 1; else common_any = 1;
+      }
+      if (first >= 0 && common_any) enc = OBX_ENC_STRING_DIFF;
+      else if (first >= 0 && nch <= 16) enc = OBX_ENC_HEX_PACKING;
+      else enc = OBX_ENC_RAW;
+      (void)vary_any;
+    }
     else enc = OBX_ENC_RAW;
     if (enc == OBX_ENC_INTEGER_BASE_DIFF) {
       /* int-diff only pays when delta width < raw width; fall back to raw
