@@ -149,6 +149,10 @@ __device__ __forceinline__ col_ctx make_col_ctx(const dev_col &c,
         x.sext_sh = (uint8_t)(64 - 8 * ((c.flags & OBX_DF_SIGNED) ? c.tss
                                                                   : c.width));
       if (x.sext_sh == 0 || x.sext_sh >= 64) x.sext_sh = 0;
+      /* aligned 8-byte raw (Q1's l_extendedprice shape): one aligned
+         64-bit load replaces the generic bit-funnel read */
+      if (x.W == 64 && x.sext_sh == 0 && ((blk_bit + x.data_bit) & 63) == 0)
+        x.kind = 5;
       break;
     case OBX_D_DICT:
       x.kind = 1;
@@ -220,6 +224,15 @@ __device__ __forceinline__ int64_t ctx_value(const blk_view &bv,
     case 3: { /* CONST, no exceptions */
       if (x.count == 0) { null_out = true; return 0; }
       return x.base;
+    }
+    case 5: { /* aligned raw64 */
+      if (x.has_ext && bit_read_at(bv.base,
+                                   bv.rbase_bit + x.ext_bit +
+                                       (uint64_t)r * x.ext_w, x.ext_w)) {
+        null_out = true; return 0;
+      }
+      const uint64_t byte0 = (bv.rbase_bit + x.data_bit) >> 3;
+      return *(const int64_t *)(bv.base + byte0 + ((uint64_t)r << 3));
     }
     default:
       return 0; /* caller uses the slow path for kind 4 */
