@@ -177,3 +177,26 @@ def test_jit_group_spill_past_wg_table():
     assert sorted(abi.result_rows(res_gpu, 2)) == sorted(
         abi.result_rows(res_cpu, 2))
     eng.close()
+
+
+def test_hex_nibble_order_matches_reference():
+    """Recreates TEST(ObHexStringMap, store_order) from the reference's
+    unittest/storage/blocksstable/encoding/test_hex.cpp:17-40: packing the
+    digits '0'..'9' and hex-printing the packed bytes must reproduce the
+    input — pinning both the ascending-char index map and the
+    high-nibble-first pack order."""
+    schema = oracle.make_schema([(abi.T_CHAR, 0, 0, 8)])
+    a = np.frombuffer(b"01234567", dtype=np.uint8).copy()
+    blk = oracle.encode_block(schema, [a], [abi.ENC_HEX], None)
+    pb = pymodel.Block(blk, [(abi.T_CHAR, 0, 0, 8)])
+    ch = pb.col_headers[0]
+    base = pb.meta_base + ch["offset"]
+    import struct
+    ver, nch, ssize = struct.unpack_from("<BBH", pb.data, base)
+    assert nch == 8 and ssize == 8
+    chars = bytes(pb.data[base + 4:base + 4 + nch])
+    assert chars == b"01234567"  # build_index: ascending char order
+    row = bytes(pb.data[base + ch["length"]:base + ch["length"] + 4])
+    # '0'..'7' map to nibbles 0..7; high nibble first => hex print equals
+    # the input string, exactly as the reference test asserts
+    assert row.hex() == "01234567"
