@@ -71,6 +71,9 @@ enum obx_encoding {
                                     (ob_string_diff_encoder.h) */
   OBX_ENC_HEX_PACKING = 6,       /* fixed char, <=16 distinct chars: nibble
                                     packing (ob_hex_string_encoder.h) */
+  OBX_ENC_COLUMN_EQUAL = 8,      /* equals the nearest previous same-shape
+                                    column except listed exception rows
+                                    (ob_column_equal_encoder.h) */
   OBX_ENC_MAX = 10,
   OBX_ENC_AUTO = 255,            /* writer picks (cost-ranked like
                                     ob_encoding_util.h:270-303) */

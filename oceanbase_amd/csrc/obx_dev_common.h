@@ -424,6 +424,38 @@ __device__ __noinline__ int64_t col_value(const blk_view bv,
 }
 
 /* generic VALUE-mode leaf (operands order-mapped by the host for char) */
+/* col_value with block context: handles COLUMN_EQUAL (exception rows,
+ * else the reference column's value — one level deep by construction,
+ * ob_column_equal_encoder.h; cold generic path). */
+__device__ __noinline__ int64_t col_value2(const blk_view bv,
+                                           const dev_block &blk,
+                                           const dev_col &c, uint32_t r,
+                                           bool &null_out) {
+  if (c.enc == OBX_D_EQUAL) {
+    null_out = false;
+#pragma clang loop unroll(disable)
+    for (uint32_t i = 0; i < c.runs; i++) {
+      uint64_t rid = bit_read(bv, (c.dict_byte + (uint64_t)i * c.rib) * 8,
+                              (uint32_t)c.rib * 8);
+      if (rid == r) {
+        if (bit_read(bv, c.aux_byte * 8 + i, 1)) {
+          null_out = true;
+          return 0;
+        }
+        uint64_t v = bit_read(
+            bv,
+            (c.aux_byte + (c.runs + 7) / 8 + (uint64_t)i * c.datum_len) * 8,
+            (uint32_t)c.datum_len * 8);
+        if (c.flags & OBX_DF_STRING) return (int64_t)v;
+        return sext(v, (c.flags & OBX_DF_SIGNED) ? c.tss : c.datum_len);
+      }
+      if (rid > r) break;
+    }
+    return col_value(bv, blk.cols[c.width], r, null_out); /* width=ref col */
+  }
+  return col_value(bv, c, r, null_out);
+}
+
 __device__ __forceinline__ bool leaf_value_match(const dev_leaf &lf, int64_t v,
                                                  bool isn) {
   if (lf.op == 8) return isn;
@@ -471,12 +503,12 @@ __device__ __noinline__ bool leaf_match(const blk_view bv,
     }
     case OBX_LEAF_NULL: {
       bool isn;
-      (void)col_value(bv, c, r, isn);
+      (void)col_value2(bv, blk, c, r, isn);
       return isn != (bool)blf.invert;
     }
     default: {
       bool isn;
-      int64_t v = col_value(bv, c, r, isn);
+      int64_t v = col_value2(bv, blk, c, r, isn);
       return leaf_value_match(plf, v, isn);
     }
   }

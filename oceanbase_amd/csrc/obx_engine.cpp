@@ -283,6 +283,16 @@ static int parse_block(const obx_col_schema *cols, uint16_t n_cols,
         else dc->data_bit = (data0 + (ext_bits + 7) / 8) * 8;
         break;
       }
+      case OBX_COL_EQUAL: {
+        const obx_coleq_meta *em = (const obx_coleq_meta *)col_host;
+        if (em->ref_col >= n_cols) return OBX_INVALID_ARGUMENT;
+        dc->runs = em->exc_cnt;
+        dc->rib = em->rib;
+        dc->width = (uint8_t)em->ref_col; /* ref column index */
+        dc->dict_byte = col_base + sizeof(obx_coleq_meta); /* exc row_ids */
+        dc->aux_byte = dc->dict_byte + (uint64_t)em->exc_cnt * em->rib;
+        break;
+      }
       case OBX_COL_HEX_PACKING: {
         const obx_hex_meta *hm = (const obx_hex_meta *)col_host;
         dc->width = (uint8_t)((hm->string_size + 1) / 2); /* row stride B */
@@ -356,6 +366,7 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
         case OBX_D_DICT: cls = 1; break;
         case OBX_D_INTDIFF: cls = 2; break;
         case OBX_D_CONST: cls = dc.runs == 0 ? 3 : 4; break;
+        case OBX_D_EQUAL: cls = 5; break; /* span: no device group keys */
         default: cls = 4; break;
       }
       if (dc.flags & OBX_DF_STRING) cls |= 0x10;
@@ -512,6 +523,17 @@ static int prep_query(obx_gpu_ctx *ctx, obx_handle &h,
           da->ic = need(e->col_c);
           da->one_c = P10[h.cols[e->col_c].scale];
         }
+      }
+    }
+    /* group keys decode via col_value (no block context): COLUMN_EQUAL
+       group columns would misdecode — reject them (value/filter columns
+       on COLUMN_EQUAL are fully supported via col_value2) */
+    if (!h.col_class.empty()) {
+      for (uint32_t g = 0; g < ph.n_group_cols; g++) {
+        uint16_t col = ph.need_cols[ph.group_idx[g]];
+        for (uint32_t b2 = 0; b2 < h.n_blocks; b2++)
+          if ((h.col_class[(size_t)b2 * h.n_cols + col] & 0xF) == 5)
+            return OBX_NOT_SUPPORTED;
       }
     }
   }

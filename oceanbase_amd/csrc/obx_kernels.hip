@@ -365,7 +365,7 @@ __device__ void scan_filter_agg_body(
           case 0: { /* COUNT(col) */
             P3_LOOP({
               bool na;
-              (void)((ca.kind == 4) ? col_value(bv, *da, r, na)
+              (void)((ca.kind == 4) ? col_value2(bv, cur, *da, r, na)
                                     : ctx_value(bv, ca, r, na));
               if (!na)
                 atomicAdd(&tab.cell[s][a][stripe][0], 1ull);
@@ -375,7 +375,7 @@ __device__ void scan_filter_agg_body(
           case 1: { /* SUM */
             P3_LOOP({
               bool na;
-              int64_t va = (ca.kind == 4) ? col_value(bv, *da, r, na)
+              int64_t va = (ca.kind == 4) ? col_value2(bv, cur, *da, r, na)
                                           : ctx_value(bv, ca, r, na);
               if (!na)
                 lds_acc_i128(tab.cell[s][a][stripe], i128_from_i64(va));
@@ -386,7 +386,7 @@ __device__ void scan_filter_agg_body(
             const bool is_min = ag.kind == 2;
             P3_LOOP({
               bool na;
-              int64_t va = (ca.kind == 4) ? col_value(bv, *da, r, na)
+              int64_t va = (ca.kind == 4) ? col_value2(bv, cur, *da, r, na)
                                           : ctx_value(bv, ca, r, na);
               if (!na) {
                 cas_minmax(&tab.cell[s][a][stripe][0], va, is_min);
@@ -399,11 +399,11 @@ __device__ void scan_filter_agg_body(
             if (fuse_p3) {
               P3_LOOP({
                 bool na, nb, nc;
-                int64_t va = (ca.kind == 4) ? col_value(bv, *da, r, na)
+                int64_t va = (ca.kind == 4) ? col_value2(bv, cur, *da, r, na)
                                             : ctx_value(bv, ca, r, na);
-                int64_t vb = (cb.kind == 4) ? col_value(bv, *db, r, nb)
+                int64_t vb = (cb.kind == 4) ? col_value2(bv, cur, *db, r, nb)
                                             : ctx_value(bv, cb, r, nb);
-                int64_t vc = (cc.kind == 4) ? col_value(bv, *dc2, r, nc)
+                int64_t vc = (cc.kind == 4) ? col_value2(bv, cur, *dc2, r, nc)
                                             : ctx_value(bv, cc, r, nc);
                 if (!na && !nb) {
                   i128v p2 = i128_mul_i64(va, ag.one_b - vb);
@@ -416,9 +416,9 @@ __device__ void scan_filter_agg_body(
             } else {
               P3_LOOP({
                 bool na, nb;
-                int64_t va = (ca.kind == 4) ? col_value(bv, *da, r, na)
+                int64_t va = (ca.kind == 4) ? col_value2(bv, cur, *da, r, na)
                                             : ctx_value(bv, ca, r, na);
-                int64_t vb = (cb.kind == 4) ? col_value(bv, *db, r, nb)
+                int64_t vb = (cb.kind == 4) ? col_value2(bv, cur, *db, r, nb)
                                             : ctx_value(bv, cb, r, nb);
                 if (!na && !nb)
                   lds_acc_i128(tab.cell[s][a][stripe],
@@ -430,11 +430,11 @@ __device__ void scan_filter_agg_body(
           case 5: { /* SUM_PROD3 standalone */
             P3_LOOP({
               bool na, nb, nc;
-              int64_t va = (ca.kind == 4) ? col_value(bv, *da, r, na)
+              int64_t va = (ca.kind == 4) ? col_value2(bv, cur, *da, r, na)
                                           : ctx_value(bv, ca, r, na);
-              int64_t vb = (cb.kind == 4) ? col_value(bv, *db, r, nb)
+              int64_t vb = (cb.kind == 4) ? col_value2(bv, cur, *db, r, nb)
                                           : ctx_value(bv, cb, r, nb);
-              int64_t vc = (cc.kind == 4) ? col_value(bv, *dc2, r, nc)
+              int64_t vc = (cc.kind == 4) ? col_value2(bv, cur, *dc2, r, nc)
                                           : ctx_value(bv, cc, r, nc);
               if (!na && !nb && !nc)
                 lds_acc_i128(tab.cell[s][a][stripe],
@@ -447,9 +447,9 @@ __device__ void scan_filter_agg_body(
           case 6: { /* SUM_MUL */
             P3_LOOP({
               bool na, nb;
-              int64_t va = (ca.kind == 4) ? col_value(bv, *da, r, na)
+              int64_t va = (ca.kind == 4) ? col_value2(bv, cur, *da, r, na)
                                           : ctx_value(bv, ca, r, na);
-              int64_t vb = (cb.kind == 4) ? col_value(bv, *db, r, nb)
+              int64_t vb = (cb.kind == 4) ? col_value2(bv, cur, *db, r, nb)
                                           : ctx_value(bv, cb, r, nb);
               if (!na && !nb)
                 lds_acc_i128(tab.cell[s][a][stripe], i128_mul_i64(va, vb));
@@ -777,7 +777,7 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_decode(
     const uint64_t row_start = dev_block_row_start(&cur);
     for (uint32_t r = tid; r < rows; r += WG) {
       bool isn;
-      int64_t v = col_value(bv, cur.cols[col], r, isn);
+      int64_t v = col_value2(bv, cur, cur.cols[col], r, isn);
       uint64_t uv = isn ? 0 : (uint64_t)v;
       uint8_t *dst = out + (row_start + r) * datum_len;
       for (uint32_t i = 0; i < datum_len; i++)
@@ -1126,13 +1126,13 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_agg_pass(
       bool na = false, nb = false, nc = false;
       int64_t va = 0, vb = 0, vc = 0;
       if (ag.ia != 0xFF)
-        va = (ca.kind == 4) ? col_value(bv, *da, r, na)
+        va = (ca.kind == 4) ? col_value2(bv, cur, *da, r, na)
                             : ctx_value(bv, ca, r, na);
       if (ag.ib != 0xFF)
-        vb = (cb.kind == 4) ? col_value(bv, *db, r, nb)
+        vb = (cb.kind == 4) ? col_value2(bv, cur, *db, r, nb)
                             : ctx_value(bv, cb, r, nb);
       if (dc2)
-        vc = (cc.kind == 4) ? col_value(bv, *dc2, r, nc)
+        vc = (cc.kind == 4) ? col_value2(bv, cur, *dc2, r, nc)
                             : ctx_value(bv, cc, r, nc);
       switch (ag.kind) {
         case 0: /* COUNT(col) */

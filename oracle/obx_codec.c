@@ -250,7 +250,9 @@ static uint64_t diff_get(void *c, uint32_t r) {
 static int encode_column(enc_buf *mb, obx_col_header *ch,
                          const obx_col_schema *cs, const uint8_t *data,
                          const uint8_t *nulls, uint32_t rows, uint8_t enc,
-                         int block_ext_bit) {
+                         int block_ext_bit, const obx_col_schema *all_cols,
+                         const uint8_t *const *all_data,
+                         const uint8_t *const *all_nulls, uint16_t col_idx) {
   const int len = cs->len;
   const int sc = obx_store_class(cs->obj_type);
   const int64_t tss = obx_type_store_size(cs->obj_type); /* -1 for dec/char */
@@ -693,6 +695,65 @@ static int encode_column(enc_buf *mb, obx_col_header *ch,
       if (rc) return rc;
       break;
     }
+    case OBX_ENC_COLUMN_EQUAL: {
+      /* ObColumnEqualEncoder (ob_column_equal_encoder.{h,cpp}): equals the
+         nearest previous column with the same shape except at exception
+         rows; is_datum_equal treats both-null as equal (h:83-101). */
+      int ref = -1;
+      for (int j = (int)col_idx - 1; j >= 0; j--) {
+        if (all_cols[j].len == cs->len &&
+            obx_store_class(all_cols[j].obj_type) == sc) { ref = j; break; }
+      }
+      if (ref < 0 || !all_data) return OBX_NOT_SUPPORTED;
+      const uint8_t *rd = all_data[ref];
+      const uint8_t *rn = all_nulls ? all_nulls[ref] : NULL;
+      uint32_t exc = 0;
+      for (uint32_t r = 0; r < rows; r++) {
+        int ln = null_at(nulls, r), rnl = null_at(rn, r);
+        if (ln != rnl ||
+            (!ln && memcmp(data + (size_t)r * len, rd + (size_t)r * len,
+                           (size_t)len) != 0))
+          exc++;
+      }
+      if (exc * 4 > rows) return OBX_NOT_SUPPORTED; /* not worth it */
+      int rib = (int)obx_byte_packed_int_size(rows ? rows - 1 : 0);
+      obx_coleq_meta em;
+      em.version = 0;
+      em.ref_col = (uint16_t)ref;
+      em.exc_cnt = (uint16_t)exc;
+      em.rib = (uint8_t)rib;
+      int64_t meta_size = (int64_t)sizeof(em) + (int64_t)exc * rib +
+                          (exc + 7) / 8 + (int64_t)exc * len;
+      if (mb->len + meta_size > mb->cap) return OBX_BUF_NOT_ENOUGH;
+      uint8_t *mp = mb->p + mb->len;
+      memcpy(mp, &em, sizeof(em)); mp += sizeof(em);
+      uint8_t *rid_p = mp;
+      uint8_t *nb_p = rid_p + (size_t)exc * rib;
+      uint8_t *dat_p = nb_p + (exc + 7) / 8;
+      memset(nb_p, 0, (exc + 7) / 8);
+      uint32_t k = 0;
+      for (uint32_t r = 0; r < rows; r++) {
+        int ln = null_at(nulls, r), rnl = null_at(rn, r);
+        if (ln == rnl &&
+            (ln || memcmp(data + (size_t)r * len, rd + (size_t)r * len,
+                          (size_t)len) == 0))
+          continue;
+        memcpy(rid_p + (size_t)k * rib, &r, (size_t)rib);
+        if (ln) {
+          nb_p[k / 8] |= (uint8_t)(1u << (k % 8));
+          memset(dat_p + (size_t)k * len, 0, (size_t)len);
+        } else {
+          memcpy(dat_p + (size_t)k * len, data + (size_t)r * len,
+                 (size_t)len);
+        }
+        k++;
+      }
+      mb->len += meta_size;
+      ch->type = OBX_COL_EQUAL;
+      ch->attr |= OBX_COL_ATTR_FIX_LENGTH;
+      ch->length = (uint32_t)meta_size;
+      break;
+    }
     default:
       return OBX_NOT_SUPPORTED;
   }
@@ -729,7 +790,8 @@ int64_t obx_encode_block(const obx_col_schema *cols, uint16_t n_cols,
     uint8_t enc = enc_request ? enc_request[c] : OBX_ENC_AUTO;
     int rc = encode_column(&mb, &chp[c], &cols[c], col_data[c],
                            null_bitmaps ? null_bitmaps[c] : NULL, row_count,
-                           enc, ext_bit ? 1 : 0);
+                           enc, ext_bit ? 1 : 0,
+                           cols, col_data, null_bitmaps, c);
     if (rc) return rc;
   }
 
@@ -801,13 +863,29 @@ typedef struct col_dec {
   const uint8_t *str_chars;  /* hex char array */
   const uint8_t *str_descs;  /* diff descs */
   const uint8_t *str_common; /* common bytes */
+  /* COLUMN_EQUAL: block context for the nested ref-column decode */
+  const obx_coleq_meta *em;
+  const uint8_t *exc_rid, *exc_nb, *exc_dat;
+  const obx_col_header *chp0; /* column header array base */
+  const uint8_t *meta_region;
 } col_dec;
+
+struct col_dec;
+static int col_dec_row(const col_dec *d, const obx_micro_header *h,
+                       const obx_col_schema *cs, uint32_t r,
+                       int64_t *out, int *is_null);
 
 static int col_dec_init(col_dec *d, const obx_micro_header *h,
                         const obx_col_header *ch, const uint8_t *meta_region) {
   memset(d, 0, sizeof(*d));
   d->ch = ch;
   d->meta = meta_region + ch->offset;
+  /* block context for span encodings (COLUMN_EQUAL's ref decode); the
+     header array immediately precedes the meta region */
+  d->chp0 = (const obx_col_header *)(meta_region -
+                                     (size_t)h->column_count *
+                                         sizeof(obx_col_header));
+  d->meta_region = meta_region;
   int ext = (ch->attr & OBX_COL_ATTR_HAS_EXTEND_VALUE) ? 1 : 0;
   int evb = obx_hdr_extend_value_bit(h);
   switch (ch->type) {
@@ -871,6 +949,16 @@ static int col_dec_init(col_dec *d, const obx_micro_header *h,
       d->data = d->meta + ch->length;
       int64_t ext_bits = ext ? (int64_t)evb * h->row_count : 0;
       d->fix_off = (ext_bits + 7) / 8;
+      break;
+    }
+    case OBX_COL_EQUAL: {
+      d->em = (const obx_coleq_meta *)d->meta;
+      if (d->em->ref_col >= h->column_count) return OBX_INVALID_ARGUMENT;
+      if (d->chp0[d->em->ref_col].type == OBX_COL_EQUAL)
+        return OBX_NOT_SUPPORTED; /* one level only */
+      d->exc_rid = d->meta + sizeof(obx_coleq_meta);
+      d->exc_nb = d->exc_rid + (size_t)d->em->exc_cnt * d->em->rib;
+      d->exc_dat = d->exc_nb + (d->em->exc_cnt + 7) / 8;
       break;
     }
     case OBX_COL_STRING_DIFF: {
@@ -1011,6 +1099,31 @@ static int col_dec_row(const col_dec *d, const obx_micro_header *h,
       if (sc == OBX_SC_INT) v = obx_sign_extend(v, tss, 1);
       *out = (int64_t)v;
       return 0;
+    }
+    case OBX_COL_EQUAL: {
+      /* exception lookup (ascending row_ids), else the ref column's row */
+      for (uint32_t i = 0; i < d->em->exc_cnt; i++) {
+        uint64_t rid = 0;
+        memcpy(&rid, d->exc_rid + (size_t)i * d->em->rib,
+               (size_t)d->em->rib);
+        if (rid == r) {
+          if ((d->exc_nb[i / 8] >> (i % 8)) & 1) {
+            *is_null = 1; *out = 0; return 0;
+          }
+          uint64_t v = 0;
+          memcpy(&v, d->exc_dat + (size_t)i * cs->len, (size_t)cs->len);
+          if (sc == OBX_SC_INT) v = obx_sign_extend(v, tss, 1);
+          else if (sc == OBX_SC_DECIMAL) v = obx_sign_extend(v, cs->len, 1);
+          *out = (int64_t)v;
+          return 0;
+        }
+        if (rid > r) break;
+      }
+      col_dec rd;
+      int rc2 = col_dec_init(&rd, h, &d->chp0[d->em->ref_col],
+                             d->meta_region);
+      if (rc2) return rc2;
+      return col_dec_row(&rd, h, cs, r, out, is_null);
     }
     case OBX_COL_HEX_PACKING: {
       if (col_dec_is_null(d, h, r)) { *is_null = 1; *out = 0; return 0; }

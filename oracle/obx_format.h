@@ -97,6 +97,7 @@ enum {
   OBX_COL_INTEGER_BASE_DIFF = 4,
   OBX_COL_STRING_DIFF = 5,
   OBX_COL_HEX_PACKING = 6,
+  OBX_COL_EQUAL = 8,
   /* 5..9: string encoders, out of scope this round */
 };
 enum {
@@ -177,6 +178,22 @@ typedef struct obx_sdiff_meta {
   uint16_t string_size;  /* N */
   uint8_t diff_desc_cnt;
 } obx_sdiff_meta;        /* 5 bytes packed */
+
+/* COLUMN_EQUAL meta (ObColumnEqualMetaHeader,
+   ob_column_equal_encoder.h:24-40): this column equals ref_col except at
+   the exception rows. The reference stores exceptions through its
+   ObBitMapMeta family; our container reuses the CONST-style exception
+   list instead (ascending row_ids + fixed-size datums) — same
+   information, simpler layout, documented divergence. Exception NULLs
+   are a bitmap between the row_ids and the datums. */
+typedef struct obx_coleq_meta {
+  uint8_t version;       /* 0 */
+  uint16_t ref_col;      /* schema index of the reference column */
+  uint16_t exc_cnt;      /* exception rows */
+  uint8_t rib;           /* row-id byte width */
+  /* then: exc row_ids (rib B each, ascending); exc null bitmap
+     ceil(exc_cnt/8) B; exc datums (schema len B each, 0 for nulls) */
+} obx_coleq_meta;        /* 6 bytes packed */
 
 typedef struct obx_intdiff_meta {
   uint8_t version;       /* 0 */

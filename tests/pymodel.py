@@ -13,6 +13,7 @@ All values are returned as Python ints (already sign-extended) or None (NULL).
 import struct
 
 ENC_RAW, ENC_DICT, ENC_RLE, ENC_CONST, ENC_INT_DIFF, ENC_SDIFF, ENC_HEX = range(7)
+ENC_COLEQ = 8
 
 ATTR_FIX = 1
 ATTR_EXT = 2
@@ -276,6 +277,27 @@ class Block:
                     nib = (rp[i // 2] >> (((i + 1) % 2) * 4)) & 0xF
                     v |= chars[nib] << (8 * i)
                 out.append(v)
+        elif t == ENC_COLEQ:
+            # obx_coleq_meta: version u8, ref_col u16, exc_cnt u16, rib u8
+            ver, ref_col, exc_cnt, rib = struct.unpack_from("<BHHB", d, base)
+            p0 = base + 6
+            rids = [int.from_bytes(d[p0 + i * rib:p0 + (i + 1) * rib],
+                                   "little") for i in range(exc_cnt)]
+            nb0 = p0 + exc_cnt * rib
+            datp = nb0 + (exc_cnt + 7) // 8
+            ref_vals = self.decode_col(ref_col)
+            out = list(ref_vals)
+            for i, rid in enumerate(rids):
+                if (d[nb0 + i // 8] >> (i % 8)) & 1:
+                    out[rid] = None
+                else:
+                    v = int.from_bytes(d[datp + i * dlen:
+                                         datp + (i + 1) * dlen], "little")
+                    if sc == SC_INT:
+                        v = sign_extend(v, tss)
+                    elif sc == SC_DECIMAL:
+                        v = sign_extend(v, dlen)
+                    out[rid] = v
         else:
             raise NotImplementedError(t)
         return out

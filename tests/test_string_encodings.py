@@ -200,3 +200,88 @@ def test_hex_nibble_order_matches_reference():
     # '0'..'7' map to nibbles 0..7; high nibble first => hex print equals
     # the input string, exactly as the reference test asserts
     assert row.hex() == "01234567"
+
+
+# ---- COLUMN_EQUAL (span encoder: this column == a previous column except
+# at exception rows; ObColumnEqualEncoder, ob_column_equal_encoder.h) ----
+
+def _coleq_blockset(rows=3000, seed=41, nulls=True):
+    schema = oracle.make_schema([(abi.T_INT, 0, 0, 8), (abi.T_INT, 0, 0, 8)])
+    rng = np.random.default_rng(seed)
+    a = rng.integers(-10**9, 10**9, rows)
+    b = a.copy()
+    exc = rng.choice(rows, rows // 40, replace=False)
+    b[exc] += rng.integers(1, 100, len(exc))
+    na = np.zeros((rows + 7) // 8, dtype=np.uint8)
+    nb = np.zeros((rows + 7) // 8, dtype=np.uint8)
+    if nulls:
+        na[0] |= 1      # row 0: A null, B not -> exception
+        nb[1] |= 1      # row 8: B null only -> null exception
+    return schema, a, b, na, nb
+
+
+def test_column_equal_roundtrip_and_pymodel():
+    schema, a, b, na, nb = _coleq_blockset()
+    rows = len(a)
+    blk = oracle.encode_block(schema, [a.view(np.uint8), b.view(np.uint8)],
+                              [abi.ENC_RAW, abi.ENC_COLUMN_EQUAL], [na, nb])
+    rc, outs, nbs = oracle.decode_block(schema, 2, blk, [1])
+    got = np.frombuffer(outs[0], dtype=np.int64)
+    for r in range(rows):
+        isn = (nbs[0][r // 8] >> (r % 8)) & 1
+        want_null = bool((nb[r // 8] >> (r % 8)) & 1)
+        assert bool(isn) == want_null
+        if not isn:
+            assert got[r] == b[r]
+    pb = pymodel.Block(blk, [(abi.T_INT, 0, 0, 8), (abi.T_INT, 0, 0, 8)])
+    vals = pb.decode_col(1)
+    for r in range(rows):
+        want = None if (nb[r // 8] >> (r % 8)) & 1 else int(b[r])
+        assert vals[r] == want
+
+
+def test_column_equal_filter_and_agg_cpu():
+    schema, a, b, na, nb = _coleq_blockset()
+    blk = oracle.encode_block(schema, [a.view(np.uint8), b.view(np.uint8)],
+                              [abi.ENC_RAW, abi.ENC_COLUMN_EQUAL], [na, nb])
+    filt = abi.make_filter([dict(col=1, op=abi.OP_GT, lo=0)])
+    bits, passed = oracle.filter_block(schema, 2, blk, filt)
+    want = sum(1 for r in range(len(b))
+               if not (nb[r // 8] >> (r % 8)) & 1 and b[r] > 0)
+    assert passed == want
+
+
+@pytest.mark.gpu
+def test_column_equal_gpu_parity():
+    from oceanbase_amd.engine import GpuEngine
+    from test_gpu_parity import _manual_blockset
+    schema, a, b, na, nb = _coleq_blockset()
+    rows = len(a)
+    blocks = []
+    for s in range(0, rows, 1000):
+        e = min(s + 1000, rows)
+        naw = np.zeros((e - s + 7) // 8, dtype=np.uint8)
+        nbw = np.zeros((e - s + 7) // 8, dtype=np.uint8)
+        for r in range(s, e):
+            if (na[r // 8] >> (r % 8)) & 1:
+                naw[(r - s) // 8] |= 1 << ((r - s) % 8)
+            if (nb[r // 8] >> (r % 8)) & 1:
+                nbw[(r - s) // 8] |= 1 << ((r - s) % 8)
+        blocks.append(oracle.encode_block(
+            schema, [a[s:e].view(np.uint8), b[s:e].view(np.uint8)],
+            [abi.ENC_RAW, abi.ENC_COLUMN_EQUAL], [naw, nbw]))
+    bs = _manual_blockset(schema, blocks)
+    bs.total_rows = rows
+    eng = GpuEngine(0)
+    h = eng.load(bs)
+    # filter on the COLUMN_EQUAL column + SUM over it
+    filt = abi.make_filter([dict(col=1, op=abi.OP_GT, lo=0)])
+    agg = abi.make_agg([], [dict(kind=abi.AGG_COUNT),
+                            dict(kind=abi.AGG_SUM, col_a=1)])
+    res_gpu = eng.scan_filter_agg(h, filt, agg)
+    res_cpu = oracle.scan_filter_agg(bs, filt, agg)
+    assert res_gpu.rows_passed == res_cpu.rows_passed
+    assert sorted(abi.result_rows(res_gpu, 2)) == sorted(
+        abi.result_rows(res_cpu, 2))
+    assert eng.filter(h, filt) == res_cpu.rows_passed
+    eng.close()
