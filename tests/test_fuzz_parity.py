@@ -104,11 +104,24 @@ def test_fuzz_case(case_seed):
 
     blocks = []
     distinct = [set() for _ in range(n_cols)]
+    coleq_cols = set()
     r0 = 0
     while r0 < rows_total:
         n = min(rows_per_block, rows_total - r0)
         arrays, nulls = [], []
         for c in range(n_cols):
+            if c in coleq_cols:
+                prev = next(j for j in reversed(range(c))
+                            if tspecs[j][3] == tspecs[c][3]
+                            and _sc(tspecs[j]) == _sc(tspecs[c])
+                            and j not in coleq_cols)
+                arr = arrays[prev].copy()
+                ln = tspecs[c][3]
+                for r in rng.choice(n, max(1, n // 50), replace=False):
+                    arr[r * ln] ^= 0x5
+                arrays.append(arr)
+                nulls.append(nulls[prev])
+                continue
             arr, v, length = _gen_column(rng, tspecs[c], n, styles[c])
             if len(distinct[c]) <= 40:
                 distinct[c].update(np.unique(v)[:41].tolist())
@@ -138,7 +151,8 @@ def test_fuzz_case(case_seed):
     filt = abi.make_filter(leaves)
 
     # random aggregate plan: group by <=2 byte-compatible cols, <=5 aggs
-    gcands = [c for c in range(n_cols) if len(distinct[c]) <= 10]
+    gcands = [c for c in range(n_cols)
+              if len(distinct[c]) <= 10 and c not in coleq_cols]
     rng.shuffle(gcands)
     group = []
     klen = 0
