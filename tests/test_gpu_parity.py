@@ -349,3 +349,36 @@ def test_jit_engages_on_q1_shape():
     assert sorted(abi.result_rows(res, 4)) == sorted(
         abi.result_rows(ores, 4))
     eng.close()
+
+
+@pytest.mark.gpu
+def test_jit_multi_window_blocks():
+    """Blocks with more than 2048 rows split into LDS windows inside one
+    kernel launch; exercise that path through the JIT kernel (narrow
+    columns keep 4000-row blocks under the LDS stage size)."""
+    from oceanbase_amd.engine import GpuEngine
+    rows = 12000
+    schema = oracle.make_schema([(abi.T_CHAR, 0, 0, 1), (abi.T_INT, 0, 0, 8)])
+    rng = np.random.default_rng(77)
+    flag = rng.choice(np.frombuffer(b"ANR", dtype=np.uint8), rows)
+    qty = rng.integers(1, 50, rows, dtype=np.int64)
+    blocks = []
+    for s in range(0, rows, 4000):
+        e = s + 4000
+        blocks.append(oracle.encode_block(
+            schema, [flag[s:e].copy(), qty[s:e].view(np.uint8)],
+            [abi.ENC_DICT, abi.ENC_DICT], None))
+    bs = _manual_blockset(schema, blocks)
+    bs.total_rows = rows
+    eng = GpuEngine(0)
+    h = eng.load(bs)
+    filt = abi.make_filter([dict(col=1, op=abi.OP_LT, lo=40)])
+    agg = abi.make_agg([0], [dict(kind=abi.AGG_COUNT),
+                             dict(kind=abi.AGG_SUM, col_a=1)])
+    res_gpu = eng.scan_filter_agg(h, filt, agg)
+    assert eng.last_jit(), "expected the JIT path (narrow dict columns)"
+    res_cpu = oracle.scan_filter_agg(bs, filt, agg)
+    assert res_gpu.rows_passed == res_cpu.rows_passed
+    assert sorted(abi.result_rows(res_gpu, 2)) == sorted(
+        abi.result_rows(res_cpu, 2))
+    eng.close()
