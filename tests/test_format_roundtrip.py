@@ -258,3 +258,27 @@ def test_header_fields():
     assert pb.header_size == 64
     assert pb.row_index_byte == 0          # all columns fix-stored
     assert pb.row_data_offset == len(blk) - 16  # no var region (16 = slack)
+
+
+def test_corrupt_magic_rejected():
+    """A block whose header magic is wrong must be rejected, not decoded
+    (ObMicroBlockHeader::check_header_checksum analog)."""
+    schema = oracle.make_schema([(abi.T_INT, 0, 0, 8)])
+    v = np.arange(100, dtype=np.int64)
+    blk = bytearray(oracle.encode_block(schema, [v.view(np.uint8)],
+                                        [abi.ENC_RAW], None))
+    blk[0] ^= 0xFF  # clobber magic
+    with pytest.raises(RuntimeError):
+        oracle.decode_block(schema, 1, bytes(blk), [0])
+
+
+def test_invalid_combine_program_rejected():
+    """Malformed postfix programs (stack underflow / wrong arity) must be
+    rejected by the oracle filter."""
+    schema = oracle.make_schema([(abi.T_INT, 0, 0, 8)])
+    v = np.arange(1000, dtype=np.int64)
+    blk = oracle.encode_block(schema, [v.view(np.uint8)], [abi.ENC_RAW], None)
+    bad = abi.make_filter([dict(col=0, op=abi.OP_LT, lo=10)],
+                          prog=[0, abi.TOK_AND])  # arity underflow
+    with pytest.raises(RuntimeError):
+        oracle.filter_block(schema, 1, blk, bad)

@@ -382,3 +382,22 @@ def test_jit_multi_window_blocks():
     assert sorted(abi.result_rows(res_gpu, 2)) == sorted(
         abi.result_rows(res_cpu, 2))
     eng.close()
+
+
+@pytest.mark.gpu
+def test_pipeline_agg_path_parity(eng):
+    """OBX_PIPELINE_AGG=1 switches to the filter->group->agg kernel DAG;
+    results must match the fused/JIT path bit-exactly."""
+    import os as _os
+    li = oracle.Lineitem(4, 60000, seed=9)
+    h = eng.load(li.bs)
+    filt, agg = _q1_descs()
+    base = eng.scan_filter_agg(h, filt, agg)
+    _os.environ["OBX_PIPELINE_AGG"] = "1"
+    try:
+        piped = eng.scan_filter_agg(h, filt, agg)
+    finally:
+        del _os.environ["OBX_PIPELINE_AGG"]
+    assert base.rows_passed == piped.rows_passed
+    assert sorted(abi.result_rows(base, 6)) == sorted(
+        abi.result_rows(piped, 6))
