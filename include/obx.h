@@ -71,6 +71,8 @@ enum obx_encoding {
                                     (ob_string_diff_encoder.h) */
   OBX_ENC_HEX_PACKING = 6,       /* fixed char, <=16 distinct chars: nibble
                                     packing (ob_hex_string_encoder.h) */
+  OBX_ENC_STRING_PREFIX = 7,    /* fixed char: prefix table + suffixes
+                                    (ob_string_prefix_encoder.h) */
   OBX_ENC_COLUMN_EQUAL = 8,      /* equals the nearest previous same-shape
                                     column except listed exception rows
                                     (ob_column_equal_encoder.h) */

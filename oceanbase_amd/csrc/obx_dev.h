@@ -50,6 +50,7 @@ enum {
   OBX_D_INTDIFF = 4,
   OBX_D_SDIFF = 5,   /* STRING_DIFF: common/diff byte runs */
   OBX_D_HEX = 6,     /* HEX_PACKING: nibble-packed chars */
+  OBX_D_SPREFIX = 7, /* STRING_PREFIX: prefix table + suffixes */
   OBX_D_EQUAL = 8,   /* COLUMN_EQUAL: ref column + exception rows */
 };
 

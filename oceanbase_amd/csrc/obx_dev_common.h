@@ -338,6 +338,37 @@ __device__ __noinline__ int64_t str_value(const blk_view bv,
     }
     return (int64_t)v;
   }
+  if (c.enc == OBX_D_SPREFIX) {
+    /* prefix ref nibble + suffix (ObStringPrefixCellHeader semantics for
+       fixed-length cells); prefix lengths via the cumulative end index */
+    const bool hexp = c.entry_len > 0;
+    uint32_t ref = (uint32_t)bit_read(bv, row_byte * 8, 8) & 0xF;
+    if (ref >= c.count) return 0;
+    uint64_t pstart = 0, pend = 0;
+    if (ref > 0)
+      pstart = bit_read(bv, (c.aux_byte + (uint64_t)(ref - 1) * c.rib) * 8,
+                        (uint32_t)c.rib * 8);
+    pend = bit_read(bv, (c.aux_byte + (uint64_t)ref * c.rib) * 8,
+                    (uint32_t)c.rib * 8);
+    uint32_t plen = (uint32_t)(pend - pstart);
+    uint64_t pdata = c.aux_byte + (uint64_t)c.count * c.rib;
+#pragma clang loop unroll(disable)
+    for (uint32_t i = 0; i < plen; i++)
+      v |= bit_read(bv, (pdata + pstart + i) * 8, 8) << (8 * i);
+#pragma clang loop unroll(disable)
+    for (uint32_t i = 0; i < c.datum_len - plen; i++) {
+      uint64_t ch8;
+      if (hexp) {
+        uint64_t b8 = bit_read(bv, (row_byte + 1 + i / 2) * 8, 8);
+        ch8 = bit_read(
+            bv, (c.dict_byte + ((b8 >> (((i + 1) % 2) * 4)) & 0xF)) * 8, 8);
+      } else {
+        ch8 = bit_read(bv, (row_byte + 1 + i) * 8, 8);
+      }
+      v |= ch8 << (8 * (plen + i));
+    }
+    return (int64_t)v;
+  }
   /* STRING_DIFF */
   const bool hex = c.entry_len > 0;
   const uint64_t chars_byte = c.dict_byte + c.runs;
@@ -415,7 +446,8 @@ __device__ __noinline__ int64_t col_value(const blk_view bv,
       return (int64_t)((uint64_t)c.base + diff);
     }
     case OBX_D_HEX:
-    case OBX_D_SDIFF: {
+    case OBX_D_SDIFF:
+    case OBX_D_SPREFIX: {
       if (col_is_null_ext(bv, c, r)) { null_out = true; return 0; }
       return str_value(bv, c, r);
     }

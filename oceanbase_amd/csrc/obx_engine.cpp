@@ -283,6 +283,38 @@ static int parse_block(const obx_col_schema *cols, uint16_t n_cols,
         else dc->data_bit = (data0 + (ext_bits + 7) / 8) * 8;
         break;
       }
+      case OBX_COL_STRING_PREFIX: {
+        const obx_sprefix_meta *pm = (const obx_sprefix_meta *)col_host;
+        dc->count = pm->count;
+        dc->entry_len = pm->hex_char_cnt;
+        dc->rib = pm->pib;
+        dc->dict_byte = col_base + sizeof(obx_sprefix_meta); /* hex chars */
+        dc->aux_byte = dc->dict_byte + pm->hex_char_cnt;     /* end index */
+        /* fixed cell stride: 1 + max suffix (raw or nibble-packed) */
+        {
+          const uint8_t *ends = col_host + sizeof(obx_sprefix_meta) +
+                                pm->hex_char_cnt;
+          uint64_t prev = 0, min_pl = ~0ull;
+          for (uint32_t j = 0; j < pm->count; j++) {
+            uint64_t e2 = 0;
+            memcpy(&e2, ends + (size_t)j * pm->pib, pm->pib);
+            if (e2 - prev < min_pl) min_pl = e2 - prev;
+            prev = e2;
+          }
+          uint32_t max_suffix = pm->string_size - (uint32_t)min_pl;
+          dc->width = (uint8_t)(1 + (pm->hex_char_cnt
+                                         ? (max_suffix + 1) / 2
+                                         : max_suffix));
+        }
+        uint64_t data0 = col_base + ch->length;
+        uint64_t ext_bits = has_ext ? (uint64_t)evb * h->row_count : 0;
+        if (has_ext) {
+          dc->flags |= OBX_DF_HAS_EXT;
+          dc->ext_bit = data0 * 8;
+        }
+        dc->data_bit = (data0 + (ext_bits + 7) / 8) * 8;
+        break;
+      }
       case OBX_COL_EQUAL: {
         const obx_coleq_meta *em = (const obx_coleq_meta *)col_host;
         if (em->ref_col >= n_cols) return OBX_INVALID_ARGUMENT;

@@ -695,6 +695,104 @@ static int encode_column(enc_buf *mb, obx_col_header *ch,
       if (rc) return rc;
       break;
     }
+    case OBX_ENC_STRING_PREFIX: {
+      /* ObStringPrefixEncoder (ob_string_prefix_encoder.{h,cpp}); our
+         writer picks the longest equal-length prefix with <=15 distinct
+         values (the reference's ObMultiPrefixTree is writer policy — the
+         FORMAT carries per-prefix lengths via the end-offset index). */
+      if (sc != OBX_SC_STRING || len < 2) return OBX_NOT_SUPPORTED;
+      int first = -1;
+      for (uint32_t r = 0; r < rows; r++)
+        if (!null_at(nulls, r)) { first = (int)r; break; }
+      if (first < 0) return OBX_NOT_SUPPORTED;
+      int plen = 0;
+      uint8_t pfx[15 * 8];
+      uint32_t np = 0;
+      for (int p = len - 1; p >= 1; p--) {
+        np = 0;
+        int ok2 = 1;
+        for (uint32_t r = 0; r < rows && ok2; r++) {
+          if (null_at(nulls, r)) continue;
+          const uint8_t *s2 = data + (size_t)r * len;
+          uint32_t j = 0;
+          for (; j < np; j++)
+            if (memcmp(pfx + (size_t)j * p, s2, (size_t)p) == 0) break;
+          if (j == np) {
+            if (np >= 15) { ok2 = 0; break; }
+            memcpy(pfx + (size_t)np * p, s2, (size_t)p);
+            np++;
+          }
+        }
+        if (ok2) { plen = p; break; }
+      }
+      if (plen < 1) return OBX_NOT_SUPPORTED;
+      const int max_suffix = len - plen;
+      /* suffix hex packability */
+      uint8_t seen[256]; memset(seen, 0, sizeof(seen));
+      uint32_t nch = 0;
+      for (uint32_t r = 0; r < rows && nch <= 16; r++) {
+        if (null_at(nulls, r)) continue;
+        for (int i = plen; i < len; i++) {
+          uint8_t cc = data[(size_t)r * len + i];
+          if (!seen[cc]) { seen[cc] = 1; nch++; }
+        }
+      }
+      int use_hex = nch <= 16 && (max_suffix + 1) / 2 < max_suffix;
+      uint8_t chars[16]; uint8_t idx_of[256]; uint32_t k2 = 0;
+      if (use_hex)
+        for (int i = 0; i < 256; i++)
+          if (seen[i]) { idx_of[i] = (uint8_t)k2; chars[k2++] = (uint8_t)i; }
+      int pib = (int)obx_byte_packed_int_size((uint64_t)np * plen);
+      obx_sprefix_meta pm;
+      pm.version = 0;
+      pm.count = (uint8_t)np;
+      pm.string_size = (uint16_t)len;
+      pm.hex_char_cnt = (uint8_t)(use_hex ? nch : 0);
+      pm.pib = (uint8_t)pib;
+      int64_t meta_size = (int64_t)sizeof(pm) + (use_hex ? nch : 0) +
+                          (int64_t)np * pib + (int64_t)np * plen;
+      if (mb->len + meta_size > mb->cap) return OBX_BUF_NOT_ENOUGH;
+      uint8_t *mp = mb->p + mb->len;
+      memcpy(mp, &pm, sizeof(pm)); mp += sizeof(pm);
+      if (use_hex) { memcpy(mp, chars, nch); mp += nch; }
+      for (uint32_t j = 0; j < np; j++) { /* cumulative END offsets */
+        uint64_t end = (uint64_t)(j + 1) * plen;
+        memcpy(mp, &end, (size_t)pib); mp += pib;
+      }
+      for (uint32_t j = 0; j < np; j++) {
+        memcpy(mp, pfx + (size_t)j * plen, (size_t)plen); mp += plen;
+      }
+      mb->len += meta_size;
+      ch->type = OBX_COL_STRING_PREFIX;
+      ch->attr |= OBX_COL_ATTR_FIX_LENGTH;
+      if (null_cnt) ch->attr |= OBX_COL_ATTR_HAS_EXTEND_VALUE;
+      ch->length = (uint32_t)meta_size;
+      int stride = 1 + (use_hex ? (max_suffix + 1) / 2 : max_suffix);
+      uint8_t *scr = (uint8_t *)calloc((size_t)rows * stride + 1, 1);
+      if (!scr) return OBX_INTERNAL_ERROR;
+      for (uint32_t r = 0; r < rows; r++) {
+        if (null_at(nulls, r)) continue;
+        const uint8_t *s2 = data + (size_t)r * len;
+        uint32_t j = 0;
+        for (; j < np; j++)
+          if (memcmp(pfx + (size_t)j * plen, s2, (size_t)plen) == 0) break;
+        uint8_t *rp = scr + (size_t)r * stride;
+        rp[0] = (uint8_t)j;
+        for (int i = 0; i < max_suffix; i++) {
+          uint8_t cc = s2[plen + i];
+          if (use_hex)
+            rp[1 + i / 2] |= (uint8_t)(idx_of[cc] << (((i + 1) % 2) * 4));
+          else
+            rp[1 + i] = cc;
+        }
+      }
+      fix_store_spec s = { block_ext_bit, null_cnt > 0, 0, stride };
+      int rc = store_fix_region(mb, &s, rows, nulls, NULL, NULL, scr,
+                                stride);
+      free(scr);
+      if (rc) return rc;
+      break;
+    }
     case OBX_ENC_COLUMN_EQUAL: {
       /* ObColumnEqualEncoder (ob_column_equal_encoder.{h,cpp}): equals the
          nearest previous column with the same shape except at exception
@@ -863,6 +961,9 @@ typedef struct col_dec {
   const uint8_t *str_chars;  /* hex char array */
   const uint8_t *str_descs;  /* diff descs */
   const uint8_t *str_common; /* common bytes */
+  /* STRING_PREFIX */
+  const obx_sprefix_meta *pm;
+  const uint8_t *pfx_ends, *pfx_data;
   /* COLUMN_EQUAL: block context for the nested ref-column decode */
   const obx_coleq_meta *em;
   const uint8_t *exc_rid, *exc_nb, *exc_dat;
@@ -951,6 +1052,16 @@ static int col_dec_init(col_dec *d, const obx_micro_header *h,
       d->fix_off = (ext_bits + 7) / 8;
       break;
     }
+    case OBX_COL_STRING_PREFIX: {
+      d->pm = (const obx_sprefix_meta *)d->meta;
+      d->str_chars = d->meta + sizeof(obx_sprefix_meta);
+      d->pfx_ends = d->str_chars + d->pm->hex_char_cnt;
+      d->pfx_data = d->pfx_ends + (size_t)d->pm->count * d->pm->pib;
+      d->data = d->meta + ch->length;
+      int64_t ext_bits = ext ? (int64_t)evb * h->row_count : 0;
+      d->fix_off = (ext_bits + 7) / 8;
+      break;
+    }
     case OBX_COL_EQUAL: {
       d->em = (const obx_coleq_meta *)d->meta;
       if (d->em->ref_col >= h->column_count) return OBX_INVALID_ARGUMENT;
@@ -987,6 +1098,7 @@ static inline int col_dec_is_null(const col_dec *d, const obx_micro_header *h,
     case OBX_COL_INTEGER_BASE_DIFF:
     case OBX_COL_HEX_PACKING:
     case OBX_COL_STRING_DIFF:
+    case OBX_COL_STRING_PREFIX:
       if (!(d->ch->attr & OBX_COL_ATTR_HAS_EXTEND_VALUE)) return 0;
       return obx_bs_get(d->data, (int64_t)r * evb, evb) != 0;
     default:
@@ -1097,6 +1209,49 @@ static int col_dec_row(const col_dec *d, const obx_micro_header *h,
       uint64_t v = 0;
       memcpy(&v, d->dict_pay + ref * d->dm->data_size, d->dm->data_size);
       if (sc == OBX_SC_INT) v = obx_sign_extend(v, tss, 1);
+      *out = (int64_t)v;
+      return 0;
+    }
+    case OBX_COL_STRING_PREFIX: {
+      if (col_dec_is_null(d, h, r)) { *is_null = 1; *out = 0; return 0; }
+      int n = d->pm->string_size;
+      /* prefix lengths from the cumulative end-offset index */
+      int use_hex = d->pm->hex_char_cnt > 0;
+      /* max suffix for the fixed stride: n - min prefix length */
+      uint64_t e0 = 0;
+      memcpy(&e0, d->pfx_ends, d->pm->pib);
+      uint64_t min_pl = e0, prev = 0;
+      for (uint32_t j = 0; j < d->pm->count; j++) {
+        uint64_t e2 = 0;
+        memcpy(&e2, d->pfx_ends + (size_t)j * d->pm->pib, d->pm->pib);
+        if (e2 - prev < min_pl) min_pl = e2 - prev;
+        prev = e2;
+      }
+      int max_suffix = n - (int)min_pl;
+      int stride = 1 + (use_hex ? (max_suffix + 1) / 2 : max_suffix);
+      const uint8_t *rp = d->data + d->fix_off + (size_t)r * stride;
+      uint32_t ref = rp[0] & 0xF;
+      if (ref >= d->pm->count) return OBX_INVALID_ARGUMENT;
+      uint64_t pstart = 0, pend = 0;
+      if (ref > 0)
+        memcpy(&pstart, d->pfx_ends + (size_t)(ref - 1) * d->pm->pib,
+               d->pm->pib);
+      memcpy(&pend, d->pfx_ends + (size_t)ref * d->pm->pib, d->pm->pib);
+      int plen = (int)(pend - pstart);
+      uint64_t v = 0;
+      for (int i = 0; i < plen; i++)
+        v |= (uint64_t)d->pfx_data[pstart + i] << (8 * i);
+      for (int i = 0; i < n - plen; i++) {
+        uint8_t cc;
+        if (use_hex) {
+          uint8_t nib = (uint8_t)((rp[1 + i / 2] >>
+                                   (((i + 1) % 2) * 4)) & 0xF);
+          cc = d->str_chars[nib];
+        } else {
+          cc = rp[1 + i];
+        }
+        v |= (uint64_t)cc << (8 * (plen + i));
+      }
       *out = (int64_t)v;
       return 0;
     }

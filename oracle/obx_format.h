@@ -97,6 +97,7 @@ enum {
   OBX_COL_INTEGER_BASE_DIFF = 4,
   OBX_COL_STRING_DIFF = 5,
   OBX_COL_HEX_PACKING = 6,
+  OBX_COL_STRING_PREFIX = 7,
   OBX_COL_EQUAL = 8,
   /* 5..9: string encoders, out of scope this round */
 };
@@ -178,6 +179,28 @@ typedef struct obx_sdiff_meta {
   uint16_t string_size;  /* N */
   uint8_t diff_desc_cnt;
 } obx_sdiff_meta;        /* 5 bytes packed */
+
+/* STRING_PREFIX meta (ObStringPrefixMetaHeader,
+   ob_string_prefix_encoder.h:64-105): a table of <=15 prefixes; each row
+   stores a prefix ref + its suffix. Restated for the fixed char(N<=8)
+   boundary with FIXED-STRIDE cells (the reference's var-cell len_ field
+   is derivable for fixed-length strings, so cells carry only the ref
+   nibble; suffixes are nibble-packed like HEX_PACKING when the suffix
+   chars fit 16 — the odd_ nibble is implicit for fixed lengths).
+   Layout after the header: hex_char_array (hex_char_cnt B), prefix END
+   offsets (count entries x pib B, cumulative — the reference stores
+   count-1, we keep all ends), prefix bytes concatenated. Row data (fix
+   region): stride = 1 + max_suffix (raw) or 1 + ceil(max_suffix/2)
+   (hex); byte 0 low nibble = prefix ref. Our writer emits equal-length
+   prefixes (writer policy); the decoder honors per-prefix lengths via
+   the offset index. */
+typedef struct obx_sprefix_meta {
+  uint8_t version;       /* 0 */
+  uint8_t count;         /* prefixes, 1..15 */
+  uint16_t string_size;  /* N */
+  uint8_t hex_char_cnt;  /* 0 = raw suffix bytes */
+  uint8_t pib;           /* prefix end-offset byte width */
+} obx_sprefix_meta;      /* 6 bytes packed */
 
 /* COLUMN_EQUAL meta (ObColumnEqualMetaHeader,
    ob_column_equal_encoder.h:24-40): this column equals ref_col except at

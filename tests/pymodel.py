@@ -13,6 +13,7 @@ All values are returned as Python ints (already sign-extended) or None (NULL).
 import struct
 
 ENC_RAW, ENC_DICT, ENC_RLE, ENC_CONST, ENC_INT_DIFF, ENC_SDIFF, ENC_HEX = range(7)
+ENC_SPREFIX = 7
 ENC_COLEQ = 8
 
 ATTR_FIX = 1
@@ -276,6 +277,44 @@ class Block:
                 for i in range(ssize):
                     nib = (rp[i // 2] >> (((i + 1) % 2) * 4)) & 0xF
                     v |= chars[nib] << (8 * i)
+                out.append(v)
+        elif t == ENC_SPREFIX:
+            # obx_sprefix_meta: version u8, count u8, string_size u16,
+            # hex_char_cnt u8, pib u8; then hex chars, END offsets, prefixes
+            ver, pcnt, ssize, hex_cnt, pib = struct.unpack_from(
+                "<BBHBB", d, base)
+            p0 = base + 6
+            chars = list(d[p0:p0 + hex_cnt])
+            e0 = p0 + hex_cnt
+            ends = [int.from_bytes(d[e0 + j * pib:e0 + (j + 1) * pib],
+                                   "little") for j in range(pcnt)]
+            pdata = e0 + pcnt * pib
+            plens = [ends[0]] + [ends[j] - ends[j - 1]
+                                 for j in range(1, pcnt)]
+            max_suffix = ssize - min(plens)
+            stride = 1 + ((max_suffix + 1) // 2 if hex_cnt else max_suffix)
+            data0 = base + ch["length"]
+            ext_bits = evb * rows if has_ext else 0
+            fix0 = data0 + (ext_bits + 7) // 8
+            for r in range(rows):
+                if has_ext and bs_get(d[data0:], r * evb, evb):
+                    out.append(None)
+                    continue
+                rp = d[fix0 + r * stride:fix0 + (r + 1) * stride]
+                ref = rp[0] & 0xF
+                pstart = ends[ref - 1] if ref else 0
+                plen = ends[ref] - pstart
+                v = 0
+                for i in range(plen):
+                    v |= d[pdata + pstart + i] << (8 * i)
+                for i in range(ssize - plen):
+                    if hex_cnt:
+                        nib = (rp[1 + i // 2] >>
+                               (((i + 1) % 2) * 4)) & 0xF
+                        cc = chars[nib]
+                    else:
+                        cc = rp[1 + i]
+                    v |= cc << (8 * (plen + i))
                 out.append(v)
         elif t == ENC_COLEQ:
             # obx_coleq_meta: version u8, ref_col u16, exc_cnt u16, rib u8

@@ -285,3 +285,83 @@ def test_column_equal_gpu_parity():
         abi.result_rows(res_cpu, 2))
     assert eng.filter(h, filt) == res_cpu.rows_passed
     eng.close()
+
+
+# ---- STRING_PREFIX (prefix table + per-row ref/suffix;
+# ObStringPrefixEncoder, ob_string_prefix_encoder.h) ----
+
+def _prefix_col(rows, seed, prefixes=(b"ABC", b"XYZ", b"MNO"),
+                alpha=b"0123456789", nulls_at=()):
+    rng = np.random.default_rng(seed)
+    n = len(prefixes[0]) + 3
+    a = np.zeros((rows, n), dtype=np.uint8)
+    for r in range(rows):
+        p = prefixes[rng.integers(0, len(prefixes))]
+        sfx = bytes(rng.choice(np.frombuffer(alpha, dtype=np.uint8), 3))
+        a[r] = np.frombuffer(p + sfx, dtype=np.uint8)
+    nb = np.zeros((rows + 7) // 8, dtype=np.uint8)
+    for r in nulls_at:
+        nb[r // 8] |= 1 << (r % 8)
+    return a, nb, n
+
+
+@pytest.mark.parametrize("with_nulls", [False, True])
+def test_string_prefix_roundtrip_and_pymodel(with_nulls):
+    rows = 900
+    nulls_at = (1, 500) if with_nulls else ()
+    a, nb, n = _prefix_col(rows, seed=3, nulls_at=nulls_at)
+    schema = oracle.make_schema([(abi.T_CHAR, 0, 0, n)])
+    blk = oracle.encode_block(schema, [a.reshape(-1)],
+                              [abi.ENC_STRING_PREFIX],
+                              [nb if with_nulls else None])
+    rc, outs, nbs = oracle.decode_block(schema, 1, blk, [0])
+    got = np.frombuffer(outs[0], dtype=np.uint8).reshape(rows, n)
+    for r in range(rows):
+        isn = (nbs[0][r // 8] >> (r % 8)) & 1
+        if r in nulls_at:
+            assert isn
+        else:
+            assert not isn and (got[r] == a[r]).all()
+    pb = pymodel.Block(blk, [(abi.T_CHAR, 0, 0, n)])
+    vals = pb.decode_col(0)
+    for r in range(rows):
+        want = (None if r in nulls_at
+                else int.from_bytes(a[r].tobytes(), "little"))
+        assert vals[r] == want
+
+
+@pytest.mark.gpu
+def test_string_prefix_gpu_parity():
+    from oceanbase_amd.engine import GpuEngine
+    from test_gpu_parity import _manual_blockset
+    rows = 2400
+    a, nb, n = _prefix_col(rows, seed=23, nulls_at=(7, 1800))
+    schema = oracle.make_schema([(abi.T_CHAR, 0, 0, n),
+                                 (abi.T_INT, 0, 0, 8)])
+    rng = np.random.default_rng(24)
+    q = rng.integers(1, 500, rows, dtype=np.int64)
+    blocks = []
+    for s in range(0, rows, 800):
+        e = s + 800
+        nbw = np.zeros((e - s + 7) // 8, dtype=np.uint8)
+        for r in (7, 1800):
+            if s <= r < e:
+                nbw[(r - s) // 8] |= 1 << ((r - s) % 8)
+        blocks.append(oracle.encode_block(
+            schema, [a[s:e].reshape(-1), q[s:e].view(np.uint8)],
+            [abi.ENC_STRING_PREFIX, abi.ENC_RAW], [nbw, None]))
+    bs = _manual_blockset(schema, blocks)
+    bs.total_rows = rows
+    eng = GpuEngine(0)
+    h = eng.load(bs)
+    lo = int.from_bytes(b"MNO000", "little")
+    filt = abi.make_filter([dict(col=0, op=abi.OP_GE, lo=lo)])
+    agg = abi.make_agg([], [dict(kind=abi.AGG_COUNT),
+                            dict(kind=abi.AGG_SUM, col_a=1)])
+    res_gpu = eng.scan_filter_agg(h, filt, agg)
+    res_cpu = oracle.scan_filter_agg(bs, filt, agg)
+    assert res_gpu.rows_passed == res_cpu.rows_passed
+    assert sorted(abi.result_rows(res_gpu, 2)) == sorted(
+        abi.result_rows(res_cpu, 2))
+    assert eng.filter(h, filt) == res_cpu.rows_passed
+    eng.close()
