@@ -73,7 +73,9 @@ enum obx_encoding {
                                     packing (ob_hex_string_encoder.h) */
   OBX_ENC_STRING_PREFIX = 7,    /* fixed char: prefix table + suffixes
                                     (ob_string_prefix_encoder.h) */
-  OBX_ENC_COLUMN_EQUAL = 8,      /* equals the nearest previous same-shape
+  OBX_ENC_COLUMN_EQUAL = 8,
+  OBX_ENC_COLUMN_SUBSTR = 9,     /* substring of a previous char column
+                                    (ob_inter_column_substring_encoder.h) */      /* equals the nearest previous same-shape
                                     column except listed exception rows
                                     (ob_column_equal_encoder.h) */
   OBX_ENC_MAX = 10,

@@ -6,7 +6,7 @@ reference structs it mirrors; see that header).
 import ctypes as C
 
 # ---- enums -----------------------------------------------------------------
-ENC_RAW, ENC_DICT, ENC_RLE, ENC_CONST, ENC_INT_DIFF, ENC_SDIFF, ENC_HEX, ENC_STRING_PREFIX, ENC_COLUMN_EQUAL, ENC_AUTO = 0, 1, 2, 3, 4, 5, 6, 7, 8, 255
+ENC_RAW, ENC_DICT, ENC_RLE, ENC_CONST, ENC_INT_DIFF, ENC_SDIFF, ENC_HEX, ENC_STRING_PREFIX, ENC_COLUMN_EQUAL, ENC_COLUMN_SUBSTR, ENC_AUTO = 0, 1, 2, 3, 4, 5, 6, 7, 8, 9, 255
 
 T_INT, T_INT32, T_DATE, T_CHAR, T_DECIMAL_INT = 5, 4, 19, 23, 50
 

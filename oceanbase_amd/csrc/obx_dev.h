@@ -52,6 +52,7 @@ enum {
   OBX_D_HEX = 6,     /* HEX_PACKING: nibble-packed chars */
   OBX_D_SPREFIX = 7, /* STRING_PREFIX: prefix table + suffixes */
   OBX_D_EQUAL = 8,   /* COLUMN_EQUAL: ref column + exception rows */
+  OBX_D_SUBSTR = 9,  /* COLUMN_SUBSTR: slice of ref column + exceptions */
 };
 
 /* dev_col flags */

@@ -485,6 +485,30 @@ __device__ __noinline__ int64_t col_value2(const blk_view bv,
     }
     return col_value(bv, blk.cols[c.width], r, null_out); /* width=ref col */
   }
+  if (c.enc == OBX_D_SUBSTR) {
+    null_out = false;
+#pragma clang loop unroll(disable)
+    for (uint32_t i = 0; i < c.runs; i++) {
+      uint64_t rid = bit_read(bv, (c.dict_byte + (uint64_t)i * c.rib) * 8,
+                              (uint32_t)c.rib * 8);
+      if (rid == r) {
+        if (bit_read(bv, c.aux_byte * 8 + i, 1)) {
+          null_out = true;
+          return 0;
+        }
+        return (int64_t)bit_read(
+            bv,
+            (c.aux_byte + (c.runs + 7) / 8 + (uint64_t)i * c.datum_len) * 8,
+            (uint32_t)c.datum_len * 8);
+      }
+      if (rid > r) break;
+    }
+    int64_t rv = col_value(bv, blk.cols[c.width], r, null_out);
+    if (null_out) return 0;
+    uint64_t mask2 =
+        c.datum_len >= 8 ? ~0ull : ((1ull << (8 * c.datum_len)) - 1);
+    return (int64_t)(((uint64_t)rv >> (8 * (uint32_t)c.base)) & mask2);
+  }
   return col_value(bv, c, r, null_out);
 }
 

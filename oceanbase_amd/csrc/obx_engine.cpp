@@ -315,6 +315,17 @@ static int parse_block(const obx_col_schema *cols, uint16_t n_cols,
         dc->data_bit = (data0 + (ext_bits + 7) / 8) * 8;
         break;
       }
+      case OBX_COL_SUBSTR: {
+        const obx_substr_meta *sm3 = (const obx_substr_meta *)col_host;
+        if (sm3->ref_col >= n_cols) return OBX_INVALID_ARGUMENT;
+        dc->runs = sm3->exc_cnt;
+        dc->rib = sm3->rib;
+        dc->width = (uint8_t)sm3->ref_col; /* ref column index */
+        dc->base = sm3->start_pos;
+        dc->dict_byte = col_base + sizeof(obx_substr_meta); /* exc rids */
+        dc->aux_byte = dc->dict_byte + (uint64_t)sm3->exc_cnt * sm3->rib;
+        break;
+      }
       case OBX_COL_EQUAL: {
         const obx_coleq_meta *em = (const obx_coleq_meta *)col_host;
         if (em->ref_col >= n_cols) return OBX_INVALID_ARGUMENT;
@@ -398,7 +409,8 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
         case OBX_D_DICT: cls = 1; break;
         case OBX_D_INTDIFF: cls = 2; break;
         case OBX_D_CONST: cls = dc.runs == 0 ? 3 : 4; break;
-        case OBX_D_EQUAL: cls = 5; break; /* span: no device group keys */
+        case OBX_D_EQUAL: case OBX_D_SUBSTR:
+          cls = 5; break; /* span: no device group keys */
         default: cls = 4; break;
       }
       if (dc.flags & OBX_DF_STRING) cls |= 0x10;

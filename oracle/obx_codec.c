@@ -793,6 +793,77 @@ static int encode_column(enc_buf *mb, obx_col_header *ch,
       if (rc) return rc;
       break;
     }
+    case OBX_ENC_COLUMN_SUBSTR: {
+      /* ObInterColSubStrEncoder (ob_inter_column_substring_encoder.{h,cpp})
+         restated with same_start + fix_length always set: find the best
+         uniform start slice of the nearest previous wider char column;
+         rows that are not that substring (or differ in null-ness) go to
+         the exception list. */
+      if (sc != OBX_SC_STRING) return OBX_NOT_SUPPORTED;
+      int ref = -1;
+      for (int j = (int)col_idx - 1; j >= 0; j--) {
+        if (obx_store_class(all_cols[j].obj_type) == OBX_SC_STRING &&
+            all_cols[j].len >= len) { ref = j; break; }
+      }
+      if (ref < 0 || !all_data) return OBX_NOT_SUPPORTED;
+      const uint8_t *rd = all_data[ref];
+      const uint8_t *rn = all_nulls ? all_nulls[ref] : NULL;
+      const int rlen = all_cols[ref].len;
+      uint32_t best_exc = rows + 1; int best_start = 0;
+      for (int st = 0; st + len <= rlen; st++) {
+        uint32_t exc2 = 0;
+        for (uint32_t r = 0; r < rows; r++) {
+          int ln = null_at(nulls, r), rnl = null_at(rn, r);
+          if (ln != rnl ||
+              (!ln && memcmp(data + (size_t)r * len,
+                             rd + (size_t)r * rlen + st, (size_t)len) != 0))
+            exc2++;
+        }
+        if (exc2 < best_exc) { best_exc = exc2; best_start = st; }
+      }
+      if (best_exc * 4 > rows) return OBX_NOT_SUPPORTED;
+      int rib = (int)obx_byte_packed_int_size(rows ? rows - 1 : 0);
+      obx_substr_meta sm3;
+      sm3.version = 0;
+      sm3.attr = 3; /* same_start | fix_length */
+      sm3.start_pos = (uint16_t)best_start;
+      sm3.ref_col = (uint16_t)ref;
+      sm3.exc_cnt = (uint16_t)best_exc;
+      sm3.rib = (uint8_t)rib;
+      sm3.ref_len = (uint8_t)rlen;
+      int64_t meta_size = (int64_t)sizeof(sm3) + (int64_t)best_exc * rib +
+                          (best_exc + 7) / 8 + (int64_t)best_exc * len;
+      if (mb->len + meta_size > mb->cap) return OBX_BUF_NOT_ENOUGH;
+      uint8_t *mp = mb->p + mb->len;
+      memcpy(mp, &sm3, sizeof(sm3)); mp += sizeof(sm3);
+      uint8_t *rid_p = mp;
+      uint8_t *nb_p = rid_p + (size_t)best_exc * rib;
+      uint8_t *dat_p = nb_p + (best_exc + 7) / 8;
+      memset(nb_p, 0, (best_exc + 7) / 8);
+      uint32_t k3 = 0;
+      for (uint32_t r = 0; r < rows; r++) {
+        int ln = null_at(nulls, r), rnl = null_at(rn, r);
+        if (ln == rnl &&
+            (ln || memcmp(data + (size_t)r * len,
+                          rd + (size_t)r * rlen + best_start,
+                          (size_t)len) == 0))
+          continue;
+        memcpy(rid_p + (size_t)k3 * rib, &r, (size_t)rib);
+        if (ln) {
+          nb_p[k3 / 8] |= (uint8_t)(1u << (k3 % 8));
+          memset(dat_p + (size_t)k3 * len, 0, (size_t)len);
+        } else {
+          memcpy(dat_p + (size_t)k3 * len, data + (size_t)r * len,
+                 (size_t)len);
+        }
+        k3++;
+      }
+      mb->len += meta_size;
+      ch->type = OBX_COL_SUBSTR;
+      ch->attr |= OBX_COL_ATTR_FIX_LENGTH;
+      ch->length = (uint32_t)meta_size;
+      break;
+    }
     case OBX_ENC_COLUMN_EQUAL: {
       /* ObColumnEqualEncoder (ob_column_equal_encoder.{h,cpp}): equals the
          nearest previous column with the same shape except at exception
@@ -964,8 +1035,9 @@ typedef struct col_dec {
   /* STRING_PREFIX */
   const obx_sprefix_meta *pm;
   const uint8_t *pfx_ends, *pfx_data;
-  /* COLUMN_EQUAL: block context for the nested ref-column decode */
+  /* COLUMN_EQUAL / COLUMN_SUBSTR: block context for the ref decode */
   const obx_coleq_meta *em;
+  const obx_substr_meta *ssm;
   const uint8_t *exc_rid, *exc_nb, *exc_dat;
   const obx_col_header *chp0; /* column header array base */
   const uint8_t *meta_region;
@@ -1060,6 +1132,17 @@ static int col_dec_init(col_dec *d, const obx_micro_header *h,
       d->data = d->meta + ch->length;
       int64_t ext_bits = ext ? (int64_t)evb * h->row_count : 0;
       d->fix_off = (ext_bits + 7) / 8;
+      break;
+    }
+    case OBX_COL_SUBSTR: {
+      d->ssm = (const obx_substr_meta *)d->meta;
+      if (d->ssm->ref_col >= h->column_count) return OBX_INVALID_ARGUMENT;
+      uint8_t rt = d->chp0[d->ssm->ref_col].type;
+      if (rt == OBX_COL_EQUAL || rt == OBX_COL_SUBSTR)
+        return OBX_NOT_SUPPORTED; /* one level only */
+      d->exc_rid = d->meta + sizeof(obx_substr_meta);
+      d->exc_nb = d->exc_rid + (size_t)d->ssm->exc_cnt * d->ssm->rib;
+      d->exc_dat = d->exc_nb + (d->ssm->exc_cnt + 7) / 8;
       break;
     }
     case OBX_COL_EQUAL: {
@@ -1253,6 +1336,38 @@ static int col_dec_row(const col_dec *d, const obx_micro_header *h,
         v |= (uint64_t)cc << (8 * (plen + i));
       }
       *out = (int64_t)v;
+      return 0;
+    }
+    case OBX_COL_SUBSTR: {
+      for (uint32_t i = 0; i < d->ssm->exc_cnt; i++) {
+        uint64_t rid = 0;
+        memcpy(&rid, d->exc_rid + (size_t)i * d->ssm->rib,
+               (size_t)d->ssm->rib);
+        if (rid == r) {
+          if ((d->exc_nb[i / 8] >> (i % 8)) & 1) {
+            *is_null = 1; *out = 0; return 0;
+          }
+          uint64_t v = 0;
+          memcpy(&v, d->exc_dat + (size_t)i * cs->len, (size_t)cs->len);
+          *out = (int64_t)v;
+          return 0;
+        }
+        if (rid > r) break;
+      }
+      col_dec rd;
+      int rc2 = col_dec_init(&rd, h, &d->chp0[d->ssm->ref_col],
+                             d->meta_region);
+      if (rc2) return rc2;
+      int64_t rv = 0; int rnull = 0;
+      obx_col_schema rcs; /* the REF column's schema shape for decode */
+      rcs.obj_type = d->chp0[d->ssm->ref_col].obj_type;
+      rcs.scale = 0; rcs.precision = 0;
+      rcs.len = d->ssm->ref_len;
+      rc2 = col_dec_row(&rd, h, &rcs, r, &rv, &rnull);
+      if (rc2) return rc2;
+      if (rnull) { *is_null = 1; *out = 0; return 0; }
+      uint64_t mask2 = cs->len >= 8 ? ~0ull : ((1ull << (8 * cs->len)) - 1);
+      *out = (int64_t)(((uint64_t)rv >> (8 * d->ssm->start_pos)) & mask2);
       return 0;
     }
     case OBX_COL_EQUAL: {

@@ -99,6 +99,7 @@ enum {
   OBX_COL_HEX_PACKING = 6,
   OBX_COL_STRING_PREFIX = 7,
   OBX_COL_EQUAL = 8,
+  OBX_COL_SUBSTR = 9,
   /* 5..9: string encoders, out of scope this round */
 };
 enum {
@@ -217,6 +218,25 @@ typedef struct obx_coleq_meta {
   /* then: exc row_ids (rib B each, ascending); exc null bitmap
      ceil(exc_cnt/8) B; exc datums (schema len B each, 0 for nulls) */
 } obx_coleq_meta;        /* 6 bytes packed */
+
+/* COLUMN_SUBSTR meta (ObInterColSubStrMetaHeader,
+   ob_inter_column_substring_encoder.h:18-55): this column is a substring
+   of ref_col. Restated for the fixed boundary with is_same_start_pos and
+   is_fix_length always set (start_pos and the schema length describe the
+   slice), so rows carry no per-row start/len; exceptions use the same
+   CONST-style list as our COLUMN_EQUAL (ascending row_ids + null bitmap
+   + datums — the reference's EXCEPTION_START_POS/EXT_START_POS markers
+   folded into the list). */
+typedef struct obx_substr_meta {
+  uint8_t version;       /* 0 */
+  uint8_t attr;          /* bit0 same_start (always 1), bit1 fix_len (1) */
+  uint16_t start_pos;
+  uint16_t ref_col;
+  uint16_t exc_cnt;
+  uint8_t rib;
+  uint8_t ref_len;       /* ref column datum bytes (decoder has no schema
+                            array in scope for the nested decode) */
+} obx_substr_meta;       /* 10 bytes packed */
 
 typedef struct obx_intdiff_meta {
   uint8_t version;       /* 0 */

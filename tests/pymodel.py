@@ -15,6 +15,7 @@ import struct
 ENC_RAW, ENC_DICT, ENC_RLE, ENC_CONST, ENC_INT_DIFF, ENC_SDIFF, ENC_HEX = range(7)
 ENC_SPREFIX = 7
 ENC_COLEQ = 8
+ENC_SUBSTR = 9
 
 ATTR_FIX = 1
 ATTR_EXT = 2
@@ -337,6 +338,26 @@ class Block:
                     elif sc == SC_DECIMAL:
                         v = sign_extend(v, dlen)
                     out[rid] = v
+        elif t == ENC_SUBSTR:
+            # obx_substr_meta: version u8, attr u8, start u16, ref u16,
+            # exc_cnt u16, rib u8, ref_len u8
+            (ver, attr, start, ref_col, exc_cnt, rib,
+             ref_len) = struct.unpack_from("<BBHHHBB", d, base)
+            p0 = base + 10
+            rids = [int.from_bytes(d[p0 + i * rib:p0 + (i + 1) * rib],
+                                   "little") for i in range(exc_cnt)]
+            nb0 = p0 + exc_cnt * rib
+            datp = nb0 + (exc_cnt + 7) // 8
+            ref_vals = self.decode_col(ref_col)
+            mask = (1 << (8 * dlen)) - 1
+            out = [None if v is None else (v >> (8 * start)) & mask
+                   for v in ref_vals]
+            for i, rid in enumerate(rids):
+                if (d[nb0 + i // 8] >> (i % 8)) & 1:
+                    out[rid] = None
+                else:
+                    out[rid] = int.from_bytes(
+                        d[datp + i * dlen:datp + (i + 1) * dlen], "little")
         else:
             raise NotImplementedError(t)
         return out

@@ -365,3 +365,72 @@ def test_string_prefix_gpu_parity():
         abi.result_rows(res_cpu, 2))
     assert eng.filter(h, filt) == res_cpu.rows_passed
     eng.close()
+
+
+# ---- COLUMN_SUBSTR (this column is a fixed slice of a previous char
+# column except at exception rows; ObInterColSubStrEncoder) ----
+
+def _substr_blockset(rows=2100, seed=51):
+    schema = oracle.make_schema([(abi.T_CHAR, 0, 0, 6), (abi.T_CHAR, 0, 0, 3)])
+    rng = np.random.default_rng(seed)
+    alpha = np.frombuffer(b"ABCDEFGH", dtype=np.uint8)
+    a = rng.choice(alpha, (rows, 6))
+    b = a[:, 2:5].copy()
+    for r in rng.choice(rows, rows // 50, replace=False):
+        b[r, 0] ^= 1
+    na = np.zeros((rows + 7) // 8, dtype=np.uint8); na[0] |= 1
+    nb = np.zeros((rows + 7) // 8, dtype=np.uint8); nb[0] |= 3
+    return schema, a, b, na, nb
+
+
+def test_column_substr_roundtrip_and_pymodel():
+    schema, a, b, na, nb = _substr_blockset()
+    rows = len(a)
+    blk = oracle.encode_block(schema, [a.reshape(-1), b.reshape(-1)],
+                              [abi.ENC_RAW, abi.ENC_COLUMN_SUBSTR], [na, nb])
+    rc, outs, nbs = oracle.decode_block(schema, 2, blk, [1])
+    got = np.frombuffer(outs[0], dtype=np.uint8).reshape(rows, 3)
+    for r in range(rows):
+        isn = (nbs[0][r // 8] >> (r % 8)) & 1
+        assert bool(isn) == bool((nb[r // 8] >> (r % 8)) & 1)
+        if not isn:
+            assert (got[r] == b[r]).all()
+    pb = pymodel.Block(blk, [(abi.T_CHAR, 0, 0, 6), (abi.T_CHAR, 0, 0, 3)])
+    vals = pb.decode_col(1)
+    for r in range(rows):
+        want = (None if (nb[r // 8] >> (r % 8)) & 1
+                else int.from_bytes(b[r].tobytes(), "little"))
+        assert vals[r] == want
+
+
+@pytest.mark.gpu
+def test_column_substr_gpu_parity():
+    from oceanbase_amd.engine import GpuEngine
+    from test_gpu_parity import _manual_blockset
+    schema, a, b, na, nb = _substr_blockset()
+    rows = len(a)
+    blocks = []
+    for s in range(0, rows, 700):
+        e = min(s + 700, rows)
+        naw = np.zeros((e - s + 7) // 8, dtype=np.uint8)
+        nbw = np.zeros((e - s + 7) // 8, dtype=np.uint8)
+        for r in range(s, e):
+            if (na[r // 8] >> (r % 8)) & 1:
+                naw[(r - s) // 8] |= 1 << ((r - s) % 8)
+            if (nb[r // 8] >> (r % 8)) & 1:
+                nbw[(r - s) // 8] |= 1 << ((r - s) % 8)
+        blocks.append(oracle.encode_block(
+            schema, [a[s:e].reshape(-1), b[s:e].reshape(-1)],
+            [abi.ENC_RAW, abi.ENC_COLUMN_SUBSTR], [naw, nbw]))
+    bs = _manual_blockset(schema, blocks)
+    bs.total_rows = rows
+    eng = GpuEngine(0)
+    h = eng.load(bs)
+    lo = int.from_bytes(b"D" + b"\x00\x00", "little")
+    filt = abi.make_filter([dict(col=1, op=abi.OP_GE, lo=lo)])
+    agg = abi.make_agg([], [dict(kind=abi.AGG_COUNT)])
+    res_gpu = eng.scan_filter_agg(h, filt, agg)
+    res_cpu = oracle.scan_filter_agg(bs, filt, agg)
+    assert res_gpu.rows_passed == res_cpu.rows_passed
+    assert eng.filter(h, filt) == res_cpu.rows_passed
+    eng.close()
