@@ -21,6 +21,7 @@ TYPES = [
     (abi.T_DATE, 0, 0, 4),
     (abi.T_DECIMAL_INT, 2, 15, 8),
     (abi.T_CHAR, 0, 0, 1),
+    (abi.T_CHAR, 0, 0, 4),
 ]
 ENCS = [abi.ENC_RAW, abi.ENC_DICT, abi.ENC_RLE, abi.ENC_INT_DIFF,
         abi.ENC_AUTO]
@@ -30,8 +31,22 @@ def _gen_column(rng, tspec, rows, style):
     t, scale, prec, length = tspec
     if t == abi.T_CHAR:
         alpha = np.frombuffer(b"ABCDNRXYZ", dtype=np.uint8)
-        vals = rng.choice(alpha[: rng.integers(2, 9)], rows)
-        return vals, np.int64(vals[:]), length
+        if length == 1:
+            vals = rng.choice(alpha[: rng.integers(2, 9)], rows)
+            return vals, np.int64(vals[:]), length
+        # multi-byte chars: common-prefix + small-alphabet suffix shapes
+        # (exercises DICT/STRING_DIFF/HEX_PACKING/STRING_PREFIX writers)
+        a2 = np.empty((rows, length), dtype=np.uint8)
+        n_pref = int(rng.integers(1, 4))
+        prefs = [bytes(rng.choice(alpha, length - 1)) for _ in range(n_pref)]
+        pick = rng.integers(0, n_pref, rows)
+        for r in range(rows):
+            a2[r, :length - 1] = np.frombuffer(prefs[pick[r]],
+                                               dtype=np.uint8)
+        a2[:, length - 1] = rng.choice(alpha[:4], rows)
+        v = np.array([int.from_bytes(a2[r].tobytes(), "little")
+                      for r in range(rows)], dtype=np.int64)
+        return a2.reshape(-1), v, length
     if style == 0:      # tiny domain (dict/rle friendly)
         dom = rng.integers(-50, 50, rng.integers(2, 12)).astype(np.int64)
         v = rng.choice(dom, rows)
@@ -56,6 +71,8 @@ def _encoding_for(rng, tspec, enc, style):
     t = tspec[0]
     if enc == abi.ENC_INT_DIFF and t == abi.T_CHAR:
         # chars can't int-diff: exercise the string transforms instead
+        if tspec[3] > 1 and style == 3:
+            return abi.ENC_STRING_PREFIX
         return abi.ENC_HEX if style % 2 else abi.ENC_SDIFF
     if enc == abi.ENC_INT_DIFF and t == abi.T_DECIMAL_INT:
         return abi.ENC_RAW
