@@ -86,7 +86,10 @@ def result_to_bytes(res):
 
 
 def merge_results(blobs, n_aggs):
-    """Merge serialized AggResults (Python ints: exact 256-bit adds)."""
+    """Merge serialized AggResults (Python ints: exact 256-bit adds).
+
+    Sum-family aggregates only (COUNT/SUM/PROD sums — every bench
+    workload's shape); MIN/MAX cells would need min/max folds here."""
     groups = {}
     scanned = passed = 0
     for blob in blobs:
