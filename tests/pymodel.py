@@ -222,7 +222,8 @@ class Block:
                     f0 = data0 + (ext_bits + 7) // 8
                     diff = int.from_bytes(d[f0 + r * k:f0 + (r + 1) * k],
                                           "little")
-                out.append(bval + diff)
+                s = (bval + diff) & ((1 << 64) - 1)  # mod-2^64 wrap
+                out.append(sign_extend(s, 8) if sc == SC_INT else s)
         elif t == ENC_SDIFF:
             # obx_sdiff_meta: version u8, hex_char_cnt u8, string_size u16,
             # diff_desc_cnt u8; then descs, hex chars, common bytes
