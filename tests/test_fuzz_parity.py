@@ -7,6 +7,7 @@ tests (ObRowGenerate in test_column_decoder.h) but adversarially across the
 whole plan space the engine supports.
 """
 import ctypes as Ct
+import os
 
 import numpy as np
 import pytest
@@ -105,7 +106,8 @@ def _blockset(schema, blocks):
     return bs
 
 
-@pytest.mark.parametrize("case_seed", range(30))
+@pytest.mark.parametrize("case_seed",
+                         range(int(os.environ.get("OBX_FUZZ_SEEDS", "30"))))
 def test_fuzz_case(case_seed):
     from oceanbase_amd.engine import GpuEngine
     rng = np.random.default_rng(1000 + case_seed)
