@@ -907,8 +907,12 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
         g->cells[a].limb[0] = live[i]->count;
         g->cells[a].limb[1] = g->cells[a].limb[2] = g->cells[a].limb[3] = 0;
       } else if (kind == OBX_AGG_MIN || kind == OBX_AGG_MAX) {
-        /* sign-extend the int64 min/max into the 256-bit cell */
-        int64_t v = (int64_t)live[i]->cells[a][0];
+        /* sign-extend the int64 min/max into the 256-bit cell; a group
+           with NO non-null values keeps the init sentinel and its valid
+           flag (cells[a][1]) unset — report 0, matching the oracle's
+           empty-MIN/MAX convention (SQL NULL surfaces via row_count /
+           COUNT(col) at the caller) */
+        int64_t v = live[i]->cells[a][1] ? (int64_t)live[i]->cells[a][0] : 0;
         g->cells[a].limb[0] = (uint64_t)v;
         uint64_t s = v < 0 ? ~0ull : 0ull;
         g->cells[a].limb[1] = g->cells[a].limb[2] = g->cells[a].limb[3] = s;
