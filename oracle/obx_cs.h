@@ -1,0 +1,82 @@
+/*
+ * obx_cs.h — CS (column-store, OceanBase 4.3 "cs_encoding") format layer,
+ * round-2 row of SURVEY §8(f): the INTEGER STREAM restatement.
+ *
+ * TEST INFRASTRUCTURE (oracle): restates, file-by-file,
+ *   ob_stream_encoding_struct.{h,cpp} — ObIntegerStreamMeta field layout
+ *     and its serialize format (i8 version/attr/type/width + conditional
+ *     vi64 base / vi64 null_replace / i8 precision_width + i8
+ *     pfor_packing_type for V2 metas, :20-95)
+ *   deps/oblib/src/lib/utility/serialization.h:297 — encode_vi64: 7-bit
+ *     little-endian groups with a 0x80 continuation bit (a negative value
+ *     always serializes as 10 groups)
+ *   ob_integer_stream_encoder.cpp:93-155 — datum→uint conversion (null →
+ *     replace value, base subtraction FIRST) and RAW width-packed output
+ *   ob_stream_encoding_struct.cpp:96+ — build_signed_stream_meta's
+ *     base/width selection (range = max-min, smallest of 1/2/4/8 bytes)
+ *
+ * Scope of this slice: stream meta + RAW encoding type. The delta/zigzag/
+ * PFoR codec types (ObIntegerStream::EncodingType 2-8, backed by the
+ * deps/oblib ObCodec family) and the CS micro-block transformer/column
+ * layer are the continuation rows.
+ */
+#ifndef OBX_CS_H_
+#define OBX_CS_H_
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ObIntegerStream::Attribute */
+enum {
+  OBX_CS_USE_BASE = 0x1,
+  OBX_CS_REPLACE_NULL = 0x2,
+  OBX_CS_DECIMAL_INT = 0x4,
+};
+
+/* ObIntegerStream::EncodingType (subset implemented: RAW) */
+enum {
+  OBX_CS_ENC_RAW = 1,
+};
+
+/* in-memory mirror of ObIntegerStreamMeta (the serialized form is
+ * variable-length; see obx_cs_int_meta_enc) */
+typedef struct obx_cs_int_meta {
+  uint8_t version;   /* 1 = V2 (writes pfor_packing_type) */
+  uint8_t attr;
+  uint8_t type;      /* OBX_CS_ENC_* */
+  uint8_t width_tag; /* 0/1/2/3 -> 1/2/4/8 bytes (UintWidth) */
+  uint64_t base;
+  uint64_t null_replaced;
+  uint8_t precision_width_tag;
+  uint8_t pfor_packing_type; /* 0 = CPU_ARCH_INDEPENDANT_SCALAR */
+} obx_cs_int_meta;
+
+/* OceanBase vi64 varint. Returns bytes written / read, or -1. */
+int obx_cs_vi64_enc(uint8_t *buf, size_t cap, int64_t v);
+int obx_cs_vi64_dec(const uint8_t *buf, size_t len, int64_t *out);
+
+/* serialize / deserialize the stream meta; returns byte count or -1 */
+int obx_cs_int_meta_enc(const obx_cs_int_meta *m, uint8_t *buf, size_t cap);
+int obx_cs_int_meta_dec(const uint8_t *buf, size_t len, obx_cs_int_meta *m);
+
+/* Encode a signed-int64 column slice as [meta][RAW width-packed stream].
+ * nulls: optional bitmap (bit r set = NULL; replaced in-stream with the
+ * base value, REPLACE_NULL semantics left to the column layer). Returns
+ * total bytes or -1. */
+int64_t obx_cs_int_stream_enc(const int64_t *vals, const uint8_t *nulls,
+                              uint32_t rows, uint8_t *buf, size_t cap);
+
+/* Decode [meta][stream] back to int64 values (base re-applied). null
+ * positions decode to the stored replace value. Returns consumed bytes
+ * or -1. */
+int64_t obx_cs_int_stream_dec(const uint8_t *buf, size_t len, uint32_t rows,
+                              int64_t *out, obx_cs_int_meta *meta_out);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* OBX_CS_H_ */

@@ -1,0 +1,111 @@
+"""CS (column-store) integer-stream layer (first
+slice): ObIntegerStreamMeta serialize format, the OceanBase vi64 varint,
+and the RAW width-packed stream — restated in oracle/obx_cs.c and pinned
+here at byte level (the serialize format is replica-checksummed in the
+reference, ob_stream_encoding_struct.cpp:20-95, so the bytes ARE the
+contract)."""
+import ctypes as C
+import os
+import sys
+
+import numpy as np
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from oceanbase_amd import oracle  # noqa: E402  (builds liboracle.so)
+
+_lib = C.CDLL(os.path.join(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))), "oracle", "liboracle.so"))
+_lib.obx_cs_vi64_enc.restype = C.c_int
+_lib.obx_cs_vi64_enc.argtypes = [C.POINTER(C.c_uint8), C.c_size_t, C.c_int64]
+_lib.obx_cs_vi64_dec.restype = C.c_int
+_lib.obx_cs_vi64_dec.argtypes = [C.POINTER(C.c_uint8), C.c_size_t,
+                                 C.POINTER(C.c_int64)]
+_lib.obx_cs_int_stream_enc.restype = C.c_int64
+_lib.obx_cs_int_stream_enc.argtypes = [
+    C.POINTER(C.c_int64), C.POINTER(C.c_uint8), C.c_uint32,
+    C.POINTER(C.c_uint8), C.c_size_t]
+_lib.obx_cs_int_stream_dec.restype = C.c_int64
+_lib.obx_cs_int_stream_dec.argtypes = [
+    C.POINTER(C.c_uint8), C.c_size_t, C.c_uint32, C.POINTER(C.c_int64),
+    C.c_void_p]
+
+
+def _vi64(v):
+    buf = (C.c_uint8 * 10)()
+    n = _lib.obx_cs_vi64_enc(buf, 10, v)
+    assert n > 0
+    return bytes(buf[:n])
+
+
+def test_vi64_matches_reference_format():
+    """encode_vi64 (serialization.h:297): 7-bit LE groups, 0x80
+    continuation; negatives cast through uint64 -> always 10 bytes."""
+    assert _vi64(0) == b"\x00"
+    assert _vi64(0x7F) == b"\x7f"
+    assert _vi64(0x80) == b"\x80\x01"
+    assert _vi64(300) == b"\xac\x02"
+    assert len(_vi64(-1)) == 10
+    for v in (0, 1, 127, 128, 300, 2**32, 2**62, -1, -2**62):
+        b = _vi64(v)
+        out = C.c_int64()
+        n = _lib.obx_cs_vi64_dec((C.c_uint8 * len(b))(*b), len(b),
+                                 C.byref(out))
+        assert n == len(b) and out.value == v
+
+
+def _roundtrip(vals, nulls=None):
+    rows = len(vals)
+    v = np.asarray(vals, dtype=np.int64)
+    cap = 64 + rows * 8
+    buf = (C.c_uint8 * cap)()
+    nb = None
+    if nulls is not None:
+        nba = np.zeros((rows + 7) // 8, dtype=np.uint8)
+        for r in nulls:
+            nba[r // 8] |= 1 << (r % 8)
+        nb = nba.ctypes.data_as(C.POINTER(C.c_uint8))
+    n = _lib.obx_cs_int_stream_enc(
+        v.ctypes.data_as(C.POINTER(C.c_int64)), nb, rows, buf, cap)
+    assert n > 0
+    out = np.zeros(rows, dtype=np.int64)
+    m = _lib.obx_cs_int_stream_dec(buf, n, rows,
+                                   out.ctypes.data_as(C.POINTER(C.c_int64)),
+                                   None)
+    assert m == n
+    return bytes(buf[:n]), out
+
+
+@pytest.mark.parametrize("case", [
+    [0, 1, 2, 3],
+    [1000, 1001, 1255],                       # 1-byte range off a base
+    [-5, 300, 7],                             # 2-byte range, negative base
+    [2**40, 2**40 + 2**33],                   # 8-byte range
+    [-2**62, 2**62 - 1],
+    [42] * 100,
+])
+def test_stream_roundtrip(case):
+    blob, out = _roundtrip(case)
+    assert list(out) == case
+
+
+def test_stream_null_replace_uses_base():
+    """nulls occupy a slot with the base value (encoder :108-112: when no
+    explicit replace value, base 'makes int small')."""
+    blob, out = _roundtrip([10, 20, 30, 40], nulls=[2])
+    assert list(out) == [10, 20, 10, 40]  # null slot decodes to base=min
+
+
+def test_meta_bytes_pinned():
+    """Byte-level pin of the serialized meta for a known stream: version 1
+    (V2), attr USE_BASE, type RAW, width 1-byte-range tag, vi64 base,
+    pfor_packing_type 0."""
+    blob, _ = _roundtrip([1000, 1001, 1255])
+    # [version=1][attr=1][type=1][width_tag=0][vi64(1000)=E8 07][pfor=0]
+    assert blob[:4] == bytes([1, 1, 1, 0])
+    assert blob[4:6] == b"\xe8\x07"
+    assert blob[6] == 0
+    # stream: 3 x 1-byte deltas off base 1000
+    assert blob[7:10] == bytes([0, 1, 255])
+    assert len(blob) == 10
