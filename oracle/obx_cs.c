@@ -157,3 +157,97 @@ int64_t obx_cs_int_stream_dec(const uint8_t *buf, size_t len, uint32_t rows,
   if (meta_out) *meta_out = m;
   return (int64_t)pos;
 }
+
+int obx_cs_vi32_enc(uint8_t *buf, size_t cap, int32_t v) {
+  uint32_t u = (uint32_t)v;
+  int n = 0;
+  while (u > 0x7f) {
+    if ((size_t)n >= cap) return -1;
+    buf[n++] = (uint8_t)(u | 0x80);
+    u >>= 7;
+  }
+  if ((size_t)n >= cap) return -1;
+  buf[n++] = (uint8_t)(u & 0x7f);
+  return n;
+}
+
+int obx_cs_vi32_dec(const uint8_t *buf, size_t len, int32_t *out) {
+  uint32_t u = 0;
+  int shift = 0, n = 0;
+  for (;;) {
+    if ((size_t)n >= len || shift > 31) return -1;
+    uint8_t b = buf[n++];
+    u |= (uint32_t)(b & 0x7f) << shift;
+    if (!(b & 0x80)) break;
+    shift += 7;
+  }
+  *out = (int32_t)u;
+  return n;
+}
+
+int obx_cs_str_meta_enc(const obx_cs_str_meta *m, uint8_t *buf, size_t cap) {
+  size_t pos = 0;
+  if (cap < 2) return -1;
+  buf[pos++] = m->version;
+  buf[pos++] = m->attr;
+  int n = obx_cs_vi32_enc(buf + pos, cap - pos,
+                          (int32_t)m->uncompressed_len);
+  if (n < 0) return -1;
+  pos += n;
+  if (m->attr & OBX_CS_STR_FIXED_LEN) {
+    n = obx_cs_vi32_enc(buf + pos, cap - pos, (int32_t)m->fixed_str_len);
+    if (n < 0) return -1;
+    pos += n;
+  }
+  return (int)pos;
+}
+
+int obx_cs_str_meta_dec(const uint8_t *buf, size_t len, obx_cs_str_meta *m) {
+  size_t pos = 0;
+  memset(m, 0, sizeof(*m));
+  if (len < 2) return -1;
+  m->version = buf[pos++];
+  m->attr = buf[pos++];
+  int32_t v;
+  int n = obx_cs_vi32_dec(buf + pos, len - pos, &v);
+  if (n < 0) return -1;
+  m->uncompressed_len = (uint32_t)v;
+  pos += n;
+  if (m->attr & OBX_CS_STR_FIXED_LEN) {
+    n = obx_cs_vi32_dec(buf + pos, len - pos, &v);
+    if (n < 0) return -1;
+    m->fixed_str_len = (uint32_t)v;
+    pos += n;
+  }
+  return (int)pos;
+}
+
+int64_t obx_cs_str_stream_enc_fixed(const uint8_t *bytes, uint32_t rows,
+                                    uint32_t fixed_len, uint8_t *buf,
+                                    size_t cap) {
+  obx_cs_str_meta m;
+  memset(&m, 0, sizeof(m));
+  m.version = 0;
+  m.attr = OBX_CS_STR_FIXED_LEN;
+  m.uncompressed_len = rows * fixed_len;
+  m.fixed_str_len = fixed_len;
+  int hn = obx_cs_str_meta_enc(&m, buf, cap);
+  if (hn < 0) return -1;
+  size_t pos = (size_t)hn;
+  if (pos + (size_t)rows * fixed_len > cap) return -1;
+  memcpy(buf + pos, bytes, (size_t)rows * fixed_len);
+  return (int64_t)(pos + (size_t)rows * fixed_len);
+}
+
+int64_t obx_cs_str_stream_dec_fixed(const uint8_t *buf, size_t len,
+                                    uint32_t rows, uint32_t *fixed_len_out,
+                                    const uint8_t **bytes_out) {
+  obx_cs_str_meta m;
+  int hn = obx_cs_str_meta_dec(buf, len, &m);
+  if (hn < 0 || !(m.attr & OBX_CS_STR_FIXED_LEN)) return -1;
+  if (m.uncompressed_len != rows * m.fixed_str_len) return -1;
+  if ((size_t)hn + m.uncompressed_len > len) return -1;
+  *fixed_len_out = m.fixed_str_len;
+  *bytes_out = buf + hn;
+  return (int64_t)hn + m.uncompressed_len;
+}

@@ -76,6 +76,40 @@ int64_t obx_cs_int_stream_enc(const int64_t *vals, const uint8_t *nulls,
 int64_t obx_cs_int_stream_dec(const uint8_t *buf, size_t len, uint32_t rows,
                               int64_t *out, obx_cs_int_meta *meta_out);
 
+/* ---- string stream (ObStringStreamMeta, ob_stream_encoding_struct.h:341
+ * and its DEFINE_SERIALIZE at .cpp:248: i8 version, i8 attr,
+ * vi32 uncompressed_len, vi32 fixed_str_len when IS_FIXED_LEN_STRING).
+ * The byte stream itself is the concatenated string bytes; var-length
+ * columns pair it with a companion INTEGER stream of end offsets
+ * (ObStringStreamEncoder), which callers build with
+ * obx_cs_int_stream_enc. ---- */
+
+enum {
+  OBX_CS_STR_ZERO_LEN_NULL = 0x1,
+  OBX_CS_STR_FIXED_LEN = 0x2,
+};
+
+typedef struct obx_cs_str_meta {
+  uint8_t version;
+  uint8_t attr;
+  uint32_t uncompressed_len;
+  uint32_t fixed_str_len;
+} obx_cs_str_meta;
+
+int obx_cs_vi32_enc(uint8_t *buf, size_t cap, int32_t v);
+int obx_cs_vi32_dec(const uint8_t *buf, size_t len, int32_t *out);
+int obx_cs_str_meta_enc(const obx_cs_str_meta *m, uint8_t *buf, size_t cap);
+int obx_cs_str_meta_dec(const uint8_t *buf, size_t len, obx_cs_str_meta *m);
+
+/* Fixed-length string stream: [meta][rows*fixed_len bytes]. Returns total
+ * bytes or -1. Decode verifies the meta and returns a pointer offset. */
+int64_t obx_cs_str_stream_enc_fixed(const uint8_t *bytes, uint32_t rows,
+                                    uint32_t fixed_len, uint8_t *buf,
+                                    size_t cap);
+int64_t obx_cs_str_stream_dec_fixed(const uint8_t *buf, size_t len,
+                                    uint32_t rows, uint32_t *fixed_len_out,
+                                    const uint8_t **bytes_out);
+
 #ifdef __cplusplus
 }
 #endif

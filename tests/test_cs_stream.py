@@ -109,3 +109,65 @@ def test_meta_bytes_pinned():
     # stream: 3 x 1-byte deltas off base 1000
     assert blob[7:10] == bytes([0, 1, 255])
     assert len(blob) == 10
+
+
+# ---- string stream (ObStringStreamMeta + fixed byte stream; var-length
+# pairs it with a companion integer stream of end offsets) ----
+
+_lib.obx_cs_str_stream_enc_fixed.restype = C.c_int64
+_lib.obx_cs_str_stream_enc_fixed.argtypes = [
+    C.POINTER(C.c_uint8), C.c_uint32, C.c_uint32, C.POINTER(C.c_uint8),
+    C.c_size_t]
+_lib.obx_cs_str_stream_dec_fixed.restype = C.c_int64
+_lib.obx_cs_str_stream_dec_fixed.argtypes = [
+    C.POINTER(C.c_uint8), C.c_size_t, C.c_uint32, C.POINTER(C.c_uint32),
+    C.POINTER(C.POINTER(C.c_uint8))]
+
+
+def test_string_stream_fixed_roundtrip_and_meta_bytes():
+    rows, ln = 300, 4
+    rng = np.random.default_rng(6)
+    data = rng.integers(65, 91, rows * ln).astype(np.uint8)
+    cap = 64 + rows * ln
+    buf = (C.c_uint8 * cap)()
+    n = _lib.obx_cs_str_stream_enc_fixed(
+        data.ctypes.data_as(C.POINTER(C.c_uint8)), rows, ln, buf, cap)
+    assert n > 0
+    blob = bytes(buf[:n])
+    # meta pin: [version=0][attr=FIXED(2)][vi32(1200)=B0 09][vi32(4)=04]
+    assert blob[:2] == bytes([0, 2])
+    assert blob[2:4] == b"\xb0\x09"
+    assert blob[4] == 4
+    flo = C.c_uint32()
+    bp = C.POINTER(C.c_uint8)()
+    m = _lib.obx_cs_str_stream_dec_fixed(buf, n, rows, C.byref(flo),
+                                         C.byref(bp))
+    assert m == n and flo.value == ln
+    got = bytes(C.cast(bp, C.POINTER(C.c_uint8 * (rows * ln))).contents)
+    assert got == data.tobytes()
+
+
+def test_var_string_as_bytes_plus_offset_stream():
+    """Var-length column shape: byte stream of concatenated strings + a
+    companion integer stream of END offsets (the CS column layer's
+    pairing; offsets use the integer stream already restated above)."""
+    strings = [b"a", b"", b"hello", b"OB", b"cs-format"]
+    blob_bytes = b"".join(strings)
+    ends = np.cumsum([len(s) for s in strings]).astype(np.int64)
+    cap = 64 + len(ends) * 8
+    obuf = (C.c_uint8 * cap)()
+    n = _lib.obx_cs_int_stream_enc(
+        ends.ctypes.data_as(C.POINTER(C.c_int64)), None, len(ends), obuf,
+        cap)
+    assert n > 0
+    out = np.zeros(len(ends), dtype=np.int64)
+    m = _lib.obx_cs_int_stream_dec(obuf, n, len(ends),
+                                   out.ctypes.data_as(C.POINTER(C.c_int64)),
+                                   None)
+    assert m == n and list(out) == list(ends)
+    # reconstruct strings
+    got, prev = [], 0
+    for e in out:
+        got.append(blob_bytes[prev:e])
+        prev = e
+    assert got == strings
