@@ -3,6 +3,7 @@
 #include "obx_cs.h"
 
 #include <string.h>
+#include <stdlib.h>
 
 int obx_cs_vi64_enc(uint8_t *buf, size_t cap, int64_t v) {
   /* serialization::encode_vi64: 7-bit LE groups, 0x80 continuation */
@@ -102,6 +103,16 @@ static inline int null_at(const uint8_t *nulls, uint32_t r) {
 
 int64_t obx_cs_int_stream_enc(const int64_t *vals, const uint8_t *nulls,
                               uint32_t rows, uint8_t *buf, size_t cap) {
+  return obx_cs_int_stream_enc2(vals, nulls, rows, OBX_CS_ENC_RAW, buf,
+                                cap);
+}
+
+int64_t obx_cs_dzr_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                       uint8_t *out, size_t cap);
+
+int64_t obx_cs_int_stream_enc2(const int64_t *vals, const uint8_t *nulls,
+                               uint32_t rows, uint8_t enc_type,
+                               uint8_t *buf, size_t cap) {
   if (!vals || !rows || !buf) return -1;
   /* build_signed_stream_meta: base = min, width covers range = max-min */
   int64_t mn = 0, mx = 0;
@@ -117,7 +128,7 @@ int64_t obx_cs_int_stream_enc(const int64_t *vals, const uint8_t *nulls,
   obx_cs_int_meta m;
   memset(&m, 0, sizeof(m));
   m.version = 1; /* V2 */
-  m.type = OBX_CS_ENC_RAW;
+  m.type = enc_type;
   m.attr = OBX_CS_USE_BASE;
   m.base = (uint64_t)mn;
   m.width_tag = range <= 0xFF ? 0 : range <= 0xFFFF ? 1
@@ -127,14 +138,31 @@ int64_t obx_cs_int_stream_enc(const int64_t *vals, const uint8_t *nulls,
   if (hn < 0) return -1;
   size_t pos = (size_t)hn;
   uint32_t wb = WIDTH_BYTES[m.width_tag];
-  if (pos + (size_t)rows * wb > cap) return -1;
+  if (pos + (size_t)rows * wb + 32 > cap) return -1;
+  /* datum->uint conversion: null -> replace value (== base here: "make
+     int small", ob_integer_stream_encoder.cpp:108-112), base FIRST */
+  uint8_t *packed = buf + pos; /* RAW writes in place */
+  uint8_t *tmp = NULL;
+  if (enc_type != OBX_CS_ENC_RAW) {
+    tmp = (uint8_t *)malloc((size_t)rows * wb);
+    if (!tmp) return -1;
+    packed = tmp;
+  }
   for (uint32_t r = 0; r < rows; r++) {
-    /* null -> replace value (== base here: "make int small",
-       ob_integer_stream_encoder.cpp:108-112), base subtracted FIRST */
     uint64_t ele = null_at(nulls, r) ? m.base : (uint64_t)vals[r];
     ele -= m.base;
-    memcpy(buf + pos, &ele, wb);
-    pos += wb;
+    memcpy(packed + (size_t)r * wb, &ele, wb);
+  }
+  if (enc_type == OBX_CS_ENC_RAW) {
+    pos += (size_t)rows * wb;
+  } else if (enc_type == OBX_CS_ENC_DELTA_ZIGZAG_RLE) {
+    int64_t n = obx_cs_dzr_enc(packed, rows, wb, buf + pos, cap - pos);
+    free(tmp);
+    if (n < 0) return -1;
+    pos += (size_t)n;
+  } else {
+    free(tmp);
+    return -1;
   }
   return (int64_t)pos;
 }
@@ -143,16 +171,32 @@ int64_t obx_cs_int_stream_dec(const uint8_t *buf, size_t len, uint32_t rows,
                               int64_t *out, obx_cs_int_meta *meta_out) {
   obx_cs_int_meta m;
   int hn = obx_cs_int_meta_dec(buf, len, &m);
-  if (hn < 0 || m.type != OBX_CS_ENC_RAW) return -1;
+  if (hn < 0) return -1;
   size_t pos = (size_t)hn;
   uint32_t wb = WIDTH_BYTES[m.width_tag];
-  if (pos + (size_t)rows * wb > len) return -1;
   uint64_t base = (m.attr & OBX_CS_USE_BASE) ? m.base : 0;
-  for (uint32_t r = 0; r < rows; r++) {
-    uint64_t ele = 0;
-    memcpy(&ele, buf + pos, wb);
-    pos += wb;
-    out[r] = (int64_t)(ele + base);
+  if (m.type == OBX_CS_ENC_RAW) {
+    if (pos + (size_t)rows * wb > len) return -1;
+    for (uint32_t r = 0; r < rows; r++) {
+      uint64_t ele = 0;
+      memcpy(&ele, buf + pos, wb);
+      pos += wb;
+      out[r] = (int64_t)(ele + base);
+    }
+  } else if (m.type == OBX_CS_ENC_DELTA_ZIGZAG_RLE) {
+    uint8_t *packed = (uint8_t *)malloc((size_t)rows * wb);
+    if (!packed) return -1;
+    int64_t n = obx_cs_dzr_dec(buf + pos, len - pos, rows, wb, packed);
+    if (n < 0) { free(packed); return -1; }
+    pos += (size_t)n;
+    for (uint32_t r = 0; r < rows; r++) {
+      uint64_t ele = 0;
+      memcpy(&ele, packed + (size_t)r * wb, wb);
+      out[r] = (int64_t)(ele + base);
+    }
+    free(packed);
+  } else {
+    return -1;
   }
   if (meta_out) *meta_out = m;
   return (int64_t)pos;
@@ -250,4 +294,223 @@ int64_t obx_cs_str_stream_dec_fixed(const uint8_t *buf, size_t len,
   *fixed_len_out = m.fixed_str_len;
   *bytes_out = buf + hn;
   return (int64_t)hn + m.uncompressed_len;
+}
+
+/* ====================================================================== */
+/* DELTA_ZIGZAG_RLE codec (ObDeltaZigzagRleInner, deps/oblib/src/lib/
+ * codec/ob_delta_zigzag_rle.h:22-318, over the ObBitUtils protocol of
+ * ob_bp_util.h:128-340 and ObZigZag:74). Bit stream semantics mirrored
+ * exactly: little-endian bit accumulation (bw += v << br), e_slide
+ * writes 8 raw bytes and advances br/8, put<u64> splits >45-bit values
+ * as (high width-32, slide, low 32), flush advances ceil(br/8).
+ * Repeat runs: 1..18 as single '1' bits; longer as
+ * [4b zero flag][3b zero][3b byte-cnt-1][byte-cnt*8b of cnt-18].
+ * Deltas (zigzag at element width): 0 -> nothing; < 2^(N2-1) ->
+ * [delta<<2|0b10] in N2+2 bits; N3/N4 likewise with 0b100/0b1000; else
+ * [4b zero][3b byte-cnt-1][byte-cnt*8b delta]. */
+
+static const uint32_t DZR_N2[4] = {3, 6, 6, 6};
+static const uint32_t DZR_N3[4] = {5, 12, 10, 12};
+static const uint32_t DZR_N4[4] = {9, 17, 17, 20};
+#define DZR_BASE_REPEAT 18
+
+typedef struct {
+  uint64_t bw;
+  uint32_t br;
+  uint8_t *op;
+  uint8_t *end;
+} dzr_e;
+
+static inline int dzr_e_room(dzr_e *e, size_t n) {
+  return (size_t)(e->end - e->op) >= n + 16;
+}
+static inline void dzr_put(dzr_e *e, uint32_t width, uint64_t v) {
+  e->bw += v << e->br;
+  e->br += width;
+}
+static inline void dzr_slide(dzr_e *e) {
+  memcpy(e->op, &e->bw, 8);
+  e->op += e->br >> 3;
+  e->bw >>= (e->br & ~7u);
+  e->br &= 7;
+}
+static inline void dzr_put64(dzr_e *e, uint32_t width, uint64_t v) {
+  if (width > 45) { /* put<uint64_t> split, ob_bp_util.h:300-310 */
+    dzr_put(e, width - 32, v >> 32);
+    dzr_slide(e);
+    dzr_put(e, 32, (uint32_t)v);
+  } else {
+    dzr_put(e, width, v);
+  }
+}
+
+static inline uint32_t dzr_bits_u64(uint64_t v) { /* gccbits */
+  uint32_t n = 0;
+  while (v) { n++; v >>= 1; }
+  return n ? n : 1;
+}
+
+static inline uint64_t dzr_zz_enc(uint64_t v, uint32_t wbits) {
+  uint64_t m = wbits >= 64 ? ~0ull : ((1ull << wbits) - 1);
+  v &= m;
+  uint64_t sign = (v >> (wbits - 1)) & 1;
+  return (((v << 1) ^ (sign ? m : 0)) & m);
+}
+static inline uint64_t dzr_zz_dec(uint64_t v, uint32_t wbits) {
+  uint64_t m = wbits >= 64 ? ~0ull : ((1ull << wbits) - 1);
+  v &= m;
+  return (((v >> 1) ^ (0ull - (v & 1))) & m);
+}
+
+static inline uint64_t dzr_load(const uint8_t *p, uint32_t wb) {
+  uint64_t v = 0;
+  memcpy(&v, p, wb);
+  return v;
+}
+
+/* emit pending repeat count + one (nonzero or final-zero) delta */
+static int dzr_emit(dzr_e *e, uint64_t r, uint64_t delta_zz, int wi) {
+  if (!dzr_e_room(e, 32)) return -1;
+  if (r > DZR_BASE_REPEAT) {
+    r -= DZR_BASE_REPEAT;
+    uint32_t b = (dzr_bits_u64(r) + 7) >> 3;
+    dzr_put(e, 4 + 3 + 3, (uint64_t)(b - 1) << (4 + 3));
+    dzr_put64(e, b << 3, r);
+    dzr_slide(e);
+  } else {
+    while (r--) {
+      dzr_put(e, 1, 1);
+      dzr_slide(e);
+    }
+  }
+  const uint32_t N2 = DZR_N2[wi], N3 = DZR_N3[wi], N4 = DZR_N4[wi];
+  if (delta_zz == 0) {
+    /* nothing: only legal as the final element (decode stops by count) */
+  } else if (delta_zz < (1ull << (N2 - 1))) {
+    dzr_put(e, N2 + 2, (delta_zz << 2) | 2);
+  } else if (delta_zz < (1ull << (N3 - 1))) {
+    dzr_put(e, N3 + 3, (delta_zz << 3) | 4);
+  } else if (delta_zz < (1ull << (N4 - 1))) {
+    dzr_put(e, N4 + 4, (delta_zz << 4) | 8);
+  } else {
+    uint32_t b = (dzr_bits_u64(delta_zz) + 7) >> 3;
+    dzr_put(e, 4 + 3, (uint64_t)(b - 1) << 4);
+    dzr_put64(e, b << 3, delta_zz);
+  }
+  dzr_slide(e);
+  return 0;
+}
+
+int64_t obx_cs_dzr_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                       uint8_t *out, size_t cap) {
+  int wi = wb == 1 ? 0 : wb == 2 ? 1 : wb == 4 ? 2 : 3;
+  uint32_t wbits = wb * 8;
+  uint64_t wmask = wbits >= 64 ? ~0ull : ((1ull << wbits) - 1);
+  dzr_e e = {0, 0, out, out + cap};
+  if (cap < 24) return -1;
+  uint64_t start = 0, pending = 0;
+  for (uint32_t i = 0; i < count; i++) {
+    uint64_t v = dzr_load(in + (size_t)i * wb, wb);
+    uint64_t delta = (v - start) & wmask;
+    start = v;
+    if (delta == 0) {
+      pending++;
+      continue;
+    }
+    if (dzr_emit(&e, pending, dzr_zz_enc(delta, wbits), wi)) return -1;
+    pending = 0;
+  }
+  if (pending) {
+    if (dzr_emit(&e, pending, 0, wi)) return -1;
+  }
+  /* flush (ob_bp_util.h:236): write bw, advance ceil(br/8) */
+  if (!dzr_e_room(&e, 8)) return -1;
+  memcpy(e.op, &e.bw, 8);
+  e.op += (e.br + 7) >> 3;
+  return (int64_t)(e.op - out);
+}
+
+typedef struct {
+  uint64_t bw;
+  uint32_t br;
+  const uint8_t *ip;
+  const uint8_t *end;
+} dzr_d;
+
+static inline void dzr_d_slide(dzr_d *d) {
+  d->ip += d->br >> 3;
+  if (d->ip + 8 <= d->end) {
+    memcpy(&d->bw, d->ip, 8);
+  } else {
+    uint8_t tmp[8] = {0};
+    if (d->ip < d->end) memcpy(tmp, d->ip, (size_t)(d->end - d->ip));
+    memcpy(&d->bw, tmp, 8);
+  }
+  d->br &= 7;
+}
+static inline uint64_t dzr_get(dzr_d *d, uint32_t b) { /* get57 semantics */
+  uint64_t v = (d->bw >> d->br);
+  v &= (b >= 64) ? ~0ull : ((1ull << b) - 1);
+  d->br += b;
+  return v;
+}
+
+int64_t obx_cs_dzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                       uint32_t wb, uint8_t *out) {
+  int wi = wb == 1 ? 0 : wb == 2 ? 1 : wb == 4 ? 2 : 3;
+  uint32_t wbits = wb * 8;
+  uint64_t wmask = wbits >= 64 ? ~0ull : ((1ull << wbits) - 1);
+  const uint32_t N2 = DZR_N2[wi], N3 = DZR_N3[wi], N4 = DZR_N4[wi];
+  dzr_d d = {0, 0, in, in + in_len};
+  uint64_t start = 0;
+  uint32_t done = 0;
+  dzr_d_slide(&d);
+  while (done < count) {
+    uint64_t peek = d.bw >> d.br;
+    uint64_t delta;
+    if (peek & 1) {
+      d.br += 1;
+      delta = 0;
+    } else if (peek & 2) {
+      d.br += N2 + 2;
+      delta = (peek >> 2) & ((1ull << N2) - 1);
+    } else if (peek & 4) {
+      d.br += N3 + 3;
+      delta = (peek >> 3) & ((1ull << N3) - 1);
+    } else if (peek & 8) {
+      d.br += N4 + 4;
+      delta = (peek >> 4) & ((1ull << N4) - 1);
+    } else {
+      uint32_t f = (uint32_t)dzr_get(&d, 4 + 3);
+      uint32_t b = f >> 4;
+      if (b == 0) { /* long repeat count */
+        b = (uint32_t)dzr_get(&d, 3);
+        uint64_t r = dzr_get(&d, (b + 1) << 3); /* get<uint32_t>=get57 */
+        dzr_d_slide(&d);
+        r += DZR_BASE_REPEAT;
+        while (r-- && done < count) {
+          memcpy(out + (size_t)done * wb, &start, wb);
+          done++;
+        }
+        continue;
+      }
+      if (b == 1) return -1; /* overflow: "can not be" */
+      uint32_t bits = (b + 1) << 3;
+      if (wb == 8 && bits > 45) { /* get<uint64_t> split */
+        uint64_t hi = dzr_get(&d, bits - 32);
+        dzr_d_slide(&d);
+        uint64_t lo = dzr_get(&d, 32);
+        delta = (hi << 32) | lo;
+      } else {
+        delta = dzr_get(&d, bits);
+      }
+    }
+    start = (start + dzr_zz_dec(delta, wbits)) & wmask;
+    memcpy(out + (size_t)done * wb, &start, wb);
+    done++;
+    dzr_d_slide(&d);
+  }
+  /* align (ob_bp_util.h:152) */
+  d.ip += (d.br + 7) >> 3;
+  return (int64_t)(d.ip - in);
 }

@@ -37,9 +37,10 @@ enum {
   OBX_CS_DECIMAL_INT = 0x4,
 };
 
-/* ObIntegerStream::EncodingType (subset implemented: RAW) */
+/* ObIntegerStream::EncodingType (subset implemented) */
 enum {
   OBX_CS_ENC_RAW = 1,
+  OBX_CS_ENC_DELTA_ZIGZAG_RLE = 4,
 };
 
 /* in-memory mirror of ObIntegerStreamMeta (the serialized form is
@@ -69,6 +70,12 @@ int obx_cs_int_meta_dec(const uint8_t *buf, size_t len, obx_cs_int_meta *m);
  * total bytes or -1. */
 int64_t obx_cs_int_stream_enc(const int64_t *vals, const uint8_t *nulls,
                               uint32_t rows, uint8_t *buf, size_t cap);
+
+/* like obx_cs_int_stream_enc with an explicit stream encoding type
+ * (OBX_CS_ENC_RAW or OBX_CS_ENC_DELTA_ZIGZAG_RLE) */
+int64_t obx_cs_int_stream_enc2(const int64_t *vals, const uint8_t *nulls,
+                               uint32_t rows, uint8_t enc_type,
+                               uint8_t *buf, size_t cap);
 
 /* Decode [meta][stream] back to int64 values (base re-applied). null
  * positions decode to the stored replace value. Returns consumed bytes
@@ -109,6 +116,14 @@ int64_t obx_cs_str_stream_enc_fixed(const uint8_t *bytes, uint32_t rows,
 int64_t obx_cs_str_stream_dec_fixed(const uint8_t *buf, size_t len,
                                     uint32_t rows, uint32_t *fixed_len_out,
                                     const uint8_t **bytes_out);
+
+/* DELTA_ZIGZAG_RLE codec (ObDeltaZigzagRleInner; see obx_cs.c header
+ * comment for the cited bit protocol). in/out are width-packed arrays
+ * (wb in {1,2,4,8}); returns encoded / consumed bytes or -1. */
+int64_t obx_cs_dzr_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                       uint8_t *out, size_t cap);
+int64_t obx_cs_dzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                       uint32_t wb, uint8_t *out);
 
 #ifdef __cplusplus
 }

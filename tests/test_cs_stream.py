@@ -171,3 +171,86 @@ def test_var_string_as_bytes_plus_offset_stream():
         got.append(blob_bytes[prev:e])
         prev = e
     assert got == strings
+
+
+# ---- DELTA_ZIGZAG_RLE codec (ObDeltaZigzagRleInner bit protocol) ----
+
+_lib.obx_cs_dzr_enc.restype = C.c_int64
+_lib.obx_cs_dzr_enc.argtypes = [C.POINTER(C.c_uint8), C.c_uint32,
+                                C.c_uint32, C.POINTER(C.c_uint8), C.c_size_t]
+_lib.obx_cs_dzr_dec.restype = C.c_int64
+_lib.obx_cs_dzr_dec.argtypes = [C.POINTER(C.c_uint8), C.c_size_t,
+                                C.c_uint32, C.c_uint32,
+                                C.POINTER(C.c_uint8)]
+_lib.obx_cs_int_stream_enc2.restype = C.c_int64
+_lib.obx_cs_int_stream_enc2.argtypes = [
+    C.POINTER(C.c_int64), C.POINTER(C.c_uint8), C.c_uint32, C.c_uint8,
+    C.POINTER(C.c_uint8), C.c_size_t]
+
+
+def _dzr_rt(arr, wb):
+    dt = {1: np.uint8, 2: np.uint16, 4: np.uint32, 8: np.uint64}[wb]
+    packed = np.asarray(arr).astype(dt)
+    inb = packed.tobytes()
+    cap = len(inb) * 3 + 64
+    out = (C.c_uint8 * cap)()
+    n = _lib.obx_cs_dzr_enc((C.c_uint8 * len(inb)).from_buffer_copy(inb),
+                            len(packed), wb, out, cap)
+    assert n > 0
+    dec = (C.c_uint8 * len(inb))()
+    m = _lib.obx_cs_dzr_dec(out, n, len(packed), wb, dec)
+    assert m == n
+    assert np.frombuffer(bytes(dec), dtype=dt).tolist() == packed.tolist()
+    return bytes(out[:n])
+
+
+def test_dzr_hand_vectors():
+    """Bit-layout pins computed by hand from the reference protocol:
+    - 25 equal elements -> long repeat: 10-bit header (flag 0000, 3x0,
+      byte-cnt-1=0) + 8 bits of (25-18)=7 -> bytes 00 1C + flush 00
+    - 3 equal elements -> three single '1' bits -> byte 0x07"""
+    assert _dzr_rt([0] * 25, 8) == bytes([0x00, 0x1C, 0x00])
+    assert _dzr_rt([0] * 3, 4) == bytes([0x07])
+    # [0, 1] on u64: element 0 is a zero-delta run (one '1' bit), then
+    # delta 1 -> zigzag 2 -> N2(6)+2 bits (2<<2)|2 = 0b001010; stream
+    # bits LSB-first: 1 | (0b001010 << 1) = 0b0010101 = 0x15
+    assert _dzr_rt([0, 1], 8)[0] == 0x15
+
+
+@pytest.mark.parametrize("wb", [1, 2, 4, 8])
+def test_dzr_roundtrip(wb):
+    rng = np.random.default_rng(wb)
+    lim = 1 << min(8 * wb, 63)
+    for trial in range(40):
+        n = int(rng.integers(1, 500))
+        style = trial % 4
+        if style == 0:
+            v = rng.integers(0, lim, n)
+        elif style == 1:
+            v = np.repeat(rng.integers(0, 50, n // 25 + 1), 25)[:n]
+        elif style == 2:
+            v = np.cumsum(rng.integers(0, 3, n)).astype(np.uint64)
+        else:
+            v = np.full(n, int(rng.integers(0, lim)))
+        _dzr_rt(v, wb)
+
+
+def test_int_stream_dzr_type():
+    """Stream layer with EncodingType DELTA_ZIGZAG_RLE (meta type 4):
+    monotone-ish data compresses well and round-trips."""
+    rows = 2000
+    rng = np.random.default_rng(44)
+    v = (10_000 + np.cumsum(rng.integers(0, 3, rows))).astype(np.int64)
+    cap = 64 + rows * 8
+    buf = (C.c_uint8 * cap)()
+    n = _lib.obx_cs_int_stream_enc2(
+        v.ctypes.data_as(C.POINTER(C.c_int64)), None, rows, 4, buf, cap)
+    assert n > 0
+    assert bytes(buf[:7])[2] == 4  # meta type byte = DELTA_ZIGZAG_RLE
+    out = np.zeros(rows, dtype=np.int64)
+    m = _lib.obx_cs_int_stream_dec(buf, n, rows,
+                                   out.ctypes.data_as(C.POINTER(C.c_int64)),
+                                   None)
+    assert m == n and (out == v).all()
+    # and it actually compresses: far below the 2-byte RAW encoding
+    assert n < rows  # < 1 byte/row on this data
