@@ -155,8 +155,12 @@ int64_t obx_cs_int_stream_enc2(const int64_t *vals, const uint8_t *nulls,
   }
   if (enc_type == OBX_CS_ENC_RAW) {
     pos += (size_t)rows * wb;
-  } else if (enc_type == OBX_CS_ENC_DELTA_ZIGZAG_RLE) {
-    int64_t n = obx_cs_dzr_enc(packed, rows, wb, buf + pos, cap - pos);
+  } else if (enc_type == OBX_CS_ENC_DELTA_ZIGZAG_RLE ||
+             enc_type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE) {
+    int64_t n = enc_type == OBX_CS_ENC_DELTA_ZIGZAG_RLE
+                    ? obx_cs_dzr_enc(packed, rows, wb, buf + pos, cap - pos)
+                    : obx_cs_ddzr_enc(packed, rows, wb, buf + pos,
+                                      cap - pos);
     free(tmp);
     if (n < 0) return -1;
     pos += (size_t)n;
@@ -183,10 +187,14 @@ int64_t obx_cs_int_stream_dec(const uint8_t *buf, size_t len, uint32_t rows,
       pos += wb;
       out[r] = (int64_t)(ele + base);
     }
-  } else if (m.type == OBX_CS_ENC_DELTA_ZIGZAG_RLE) {
+  } else if (m.type == OBX_CS_ENC_DELTA_ZIGZAG_RLE ||
+             m.type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE) {
     uint8_t *packed = (uint8_t *)malloc((size_t)rows * wb);
     if (!packed) return -1;
-    int64_t n = obx_cs_dzr_dec(buf + pos, len - pos, rows, wb, packed);
+    int64_t n = m.type == OBX_CS_ENC_DELTA_ZIGZAG_RLE
+                    ? obx_cs_dzr_dec(buf + pos, len - pos, rows, wb, packed)
+                    : obx_cs_ddzr_dec(buf + pos, len - pos, rows, wb,
+                                      packed);
     if (n < 0) { free(packed); return -1; }
     pos += (size_t)n;
     for (uint32_t r = 0; r < rows; r++) {
@@ -401,23 +409,31 @@ static int dzr_emit(dzr_e *e, uint64_t r, uint64_t delta_zz, int wi) {
   return 0;
 }
 
-int64_t obx_cs_dzr_enc(const uint8_t *in, uint32_t count, uint32_t wb,
-                       uint8_t *out, size_t cap) {
+/* order 1 = DELTA_ZIGZAG_RLE; order 2 = DOUBLE_DELTA_ZIGZAG_RLE
+   (ob_double_delta_zigzag_rle.h: dd = delta - prev_delta; repeat runs
+   are constant-SLOPE spans, decoded as start += pd each) */
+static int64_t dzr_enc_core(const uint8_t *in, uint32_t count, uint32_t wb,
+                            uint8_t *out, size_t cap, int order) {
   int wi = wb == 1 ? 0 : wb == 2 ? 1 : wb == 4 ? 2 : 3;
   uint32_t wbits = wb * 8;
   uint64_t wmask = wbits >= 64 ? ~0ull : ((1ull << wbits) - 1);
   dzr_e e = {0, 0, out, out + cap};
   if (cap < 24) return -1;
-  uint64_t start = 0, pending = 0;
+  uint64_t start = 0, pd = 0, pending = 0;
   for (uint32_t i = 0; i < count; i++) {
     uint64_t v = dzr_load(in + (size_t)i * wb, wb);
-    uint64_t delta = (v - start) & wmask;
+    uint64_t d1 = (v - start) & wmask;
     start = v;
-    if (delta == 0) {
+    uint64_t dd = d1;
+    if (order == 2) {
+      dd = (d1 - pd) & wmask;
+      pd = d1;
+    }
+    if (dd == 0) {
       pending++;
       continue;
     }
-    if (dzr_emit(&e, pending, dzr_zz_enc(delta, wbits), wi)) return -1;
+    if (dzr_emit(&e, pending, dzr_zz_enc(dd, wbits), wi)) return -1;
     pending = 0;
   }
   if (pending) {
@@ -428,6 +444,16 @@ int64_t obx_cs_dzr_enc(const uint8_t *in, uint32_t count, uint32_t wb,
   memcpy(e.op, &e.bw, 8);
   e.op += (e.br + 7) >> 3;
   return (int64_t)(e.op - out);
+}
+
+int64_t obx_cs_dzr_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                       uint8_t *out, size_t cap) {
+  return dzr_enc_core(in, count, wb, out, cap, 1);
+}
+
+int64_t obx_cs_ddzr_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                        uint8_t *out, size_t cap) {
+  return dzr_enc_core(in, count, wb, out, cap, 2);
 }
 
 typedef struct {
@@ -455,14 +481,15 @@ static inline uint64_t dzr_get(dzr_d *d, uint32_t b) { /* get57 semantics */
   return v;
 }
 
-int64_t obx_cs_dzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
-                       uint32_t wb, uint8_t *out) {
+static int64_t dzr_dec_core(const uint8_t *in, size_t in_len,
+                            uint32_t count, uint32_t wb, uint8_t *out,
+                            int order) {
   int wi = wb == 1 ? 0 : wb == 2 ? 1 : wb == 4 ? 2 : 3;
   uint32_t wbits = wb * 8;
   uint64_t wmask = wbits >= 64 ? ~0ull : ((1ull << wbits) - 1);
   const uint32_t N2 = DZR_N2[wi], N3 = DZR_N3[wi], N4 = DZR_N4[wi];
   dzr_d d = {0, 0, in, in + in_len};
-  uint64_t start = 0;
+  uint64_t start = 0, pd = 0;
   uint32_t done = 0;
   dzr_d_slide(&d);
   while (done < count) {
@@ -489,6 +516,7 @@ int64_t obx_cs_dzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
         dzr_d_slide(&d);
         r += DZR_BASE_REPEAT;
         while (r-- && done < count) {
+          if (order == 2) start = (start + pd) & wmask;
           memcpy(out + (size_t)done * wb, &start, wb);
           done++;
         }
@@ -505,7 +533,12 @@ int64_t obx_cs_dzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
         delta = dzr_get(&d, bits);
       }
     }
-    start = (start + dzr_zz_dec(delta, wbits)) & wmask;
+    if (order == 2) {
+      pd = (pd + dzr_zz_dec(delta, wbits)) & wmask;
+      start = (start + pd) & wmask;
+    } else {
+      start = (start + dzr_zz_dec(delta, wbits)) & wmask;
+    }
     memcpy(out + (size_t)done * wb, &start, wb);
     done++;
     dzr_d_slide(&d);
@@ -513,4 +546,14 @@ int64_t obx_cs_dzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
   /* align (ob_bp_util.h:152) */
   d.ip += (d.br + 7) >> 3;
   return (int64_t)(d.ip - in);
+}
+
+int64_t obx_cs_dzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                       uint32_t wb, uint8_t *out) {
+  return dzr_dec_core(in, in_len, count, wb, out, 1);
+}
+
+int64_t obx_cs_ddzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                        uint32_t wb, uint8_t *out) {
+  return dzr_dec_core(in, in_len, count, wb, out, 2);
 }

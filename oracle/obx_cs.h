@@ -40,6 +40,7 @@ enum {
 /* ObIntegerStream::EncodingType (subset implemented) */
 enum {
   OBX_CS_ENC_RAW = 1,
+  OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE = 2,
   OBX_CS_ENC_DELTA_ZIGZAG_RLE = 4,
 };
 
@@ -124,6 +125,10 @@ int64_t obx_cs_dzr_enc(const uint8_t *in, uint32_t count, uint32_t wb,
                        uint8_t *out, size_t cap);
 int64_t obx_cs_dzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
                        uint32_t wb, uint8_t *out);
+int64_t obx_cs_ddzr_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                        uint8_t *out, size_t cap);
+int64_t obx_cs_ddzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                        uint32_t wb, uint8_t *out);
 
 #ifdef __cplusplus
 }
