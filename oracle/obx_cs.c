@@ -156,11 +156,15 @@ int64_t obx_cs_int_stream_enc2(const int64_t *vals, const uint8_t *nulls,
   if (enc_type == OBX_CS_ENC_RAW) {
     pos += (size_t)rows * wb;
   } else if (enc_type == OBX_CS_ENC_DELTA_ZIGZAG_RLE ||
-             enc_type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE) {
+             enc_type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE ||
+             enc_type == OBX_CS_ENC_DELTA_ZIGZAG_PFOR) {
     int64_t n = enc_type == OBX_CS_ENC_DELTA_ZIGZAG_RLE
                     ? obx_cs_dzr_enc(packed, rows, wb, buf + pos, cap - pos)
-                    : obx_cs_ddzr_enc(packed, rows, wb, buf + pos,
-                                      cap - pos);
+                : enc_type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE
+                    ? obx_cs_ddzr_enc(packed, rows, wb, buf + pos,
+                                      cap - pos)
+                    : obx_cs_dzp_enc(packed, rows, wb, buf + pos,
+                                     cap - pos);
     free(tmp);
     if (n < 0) return -1;
     pos += (size_t)n;
@@ -188,13 +192,17 @@ int64_t obx_cs_int_stream_dec(const uint8_t *buf, size_t len, uint32_t rows,
       out[r] = (int64_t)(ele + base);
     }
   } else if (m.type == OBX_CS_ENC_DELTA_ZIGZAG_RLE ||
-             m.type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE) {
+             m.type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE ||
+             m.type == OBX_CS_ENC_DELTA_ZIGZAG_PFOR) {
     uint8_t *packed = (uint8_t *)malloc((size_t)rows * wb);
     if (!packed) return -1;
     int64_t n = m.type == OBX_CS_ENC_DELTA_ZIGZAG_RLE
                     ? obx_cs_dzr_dec(buf + pos, len - pos, rows, wb, packed)
-                    : obx_cs_ddzr_dec(buf + pos, len - pos, rows, wb,
-                                      packed);
+                : m.type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE
+                    ? obx_cs_ddzr_dec(buf + pos, len - pos, rows, wb,
+                                      packed)
+                    : obx_cs_dzp_dec(buf + pos, len - pos, rows, wb,
+                                     packed);
     if (n < 0) { free(packed); return -1; }
     pos += (size_t)n;
     for (uint32_t r = 0; r < rows; r++) {
@@ -556,4 +564,210 @@ int64_t obx_cs_dzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
 int64_t obx_cs_ddzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
                         uint32_t wb, uint8_t *out) {
   return dzr_dec_core(in, in_len, count, wb, out, 2);
+}
+
+/* ====================================================================== */
+/* DELTA_ZIGZAG_PFOR codec (ObDeltaZigzagFixedPfor, ob_delta_zigzag_pfor.h
+ * + ObSIMDFixedPFor::__encode_array, ob_simd_fixed_pfor.h:30-200, in the
+ * CPU_ARCH_INDEPENDANT_SCALAR layout = flat LSB-first bit packing,
+ * ob_bp_helpers.h:1378 scalar_bit_packing over the generated sequential
+ * pack layout). Frame: 128-value blocks of zigzag deltas; per block the
+ * cost model (find_most_fit_bx) picks a low width b with exceptions:
+ * header [b] (no exceptions) or [0x80|b][bx]; with exceptions the block
+ * is [16-byte exception bitmap][exceptions' high parts (v>>b) packed at
+ * bx bits, byte-rounded][all values' low b bits packed]. The tail
+ * (<128 values) is ObSimpleBitPacking: [max-bits byte][flat packed]. */
+
+#define DZP_BLOCK 128
+
+static inline uint32_t dzp_bits0(uint64_t v) { /* gccbits: 0 for 0 */
+  uint32_t n = 0;
+  while (v) { n++; v >>= 1; }
+  return n;
+}
+
+/* flat LSB-first bit packing, byte-rounded at the end (up to 71 bits per
+   element span handled through a 128-bit accumulator; oracle clarity
+   over speed) */
+static size_t dzp_pack(const uint64_t *v, uint32_t n, uint32_t b,
+                       uint8_t *out) {
+  if (b == 0) return 0;
+  size_t total = ((size_t)n * b + 7) / 8;
+  memset(out, 0, total + 8);
+  uint64_t bitpos = 0;
+  for (uint32_t i = 0; i < n; i++) {
+    uint64_t val = b >= 64 ? v[i] : (v[i] & ((1ull << b) - 1));
+    size_t byte = bitpos >> 3;
+    uint32_t sh = (uint32_t)(bitpos & 7);
+    unsigned __int128 acc = (unsigned __int128)val << sh;
+    for (uint32_t k = 0; k * 8 < sh + b; k++)
+      out[byte + k] |= (uint8_t)(acc >> (8 * k));
+    bitpos += b;
+  }
+  return total;
+}
+
+static void dzp_unpack(const uint8_t *in, uint32_t n, uint32_t b,
+                       uint64_t *out) {
+  if (b == 0) {
+    memset(out, 0, (size_t)n * 8);
+    return;
+  }
+  uint64_t bitpos = 0;
+  for (uint32_t i = 0; i < n; i++) {
+    size_t byte = bitpos >> 3;
+    uint32_t sh = (uint32_t)(bitpos & 7);
+    unsigned __int128 acc = 0;
+    for (uint32_t k = 0; k * 8 < sh + b; k++)
+      acc |= (unsigned __int128)in[byte + k] << (8 * k);
+    uint64_t v = (uint64_t)(acc >> sh);
+    if (b < 64) v &= (1ull << b) - 1;
+    out[i] = v;
+    bitpos += b;
+  }
+}
+
+/* find_most_fit_bx (ob_simd_fixed_pfor.h:30-87) */
+static void dzp_fit(const uint64_t *v, uint32_t n, uint32_t wbits,
+                    uint32_t *b_out, uint32_t *bx_out) {
+  uint32_t cnt[65] = {0};
+  uint64_t u = 0;
+  for (uint32_t i = 0; i < n; i++) {
+    cnt[dzp_bits0(v[i])]++;
+    u |= v[i];
+  }
+  int32_t b = (int32_t)dzp_bits0(u);
+  uint32_t bx = (uint32_t)b;
+  int32_t ml = (int32_t)(((uint64_t)n * b + 7) / 8) + 1;
+  uint32_t x = cnt[b];
+  uint32_t bmp8 = (n + 7) / 8;
+  for (int32_t i = b - 1; i >= 0; --i) {
+    int32_t l = (int32_t)(2 + bmp8 + (((uint64_t)x * (bx - i) + 7) / 8) +
+                          (((uint64_t)n * i + 7) / 8));
+    x += cnt[i];
+    if (l < ml) {
+      ml = l;
+      b = i;
+    }
+  }
+  *b_out = (uint32_t)b;
+  *bx_out = bx - (uint32_t)b;
+  (void)wbits;
+}
+
+int64_t obx_cs_dzp_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                       uint8_t *out, size_t cap) {
+  uint32_t wbits = wb * 8;
+  uint64_t wmask = wbits >= 64 ? ~0ull : ((1ull << wbits) - 1);
+  uint8_t *op = out;
+  uint8_t *end = out + cap;
+  uint64_t start = 0;
+  uint64_t zz[DZP_BLOCK];
+  uint32_t done = 0;
+  while (count - done >= DZP_BLOCK) {
+    for (uint32_t i = 0; i < DZP_BLOCK; i++) {
+      uint64_t v = dzr_load(in + (size_t)(done + i) * wb, wb);
+      zz[i] = dzr_zz_enc((v - start) & wmask, wbits);
+      start = v;
+    }
+    uint32_t b, bx;
+    dzp_fit(zz, DZP_BLOCK, wbits, &b, &bx);
+    if ((size_t)(end - op) < 2 + 16 + (size_t)DZP_BLOCK * wb + 24)
+      return -1;
+    if (bx == 0) {
+      *op++ = (uint8_t)b;
+      op += dzp_pack(zz, DZP_BLOCK, b, op);
+    } else {
+      *op++ = (uint8_t)(0x80 | b);
+      *op++ = (uint8_t)bx;
+      uint64_t msk = b >= 64 ? ~0ull : ((1ull << b) - 1);
+      uint64_t lowv[DZP_BLOCK], exc[DZP_BLOCK];
+      uint64_t xmap[DZP_BLOCK / 64] = {0, 0};
+      uint32_t xn = 0;
+      for (uint32_t i = 0; i < DZP_BLOCK; i++) {
+        lowv[i] = zz[i] & msk;
+        if (zz[i] > msk) {
+          xmap[i >> 6] |= 1ull << (i & 63);
+          exc[xn++] = zz[i] >> b;
+        }
+      }
+      memcpy(op, xmap, DZP_BLOCK / 8);
+      op += DZP_BLOCK / 8;
+      op += dzp_pack(exc, xn, bx, op);
+      op += dzp_pack(lowv, DZP_BLOCK, b, op);
+    }
+    done += DZP_BLOCK;
+  }
+  if (done < count) { /* tail: ObSimpleBitPacking (1-byte maxbits) */
+    uint32_t rem = count - done;
+    uint64_t u = 0;
+    for (uint32_t i = 0; i < rem; i++) {
+      uint64_t v = dzr_load(in + (size_t)(done + i) * wb, wb);
+      zz[i] = dzr_zz_enc((v - start) & wmask, wbits);
+      start = v;
+      u |= zz[i];
+    }
+    uint32_t b = dzp_bits0(u);
+    if ((size_t)(end - op) < 1 + (size_t)rem * wb + 16) return -1;
+    *op++ = (uint8_t)b;
+    op += dzp_pack(zz, rem, b, op);
+  }
+  return (int64_t)(op - out);
+}
+
+int64_t obx_cs_dzp_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                       uint32_t wb, uint8_t *out) {
+  uint32_t wbits = wb * 8;
+  uint64_t wmask = wbits >= 64 ? ~0ull : ((1ull << wbits) - 1);
+  const uint8_t *ip = in;
+  const uint8_t *end = in + in_len;
+  uint64_t start = 0;
+  uint64_t zz[DZP_BLOCK], exc[DZP_BLOCK];
+  uint32_t done = 0;
+  while (count - done >= DZP_BLOCK) {
+    if (ip >= end) return -1;
+    uint8_t h = *ip++;
+    /* 7-bit width field; the VByte variant is commented out in the
+       reference encoder (ob_simd_fixed_pfor.h:177) */
+    uint32_t b = h & 0x7F, bx = 0;
+    if (h & 0x80) {
+      if (ip >= end) return -1;
+      bx = *ip++;
+    }
+    if (bx == 0) {
+      dzp_unpack(ip, DZP_BLOCK, b, zz);
+      ip += ((size_t)DZP_BLOCK * b + 7) / 8;
+    } else {
+      uint64_t xmap[2];
+      if (ip + 16 > end) return -1;
+      memcpy(xmap, ip, 16);
+      ip += 16;
+      uint32_t xn = (uint32_t)(__builtin_popcountll(xmap[0]) +
+                               __builtin_popcountll(xmap[1]));
+      dzp_unpack(ip, xn, bx, exc);
+      ip += ((size_t)xn * bx + 7) / 8;
+      dzp_unpack(ip, DZP_BLOCK, b, zz);
+      ip += ((size_t)DZP_BLOCK * b + 7) / 8;
+      uint32_t xi = 0;
+      for (uint32_t i = 0; i < DZP_BLOCK; i++)
+        if ((xmap[i >> 6] >> (i & 63)) & 1) zz[i] |= exc[xi++] << b;
+    }
+    for (uint32_t i = 0; i < DZP_BLOCK; i++) {
+      start = (start + dzr_zz_dec(zz[i], wbits)) & wmask;
+      memcpy(out + (size_t)(done + i) * wb, &start, wb);
+    }
+    done += DZP_BLOCK;
+  }
+  if (done < count) {
+    uint32_t rem = count - done;
+    if (ip >= end) return -1;
+    uint32_t b = *ip++;
+    dzp_unpack(ip, rem, b, zz);
+    ip += ((size_t)rem * b + 7) / 8;
+    for (uint32_t i = 0; i < rem; i++) {
+      start = (start + dzr_zz_dec(zz[i], wbits)) & wmask;
+      memcpy(out + (size_t)(done + i) * wb, &start, wb);
+    }
+  }
+  return (int64_t)(ip - in);
 }

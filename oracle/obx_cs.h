@@ -42,6 +42,7 @@ enum {
   OBX_CS_ENC_RAW = 1,
   OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE = 2,
   OBX_CS_ENC_DELTA_ZIGZAG_RLE = 4,
+  OBX_CS_ENC_DELTA_ZIGZAG_PFOR = 5,
 };
 
 /* in-memory mirror of ObIntegerStreamMeta (the serialized form is
@@ -129,6 +130,11 @@ int64_t obx_cs_ddzr_enc(const uint8_t *in, uint32_t count, uint32_t wb,
                         uint8_t *out, size_t cap);
 int64_t obx_cs_ddzr_dec(const uint8_t *in, size_t in_len, uint32_t count,
                         uint32_t wb, uint8_t *out);
+/* DELTA_ZIGZAG_PFOR (128-value PFoR frames + SimpleBitPacking tail) */
+int64_t obx_cs_dzp_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                       uint8_t *out, size_t cap);
+int64_t obx_cs_dzp_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                       uint32_t wb, uint8_t *out);
 
 #ifdef __cplusplus
 }

@@ -322,3 +322,73 @@ def test_int_stream_ddzr_type():
                                    out.ctypes.data_as(C.POINTER(C.c_int64)),
                                    None)
     assert m == n and (out == v).all()
+
+
+# ---- DELTA_ZIGZAG_PFOR (128-value PFoR frames; ObDeltaZigzagFixedPfor +
+# ObSIMDFixedPFor scalar layout) ----
+
+_lib.obx_cs_dzp_enc.restype = C.c_int64
+_lib.obx_cs_dzp_enc.argtypes = _lib.obx_cs_dzr_enc.argtypes
+_lib.obx_cs_dzp_dec.restype = C.c_int64
+_lib.obx_cs_dzp_dec.argtypes = _lib.obx_cs_dzr_dec.argtypes
+
+
+def _dzp_rt(arr, wb):
+    dt = {1: np.uint8, 2: np.uint16, 4: np.uint32, 8: np.uint64}[wb]
+    packed = np.asarray(arr).astype(dt)
+    inb = packed.tobytes()
+    cap = len(inb) * 3 + 128
+    out = (C.c_uint8 * cap)()
+    n = _lib.obx_cs_dzp_enc((C.c_uint8 * len(inb)).from_buffer_copy(inb),
+                            len(packed), wb, out, cap)
+    assert n > 0
+    dec = (C.c_uint8 * len(inb))()
+    m = _lib.obx_cs_dzp_dec(out, n, len(packed), wb, dec)
+    assert m == n
+    assert np.frombuffer(bytes(dec), dtype=dt).tolist() == packed.tolist()
+    return bytes(out[:n])
+
+
+def test_dzp_hand_vectors():
+    """Constant data: every block's deltas are all zero -> b=0, bx=0 ->
+    one header byte per 128-value block and [maxbits=0] for the tail."""
+    assert _dzp_rt([0] * 128, 4) == bytes([0x00])
+    assert _dzp_rt([0] * 130, 4) == bytes([0x00, 0x00])
+
+
+@pytest.mark.parametrize("wb", [1, 2, 4, 8])
+def test_dzp_roundtrip(wb):
+    rng = np.random.default_rng(200 + wb)
+    lim = 1 << min(8 * wb, 63)
+    for trial in range(40):
+        n = int(rng.integers(1, 700))
+        style = trial % 4
+        if style == 0:
+            v = rng.integers(0, lim, n)
+        elif style == 1:  # smooth with outliers: the PFoR sweet spot
+            v = np.cumsum(rng.integers(0, 5, n)).astype(np.uint64)
+            for r in rng.choice(n, max(1, n // 30), replace=False):
+                v[r] = int(rng.integers(0, lim))
+        elif style == 2:
+            v = np.full(n, int(rng.integers(0, lim)))
+        else:
+            v = np.cumsum(rng.integers(0, 2, n)).astype(np.uint64)
+        _dzp_rt(v, wb)
+
+
+def test_int_stream_dzp_type():
+    rows = 1000
+    rng = np.random.default_rng(55)
+    v = (10**9 + np.cumsum(rng.integers(0, 7, rows))).astype(np.int64)
+    v[::97] += 10**6  # outliers -> exception path
+    cap = 64 + rows * 8
+    buf = (C.c_uint8 * cap)()
+    n = _lib.obx_cs_int_stream_enc2(
+        v.ctypes.data_as(C.POINTER(C.c_int64)), None, rows, 5, buf, cap)
+    assert n > 0
+    out = np.zeros(rows, dtype=np.int64)
+    m = _lib.obx_cs_int_stream_dec(buf, n, rows,
+                                   out.ctypes.data_as(C.POINTER(C.c_int64)),
+                                   None)
+    assert m == n and (out == v).all()
+    assert n < rows * 2  # far below 8-byte raw
