@@ -157,14 +157,22 @@ int64_t obx_cs_int_stream_enc2(const int64_t *vals, const uint8_t *nulls,
     pos += (size_t)rows * wb;
   } else if (enc_type == OBX_CS_ENC_DELTA_ZIGZAG_RLE ||
              enc_type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE ||
-             enc_type == OBX_CS_ENC_DELTA_ZIGZAG_PFOR) {
+             enc_type == OBX_CS_ENC_DELTA_ZIGZAG_PFOR ||
+             enc_type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_PFOR ||
+             enc_type == OBX_CS_ENC_SIMD_FIXEDPFOR) {
     int64_t n = enc_type == OBX_CS_ENC_DELTA_ZIGZAG_RLE
                     ? obx_cs_dzr_enc(packed, rows, wb, buf + pos, cap - pos)
                 : enc_type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE
                     ? obx_cs_ddzr_enc(packed, rows, wb, buf + pos,
                                       cap - pos)
-                    : obx_cs_dzp_enc(packed, rows, wb, buf + pos,
-                                     cap - pos);
+                : enc_type == OBX_CS_ENC_DELTA_ZIGZAG_PFOR
+                    ? obx_cs_dzp_enc(packed, rows, wb, buf + pos,
+                                     cap - pos)
+                : enc_type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_PFOR
+                    ? obx_cs_ddzp_enc(packed, rows, wb, buf + pos,
+                                      cap - pos)
+                    : obx_cs_fpfor_enc(packed, rows, wb, buf + pos,
+                                       cap - pos);
     free(tmp);
     if (n < 0) return -1;
     pos += (size_t)n;
@@ -193,7 +201,9 @@ int64_t obx_cs_int_stream_dec(const uint8_t *buf, size_t len, uint32_t rows,
     }
   } else if (m.type == OBX_CS_ENC_DELTA_ZIGZAG_RLE ||
              m.type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE ||
-             m.type == OBX_CS_ENC_DELTA_ZIGZAG_PFOR) {
+             m.type == OBX_CS_ENC_DELTA_ZIGZAG_PFOR ||
+             m.type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_PFOR ||
+             m.type == OBX_CS_ENC_SIMD_FIXEDPFOR) {
     uint8_t *packed = (uint8_t *)malloc((size_t)rows * wb);
     if (!packed) return -1;
     int64_t n = m.type == OBX_CS_ENC_DELTA_ZIGZAG_RLE
@@ -201,8 +211,13 @@ int64_t obx_cs_int_stream_dec(const uint8_t *buf, size_t len, uint32_t rows,
                 : m.type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE
                     ? obx_cs_ddzr_dec(buf + pos, len - pos, rows, wb,
                                       packed)
-                    : obx_cs_dzp_dec(buf + pos, len - pos, rows, wb,
-                                     packed);
+                : m.type == OBX_CS_ENC_DELTA_ZIGZAG_PFOR
+                    ? obx_cs_dzp_dec(buf + pos, len - pos, rows, wb, packed)
+                : m.type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_PFOR
+                    ? obx_cs_ddzp_dec(buf + pos, len - pos, rows, wb,
+                                      packed)
+                    : obx_cs_fpfor_dec(buf + pos, len - pos, rows, wb,
+                                       packed);
     if (n < 0) { free(packed); return -1; }
     pos += (size_t)n;
     for (uint32_t r = 0; r < rows; r++) {
@@ -655,20 +670,34 @@ static void dzp_fit(const uint64_t *v, uint32_t n, uint32_t wbits,
   (void)wbits;
 }
 
-int64_t obx_cs_dzp_enc(const uint8_t *in, uint32_t count, uint32_t wb,
-                       uint8_t *out, size_t cap) {
+/* transform: 1 = delta+zigzag (DELTA_ZIGZAG_PFOR), 2 = double-delta+
+   zigzag (DOUBLE_DELTA_ZIGZAG_PFOR), 0 = none (SIMD_FIXEDPFOR; the
+   reference frames EVERYTHING in 128s over padded buffers — our
+   restatement zero-pads the final partial frame, documented divergence
+   from the in-place over-read) */
+static int64_t dzp_enc_core(const uint8_t *in, uint32_t count, uint32_t wb,
+                            uint8_t *out, size_t cap, int transform) {
   uint32_t wbits = wb * 8;
   uint64_t wmask = wbits >= 64 ? ~0ull : ((1ull << wbits) - 1);
   uint8_t *op = out;
   uint8_t *end = out + cap;
-  uint64_t start = 0;
+  uint64_t start = 0, pd = 0;
   uint64_t zz[DZP_BLOCK];
   uint32_t done = 0;
   while (count - done >= DZP_BLOCK) {
     for (uint32_t i = 0; i < DZP_BLOCK; i++) {
       uint64_t v = dzr_load(in + (size_t)(done + i) * wb, wb);
-      zz[i] = dzr_zz_enc((v - start) & wmask, wbits);
-      start = v;
+      if (transform == 0) {
+        zz[i] = v;
+      } else if (transform == 1) {
+        zz[i] = dzr_zz_enc((v - start) & wmask, wbits);
+        start = v;
+      } else {
+        uint64_t d1 = (v - start) & wmask;
+        start = v;
+        zz[i] = dzr_zz_enc((d1 - pd) & wmask, wbits);
+        pd = d1;
+      }
     }
     uint32_t b, bx;
     dzp_fit(zz, DZP_BLOCK, wbits, &b, &bx);
@@ -703,8 +732,17 @@ int64_t obx_cs_dzp_enc(const uint8_t *in, uint32_t count, uint32_t wb,
     uint64_t u = 0;
     for (uint32_t i = 0; i < rem; i++) {
       uint64_t v = dzr_load(in + (size_t)(done + i) * wb, wb);
-      zz[i] = dzr_zz_enc((v - start) & wmask, wbits);
-      start = v;
+      if (transform == 0) {
+        zz[i] = v;
+      } else if (transform == 1) {
+        zz[i] = dzr_zz_enc((v - start) & wmask, wbits);
+        start = v;
+      } else {
+        uint64_t d1 = (v - start) & wmask;
+        start = v;
+        zz[i] = dzr_zz_enc((d1 - pd) & wmask, wbits);
+        pd = d1;
+      }
       u |= zz[i];
     }
     uint32_t b = dzp_bits0(u);
@@ -715,13 +753,29 @@ int64_t obx_cs_dzp_enc(const uint8_t *in, uint32_t count, uint32_t wb,
   return (int64_t)(op - out);
 }
 
-int64_t obx_cs_dzp_dec(const uint8_t *in, size_t in_len, uint32_t count,
-                       uint32_t wb, uint8_t *out) {
+int64_t obx_cs_dzp_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                       uint8_t *out, size_t cap) {
+  return dzp_enc_core(in, count, wb, out, cap, 1);
+}
+
+int64_t obx_cs_ddzp_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                        uint8_t *out, size_t cap) {
+  return dzp_enc_core(in, count, wb, out, cap, 2);
+}
+
+int64_t obx_cs_fpfor_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                         uint8_t *out, size_t cap) {
+  return dzp_enc_core(in, count, wb, out, cap, 0);
+}
+
+static int64_t dzp_dec_core(const uint8_t *in, size_t in_len,
+                            uint32_t count, uint32_t wb, uint8_t *out,
+                            int transform) {
   uint32_t wbits = wb * 8;
   uint64_t wmask = wbits >= 64 ? ~0ull : ((1ull << wbits) - 1);
   const uint8_t *ip = in;
   const uint8_t *end = in + in_len;
-  uint64_t start = 0;
+  uint64_t start = 0, pd = 0;
   uint64_t zz[DZP_BLOCK], exc[DZP_BLOCK];
   uint32_t done = 0;
   while (count - done >= DZP_BLOCK) {
@@ -753,7 +807,14 @@ int64_t obx_cs_dzp_dec(const uint8_t *in, size_t in_len, uint32_t count,
         if ((xmap[i >> 6] >> (i & 63)) & 1) zz[i] |= exc[xi++] << b;
     }
     for (uint32_t i = 0; i < DZP_BLOCK; i++) {
-      start = (start + dzr_zz_dec(zz[i], wbits)) & wmask;
+      if (transform == 0) {
+        start = zz[i] & wmask;
+      } else if (transform == 1) {
+        start = (start + dzr_zz_dec(zz[i], wbits)) & wmask;
+      } else {
+        pd = (pd + dzr_zz_dec(zz[i], wbits)) & wmask;
+        start = (start + pd) & wmask;
+      }
       memcpy(out + (size_t)(done + i) * wb, &start, wb);
     }
     done += DZP_BLOCK;
@@ -765,9 +826,31 @@ int64_t obx_cs_dzp_dec(const uint8_t *in, size_t in_len, uint32_t count,
     dzp_unpack(ip, rem, b, zz);
     ip += ((size_t)rem * b + 7) / 8;
     for (uint32_t i = 0; i < rem; i++) {
-      start = (start + dzr_zz_dec(zz[i], wbits)) & wmask;
+      if (transform == 0) {
+        start = zz[i] & wmask;
+      } else if (transform == 1) {
+        start = (start + dzr_zz_dec(zz[i], wbits)) & wmask;
+      } else {
+        pd = (pd + dzr_zz_dec(zz[i], wbits)) & wmask;
+        start = (start + pd) & wmask;
+      }
       memcpy(out + (size_t)(done + i) * wb, &start, wb);
     }
   }
   return (int64_t)(ip - in);
+}
+
+int64_t obx_cs_dzp_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                       uint32_t wb, uint8_t *out) {
+  return dzp_dec_core(in, in_len, count, wb, out, 1);
+}
+
+int64_t obx_cs_ddzp_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                        uint32_t wb, uint8_t *out) {
+  return dzp_dec_core(in, in_len, count, wb, out, 2);
+}
+
+int64_t obx_cs_fpfor_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                         uint32_t wb, uint8_t *out) {
+  return dzp_dec_core(in, in_len, count, wb, out, 0);
 }

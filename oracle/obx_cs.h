@@ -42,7 +42,9 @@ enum {
   OBX_CS_ENC_RAW = 1,
   OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE = 2,
   OBX_CS_ENC_DELTA_ZIGZAG_RLE = 4,
+  OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_PFOR = 3,
   OBX_CS_ENC_DELTA_ZIGZAG_PFOR = 5,
+  OBX_CS_ENC_SIMD_FIXEDPFOR = 6,
 };
 
 /* in-memory mirror of ObIntegerStreamMeta (the serialized form is
@@ -135,6 +137,14 @@ int64_t obx_cs_dzp_enc(const uint8_t *in, uint32_t count, uint32_t wb,
                        uint8_t *out, size_t cap);
 int64_t obx_cs_dzp_dec(const uint8_t *in, size_t in_len, uint32_t count,
                        uint32_t wb, uint8_t *out);
+int64_t obx_cs_ddzp_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                        uint8_t *out, size_t cap);
+int64_t obx_cs_ddzp_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                        uint32_t wb, uint8_t *out);
+int64_t obx_cs_fpfor_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                         uint8_t *out, size_t cap);
+int64_t obx_cs_fpfor_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                         uint32_t wb, uint8_t *out);
 
 #ifdef __cplusplus
 }
