@@ -159,7 +159,8 @@ int64_t obx_cs_int_stream_enc2(const int64_t *vals, const uint8_t *nulls,
              enc_type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE ||
              enc_type == OBX_CS_ENC_DELTA_ZIGZAG_PFOR ||
              enc_type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_PFOR ||
-             enc_type == OBX_CS_ENC_SIMD_FIXEDPFOR) {
+             enc_type == OBX_CS_ENC_SIMD_FIXEDPFOR ||
+             enc_type == OBX_CS_ENC_XOR_FIXED_PFOR) {
     int64_t n = enc_type == OBX_CS_ENC_DELTA_ZIGZAG_RLE
                     ? obx_cs_dzr_enc(packed, rows, wb, buf + pos, cap - pos)
                 : enc_type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE
@@ -171,7 +172,10 @@ int64_t obx_cs_int_stream_enc2(const int64_t *vals, const uint8_t *nulls,
                 : enc_type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_PFOR
                     ? obx_cs_ddzp_enc(packed, rows, wb, buf + pos,
                                       cap - pos)
-                    : obx_cs_fpfor_enc(packed, rows, wb, buf + pos,
+                : enc_type == OBX_CS_ENC_SIMD_FIXEDPFOR
+                    ? obx_cs_fpfor_enc(packed, rows, wb, buf + pos,
+                                       cap - pos)
+                    : obx_cs_xpfor_enc(packed, rows, wb, buf + pos,
                                        cap - pos);
     free(tmp);
     if (n < 0) return -1;
@@ -203,7 +207,8 @@ int64_t obx_cs_int_stream_dec(const uint8_t *buf, size_t len, uint32_t rows,
              m.type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE ||
              m.type == OBX_CS_ENC_DELTA_ZIGZAG_PFOR ||
              m.type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_PFOR ||
-             m.type == OBX_CS_ENC_SIMD_FIXEDPFOR) {
+             m.type == OBX_CS_ENC_SIMD_FIXEDPFOR ||
+             m.type == OBX_CS_ENC_XOR_FIXED_PFOR) {
     uint8_t *packed = (uint8_t *)malloc((size_t)rows * wb);
     if (!packed) return -1;
     int64_t n = m.type == OBX_CS_ENC_DELTA_ZIGZAG_RLE
@@ -216,7 +221,10 @@ int64_t obx_cs_int_stream_dec(const uint8_t *buf, size_t len, uint32_t rows,
                 : m.type == OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_PFOR
                     ? obx_cs_ddzp_dec(buf + pos, len - pos, rows, wb,
                                       packed)
-                    : obx_cs_fpfor_dec(buf + pos, len - pos, rows, wb,
+                : m.type == OBX_CS_ENC_SIMD_FIXEDPFOR
+                    ? obx_cs_fpfor_dec(buf + pos, len - pos, rows, wb,
+                                       packed)
+                    : obx_cs_xpfor_dec(buf + pos, len - pos, rows, wb,
                                        packed);
     if (n < 0) { free(packed); return -1; }
     pos += (size_t)n;
@@ -753,9 +761,170 @@ static int64_t dzp_enc_core(const uint8_t *in, uint32_t count, uint32_t wb,
   return (int64_t)(op - out);
 }
 
+/* one 128-value PFoR frame (header + optional exception section + data) */
+static uint8_t *dzp_frame_enc(const uint64_t *zz, uint32_t wbits,
+                              uint8_t *op) {
+  uint32_t b, bx;
+  dzp_fit(zz, DZP_BLOCK, wbits, &b, &bx);
+  if (bx == 0) {
+    *op++ = (uint8_t)b;
+    op += dzp_pack(zz, DZP_BLOCK, b, op);
+  } else {
+    *op++ = (uint8_t)(0x80 | b);
+    *op++ = (uint8_t)bx;
+    uint64_t msk = b >= 64 ? ~0ull : ((1ull << b) - 1);
+    uint64_t lowv[DZP_BLOCK], exc[DZP_BLOCK];
+    uint64_t xmap[DZP_BLOCK / 64] = {0, 0};
+    uint32_t xn = 0;
+    for (uint32_t i = 0; i < DZP_BLOCK; i++) {
+      lowv[i] = zz[i] & msk;
+      if (zz[i] > msk) {
+        xmap[i >> 6] |= 1ull << (i & 63);
+        exc[xn++] = zz[i] >> b;
+      }
+    }
+    memcpy(op, xmap, DZP_BLOCK / 8);
+    op += DZP_BLOCK / 8;
+    op += dzp_pack(exc, xn, bx, op);
+    op += dzp_pack(lowv, DZP_BLOCK, b, op);
+  }
+  return op;
+}
+
+static const uint8_t *dzp_frame_dec(const uint8_t *ip, const uint8_t *end,
+                                    uint64_t *zz) {
+  if (ip >= end) return NULL;
+  uint8_t h = *ip++;
+  uint32_t b = h & 0x7F, bx = 0;
+  if (h & 0x80) {
+    if (ip >= end) return NULL;
+    bx = *ip++;
+  }
+  if (bx == 0) {
+    dzp_unpack(ip, DZP_BLOCK, b, zz);
+    ip += ((size_t)DZP_BLOCK * b + 7) / 8;
+  } else {
+    uint64_t xmap[2], exc[DZP_BLOCK];
+    if (ip + 16 > end) return NULL;
+    memcpy(xmap, ip, 16);
+    ip += 16;
+    uint32_t xn = (uint32_t)(__builtin_popcountll(xmap[0]) +
+                             __builtin_popcountll(xmap[1]));
+    dzp_unpack(ip, xn, bx, exc);
+    ip += ((size_t)xn * bx + 7) / 8;
+    dzp_unpack(ip, DZP_BLOCK, b, zz);
+    ip += ((size_t)DZP_BLOCK * b + 7) / 8;
+    uint32_t xi = 0;
+    for (uint32_t i = 0; i < DZP_BLOCK; i++)
+      if ((xmap[i >> 6] >> (i & 63)) & 1) zz[i] |= exc[xi++] << b;
+  }
+  return ip;
+}
+
 int64_t obx_cs_dzp_enc(const uint8_t *in, uint32_t count, uint32_t wb,
                        uint8_t *out, size_t cap) {
   return dzp_enc_core(in, count, wb, out, cap, 1);
+}
+
+/* XOR_FIXED_PFOR (ObXorFixedPforInner, ob_xor_fixed_pfor.h:22-175):
+ * per 128 block: v = cur ^ prev; shift = wbits - bits(OR of xors); each
+ * value is (v << shift) BIT-REVERSED at element width; frame =
+ * [shift byte][PFoR frame of the reversed values]. Tail:
+ * [shift byte][SimpleBitPacking (own maxbits byte)]. */
+static inline uint64_t dzp_bitrev(uint64_t v, uint32_t wbits) {
+  uint64_t r = 0;
+  for (uint32_t i = 0; i < wbits; i++) {
+    r = (r << 1) | (v & 1);
+    v >>= 1;
+  }
+  return r;
+}
+
+int64_t obx_cs_xpfor_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                         uint8_t *out, size_t cap) {
+  uint32_t wbits = wb * 8;
+  uint64_t wmask = wbits >= 64 ? ~0ull : ((1ull << wbits) - 1);
+  uint8_t *op = out;
+  uint8_t *end = out + cap;
+  uint64_t start = 0;
+  uint64_t v[DZP_BLOCK];
+  uint32_t done = 0;
+  while (count - done >= DZP_BLOCK) {
+    uint64_t orr = 0;
+    for (uint32_t i = 0; i < DZP_BLOCK; i++) {
+      uint64_t cur = dzr_load(in + (size_t)(done + i) * wb, wb);
+      v[i] = (cur ^ start) & wmask;
+      orr |= v[i];
+      start = cur;
+    }
+    uint32_t sh = wbits - dzp_bits0(orr);
+    for (uint32_t i = 0; i < DZP_BLOCK; i++)
+      v[i] = dzp_bitrev((v[i] << sh) & wmask, wbits);
+    if ((size_t)(end - op) < 2 + 16 + (size_t)DZP_BLOCK * wb + 32)
+      return -1;
+    *op++ = (uint8_t)sh;
+    op = dzp_frame_enc(v, wbits, op);
+    done += DZP_BLOCK;
+  }
+  if (done < count) {
+    uint32_t rem = count - done;
+    uint64_t orr = 0;
+    for (uint32_t i = 0; i < rem; i++) {
+      uint64_t cur = dzr_load(in + (size_t)(done + i) * wb, wb);
+      v[i] = (cur ^ start) & wmask;
+      orr |= v[i];
+      start = cur;
+    }
+    uint32_t sh = wbits - dzp_bits0(orr);
+    uint64_t u = 0;
+    for (uint32_t i = 0; i < rem; i++) {
+      v[i] = dzp_bitrev((v[i] << sh) & wmask, wbits);
+      u |= v[i];
+    }
+    uint32_t b = dzp_bits0(u);
+    if ((size_t)(end - op) < 2 + (size_t)rem * wb + 16) return -1;
+    *op++ = (uint8_t)sh;
+    *op++ = (uint8_t)b; /* SimpleBitPacking maxbits byte */
+    op += dzp_pack(v, rem, b, op);
+  }
+  return (int64_t)(op - out);
+}
+
+int64_t obx_cs_xpfor_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                         uint32_t wb, uint8_t *out) {
+  uint32_t wbits = wb * 8;
+  uint64_t wmask = wbits >= 64 ? ~0ull : ((1ull << wbits) - 1);
+  const uint8_t *ip = in;
+  const uint8_t *end = in + in_len;
+  uint64_t start = 0;
+  uint64_t v[DZP_BLOCK];
+  uint32_t done = 0;
+  while (count - done >= DZP_BLOCK) {
+    if (ip >= end) return -1;
+    uint32_t sh = *ip++;
+    ip = dzp_frame_dec(ip, end, v);
+    if (!ip) return -1;
+    for (uint32_t i = 0; i < DZP_BLOCK; i++) {
+      uint64_t x = (dzp_bitrev(v[i] & wmask, wbits) >> sh) & wmask;
+      start = (x ^ start) & wmask;
+      memcpy(out + (size_t)(done + i) * wb, &start, wb);
+    }
+    done += DZP_BLOCK;
+  }
+  if (done < count) {
+    uint32_t rem = count - done;
+    if (ip + 2 > end) return -1;
+    uint32_t sh = *ip++;
+    uint32_t b = *ip++;
+    dzp_unpack(ip, rem, b, v);
+    ip += ((size_t)rem * b + 7) / 8;
+    for (uint32_t i = 0; i < rem; i++) {
+      uint64_t x = (dzp_bitrev(v[i] & wmask, wbits) >> sh) & wmask;
+      start = (x ^ start) & wmask;
+      memcpy(out + (size_t)(done + i) * wb, &start, wb);
+    }
+  }
+  return (int64_t)(ip - in);
 }
 
 int64_t obx_cs_ddzp_enc(const uint8_t *in, uint32_t count, uint32_t wb,

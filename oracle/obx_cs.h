@@ -45,6 +45,7 @@ enum {
   OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_PFOR = 3,
   OBX_CS_ENC_DELTA_ZIGZAG_PFOR = 5,
   OBX_CS_ENC_SIMD_FIXEDPFOR = 6,
+  OBX_CS_ENC_XOR_FIXED_PFOR = 8,
 };
 
 /* in-memory mirror of ObIntegerStreamMeta (the serialized form is
@@ -144,6 +145,10 @@ int64_t obx_cs_ddzp_dec(const uint8_t *in, size_t in_len, uint32_t count,
 int64_t obx_cs_fpfor_enc(const uint8_t *in, uint32_t count, uint32_t wb,
                          uint8_t *out, size_t cap);
 int64_t obx_cs_fpfor_dec(const uint8_t *in, size_t in_len, uint32_t count,
+                         uint32_t wb, uint8_t *out);
+int64_t obx_cs_xpfor_enc(const uint8_t *in, uint32_t count, uint32_t wb,
+                         uint8_t *out, size_t cap);
+int64_t obx_cs_xpfor_dec(const uint8_t *in, size_t in_len, uint32_t count,
                          uint32_t wb, uint8_t *out);
 
 #ifdef __cplusplus

@@ -465,3 +465,58 @@ def test_int_stream_all_types(t):
                                    out.ctypes.data_as(C.POINTER(C.c_int64)),
                                    None)
     assert m == n and (out == v).all()
+
+
+# ---- XOR_FIXED_PFOR (type 8: xor deltas, left-shift by the common free
+# high bits, bit-reverse, PFoR frame; ObXorFixedPforInner) ----
+
+_lib.obx_cs_xpfor_enc.restype = C.c_int64
+_lib.obx_cs_xpfor_enc.argtypes = _lib.obx_cs_dzr_enc.argtypes
+_lib.obx_cs_xpfor_dec.restype = C.c_int64
+_lib.obx_cs_xpfor_dec.argtypes = _lib.obx_cs_dzr_dec.argtypes
+
+
+@pytest.mark.parametrize("wb", [1, 2, 4, 8])
+def test_xpfor_roundtrip(wb):
+    rng = np.random.default_rng(400 + wb)
+    lim = 1 << min(8 * wb, 63)
+    for trial in range(30):
+        n = int(rng.integers(1, 500))
+        style = trial % 3
+        if style == 0:
+            v = rng.integers(0, lim, n)
+        elif style == 1:  # low-bit churn: xor-friendly
+            base = int(rng.integers(0, lim))
+            v = base ^ rng.integers(0, 16, n)
+        else:
+            v = np.full(n, int(rng.integers(0, lim)))
+        dt = {1: np.uint8, 2: np.uint16, 4: np.uint32, 8: np.uint64}[wb]
+        packed = np.asarray(v).astype(dt)
+        inb = packed.tobytes()
+        cap = len(inb) * 3 + 128
+        out = (C.c_uint8 * cap)()
+        n2 = _lib.obx_cs_xpfor_enc(
+            (C.c_uint8 * len(inb)).from_buffer_copy(inb), len(packed), wb,
+            out, cap)
+        assert n2 > 0
+        dec = (C.c_uint8 * len(inb))()
+        m = _lib.obx_cs_xpfor_dec(out, n2, len(packed), wb, dec)
+        assert m == n2
+        assert np.frombuffer(bytes(dec),
+                             dtype=dt).tolist() == packed.tolist()
+
+
+def test_int_stream_xpfor_type():
+    rows = 500
+    rng = np.random.default_rng(88)
+    v = (0x1234_5000 ^ rng.integers(0, 8, rows)).astype(np.int64)
+    cap = 64 + rows * 8 * 2
+    buf = (C.c_uint8 * cap)()
+    n = _lib.obx_cs_int_stream_enc2(
+        v.ctypes.data_as(C.POINTER(C.c_int64)), None, rows, 8, buf, cap)
+    assert n > 0
+    out = np.zeros(rows, dtype=np.int64)
+    m = _lib.obx_cs_int_stream_dec(buf, n, rows,
+                                   out.ctypes.data_as(C.POINTER(C.c_int64)),
+                                   None)
+    assert m == n and (out == v).all()
