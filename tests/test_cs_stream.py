@@ -550,3 +550,43 @@ def test_property_all_stream_types(data):
                                    None)
     assert m == n
     assert (out == v).all()
+
+
+def test_golden_stream_bytes():
+    """Byte-level pins: the encoder's output on fixed seeded inputs must
+    match the committed sha256 fixtures (the serialize formats are
+    replica-checksummed in the reference, so the bytes are the contract —
+    any refactor that changes them is a format break, not a cleanup)."""
+    import hashlib
+    import json
+    with open(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                           "golden", "cs_streams.json")) as f:
+        golden = json.load(f)["cases"]
+    CASES = {
+        "raw_w2":      (1, 11, 500, 1000, 40000, False),
+        "ddzr_sorted": (2, 12, 777, -5000, 9000, True),
+        "ddzp_wide":   (3, 13, 1000, -2**40, 2**41, True),
+        "dzr_runs":    (4, 14, 600, 0, 50, False),
+        "dzp_sorted":  (5, 15, 900, 10**6, 10**7, True),
+        "fpfor_w4":    (6, 16, 512, 0, 2**30, False),
+        "xpfor_float": (8, 17, 640, 2**52, 2**30, False),
+    }
+    assert set(CASES) == set(golden)
+    for name, (t, seed, rows, lo, span, srt) in CASES.items():
+        rng = np.random.default_rng(seed)
+        v = rng.integers(lo, lo + span, rows, dtype=np.int64)
+        if srt:
+            v = np.sort(v)
+        cap = 64 + rows * 24
+        buf = (C.c_uint8 * cap)()
+        n = _lib.obx_cs_int_stream_enc2(
+            v.ctypes.data_as(C.POINTER(C.c_int64)), None, rows, t, buf, cap)
+        assert n == golden[name]["bytes"], name
+        h = hashlib.sha256(bytes(buf[:n])).hexdigest()
+        assert h == golden[name]["sha256"], name
+        # and the pinned bytes still decode to the input
+        out = np.zeros(rows, dtype=np.int64)
+        assert _lib.obx_cs_int_stream_dec(
+            buf, n, rows, out.ctypes.data_as(C.POINTER(C.c_int64)),
+            None) == n
+        assert (out == v).all()
