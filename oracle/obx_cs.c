@@ -113,8 +113,21 @@ int64_t obx_cs_dzr_enc(const uint8_t *in, uint32_t count, uint32_t wb,
 int64_t obx_cs_int_stream_enc2(const int64_t *vals, const uint8_t *nulls,
                                uint32_t rows, uint8_t enc_type,
                                uint8_t *buf, size_t cap) {
+  return obx_cs_int_stream_enc3(vals, nulls, rows, enc_type, 0, 0, buf,
+                                cap);
+}
+
+int64_t obx_cs_int_stream_enc3(const int64_t *vals, const uint8_t *nulls,
+                               uint32_t rows, uint8_t enc_type,
+                               int use_null_replace, int64_t null_replaced,
+                               uint8_t *buf, size_t cap) {
   if (!vals || !rows || !buf) return -1;
-  /* build_signed_stream_meta: base = min, width covers range = max-min */
+  /* build_signed_stream_meta (ob_stream_encoding_struct.cpp:118-166):
+     base ONLY when min < 0 (range = max - min); for min >= 0 the width
+     covers max itself and no base is stored. When the column layer
+     chose a null-replace value it already sits adjacent to the range
+     (min-1 or max+1, ob_integer_column_encoder.cpp:190-220), so folding
+     it into the min/max scan reproduces new_int_min/new_int_max. */
   int64_t mn = 0, mx = 0;
   int any = 0;
   for (uint32_t r = 0; r < rows; r++) {
@@ -123,14 +136,30 @@ int64_t obx_cs_int_stream_enc2(const int64_t *vals, const uint8_t *nulls,
     if (!any || vals[r] > mx) mx = vals[r];
     any = 1;
   }
+  if (use_null_replace) {
+    if (!any || null_replaced < mn) mn = null_replaced;
+    if (!any || null_replaced > mx) mx = null_replaced;
+    any = 1;
+  }
   if (!any) mn = mx = 0;
-  uint64_t range = (uint64_t)mx - (uint64_t)mn;
   obx_cs_int_meta m;
   memset(&m, 0, sizeof(m));
   m.version = 1; /* V2 */
   m.type = enc_type;
-  m.attr = OBX_CS_USE_BASE;
-  m.base = (uint64_t)mn;
+  uint64_t range;
+  if (mn < 0) {
+    m.attr = OBX_CS_USE_BASE;
+    m.base = (uint64_t)mn;
+    range = (uint64_t)mx - (uint64_t)mn;
+  } else {
+    m.attr = 0;
+    m.base = 0;
+    range = (uint64_t)mx;
+  }
+  if (use_null_replace) {
+    m.attr |= OBX_CS_REPLACE_NULL;
+    m.null_replaced = (uint64_t)null_replaced;
+  }
   m.width_tag = range <= 0xFF ? 0 : range <= 0xFFFF ? 1
                 : range <= 0xFFFFFFFFull ? 2 : 3;
   m.pfor_packing_type = 0; /* CPU_ARCH_INDEPENDANT_SCALAR */
@@ -139,8 +168,10 @@ int64_t obx_cs_int_stream_enc2(const int64_t *vals, const uint8_t *nulls,
   size_t pos = (size_t)hn;
   uint32_t wb = WIDTH_BYTES[m.width_tag];
   if (pos + (size_t)rows * wb + 32 > cap) return -1;
-  /* datum->uint conversion: null -> replace value (== base here: "make
-     int small", ob_integer_stream_encoder.cpp:108-112), base FIRST */
+  /* datum->uint conversion (ob_integer_stream_encoder.cpp:105-113):
+     null -> replace value if set, else base (else 0); base subtracted
+     after */
+  uint64_t null_fill = use_null_replace ? (uint64_t)null_replaced : m.base;
   uint8_t *packed = buf + pos; /* RAW writes in place */
   uint8_t *tmp = NULL;
   if (enc_type != OBX_CS_ENC_RAW) {
@@ -149,7 +180,7 @@ int64_t obx_cs_int_stream_enc2(const int64_t *vals, const uint8_t *nulls,
     packed = tmp;
   }
   for (uint32_t r = 0; r < rows; r++) {
-    uint64_t ele = null_at(nulls, r) ? m.base : (uint64_t)vals[r];
+    uint64_t ele = null_at(nulls, r) ? null_fill : (uint64_t)vals[r];
     ele -= m.base;
     memcpy(packed + (size_t)r * wb, &ele, wb);
   }

@@ -12,8 +12,9 @@
  *     always serializes as 10 groups)
  *   ob_integer_stream_encoder.cpp:93-155 — datum→uint conversion (null →
  *     replace value, base subtraction FIRST) and RAW width-packed output
- *   ob_stream_encoding_struct.cpp:96+ — build_signed_stream_meta's
- *     base/width selection (range = max-min, smallest of 1/2/4/8 bytes)
+ *   ob_stream_encoding_struct.cpp:118-166 — build_signed_stream_meta's
+ *     base/width selection: base = min ONLY when min < 0 (width covers
+ *     range = max-min); otherwise no base, width covers max itself
  *
  * Scope of this slice: stream meta + RAW encoding type. The delta/zigzag/
  * PFoR codec types (ObIntegerStream::EncodingType 2-8, backed by the
@@ -80,6 +81,15 @@ int64_t obx_cs_int_stream_enc(const int64_t *vals, const uint8_t *nulls,
  * (OBX_CS_ENC_RAW or OBX_CS_ENC_DELTA_ZIGZAG_RLE) */
 int64_t obx_cs_int_stream_enc2(const int64_t *vals, const uint8_t *nulls,
                                uint32_t rows, uint8_t enc_type,
+                               uint8_t *buf, size_t cap);
+
+/* full column-layer form: when use_null_replace is set, null rows are
+ * stored as null_replaced (chosen by the caller adjacent to the data
+ * range per ob_integer_column_encoder.cpp:190-220, so decoders recover
+ * nulls by equality) and REPLACE_NULL + the vi64 value go in the meta */
+int64_t obx_cs_int_stream_enc3(const int64_t *vals, const uint8_t *nulls,
+                               uint32_t rows, uint8_t enc_type,
+                               int use_null_replace, int64_t null_replaced,
                                uint8_t *buf, size_t cap);
 
 /* Decode [meta][stream] back to int64 values (base re-applied). null

@@ -90,25 +90,31 @@ def test_stream_roundtrip(case):
     assert list(out) == case
 
 
-def test_stream_null_replace_uses_base():
-    """nulls occupy a slot with the base value (encoder :108-112: when no
-    explicit replace value, base 'makes int small')."""
+def test_stream_null_fill():
+    """encoder :105-113: with no explicit replace value, null slots store
+    the base when one is used (min < 0), else 0."""
+    blob, out = _roundtrip([-10, 20, 30, 40], nulls=[2])
+    assert list(out) == [-10, 20, -10, 40]  # null slot decodes to base=min
     blob, out = _roundtrip([10, 20, 30, 40], nulls=[2])
-    assert list(out) == [10, 20, 10, 40]  # null slot decodes to base=min
+    assert list(out) == [10, 20, 0, 40]  # no base (min >= 0): null -> 0
 
 
 def test_meta_bytes_pinned():
-    """Byte-level pin of the serialized meta for a known stream: version 1
-    (V2), attr USE_BASE, type RAW, width 1-byte-range tag, vi64 base,
-    pfor_packing_type 0."""
+    """Byte-level pins of the serialized meta: build_signed_stream_meta
+    (ob_stream_encoding_struct.cpp:118-166) stores a base ONLY for
+    negative minima; non-negative columns pack absolute values wide
+    enough for max."""
+    # min >= 0: no base. [version=1][attr=0][type=RAW=1][width_tag=1
+    # (max 1255 needs 2 bytes)][pfor=0] then 3 x 2-byte absolute values.
     blob, _ = _roundtrip([1000, 1001, 1255])
-    # [version=1][attr=1][type=1][width_tag=0][vi64(1000)=E8 07][pfor=0]
+    assert blob[:5] == bytes([1, 0, 1, 1, 0])
+    assert blob[5:11] == bytes([0xE8, 0x03, 0xE9, 0x03, 0xE7, 0x04])
+    assert len(blob) == 11
+    # min < 0: USE_BASE, base = -2 as vi64 (10 bytes), 1-byte deltas.
+    blob, _ = _roundtrip([-2, 1, 3])
     assert blob[:4] == bytes([1, 1, 1, 0])
-    assert blob[4:6] == b"\xe8\x07"
-    assert blob[6] == 0
-    # stream: 3 x 1-byte deltas off base 1000
-    assert blob[7:10] == bytes([0, 1, 255])
-    assert len(blob) == 10
+    assert len(blob) == 4 + 10 + 1 + 3  # meta + vi64(-2) + pfor + 3 deltas
+    assert blob[-3:] == bytes([0, 3, 5])
 
 
 # ---- string stream (ObStringStreamMeta + fixed byte stream; var-length
