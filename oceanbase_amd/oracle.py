@@ -21,6 +21,7 @@ _SO = os.path.join(_REPO, "oracle", "liboracle.so")
 def _build_if_needed():
     srcs = [os.path.join(_REPO, "oracle", f)
             for f in ("obx_codec.c", "obx_agg.c", "obx_gen.c", "obx_cs.c",
+                      "obx_cs_block.c", "obx_cs_block.h",
                       "obx_format.h", "obx_cs.h")]
     if os.path.exists(_SO) and all(
             os.path.getmtime(_SO) >= os.path.getmtime(s) for s in srcs):

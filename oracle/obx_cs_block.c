@@ -1,0 +1,498 @@
+/* CS column/block layer restatement — see obx_cs_block.h for the cited
+ * reference functions. Oracle/test infrastructure only. */
+#include "obx_cs_block.h"
+
+#include <stdlib.h>
+#include <string.h>
+
+static const uint32_t WB[4] = {1, 2, 4, 8};
+
+static inline int in_null(const uint8_t *nulls, uint32_t r) {
+  return nulls && ((nulls[r >> 3] >> (r & 7)) & 1);
+}
+
+/* get_byte_packed_int_size: smallest of 1/2/4/8 covering v */
+static inline uint32_t byte_packed(uint64_t v) {
+  return v <= 0xFF ? 1 : v <= 0xFFFF ? 2 : v <= 0xFFFFFFFFull ? 4 : 8;
+}
+
+static inline uint8_t width_tag(uint32_t wb) {
+  return wb == 1 ? 0 : wb == 2 ? 1 : wb == 4 ? 2 : 3;
+}
+
+/* ObCSEncodingUtil::get_bit_size: bits needed to store v */
+static inline uint32_t bit_size(uint64_t v) {
+  uint32_t b = 1;
+  while (v >>= 1) b++;
+  return b;
+}
+
+static int64_t codec_enc(uint8_t t, const uint8_t *in, uint32_t n,
+                         uint32_t wb, uint8_t *out, size_t cap) {
+  switch (t) {
+    case OBX_CS_ENC_DELTA_ZIGZAG_RLE: return obx_cs_dzr_enc(in, n, wb, out, cap);
+    case OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE: return obx_cs_ddzr_enc(in, n, wb, out, cap);
+    case OBX_CS_ENC_DELTA_ZIGZAG_PFOR: return obx_cs_dzp_enc(in, n, wb, out, cap);
+    case OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_PFOR: return obx_cs_ddzp_enc(in, n, wb, out, cap);
+    case OBX_CS_ENC_SIMD_FIXEDPFOR: return obx_cs_fpfor_enc(in, n, wb, out, cap);
+    case OBX_CS_ENC_XOR_FIXED_PFOR: return obx_cs_xpfor_enc(in, n, wb, out, cap);
+    default: return -1;
+  }
+}
+
+static int64_t codec_dec(uint8_t t, const uint8_t *in, size_t len,
+                         uint32_t n, uint32_t wb, uint8_t *out) {
+  switch (t) {
+    case OBX_CS_ENC_DELTA_ZIGZAG_RLE: return obx_cs_dzr_dec(in, len, n, wb, out);
+    case OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_RLE: return obx_cs_ddzr_dec(in, len, n, wb, out);
+    case OBX_CS_ENC_DELTA_ZIGZAG_PFOR: return obx_cs_dzp_dec(in, len, n, wb, out);
+    case OBX_CS_ENC_DOUBLE_DELTA_ZIGZAG_PFOR: return obx_cs_ddzp_dec(in, len, n, wb, out);
+    case OBX_CS_ENC_SIMD_FIXEDPFOR: return obx_cs_fpfor_dec(in, len, n, wb, out);
+    case OBX_CS_ENC_XOR_FIXED_PFOR: return obx_cs_xpfor_dec(in, len, n, wb, out);
+    default: return -1;
+  }
+}
+
+/* offset-array integer stream (build_offset_array_stream_meta,
+ * ob_stream_encoding_struct.cpp:182-202): no base, width covers the
+ * last (largest) end offset, values stored as-is */
+static int64_t offset_stream_enc(const uint64_t *ends, uint32_t n,
+                                 uint8_t enc_type, uint8_t *buf,
+                                 size_t cap) {
+  if (!n) return -1;
+  uint32_t wb = byte_packed(ends[n - 1]);
+  obx_cs_int_meta m;
+  memset(&m, 0, sizeof(m));
+  m.version = 1;
+  m.type = enc_type ? enc_type : OBX_CS_ENC_RAW;
+  m.width_tag = width_tag(wb);
+  int hn = obx_cs_int_meta_enc(&m, buf, cap);
+  if (hn < 0) return -1;
+  size_t pos = (size_t)hn;
+  uint8_t *packed = (uint8_t *)malloc((size_t)n * wb);
+  if (!packed) return -1;
+  for (uint32_t i = 0; i < n; i++)
+    memcpy(packed + (size_t)i * wb, &ends[i], wb);
+  int64_t dn;
+  if (m.type == OBX_CS_ENC_RAW) {
+    dn = (int64_t)n * wb;
+    if (pos + (size_t)dn > cap) { free(packed); return -1; }
+    memcpy(buf + pos, packed, (size_t)dn);
+  } else {
+    dn = codec_enc(m.type, packed, n, wb, buf + pos, cap - pos);
+  }
+  free(packed);
+  if (dn < 0) return -1;
+  return (int64_t)pos + dn;
+}
+
+static int64_t offset_stream_dec(const uint8_t *buf, size_t len, uint32_t n,
+                                 uint64_t *ends) {
+  obx_cs_int_meta m;
+  int hn = obx_cs_int_meta_dec(buf, len, &m);
+  if (hn < 0) return -1;
+  uint32_t wb = WB[m.width_tag];
+  size_t pos = (size_t)hn;
+  uint8_t *packed = (uint8_t *)malloc((size_t)n * wb);
+  if (!packed) return -1;
+  int64_t dn;
+  if (m.type == OBX_CS_ENC_RAW) {
+    dn = (int64_t)n * wb;
+    if (pos + (size_t)dn > len) { free(packed); return -1; }
+    memcpy(packed, buf + pos, (size_t)dn);
+  } else {
+    dn = codec_dec(m.type, buf + pos, len - pos, n, wb, packed);
+  }
+  if (dn < 0) { free(packed); return -1; }
+  for (uint32_t i = 0; i < n; i++) {
+    uint64_t v = 0;
+    memcpy(&v, packed + (size_t)i * wb, wb);
+    ends[i] = v;
+  }
+  free(packed);
+  return (int64_t)pos + dn;
+}
+
+int64_t obx_cs_block_enc(uint32_t rows, uint32_t ncols,
+                         const obx_cs_col_in *cols, uint8_t *buf,
+                         size_t cap) {
+  if (!rows || !ncols || ncols > OBX_CS_MAX_COLS || !cols || !buf)
+    return -1;
+  const size_t hdr_sz = sizeof(obx_cs_block_header);
+  const size_t heads = hdr_sz + sizeof(obx_cs_all_col_header) +
+                       (size_t)ncols * sizeof(obx_cs_col_header);
+  const uint32_t bitmap_size = (rows + 7) / 8;
+  if (heads + 64 > cap) return -1;
+
+  obx_cs_col_header chdr[OBX_CS_MAX_COLS];
+  memset(chdr, 0, sizeof(chdr));
+  uint64_t soff[OBX_CS_MAX_STREAMS];
+  uint32_t n_streams = 0;
+
+  /* pooled string bytes accumulate here, appended after the columns
+   * (store_all_string_data_, ob_micro_block_cs_encoder.cpp:1247-1303;
+   * NONE_COMPRESSOR path) */
+  size_t str_pool_cap = 0;
+  for (uint32_t c = 0; c < ncols; c++)
+    if (cols[c].is_string)
+      for (uint32_t r = 0; r < rows; r++)
+        str_pool_cap += cols[c].lens ? cols[c].lens[r] : 0;
+  str_pool_cap += (size_t)rows * 8 * ncols + 64; /* fixed-null padding */
+  uint8_t *pool = (uint8_t *)malloc(str_pool_cap);
+  if (!pool) return -1;
+  size_t pool_len = 0;
+
+  size_t pos = heads;
+  int64_t rc = -1;
+  for (uint32_t c = 0; c < ncols; c++) {
+    const obx_cs_col_in *ci = &cols[c];
+    uint32_t null_cnt = 0;
+    for (uint32_t r = 0; r < rows; r++)
+      if (in_null(ci->nulls, r)) null_cnt++;
+    if (n_streams + 2 > OBX_CS_MAX_STREAMS) goto fail;
+
+    if (!ci->is_string) {
+      /* ---- INTEGER column (ob_integer_column_encoder.cpp) ---- */
+      if (!ci->ivals) goto fail;
+      int64_t mn = 0, mx = 0;
+      int any = 0;
+      for (uint32_t r = 0; r < rows; r++) {
+        if (in_null(ci->nulls, r)) continue;
+        if (!any || ci->ivals[r] < mn) mn = ci->ivals[r];
+        if (!any || ci->ivals[r] > mx) mx = ci->ivals[r];
+        any = 1;
+      }
+      if (!any) mn = mx = 0;
+      /* null handling (:190-220, int64 store type): replace with a
+       * value adjacent to the range; bitmap only when the range spans
+       * the whole of int64 */
+      int use_replace = 0, use_bitmap = 0;
+      int64_t replace = 0;
+      if (null_cnt > 0) {
+        if (mn == 0) {
+          if (mx != INT64_MAX) { use_replace = 1; replace = mx + 1; }
+          else { use_replace = 1; replace = -1; }
+        } else if (mn == INT64_MIN) {
+          if (mx != INT64_MAX) { use_replace = 1; replace = mx + 1; }
+          else { use_bitmap = 1; }
+        } else {
+          use_replace = 1;
+          replace = mn - 1;
+        }
+      }
+      chdr[c].type = OBX_CS_COL_INTEGER;
+      chdr[c].obj_type = OBX_OBJ_INT;
+      if (use_bitmap) {
+        chdr[c].attrs |= OBX_CS_CA_HAS_NULL_BITMAP;
+        if (pos + bitmap_size > cap) goto fail;
+        memset(buf + pos, 0, bitmap_size);
+        for (uint32_t r = 0; r < rows; r++)
+          if (in_null(ci->nulls, r))
+            buf[pos + r / 8] |= (uint8_t)(1 << (7 - r % 8)); /* MSB-first */
+        pos += bitmap_size;
+      }
+      int64_t n = obx_cs_int_stream_enc3(
+          ci->ivals, null_cnt ? ci->nulls : NULL, rows,
+          ci->enc_type ? ci->enc_type : OBX_CS_ENC_RAW, use_replace,
+          replace, buf + pos, cap - pos);
+      if (n < 0) goto fail;
+      pos += (size_t)n;
+      soff[n_streams++] = pos;
+    } else {
+      /* ---- STRING column (ob_string_column_encoder.cpp:53-135) ---- */
+      if (!ci->lens || (!ci->bytes && str_pool_cap)) goto fail;
+      int64_t fix_size = -1;
+      int has_zero = 0, any = 0;
+      uint64_t var_total = 0;
+      for (uint32_t r = 0; r < rows; r++) {
+        if (in_null(ci->nulls, r)) continue;
+        uint32_t l = ci->lens[r];
+        var_total += l;
+        if (l == 0) has_zero = 1;
+        if (!any) fix_size = l;
+        else if (fix_size != (int64_t)l) fix_size = -2;
+        any = 1;
+      }
+      if (fix_size < 0) fix_size = -1;
+      int use_fixed = 0, use_zero_null = 0, use_bitmap = 0;
+      if (null_cnt > 0) {
+        if (has_zero) {
+          use_bitmap = 1;
+          use_fixed = fix_size >= 0;
+        } else if (fix_size >= 0) {
+          /* cost rule (:73-90, non-raw estimate) */
+          uint64_t offs_est =
+              ((uint64_t)bit_size((uint64_t)fix_size) + 1) * rows / 8;
+          if ((uint64_t)fix_size * null_cnt + bitmap_size < offs_est) {
+            use_fixed = 1;
+            use_bitmap = 1;
+          } else {
+            use_zero_null = 1;
+          }
+        } else {
+          use_zero_null = 1;
+        }
+      } else {
+        use_fixed = fix_size >= 0;
+      }
+      chdr[c].type = OBX_CS_COL_STRING;
+      chdr[c].obj_type = OBX_OBJ_VARCHAR;
+      if (use_fixed) chdr[c].attrs |= OBX_CS_CA_IS_FIXED;
+      if (use_bitmap) {
+        chdr[c].attrs |= OBX_CS_CA_HAS_NULL_BITMAP;
+        if (pos + bitmap_size > cap) goto fail;
+        memset(buf + pos, 0, bitmap_size);
+        for (uint32_t r = 0; r < rows; r++)
+          if (in_null(ci->nulls, r))
+            buf[pos + r / 8] |= (uint8_t)(1 << (7 - r % 8));
+        pos += bitmap_size;
+      }
+      obx_cs_str_meta sm;
+      memset(&sm, 0, sizeof(sm));
+      if (use_zero_null) sm.attr |= OBX_CS_STR_ZERO_LEN_NULL;
+      if (use_fixed) {
+        sm.attr |= OBX_CS_STR_FIXED_LEN;
+        sm.fixed_str_len = (uint32_t)(fix_size < 0 ? 0 : fix_size);
+        sm.uncompressed_len = sm.fixed_str_len * rows;
+      } else {
+        sm.uncompressed_len = (uint32_t)var_total;
+      }
+      int hn = obx_cs_str_meta_enc(&sm, buf + pos, cap - pos);
+      if (hn < 0) goto fail;
+      pos += (size_t)hn;
+      soff[n_streams++] = pos;
+      /* bytes -> pool (do_convert_datum_to_stream_: fixed nulls are
+       * zero-filled placeholders; var nulls contribute nothing) */
+      const uint8_t *src = ci->bytes;
+      uint64_t ends_buf_stack[1];
+      (void)ends_buf_stack;
+      if (use_fixed) {
+        uint32_t fl = sm.fixed_str_len;
+        if (pool_len + (size_t)fl * rows > str_pool_cap) goto fail;
+        for (uint32_t r = 0; r < rows; r++) {
+          if (in_null(ci->nulls, r)) {
+            memset(pool + pool_len, 0, fl);
+          } else {
+            memcpy(pool + pool_len, src, fl);
+            src += fl;
+          }
+          pool_len += fl;
+        }
+      } else {
+        uint64_t *ends = (uint64_t *)malloc((size_t)rows * 8);
+        if (!ends) goto fail;
+        uint64_t acc = 0;
+        for (uint32_t r = 0; r < rows; r++) {
+          if (!in_null(ci->nulls, r)) {
+            uint32_t l = ci->lens[r];
+            if (pool_len + l > str_pool_cap) { free(ends); goto fail; }
+            memcpy(pool + pool_len, src, l);
+            src += l;
+            pool_len += l;
+            acc += l;
+          }
+          ends[r] = acc;
+        }
+        int64_t n = offset_stream_enc(ends, rows, ci->enc_type, buf + pos,
+                                      cap - pos);
+        free(ends);
+        if (n < 0) goto fail;
+        pos += (size_t)n;
+        soff[n_streams++] = pos;
+      }
+    }
+  }
+
+  /* pooled string data */
+  if (pos + pool_len > cap) goto fail;
+  memcpy(buf + pos, pool, pool_len);
+  pos += pool_len;
+
+  /* block-tail stream-offset stream (store_stream_offsets_) */
+  uint32_t sol = 0;
+  if (n_streams) {
+    int64_t n = offset_stream_enc(soff, n_streams, OBX_CS_ENC_RAW,
+                                  buf + pos, cap - pos);
+    if (n < 0) goto fail;
+    sol = (uint32_t)n;
+    pos += (size_t)n;
+  }
+
+  /* fill headers */
+  {
+    obx_cs_block_header bh;
+    memset(&bh, 0, sizeof(bh));
+    bh.magic = OBX_CS_BLOCK_MAGIC;
+    bh.version = 1;
+    bh.header_size = (uint16_t)hdr_sz;
+    bh.row_count = rows;
+    bh.column_count = (uint16_t)ncols;
+    memcpy(buf, &bh, sizeof(bh));
+    obx_cs_all_col_header ach;
+    memset(&ach, 0, sizeof(ach));
+    ach.all_string_data_length = (uint32_t)pool_len;
+    ach.stream_offsets_length = sol;
+    ach.stream_count = (uint16_t)n_streams;
+    memcpy(buf + hdr_sz, &ach, sizeof(ach));
+    memcpy(buf + hdr_sz + sizeof(ach), chdr,
+           (size_t)ncols * sizeof(obx_cs_col_header));
+  }
+  rc = (int64_t)pos;
+fail:
+  free(pool);
+  return rc;
+}
+
+int obx_cs_block_dec(const uint8_t *buf, size_t len,
+                     obx_cs_block_view *v) {
+  memset(v, 0, sizeof(*v));
+  obx_cs_block_header bh;
+  if (len < sizeof(bh)) return -1;
+  memcpy(&bh, buf, sizeof(bh));
+  if (bh.magic != OBX_CS_BLOCK_MAGIC || bh.version != 1 ||
+      bh.column_count == 0 || bh.column_count > OBX_CS_MAX_COLS)
+    return -1;
+  size_t hp = bh.header_size;
+  if (hp + sizeof(obx_cs_all_col_header) > len) return -1;
+  memcpy(&v->ach, buf + hp, sizeof(v->ach));
+  hp += sizeof(v->ach);
+  if (hp + (size_t)bh.column_count * sizeof(obx_cs_col_header) > len)
+    return -1;
+  v->buf = buf;
+  v->len = len;
+  v->rows = bh.row_count;
+  v->ncols = bh.column_count;
+  for (uint32_t c = 0; c < v->ncols; c++)
+    memcpy(&v->col[c].h, buf + hp + c * sizeof(obx_cs_col_header),
+           sizeof(obx_cs_col_header));
+  hp += (size_t)bh.column_count * sizeof(obx_cs_col_header);
+
+  /* tail: [... pooled strings][stream-offset stream] */
+  if (v->ach.stream_count > OBX_CS_MAX_STREAMS) return -1;
+  if ((size_t)v->ach.stream_offsets_length +
+          v->ach.all_string_data_length > len)
+    return -1;
+  const size_t so_start = len - v->ach.stream_offsets_length;
+  v->all_string = buf + so_start - v->ach.all_string_data_length;
+  v->stream_count = v->ach.stream_count;
+  if (v->stream_count) {
+    uint64_t ends[OBX_CS_MAX_STREAMS];
+    if (offset_stream_dec(buf + so_start, v->ach.stream_offsets_length,
+                          v->stream_count, ends) < 0)
+      return -1;
+    for (uint32_t i = 0; i < v->stream_count; i++) {
+      if (ends[i] > so_start - v->ach.all_string_data_length) return -1;
+      v->stream_offsets[i] = (uint32_t)ends[i];
+    }
+  }
+
+  /* walk columns to bind bitmaps and stream slices */
+  const uint32_t bitmap_size = (v->rows + 7) / 8;
+  size_t pos = hp;
+  uint32_t si = 0, str_off = 0;
+  for (uint32_t c = 0; c < v->ncols; c++) {
+    obx_cs_col_view *cv = &v->col[c];
+    if (cv->h.attrs & OBX_CS_CA_HAS_NULL_BITMAP) {
+      if (pos + bitmap_size > len) return -1;
+      cv->null_bitmap = buf + pos;
+      pos += bitmap_size;
+    }
+    if (cv->h.type == OBX_CS_COL_INTEGER) {
+      if (si >= v->stream_count || v->stream_offsets[si] < pos) return -1;
+      cv->int_stream = buf + pos;
+      cv->int_stream_len = v->stream_offsets[si] - pos;
+      pos = v->stream_offsets[si++];
+    } else if (cv->h.type == OBX_CS_COL_STRING) {
+      if (si >= v->stream_count || v->stream_offsets[si] < pos) return -1;
+      int hn = obx_cs_str_meta_dec(buf + pos, v->stream_offsets[si] - pos,
+                                   &cv->sm);
+      if (hn < 0 || pos + (size_t)hn != v->stream_offsets[si]) return -1;
+      pos = v->stream_offsets[si++];
+      cv->str_data_off = str_off;
+      str_off += cv->sm.uncompressed_len;
+      if (str_off > v->ach.all_string_data_length) return -1;
+      if (!(cv->sm.attr & OBX_CS_STR_FIXED_LEN)) {
+        if (si >= v->stream_count || v->stream_offsets[si] < pos)
+          return -1;
+        cv->off_stream = buf + pos;
+        cv->off_stream_len = v->stream_offsets[si] - pos;
+        pos = v->stream_offsets[si++];
+      }
+    } else {
+      return -1;
+    }
+  }
+  return 0;
+}
+
+static inline int blk_null(const uint8_t *bm, uint32_t r) {
+  return bm && ((bm[r / 8] >> (7 - r % 8)) & 1); /* MSB-first in block */
+}
+
+int obx_cs_block_get_int(const obx_cs_block_view *v, uint32_t c,
+                         int64_t *out, uint8_t *nulls_out) {
+  if (c >= v->ncols || v->col[c].h.type != OBX_CS_COL_INTEGER) return -1;
+  const obx_cs_col_view *cv = &v->col[c];
+  obx_cs_int_meta m;
+  if (obx_cs_int_stream_dec(cv->int_stream, cv->int_stream_len, v->rows,
+                            out, &m) < 0)
+    return -1;
+  if (nulls_out) memset(nulls_out, 0, (v->rows + 7) / 8);
+  if (m.attr & OBX_CS_REPLACE_NULL) {
+    for (uint32_t r = 0; r < v->rows; r++)
+      if (out[r] == (int64_t)m.null_replaced) {
+        out[r] = 0;
+        if (nulls_out) nulls_out[r >> 3] |= (uint8_t)(1 << (r & 7));
+      }
+  } else if (cv->null_bitmap) {
+    for (uint32_t r = 0; r < v->rows; r++)
+      if (blk_null(cv->null_bitmap, r)) {
+        out[r] = 0;
+        if (nulls_out) nulls_out[r >> 3] |= (uint8_t)(1 << (r & 7));
+      }
+  }
+  return 0;
+}
+
+int64_t obx_cs_block_get_str(const obx_cs_block_view *v, uint32_t c,
+                             uint8_t *bytes_out, size_t bytes_cap,
+                             uint32_t *lens_out, uint8_t *nulls_out) {
+  if (c >= v->ncols || v->col[c].h.type != OBX_CS_COL_STRING) return -1;
+  const obx_cs_col_view *cv = &v->col[c];
+  const uint8_t *src = v->all_string + cv->str_data_off;
+  uint32_t total = cv->sm.uncompressed_len;
+  if ((size_t)total > bytes_cap) return -1;
+  if (nulls_out) memset(nulls_out, 0, (v->rows + 7) / 8);
+  if (cv->sm.attr & OBX_CS_STR_FIXED_LEN) {
+    memcpy(bytes_out, src, total);
+    for (uint32_t r = 0; r < v->rows; r++) {
+      lens_out[r] = cv->sm.fixed_str_len;
+      if (nulls_out && blk_null(cv->null_bitmap, r))
+        nulls_out[r >> 3] |= (uint8_t)(1 << (r & 7));
+    }
+  } else {
+    uint64_t *ends = (uint64_t *)malloc((size_t)v->rows * 8);
+    if (!ends) return -1;
+    if (offset_stream_dec(cv->off_stream, cv->off_stream_len, v->rows,
+                          ends) < 0) {
+      free(ends);
+      return -1;
+    }
+    memcpy(bytes_out, src, total);
+    uint64_t prev = 0;
+    for (uint32_t r = 0; r < v->rows; r++) {
+      if (ends[r] < prev || ends[r] > total) { free(ends); return -1; }
+      lens_out[r] = (uint32_t)(ends[r] - prev);
+      prev = ends[r];
+      if (nulls_out) {
+        int isnull = cv->null_bitmap
+                         ? blk_null(cv->null_bitmap, r)
+                         : ((cv->sm.attr & OBX_CS_STR_ZERO_LEN_NULL) &&
+                            lens_out[r] == 0);
+        if (isnull) nulls_out[r >> 3] |= (uint8_t)(1 << (r & 7));
+      }
+    }
+    free(ends);
+  }
+  return (int64_t)total;
+}

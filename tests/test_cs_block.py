@@ -1,0 +1,396 @@
+"""CS column/block layer (continued): block assembly
+per ObMicroBlockCSEncoder::build_block, integer columns with the
+null-replace-adjacent-to-range rule, string columns with the
+fixed-vs-var cost rule and the pooled all-string region, MSB-first
+null bitmaps, and the block-tail stream-offset stream. Restated in
+oracle/obx_cs_block.c (citations there)."""
+import ctypes as C
+import hashlib
+import os
+import sys
+
+import numpy as np
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from oceanbase_amd import oracle  # noqa: E402  (builds liboracle.so)
+
+_lib = C.CDLL(os.path.join(os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))), "oracle", "liboracle.so"))
+
+
+class ColIn(C.Structure):
+    _fields_ = [("is_string", C.c_uint8), ("enc_type", C.c_uint8),
+                ("ivals", C.POINTER(C.c_int64)),
+                ("bytes", C.POINTER(C.c_uint8)),
+                ("lens", C.POINTER(C.c_uint32)),
+                ("nulls", C.POINTER(C.c_uint8))]
+
+
+class ColView(C.Structure):
+    _fields_ = [("version", C.c_uint8), ("type", C.c_uint8),
+                ("attrs", C.c_uint8), ("obj_type", C.c_uint8),
+                ("null_bitmap", C.POINTER(C.c_uint8)),
+                ("int_stream", C.POINTER(C.c_uint8)),
+                ("int_stream_len", C.c_size_t),
+                ("sm_version", C.c_uint8), ("sm_attr", C.c_uint8),
+                ("sm_uncompressed_len", C.c_uint32),
+                ("sm_fixed_str_len", C.c_uint32),
+                ("off_stream", C.POINTER(C.c_uint8)),
+                ("off_stream_len", C.c_size_t),
+                ("str_data_off", C.c_uint32)]
+
+
+class AllColHeader(C.Structure):
+    _pack_ = 1
+    _fields_ = [("version", C.c_uint8), ("attrs", C.c_uint8),
+                ("all_string_data_length", C.c_uint32),
+                ("stream_offsets_length", C.c_uint32),
+                ("stream_count", C.c_uint16)]
+
+
+class BlockView(C.Structure):
+    _fields_ = [("buf", C.POINTER(C.c_uint8)), ("len", C.c_size_t),
+                ("rows", C.c_uint32), ("ncols", C.c_uint32),
+                ("ach", AllColHeader),
+                ("all_string", C.POINTER(C.c_uint8)),
+                ("stream_offsets", C.c_uint32 * 96),
+                ("stream_count", C.c_uint32),
+                ("col", ColView * 48)]
+
+
+_lib.obx_cs_block_enc.restype = C.c_int64
+_lib.obx_cs_block_enc.argtypes = [C.c_uint32, C.c_uint32,
+                                  C.POINTER(ColIn), C.POINTER(C.c_uint8),
+                                  C.c_size_t]
+_lib.obx_cs_block_dec.restype = C.c_int
+_lib.obx_cs_block_dec.argtypes = [C.POINTER(C.c_uint8), C.c_size_t,
+                                  C.POINTER(BlockView)]
+_lib.obx_cs_block_get_int.restype = C.c_int
+_lib.obx_cs_block_get_int.argtypes = [C.POINTER(BlockView), C.c_uint32,
+                                      C.POINTER(C.c_int64),
+                                      C.POINTER(C.c_uint8)]
+_lib.obx_cs_block_get_str.restype = C.c_int64
+_lib.obx_cs_block_get_str.argtypes = [C.POINTER(BlockView), C.c_uint32,
+                                      C.POINTER(C.c_uint8), C.c_size_t,
+                                      C.POINTER(C.c_uint32),
+                                      C.POINTER(C.c_uint8)]
+
+CA_IS_FIXED = 0x01
+CA_HAS_NULL_BITMAP = 0x02
+STR_ZERO_LEN_NULL = 0x01
+STR_FIXED_LEN = 0x02
+
+
+def _nulls_bitmap(rows, null_rows):
+    if null_rows is None:
+        return None
+    bm = np.zeros((rows + 7) // 8, dtype=np.uint8)
+    for r in null_rows:
+        bm[r // 8] |= 1 << (r % 8)
+    return bm
+
+
+def _int_col(vals, null_rows=None, enc=0):
+    rows = len(vals)
+    col = ColIn()
+    col.is_string = 0
+    col.enc_type = enc
+    v = np.asarray(vals, dtype=np.int64)
+    col.ivals = v.ctypes.data_as(C.POINTER(C.c_int64))
+    bm = _nulls_bitmap(rows, null_rows)
+    col.nulls = (bm.ctypes.data_as(C.POINTER(C.c_uint8))
+                 if bm is not None else None)
+    col._keep = (v, bm)
+    return col
+
+
+def _str_col(strings, null_rows=None, enc=0):
+    """strings: list of bytes for non-null rows in row order (null rows
+    must be represented by ANY placeholder in the list; its bytes are
+    skipped)."""
+    rows = len(strings)
+    nulls = set(null_rows or [])
+    data = b"".join(s for r, s in enumerate(strings) if r not in nulls)
+    lens = np.array([len(s) for s in strings], dtype=np.uint32)
+    col = ColIn()
+    col.is_string = 1
+    col.enc_type = enc
+    barr = np.frombuffer(data, dtype=np.uint8).copy() if data else \
+        np.zeros(1, dtype=np.uint8)
+    col.bytes = barr.ctypes.data_as(C.POINTER(C.c_uint8))
+    col.lens = lens.ctypes.data_as(C.POINTER(C.c_uint32))
+    bm = _nulls_bitmap(rows, null_rows)
+    col.nulls = (bm.ctypes.data_as(C.POINTER(C.c_uint8))
+                 if bm is not None else None)
+    col._keep = (barr, lens, bm)
+    return col
+
+
+def _enc(rows, cols):
+    arr = (ColIn * len(cols))(*cols)
+    cap = 1 << 22
+    buf = (C.c_uint8 * cap)()
+    n = _lib.obx_cs_block_enc(rows, len(cols), arr, buf, cap)
+    assert n > 0
+    return bytes(buf[:n])
+
+
+def _dec(blob):
+    buf = (C.c_uint8 * len(blob))(*blob)
+    v = BlockView()
+    assert _lib.obx_cs_block_dec(buf, len(blob), C.byref(v)) == 0
+    v._keep = buf
+    return v
+
+
+def _get_int(v, c):
+    out = np.zeros(v.rows, dtype=np.int64)
+    nb = np.zeros((v.rows + 7) // 8, dtype=np.uint8)
+    assert _lib.obx_cs_block_get_int(
+        C.byref(v), c, out.ctypes.data_as(C.POINTER(C.c_int64)),
+        nb.ctypes.data_as(C.POINTER(C.c_uint8))) == 0
+    nulls = {r for r in range(v.rows) if (nb[r // 8] >> (r % 8)) & 1}
+    return out, nulls
+
+
+def _get_str(v, c):
+    cap = 1 << 22
+    bout = (C.c_uint8 * cap)()
+    lens = np.zeros(v.rows, dtype=np.uint32)
+    nb = np.zeros((v.rows + 7) // 8, dtype=np.uint8)
+    total = _lib.obx_cs_block_get_str(
+        C.byref(v), c, bout, cap,
+        lens.ctypes.data_as(C.POINTER(C.c_uint32)),
+        nb.ctypes.data_as(C.POINTER(C.c_uint8)))
+    assert total >= 0
+    data = bytes(bout[:total])
+    rows, pos = [], 0
+    for r in range(v.rows):
+        rows.append(data[pos:pos + int(lens[r])])
+        pos += int(lens[r])
+    nulls = {r for r in range(v.rows) if (nb[r // 8] >> (r % 8)) & 1}
+    return rows, nulls
+
+
+def test_int_roundtrip_no_nulls():
+    vals = [5, -3, 1000000, 0, 42]
+    v = _dec(_enc(5, [_int_col(vals)]))
+    out, nulls = _get_int(v, 0)
+    assert list(out) == vals and nulls == set()
+    assert v.col[0].type == 0 and v.col[0].obj_type == 5  # ObIntType
+    assert v.col[0].attrs == 0
+
+
+def test_int_null_replace_min_minus_one():
+    """min > 0: replace value = min-1 (ob_integer_column_encoder.cpp:
+    214-218), recovered by equality — no bitmap stored."""
+    vals = [10, 20, 0, 30]  # row 2 is null (placeholder value ignored)
+    v = _dec(_enc(4, [_int_col(vals, null_rows=[2])]))
+    assert not (v.col[0].attrs & CA_HAS_NULL_BITMAP)
+    out, nulls = _get_int(v, 0)
+    assert nulls == {2}
+    assert [out[r] for r in (0, 1, 3)] == [10, 20, 30]
+
+
+def test_int_null_replace_max_plus_one():
+    """min == 0: the largest not-existed value (max+1) is preferred
+    (:191-198)."""
+    vals = [0, 7, 0, 3]
+    v = _dec(_enc(4, [_int_col(vals, null_rows=[2])]))
+    assert not (v.col[0].attrs & CA_HAS_NULL_BITMAP)
+    out, nulls = _get_int(v, 0)
+    assert nulls == {2}
+    assert [out[r] for r in (0, 1, 3)] == [0, 7, 3]
+
+
+def test_int_null_bitmap_full_range():
+    """range spanning all of int64 leaves no adjacent value: bitmap
+    fallback (:203-212), MSB-first bit order (ob_icolumn_cs_encoder.cpp:
+    109)."""
+    vals = [-2**63, 2**63 - 1, 0, 5]
+    v = _dec(_enc(4, [_int_col(vals, null_rows=[2])]))
+    assert v.col[0].attrs & CA_HAS_NULL_BITMAP
+    # MSB-first: row 2 sets bit (7-2)=5 of byte 0
+    assert v.col[0].null_bitmap[0] == (1 << 5)
+    out, nulls = _get_int(v, 0)
+    assert nulls == {2}
+    assert [out[r] for r in (0, 1, 3)] == [-2**63, 2**63 - 1, 5]
+
+
+def test_fixed_string_roundtrip():
+    strs = [b"AAAA", b"BBBB", b"CCCC"]
+    v = _dec(_enc(3, [_str_col(strs)]))
+    assert v.col[0].attrs & CA_IS_FIXED
+    assert v.col[0].sm_attr & STR_FIXED_LEN
+    assert v.col[0].sm_fixed_str_len == 4
+    rows, nulls = _get_str(v, 0)
+    assert rows == strs and nulls == set()
+
+
+def test_fixed_string_nulls_cost_rule_fixed_branch():
+    """few nulls on a short fixed string: padding + bitmap beats an
+    offset stream (ob_string_column_encoder.cpp:73-90) -> IS_FIXED +
+    bitmap, null cells zero-filled."""
+    rows = 1000
+    strs = [b"%04d" % (i % 100) for i in range(rows)]
+    v = _dec(_enc(rows, [_str_col(strs, null_rows=[7, 8])]))
+    assert v.col[0].attrs & CA_IS_FIXED
+    assert v.col[0].attrs & CA_HAS_NULL_BITMAP
+    out, nulls = _get_str(v, 0)
+    assert nulls == {7, 8}
+    assert out[7] == b"\x00" * 4  # zero-filled placeholder
+    assert out[9] == strs[9]
+
+
+def test_fixed_string_nulls_cost_rule_var_branch():
+    """many nulls on a long fixed string: the offset stream wins ->
+    var layout with zero-len-as-null (:90-95)."""
+    rows = 64
+    strs = [b"x" * 100 for _ in range(rows)]
+    null_rows = list(range(0, rows, 2))
+    v = _dec(_enc(rows, [_str_col(strs, null_rows=null_rows)]))
+    assert not (v.col[0].attrs & CA_IS_FIXED)
+    assert v.col[0].sm_attr & STR_ZERO_LEN_NULL
+    out, nulls = _get_str(v, 0)
+    assert nulls == set(null_rows)
+    assert out[1] == b"x" * 100 and out[0] == b""
+
+
+def test_var_string_zero_len_null():
+    strs = [b"hello", b"", b"world!!", b"xy"]  # row 1 null
+    v = _dec(_enc(4, [_str_col(strs, null_rows=[1])]))
+    assert v.col[0].sm_attr & STR_ZERO_LEN_NULL
+    assert not (v.col[0].attrs & CA_HAS_NULL_BITMAP)
+    out, nulls = _get_str(v, 0)
+    assert nulls == {1}
+    assert out[0] == b"hello" and out[2] == b"world!!" and out[3] == b"xy"
+
+
+def test_var_string_real_empty_forces_bitmap():
+    """a zero-length REAL datum cannot share the null marker: bitmap
+    (:64-71) distinguishes null from empty."""
+    strs = [b"a", b"", b"ccc", b"dd"]  # row 1 is a real empty string
+    v = _dec(_enc(4, [_str_col(strs, null_rows=[3])]))
+    assert v.col[0].attrs & CA_HAS_NULL_BITMAP
+    out, nulls = _get_str(v, 0)
+    assert nulls == {3}
+    assert out[1] == b"" and 1 not in nulls
+    assert out[0] == b"a" and out[2] == b"ccc"
+
+
+def test_mixed_block_with_codecs():
+    """multi-column block: pooled string region is shared across string
+    columns in column order; integer and offset streams can carry any
+    stream codec."""
+    rows = 500
+    rng = np.random.default_rng(3)
+    ints1 = list(rng.integers(-10**6, 10**6, rows))
+    ints2 = list(np.sort(rng.integers(0, 10**9, rows)))
+    strs1 = [bytes("k%03d" % (i % 37), "ascii") for i in range(rows)]
+    strs2 = [b"v" * int(rng.integers(0, 12)) for _ in range(rows)]
+    cols = [
+        _int_col(ints1, null_rows=[5], enc=6),        # SIMD_FIXEDPFOR
+        _str_col(strs1),                               # fixed 4
+        _int_col(ints2, enc=3),                        # DDZP
+        _str_col(strs2, null_rows=[0, 499], enc=5),    # var, DZP offsets
+    ]
+    blob = _enc(rows, cols)
+    v = _dec(blob)
+    out, nulls = _get_int(v, 0)
+    assert nulls == {5}
+    assert [int(x) for r, x in enumerate(out) if r != 5] == \
+        [x for r, x in enumerate(ints1) if r != 5]
+    srows, snulls = _get_str(v, 1)
+    assert srows == strs1 and snulls == set()
+    out2, n2 = _get_int(v, 2)
+    assert list(out2) == [int(x) for x in ints2] and n2 == set()
+    srows2, snulls2 = _get_str(v, 3)
+    assert snulls2 == {0, 499}
+    assert all(srows2[r] == strs2[r] for r in range(rows)
+               if r not in snulls2)
+    # pooled region length = sum of both string columns' bytes
+    assert v.ach.all_string_data_length == \
+        rows * 4 + sum(len(s) for r, s in enumerate(strs2)
+                       if r not in {0, 499})
+
+
+def test_header_layout_pins():
+    """ObAllColumnHeader is 12 packed bytes, ObCSColumnHeader 4
+    (ob_column_encoding_struct.h:134-169, :60-63); stream offsets are
+    absolute block positions."""
+    assert C.sizeof(AllColHeader) == 12
+    blob = _enc(3, [_int_col([1, 2, 3])])
+    # block header 16B, then all-col header: version 0, attrs 0
+    assert blob[16] == 0 and blob[17] == 0
+    v = _dec(blob)
+    assert v.ach.stream_count == 1
+    # single RAW int stream ends where the pooled strings (none) begin
+    assert v.stream_offsets[0] == len(blob) - v.ach.stream_offsets_length
+
+
+def test_block_golden_pin():
+    """sha256 pin of a fixed mixed block (bytes are the contract)."""
+    rows = 200
+    rng = np.random.default_rng(44)
+    cols = [
+        _int_col(list(rng.integers(-500, 500, rows)), null_rows=[3, 77],
+                 enc=2),
+        _str_col([bytes("s%02d" % (i % 90), "ascii") for i in range(rows)]),
+        _str_col([b"z" * int(rng.integers(1, 9)) for _ in range(rows)],
+                 null_rows=[0]),
+    ]
+    blob = _enc(rows, cols)
+    import json
+    path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "golden", "cs_block.json")
+    with open(path) as f:
+        pin = json.load(f)
+    assert len(blob) == pin["bytes"]
+    assert hashlib.sha256(blob).hexdigest() == pin["sha256"]
+
+
+from hypothesis import given, settings, strategies as st  # noqa: E402
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.data())
+def test_property_block_roundtrip(data):
+    rows = data.draw(st.integers(1, 300))
+    rng = np.random.default_rng(data.draw(st.integers(0, 2**31)))
+    ncols = data.draw(st.integers(1, 4))
+    cols, expect = [], []
+    for _ in range(ncols):
+        nulls = sorted(set(
+            int(x) for x in rng.integers(0, rows,
+                                         int(rng.integers(0, rows // 2 + 1)))))
+        if data.draw(st.booleans()):
+            lo = data.draw(st.integers(-2**50, 2**50))
+            vals = [int(x) for x in rng.integers(lo, lo + 10**6, rows)]
+            cols.append(_int_col(vals, null_rows=nulls or None))
+            expect.append(("i", vals, set(nulls)))
+        else:
+            fixed = data.draw(st.booleans())
+            if fixed:
+                ln = data.draw(st.integers(1, 8))
+                strs = [bytes(rng.integers(65, 91, ln).astype(np.uint8))
+                        for _ in range(rows)]
+            else:
+                strs = [bytes(rng.integers(97, 123,
+                                           int(rng.integers(0, 10)))
+                              .astype(np.uint8)) for _ in range(rows)]
+            cols.append(_str_col(strs, null_rows=nulls or None))
+            expect.append(("s", strs, set(nulls)))
+    v = _dec(_enc(rows, cols))
+    for c, (kind, vals, nulls) in enumerate(expect):
+        if kind == "i":
+            out, n = _get_int(v, c)
+            assert n == nulls
+            assert all(int(out[r]) == vals[r] for r in range(rows)
+                       if r not in nulls)
+        else:
+            out, n = _get_str(v, c)
+            assert n == nulls
+            assert all(out[r] == vals[r] for r in range(rows)
+                       if r not in nulls)
