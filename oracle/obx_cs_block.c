@@ -53,14 +53,15 @@ static int64_t codec_dec(uint8_t t, const uint8_t *in, size_t len,
   }
 }
 
-/* offset-array integer stream (build_offset_array_stream_meta,
- * ob_stream_encoding_struct.cpp:182-202): no base, width covers the
- * last (largest) end offset, values stored as-is */
-static int64_t offset_stream_enc(const uint64_t *ends, uint32_t n,
-                                 uint8_t enc_type, uint8_t *buf,
-                                 size_t cap) {
+/* uint stream with caller-fixed width: used for offset arrays
+ * (build_offset_array_stream_meta, ob_stream_encoding_struct.cpp:
+ * 182-202) and dict ref streams (build_unsigned_stream_meta with
+ * min 0 -> no base; the width covers ref_stream_max_value_, which can
+ * exceed the array's own max, ob_dict_column_encoder.cpp:159-181) */
+static int64_t uint_stream_enc(const uint64_t *vals, uint32_t n,
+                               uint8_t enc_type, uint32_t wb,
+                               uint8_t *buf, size_t cap) {
   if (!n) return -1;
-  uint32_t wb = byte_packed(ends[n - 1]);
   obx_cs_int_meta m;
   memset(&m, 0, sizeof(m));
   m.version = 1;
@@ -72,7 +73,7 @@ static int64_t offset_stream_enc(const uint64_t *ends, uint32_t n,
   uint8_t *packed = (uint8_t *)malloc((size_t)n * wb);
   if (!packed) return -1;
   for (uint32_t i = 0; i < n; i++)
-    memcpy(packed + (size_t)i * wb, &ends[i], wb);
+    memcpy(packed + (size_t)i * wb, &vals[i], wb);
   int64_t dn;
   if (m.type == OBX_CS_ENC_RAW) {
     dn = (int64_t)n * wb;
@@ -84,6 +85,14 @@ static int64_t offset_stream_enc(const uint64_t *ends, uint32_t n,
   free(packed);
   if (dn < 0) return -1;
   return (int64_t)pos + dn;
+}
+
+static int64_t offset_stream_enc(const uint64_t *ends, uint32_t n,
+                                 uint8_t enc_type, uint8_t *buf,
+                                 size_t cap) {
+  if (!n) return -1;
+  return uint_stream_enc(ends, n, enc_type, byte_packed(ends[n - 1]), buf,
+                         cap);
 }
 
 static int64_t offset_stream_dec(const uint8_t *buf, size_t len, uint32_t n,
@@ -149,7 +158,243 @@ int64_t obx_cs_block_enc(uint32_t rows, uint32_t ncols,
     uint32_t null_cnt = 0;
     for (uint32_t r = 0; r < rows; r++)
       if (in_null(ci->nulls, r)) null_cnt++;
-    if (n_streams + 2 > OBX_CS_MAX_STREAMS) goto fail;
+    if (n_streams + 3 > OBX_CS_MAX_STREAMS) goto fail;
+
+    if (ci->want_dict) {
+      /* ---- INT_DICT / STR_DICT (ob_dict_column_encoder.cpp,
+       * ob_int_dict_column_encoder.cpp, ob_str_dict_column_encoder.cpp):
+       * [ObDictEncodingMeta][dict value stream(s)][ref stream] ---- */
+      chdr[c].type = ci->is_string ? OBX_CS_COL_STR_DICT
+                                   : OBX_CS_COL_INT_DICT;
+      chdr[c].obj_type = ci->is_string ? OBX_OBJ_VARCHAR : OBX_OBJ_INT;
+      uint32_t nn = rows - null_cnt;
+      obx_cs_dict_meta dm;
+      memset(&dm, 0, sizeof(dm));
+      dm.attrs = OBX_CS_DICT_IS_SORTED; /* do_sort_dict_ always runs */
+      if (null_cnt) dm.attrs |= OBX_CS_DICT_HAS_NULL;
+      uint32_t *refs = (uint32_t *)malloc((size_t)rows * 4);
+      uint32_t distinct = 0;
+      int64_t *idv = NULL;
+      const uint8_t **sps = NULL; /* distinct string ptrs */
+      uint32_t *sls = NULL;       /* distinct string lens */
+      if (!refs) goto fail;
+      if (!ci->is_string) {
+        if (!ci->ivals) { free(refs); goto fail; }
+        idv = (int64_t *)malloc((size_t)(nn ? nn : 1) * 8);
+        if (!idv) { free(refs); goto fail; }
+        uint32_t k = 0;
+        for (uint32_t r = 0; r < rows; r++)
+          if (!in_null(ci->nulls, r)) idv[k++] = ci->ivals[r];
+        /* sort + unique (ht sort_dict: ascending value order) */
+        for (uint32_t i = 1; i < k; i++) { /* insertion sort is fine for
+                                              oracle-scale blocks */
+          int64_t x = idv[i];
+          uint32_t j = i;
+          while (j && idv[j - 1] > x) { idv[j] = idv[j - 1]; j--; }
+          idv[j] = x;
+        }
+        for (uint32_t i = 0; i < k; i++)
+          if (!distinct || idv[distinct - 1] != idv[i])
+            idv[distinct++] = idv[i];
+        for (uint32_t r = 0; r < rows; r++) {
+          if (in_null(ci->nulls, r)) { refs[r] = distinct; continue; }
+          uint32_t lo = 0, hi = distinct;
+          while (lo + 1 < hi) {
+            uint32_t mid = (lo + hi) / 2;
+            if (idv[mid] <= ci->ivals[r]) lo = mid; else hi = mid;
+          }
+          refs[r] = lo;
+        }
+      } else {
+        if (!ci->lens) { free(refs); goto fail; }
+        const uint8_t **rp =
+            (const uint8_t **)malloc((size_t)rows * sizeof(void *));
+        sps = (const uint8_t **)malloc((size_t)rows * sizeof(void *) + 1);
+        sls = (uint32_t *)malloc((size_t)rows * 4 + 4);
+        if (!rp || !sps || !sls) { free(rp); goto fail_dict; }
+        const uint8_t *src = ci->bytes;
+        for (uint32_t r = 0; r < rows; r++) {
+          rp[r] = NULL;
+          if (!in_null(ci->nulls, r)) { rp[r] = src; src += ci->lens[r]; }
+        }
+        /* distinct, sorted by bytewise (memcmp, then length) order */
+        for (uint32_t r = 0; r < rows; r++) {
+          if (!rp[r]) continue;
+          uint32_t l = ci->lens[r];
+          uint32_t lo = 0, hi = distinct;
+          int found = 0;
+          while (lo < hi) {
+            uint32_t mid = (lo + hi) / 2;
+            uint32_t ml = sls[mid] < l ? sls[mid] : l;
+            int cr = memcmp(sps[mid], rp[r], ml);
+            if (!cr) cr = (sls[mid] > l) - (sls[mid] < l);
+            if (cr == 0) { found = 1; lo = mid; break; }
+            if (cr < 0) lo = mid + 1; else hi = mid;
+          }
+          if (!found) {
+            memmove(sps + lo + 1, sps + lo,
+                    (size_t)(distinct - lo) * sizeof(void *));
+            memmove(sls + lo + 1, sls + lo, (size_t)(distinct - lo) * 4);
+            sps[lo] = rp[r];
+            sls[lo] = l;
+            distinct++;
+          }
+        }
+        for (uint32_t r = 0; r < rows; r++) {
+          if (!rp[r]) { refs[r] = distinct; continue; }
+          uint32_t l = ci->lens[r], lo = 0, hi = distinct;
+          while (lo < hi) {
+            uint32_t mid = (lo + hi) / 2;
+            uint32_t ml = sls[mid] < l ? sls[mid] : l;
+            int cr = memcmp(sps[mid], rp[r], ml);
+            if (!cr) cr = (sls[mid] > l) - (sls[mid] < l);
+            if (cr == 0) { lo = mid; break; }
+            if (cr < 0) lo = mid + 1; else hi = mid;
+          }
+          refs[r] = lo;
+        }
+        free(rp);
+      }
+      dm.distinct_val_cnt = distinct;
+
+      if (distinct == 0) {
+        /* all rows null: only the dict meta is stored
+         * (build_ref_encoder_ctx_ :70-76, store_dict_ref_ :225-227) */
+        dm.ref_row_cnt = 0;
+        if (pos + sizeof(dm) > cap) goto fail_dict;
+        memcpy(buf + pos, &dm, sizeof(dm));
+        pos += sizeof(dm);
+        free(refs);
+        free(idv);
+        free(sps);
+        free(sls);
+        continue;
+      }
+
+      /* const-ref trial (try_const_encoding_ref_ :139-181) */
+      uint64_t max_ref = distinct - 1 + (null_cnt ? 1 : 0);
+      uint32_t *cnt = (uint32_t *)calloc(distinct + 1, 4);
+      if (!cnt) goto fail_dict;
+      for (uint32_t r = 0; r < rows; r++) cnt[refs[r]]++;
+      uint32_t const_ref = 0, max_cnt = 0;
+      for (uint32_t i = 0; i <= distinct; i++)
+        if (cnt[i] > max_cnt) { max_cnt = cnt[i]; const_ref = i; }
+      free(cnt);
+      uint32_t ec = rows - max_cnt;
+      int use_const = 0;
+      uint64_t ref_max;
+      if (ec == 0) {
+        use_const = 1;
+        dm.attrs |= OBX_CS_DICT_CONST_REF;
+        dm.ref_row_cnt = 2;
+        ref_max = const_ref;
+      } else if (ec <= OBX_CS_DICT_MAX_EXCEPTIONS &&
+                 ec < (uint64_t)rows * OBX_CS_DICT_MAX_EXCEPTION_PCT /
+                          100) {
+        use_const = 1;
+        dm.attrs |= OBX_CS_DICT_CONST_REF;
+        dm.ref_row_cnt = 2 + 2 * ec;
+        uint32_t ex_max_row = 0;
+        for (int64_t r = (int64_t)rows - 1; r >= 0; r--)
+          if (refs[r] != const_ref) { ex_max_row = (uint32_t)r; break; }
+        ref_max = ec;
+        if (ex_max_row > ref_max) ref_max = ex_max_row;
+        if (max_ref > ref_max) ref_max = max_ref;
+      } else {
+        dm.ref_row_cnt = rows;
+        ref_max = max_ref;
+      }
+      if (pos + sizeof(dm) > cap) goto fail_dict;
+      memcpy(buf + pos, &dm, sizeof(dm));
+      pos += sizeof(dm);
+
+      /* dict value stream(s) */
+      if (!ci->is_string) {
+        int64_t n = obx_cs_int_stream_enc3(
+            idv, NULL, distinct, ci->enc_type ? ci->enc_type
+                                              : OBX_CS_ENC_RAW,
+            0, 0, buf + pos, cap - pos);
+        if (n < 0) goto fail_dict;
+        pos += (size_t)n;
+        soff[n_streams++] = pos;
+      } else {
+        uint64_t total = 0;
+        int fixed = 1;
+        for (uint32_t i = 0; i < distinct; i++) {
+          total += sls[i];
+          if (sls[i] != sls[0]) fixed = 0;
+        }
+        obx_cs_str_meta sm;
+        memset(&sm, 0, sizeof(sm));
+        sm.uncompressed_len = (uint32_t)total;
+        if (fixed) {
+          sm.attr |= OBX_CS_STR_FIXED_LEN;
+          sm.fixed_str_len = sls[0];
+        }
+        int hn = obx_cs_str_meta_enc(&sm, buf + pos, cap - pos);
+        if (hn < 0) goto fail_dict;
+        pos += (size_t)hn;
+        soff[n_streams++] = pos;
+        if (pool_len + total > str_pool_cap) goto fail_dict;
+        for (uint32_t i = 0; i < distinct; i++) {
+          memcpy(pool + pool_len, sps[i], sls[i]);
+          pool_len += sls[i];
+        }
+        if (!fixed) {
+          uint64_t *ends = (uint64_t *)malloc((size_t)distinct * 8);
+          if (!ends) goto fail_dict;
+          uint64_t acc = 0;
+          for (uint32_t i = 0; i < distinct; i++) {
+            acc += sls[i];
+            ends[i] = acc;
+          }
+          int64_t n = offset_stream_enc(ends, distinct, ci->enc_type,
+                                        buf + pos, cap - pos);
+          free(ends);
+          if (n < 0) goto fail_dict;
+          pos += (size_t)n;
+          soff[n_streams++] = pos;
+        }
+      }
+
+      /* ref stream (do_store_dict_ref_, ob_dict_column_encoder.h:58-109;
+       * const layout: [exception_cnt][const_ref][row ids][refs]) */
+      {
+        uint64_t *ra = (uint64_t *)malloc((size_t)dm.ref_row_cnt * 8);
+        if (!ra) goto fail_dict;
+        if (use_const) {
+          ra[0] = ec;
+          ra[1] = const_ref;
+          uint32_t idx = 0;
+          for (uint32_t r = 0; r < rows && idx < ec; r++)
+            if (refs[r] != const_ref) {
+              ra[2 + idx] = r;
+              ra[2 + ec + idx] = refs[r];
+              idx++;
+            }
+        } else {
+          for (uint32_t r = 0; r < rows; r++) ra[r] = refs[r];
+        }
+        int64_t n = uint_stream_enc(ra, dm.ref_row_cnt, OBX_CS_ENC_RAW,
+                                    byte_packed(ref_max), buf + pos,
+                                    cap - pos);
+        free(ra);
+        if (n < 0) goto fail_dict;
+        pos += (size_t)n;
+        soff[n_streams++] = pos;
+      }
+      free(refs);
+      free(idv);
+      free(sps);
+      free(sls);
+      continue;
+    fail_dict:
+      free(refs);
+      free(idv);
+      free(sps);
+      free(sls);
+      goto fail;
+    }
 
     if (!ci->is_string) {
       /* ---- INTEGER column (ob_integer_column_encoder.cpp) ---- */
@@ -418,6 +663,42 @@ int obx_cs_block_dec(const uint8_t *buf, size_t len,
         cv->off_stream_len = v->stream_offsets[si] - pos;
         pos = v->stream_offsets[si++];
       }
+    } else if (cv->h.type == OBX_CS_COL_INT_DICT ||
+               cv->h.type == OBX_CS_COL_STR_DICT) {
+      if (pos + sizeof(obx_cs_dict_meta) > len) return -1;
+      memcpy(&cv->dm, buf + pos, sizeof(obx_cs_dict_meta));
+      pos += sizeof(obx_cs_dict_meta);
+      if (cv->dm.distinct_val_cnt == 0) continue; /* all-null column */
+      if (cv->h.type == OBX_CS_COL_INT_DICT) {
+        if (si >= v->stream_count || v->stream_offsets[si] < pos)
+          return -1;
+        cv->int_stream = buf + pos;
+        cv->int_stream_len = v->stream_offsets[si] - pos;
+        pos = v->stream_offsets[si++];
+      } else {
+        if (si >= v->stream_count || v->stream_offsets[si] < pos)
+          return -1;
+        int hn = obx_cs_str_meta_dec(buf + pos,
+                                     v->stream_offsets[si] - pos,
+                                     &cv->sm);
+        if (hn < 0 || pos + (size_t)hn != v->stream_offsets[si])
+          return -1;
+        pos = v->stream_offsets[si++];
+        cv->str_data_off = str_off;
+        str_off += cv->sm.uncompressed_len;
+        if (str_off > v->ach.all_string_data_length) return -1;
+        if (!(cv->sm.attr & OBX_CS_STR_FIXED_LEN)) {
+          if (si >= v->stream_count || v->stream_offsets[si] < pos)
+            return -1;
+          cv->off_stream = buf + pos;
+          cv->off_stream_len = v->stream_offsets[si] - pos;
+          pos = v->stream_offsets[si++];
+        }
+      }
+      if (si >= v->stream_count || v->stream_offsets[si] < pos) return -1;
+      cv->ref_stream = buf + pos;
+      cv->ref_stream_len = v->stream_offsets[si] - pos;
+      pos = v->stream_offsets[si++];
     } else {
       return -1;
     }
@@ -429,9 +710,75 @@ static inline int blk_null(const uint8_t *bm, uint32_t r) {
   return bm && ((bm[r / 8] >> (7 - r % 8)) & 1); /* MSB-first in block */
 }
 
+/* decode a dict column's ref stream into one ref per row
+ * (do_store_dict_ref_: plain row refs, or the const layout
+ * [exception_cnt][const_ref][row ids][refs]) */
+static int dict_refs(const obx_cs_col_view *cv, uint32_t rows,
+                     uint32_t *refs) {
+  uint64_t *ra = (uint64_t *)malloc((size_t)cv->dm.ref_row_cnt * 8);
+  if (!ra) return -1;
+  if (offset_stream_dec(cv->ref_stream, cv->ref_stream_len,
+                        cv->dm.ref_row_cnt, ra) < 0) {
+    free(ra);
+    return -1;
+  }
+  if (cv->dm.attrs & OBX_CS_DICT_CONST_REF) {
+    uint64_t ec = ra[0];
+    if (cv->dm.ref_row_cnt != 2 + 2 * ec) { free(ra); return -1; }
+    for (uint32_t r = 0; r < rows; r++) refs[r] = (uint32_t)ra[1];
+    for (uint64_t i = 0; i < ec; i++) {
+      uint64_t row = ra[2 + i];
+      if (row >= rows) { free(ra); return -1; }
+      refs[row] = (uint32_t)ra[2 + ec + i];
+    }
+  } else {
+    if (cv->dm.ref_row_cnt != rows) { free(ra); return -1; }
+    for (uint32_t r = 0; r < rows; r++) refs[r] = (uint32_t)ra[r];
+  }
+  free(ra);
+  return 0;
+}
+
 int obx_cs_block_get_int(const obx_cs_block_view *v, uint32_t c,
                          int64_t *out, uint8_t *nulls_out) {
-  if (c >= v->ncols || v->col[c].h.type != OBX_CS_COL_INTEGER) return -1;
+  if (c >= v->ncols) return -1;
+  if (v->col[c].h.type == OBX_CS_COL_INT_DICT) {
+    const obx_cs_col_view *cv = &v->col[c];
+    uint32_t distinct = cv->dm.distinct_val_cnt;
+    if (nulls_out) memset(nulls_out, 0, (v->rows + 7) / 8);
+    if (distinct == 0) { /* all null */
+      for (uint32_t r = 0; r < v->rows; r++) {
+        out[r] = 0;
+        if (nulls_out) nulls_out[r >> 3] |= (uint8_t)(1 << (r & 7));
+      }
+      return 0;
+    }
+    int64_t *dv = (int64_t *)malloc((size_t)distinct * 8);
+    uint32_t *refs = (uint32_t *)malloc((size_t)v->rows * 4);
+    int rc = -1;
+    if (dv && refs &&
+        obx_cs_int_stream_dec(cv->int_stream, cv->int_stream_len,
+                              distinct, dv, NULL) >= 0 &&
+        dict_refs(cv, v->rows, refs) == 0) {
+      rc = 0;
+      for (uint32_t r = 0; r < v->rows; r++) {
+        uint32_t ref = refs[r];
+        if (ref == distinct && (cv->dm.attrs & OBX_CS_DICT_HAS_NULL)) {
+          out[r] = 0;
+          if (nulls_out) nulls_out[r >> 3] |= (uint8_t)(1 << (r & 7));
+        } else if (ref < distinct) {
+          out[r] = dv[ref];
+        } else {
+          rc = -1;
+          break;
+        }
+      }
+    }
+    free(dv);
+    free(refs);
+    return rc;
+  }
+  if (v->col[c].h.type != OBX_CS_COL_INTEGER) return -1;
   const obx_cs_col_view *cv = &v->col[c];
   obx_cs_int_meta m;
   if (obx_cs_int_stream_dec(cv->int_stream, cv->int_stream_len, v->rows,
@@ -457,7 +804,59 @@ int obx_cs_block_get_int(const obx_cs_block_view *v, uint32_t c,
 int64_t obx_cs_block_get_str(const obx_cs_block_view *v, uint32_t c,
                              uint8_t *bytes_out, size_t bytes_cap,
                              uint32_t *lens_out, uint8_t *nulls_out) {
-  if (c >= v->ncols || v->col[c].h.type != OBX_CS_COL_STRING) return -1;
+  if (c >= v->ncols) return -1;
+  if (v->col[c].h.type == OBX_CS_COL_STR_DICT) {
+    const obx_cs_col_view *cv = &v->col[c];
+    uint32_t distinct = cv->dm.distinct_val_cnt;
+    if (nulls_out) memset(nulls_out, 0, (v->rows + 7) / 8);
+    if (distinct == 0) {
+      for (uint32_t r = 0; r < v->rows; r++) {
+        lens_out[r] = 0;
+        if (nulls_out) nulls_out[r >> 3] |= (uint8_t)(1 << (r & 7));
+      }
+      return 0;
+    }
+    /* dict value starts/lens from the pooled bytes */
+    uint64_t *ends = (uint64_t *)malloc((size_t)distinct * 8);
+    uint32_t *refs = (uint32_t *)malloc((size_t)v->rows * 4);
+    int64_t rc = -1;
+    if (!ends || !refs) goto done;
+    if (cv->sm.attr & OBX_CS_STR_FIXED_LEN) {
+      for (uint32_t i = 0; i < distinct; i++)
+        ends[i] = (uint64_t)(i + 1) * cv->sm.fixed_str_len;
+    } else if (offset_stream_dec(cv->off_stream, cv->off_stream_len,
+                                 distinct, ends) < 0) {
+      goto done;
+    }
+    if (ends[distinct - 1] != cv->sm.uncompressed_len) goto done;
+    if (dict_refs(cv, v->rows, refs) != 0) goto done;
+    {
+      const uint8_t *dict_bytes = v->all_string + cv->str_data_off;
+      size_t opos = 0;
+      for (uint32_t r = 0; r < v->rows; r++) {
+        uint32_t ref = refs[r];
+        if (ref == distinct && (cv->dm.attrs & OBX_CS_DICT_HAS_NULL)) {
+          lens_out[r] = 0;
+          if (nulls_out) nulls_out[r >> 3] |= (uint8_t)(1 << (r & 7));
+        } else if (ref < distinct) {
+          uint64_t s = ref ? ends[ref - 1] : 0;
+          uint32_t l = (uint32_t)(ends[ref] - s);
+          if (opos + l > bytes_cap) goto done;
+          memcpy(bytes_out + opos, dict_bytes + s, l);
+          opos += l;
+          lens_out[r] = l;
+        } else {
+          goto done;
+        }
+      }
+      rc = (int64_t)opos;
+    }
+  done:
+    free(ends);
+    free(refs);
+    return rc;
+  }
+  if (v->col[c].h.type != OBX_CS_COL_STRING) return -1;
   const obx_cs_col_view *cv = &v->col[c];
   const uint8_t *src = v->all_string + cv->str_data_off;
   uint32_t total = cv->sm.uncompressed_len;

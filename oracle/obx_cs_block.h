@@ -36,8 +36,11 @@
  * the pooled string region is never compressed (NONE_COMPRESSOR) and the
  * block-tail offset stream is always RAW (the reference picks codecs by
  * ObCSEncodingOpt cost trials — codec CHOICE is policy, the formats are
- * what this slice pins). INT_DICT/STR_DICT/SEMISTRUCT column types are
- * the remaining CS rows.
+ * what this slice pins; dict ref streams likewise always RAW).
+ * INT_DICT/STR_DICT (ob_dict_column_encoder.cpp and subclasses) are
+ * restated including the CONST_ENCODING_REF exception-list layout and
+ * the ref width rule (width covers ref_stream_max_value_, not just the
+ * stored array's max). SEMISTRUCT remains out of scope.
  */
 #ifndef OBX_CS_BLOCK_H_
 #define OBX_CS_BLOCK_H_
@@ -56,7 +59,20 @@ extern "C" {
 enum {
   OBX_CS_COL_INTEGER = 0,
   OBX_CS_COL_STRING = 1,
+  OBX_CS_COL_INT_DICT = 2,
+  OBX_CS_COL_STR_DICT = 3,
 };
+
+/* ObDictEncodingMeta::Attribute (ob_column_encoding_struct.h:171-208) */
+enum {
+  OBX_CS_DICT_IS_SORTED = 0x1,
+  OBX_CS_DICT_HAS_NULL = 0x2,
+  OBX_CS_DICT_CONST_REF = 0x4,
+};
+
+/* ObDictColumnEncoder limits (ob_dict_column_encoder.h:21-22) */
+#define OBX_CS_DICT_MAX_EXCEPTIONS 64
+#define OBX_CS_DICT_MAX_EXCEPTION_PCT 10
 
 /* ObCSColumnHeader::Attribute */
 enum {
@@ -98,6 +114,15 @@ typedef struct obx_cs_col_header {
   uint8_t attrs;
   uint8_t obj_type;
 } obx_cs_col_header;
+
+/* ObDictEncodingMeta (ob_column_encoding_struct.h:202-205): 10 packed
+ * bytes ahead of the dict-value stream(s) */
+typedef struct obx_cs_dict_meta {
+  uint8_t version;
+  uint8_t attrs;
+  uint32_t distinct_val_cnt;
+  uint32_t ref_row_cnt;
+} obx_cs_dict_meta;
 #pragma pack(pop)
 
 /* encoder input: one column. Integer columns give ivals; string columns
@@ -107,6 +132,10 @@ typedef struct obx_cs_col_header {
 typedef struct obx_cs_col_in {
   uint8_t is_string;
   uint8_t enc_type;
+  uint8_t want_dict; /* 1 -> INT_DICT/STR_DICT (encoding CHOICE is the
+                        reference's cost-trial policy, encoder_detection_;
+                        here the caller decides, the FORMAT is what is
+                        restated) */
   const int64_t *ivals;
   const uint8_t *bytes;
   const uint32_t *lens;
@@ -120,12 +149,15 @@ int64_t obx_cs_block_enc(uint32_t rows, uint32_t ncols,
 typedef struct obx_cs_col_view {
   obx_cs_col_header h;
   const uint8_t *null_bitmap; /* MSB-first in-block bitmap or NULL */
-  const uint8_t *int_stream;  /* integer col: [meta][data] */
+  const uint8_t *int_stream;  /* integer col / dict-value int stream */
   size_t int_stream_len;
-  obx_cs_str_meta sm;         /* string col */
+  obx_cs_str_meta sm;         /* string col / str-dict value stream */
   const uint8_t *off_stream;  /* var string: offset int stream */
   size_t off_stream_len;
   uint32_t str_data_off;      /* into the pooled all-string region */
+  obx_cs_dict_meta dm;        /* dict cols */
+  const uint8_t *ref_stream;  /* dict cols: ref int stream */
+  size_t ref_stream_len;
 } obx_cs_col_view;
 
 typedef struct obx_cs_block_view {
