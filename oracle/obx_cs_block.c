@@ -408,17 +408,27 @@ int64_t obx_cs_block_enc(uint32_t rows, uint32_t ncols,
         any = 1;
       }
       if (!any) mn = mx = 0;
-      /* null handling (:190-220, int64 store type): replace with a
-       * value adjacent to the range; bitmap only when the range spans
-       * the whole of int64 */
+      /* null handling (:190-220): replace with a value adjacent to the
+       * range; bitmap only when the range spans the column's whole
+       * STORE type (INTEGER_MASK_TABLE[type_store_size_], :183-186) */
+      uint32_t sw = ci->store_width ? ci->store_width : 8;
+      int64_t ts_min =
+          sw == 8 ? INT64_MIN : -((int64_t)1 << (8 * sw - 1));
+      int64_t ts_max =
+          sw == 8 ? INT64_MAX : ((int64_t)1 << (8 * sw - 1)) - 1;
       int use_replace = 0, use_bitmap = 0;
       int64_t replace = 0;
-      if (null_cnt > 0) {
+      if (!any && null_cnt > 0) {
+        /* all datums null: build_signed_stream_meta(0, 0, replace, 0)
+         * (ob_integer_column_encoder.cpp:79-89) */
+        use_replace = 1;
+        replace = 0;
+      } else if (null_cnt > 0) {
         if (mn == 0) {
-          if (mx != INT64_MAX) { use_replace = 1; replace = mx + 1; }
+          if (mx != ts_max) { use_replace = 1; replace = mx + 1; }
           else { use_replace = 1; replace = -1; }
-        } else if (mn == INT64_MIN) {
-          if (mx != INT64_MAX) { use_replace = 1; replace = mx + 1; }
+        } else if (mn == ts_min) {
+          if (mx != ts_max) { use_replace = 1; replace = mx + 1; }
           else { use_bitmap = 1; }
         } else {
           use_replace = 1;
@@ -426,7 +436,10 @@ int64_t obx_cs_block_enc(uint32_t rows, uint32_t ncols,
         }
       }
       chdr[c].type = OBX_CS_COL_INTEGER;
-      chdr[c].obj_type = OBX_OBJ_INT;
+      /* ObObjType per store width (ob_obj_type.h: TinyInt=1, SmallInt=2,
+       * Int32=4, Int=5) */
+      chdr[c].obj_type = sw == 1 ? 1 : sw == 2 ? 2 : sw == 4 ? 4
+                                                 : OBX_OBJ_INT;
       if (use_bitmap) {
         chdr[c].attrs |= OBX_CS_CA_HAS_NULL_BITMAP;
         if (pos + bitmap_size > cap) goto fail;

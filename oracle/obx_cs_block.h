@@ -136,6 +136,13 @@ typedef struct obx_cs_col_in {
                         reference's cost-trial policy, encoder_detection_;
                         here the caller decides, the FORMAT is what is
                         restated) */
+  uint8_t store_width; /* integer column store-type size in bytes
+                          (1/2/4/8, 0 -> 8): the null-replace decision
+                          compares min/max against THIS type's bounds
+                          (ob_integer_column_encoder.cpp:183-186
+                          INTEGER_MASK_TABLE[type_store_size_]), so an
+                          int32 column whose min is INT32_MIN picks
+                          max+1, not min-1 */
   const int64_t *ivals;
   const uint8_t *bytes;
   const uint32_t *lens;

@@ -22,7 +22,7 @@ _lib = C.CDLL(os.path.join(os.path.dirname(os.path.dirname(
 
 class ColIn(C.Structure):
     _fields_ = [("is_string", C.c_uint8), ("enc_type", C.c_uint8),
-                ("want_dict", C.c_uint8),
+                ("want_dict", C.c_uint8), ("store_width", C.c_uint8),
                 ("ivals", C.POINTER(C.c_int64)),
                 ("bytes", C.POINTER(C.c_uint8)),
                 ("lens", C.POINTER(C.c_uint32)),
@@ -115,12 +115,13 @@ def _nulls_bitmap(rows, null_rows):
     return bm
 
 
-def _int_col(vals, null_rows=None, enc=0, dict_=False):
+def _int_col(vals, null_rows=None, enc=0, dict_=False, store_width=8):
     rows = len(vals)
     col = ColIn()
     col.is_string = 0
     col.enc_type = enc
     col.want_dict = 1 if dict_ else 0
+    col.store_width = store_width
     v = np.asarray(vals, dtype=np.int64)
     col.ivals = v.ctypes.data_as(C.POINTER(C.c_int64))
     bm = _nulls_bitmap(rows, null_rows)

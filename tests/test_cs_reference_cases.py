@@ -219,3 +219,86 @@ def test_rle_shape(enc_type):
     vals = vals[:600]
     out, m, _ = _run(vals, enc_type=enc_type)
     assert [int(x) for x in out] == vals
+
+
+# ---- block/column-level cases recreated from test_cs_encoder.cpp ----
+
+from test_cs_block import (  # noqa: E402
+    _enc as blk_enc, _dec as blk_dec, _get_int as blk_get_int,
+    _int_col as blk_int_col, CA_HAS_NULL_BITMAP,
+)
+
+
+def _col_meta(v, c):
+    """decode the column's stream meta from the block view."""
+    cv = v.col[c]
+    m = IntMeta()
+    out = np.zeros(v.rows, dtype=np.int64)
+    n = _lib.obx_cs_int_stream_dec(
+        C.cast(cv.int_stream, C.POINTER(C.c_uint8)), cv.int_stream_len,
+        v.rows, out.ctypes.data_as(C.POINTER(C.c_int64)), C.byref(m))
+    assert n > 0
+    return m
+
+
+def test_cs_encoder_integer_cases():
+    """TestCSEncoder::test_integer_encoder cases <1>-<7> on an int32
+    column (store_width 4): the exact width/base/replace expectations
+    the reference asserts."""
+    i32min, i32max = -2**31, 2**31 - 1
+    # <1> -50..49: width 1, base -50, no replace
+    v = blk_dec(blk_enc(100, [blk_int_col(list(range(-50, 50)),
+                                          store_width=4)]))
+    m = _col_meta(v, 0)
+    assert m.width_tag == 0 and (m.attr & USE_BASE)
+    assert m.base == ((-50) & ((1 << 64) - 1))
+    assert not (m.attr & REPLACE_NULL)
+    # <2> INT32_MIN/NULL/-1: width 4, replace value 0 (min == store min
+    # -> largest-not-existed = max+1 = 0), base INT32_MIN
+    v = blk_dec(blk_enc(3, [blk_int_col([i32min, 0, -1], null_rows=[1],
+                                        store_width=4)]))
+    m = _col_meta(v, 0)
+    assert m.width_tag == 2 and (m.attr & REPLACE_NULL)
+    assert m.null_replaced == 0
+    assert m.base == (i32min & ((1 << 64) - 1))
+    out, nulls = blk_get_int(v, 0)
+    assert nulls == {1} and out[0] == i32min and out[2] == -1
+    # <3> full int32 span + null: bitmap (no adjacent value left),
+    # width 4, base INT32_MIN
+    vals = [i32min, i32max, 0] + list(range(-100, 100))
+    v = blk_dec(blk_enc(len(vals), [blk_int_col(vals, null_rows=[2],
+                                                store_width=4)]))
+    assert v.col[0].attrs & CA_HAS_NULL_BITMAP
+    m = _col_meta(v, 0)
+    assert not (m.attr & REPLACE_NULL) and m.width_tag == 2
+    assert m.base == (i32min & ((1 << 64) - 1))
+    out, nulls = blk_get_int(v, 0)
+    assert nulls == {2} and out[0] == i32min and out[1] == i32max
+    # <4> 0/NULL/INT32_MAX: replace -1 (max == store max, min == 0)
+    v = blk_dec(blk_enc(3, [blk_int_col([0, 0, i32max], null_rows=[1],
+                                        store_width=4)]))
+    m = _col_meta(v, 0)
+    assert (m.attr & REPLACE_NULL)
+    assert m.null_replaced == ((-1) & ((1 << 64) - 1))
+    # <5> 0/NULL/INT32_MAX-1: replace INT32_MAX (max+1)
+    v = blk_dec(blk_enc(3, [blk_int_col([0, 0, i32max - 1],
+                                        null_rows=[1], store_width=4)]))
+    m = _col_meta(v, 0)
+    assert (m.attr & REPLACE_NULL) and m.null_replaced == i32max
+    # <6> 0..999 monotonic: width 2, no base, no bitmap, no replace
+    v = blk_dec(blk_enc(1000, [blk_int_col(list(range(1000)),
+                                           store_width=4)]))
+    m = _col_meta(v, 0)
+    assert m.width_tag == 1
+    assert not (m.attr & (USE_BASE | REPLACE_NULL))
+    assert not (v.col[0].attrs & CA_HAS_NULL_BITMAP)
+    # <7> all null: width 1, replace value 0, no base, no bitmap
+    v = blk_dec(blk_enc(1000, [blk_int_col([0] * 1000,
+                                           null_rows=list(range(1000)),
+                                           store_width=4)]))
+    m = _col_meta(v, 0)
+    assert m.width_tag == 0 and (m.attr & REPLACE_NULL)
+    assert m.null_replaced == 0 and not (m.attr & USE_BASE)
+    assert not (v.col[0].attrs & CA_HAS_NULL_BITMAP)
+    out, nulls = blk_get_int(v, 0)
+    assert nulls == set(range(1000))
