@@ -302,3 +302,69 @@ def test_cs_encoder_integer_cases():
     assert not (v.col[0].attrs & CA_HAS_NULL_BITMAP)
     out, nulls = blk_get_int(v, 0)
     assert nulls == set(range(1000))
+
+
+def test_cs_encoder_dict_const_ref_cases():
+    """TestCSEncoder::test_dict_const_ref_encoder: five columns with 64
+    exceptions each probing the const-ref decision from both sides
+    (value-const vs null-const vs too-spread)."""
+    from test_cs_block import _str_col as blk_str_col
+    rows, ec = 1000, 64
+    base = rows - ec
+    # col0: 936 x 1000 then 0..63 -> const ref (value const), dict 65
+    c0 = [1000] * base + list(range(ec))
+    # col1: 936 x "a"*10 then 64 x "a"*20 -> str const ref, const ref 0
+    c1 = [b"a" * 10] * base + [b"a" * 20] * ec
+    # col2: i%2+1000 with 64 trailing nulls -> too spread, no const
+    c2 = [i % 2 + 1000 for i in range(base)] + [0] * ec
+    c2_nulls = list(range(base, rows))
+    # col3: 936 nulls then "a"*100 / "a"*101 -> NULL is the const value
+    c3 = [b""] * base + [b"a" * (100 + i % 2) for i in range(ec)]
+    c3_nulls = list(range(base))
+    # col4: 936 x 1000 then 64 nulls -> const 1000, null exceptions
+    c4 = [1000] * base + [0] * ec
+    c4_nulls = list(range(base, rows))
+    v = blk_dec(blk_enc(rows, [
+        blk_int_col(c0, dict_=True),
+        blk_str_col(c1, dict_=True),
+        blk_int_col(c2, null_rows=c2_nulls, dict_=True),
+        blk_str_col(c3, null_rows=c3_nulls, dict_=True),
+        blk_int_col(c4, null_rows=c4_nulls, dict_=True),
+    ]))
+    CONST = 0x4
+    # col0: const, ec exceptions, dict of 65
+    assert v.col[0].dm_attrs & CONST
+    assert v.col[0].dm_ref_row_cnt == 2 + 2 * ec
+    assert v.col[0].dm_distinct == 1 + ec
+    # col1: const, 2 distinct, the shorter string sorts first (ref 0
+    # is the const -- the reference asserts const_node_.dict_ref_ == 0)
+    assert v.col[1].dm_attrs & CONST
+    assert v.col[1].dm_distinct == 2
+    assert v.col[1].dm_ref_row_cnt == 2 + 2 * ec
+    # col2: 468/468/64 split -> exceptions over 10% -> plain refs
+    assert not (v.col[2].dm_attrs & CONST)
+    assert v.col[2].dm_ref_row_cnt == rows
+    # col3: null is the const value (the reference asserts max_ref_ == 2
+    # and const ref == 2 == distinct_val_cnt)
+    assert v.col[3].dm_attrs & CONST
+    assert v.col[3].dm_distinct == 2
+    assert v.col[3].dm_ref_row_cnt == 2 + 2 * ec
+    # col4: const 1000 with null exceptions (max_ref 1, const ref 0)
+    assert v.col[4].dm_attrs & CONST
+    assert v.col[4].dm_distinct == 1
+    assert v.col[4].dm_ref_row_cnt == 2 + 2 * ec
+    # roundtrips
+    out, nulls = blk_get_int(v, 0)
+    assert list(out) == c0 and nulls == set()
+    out, nulls = blk_get_int(v, 2)
+    assert nulls == set(c2_nulls)
+    assert [int(x) for x in out[:base]] == c2[:base]
+    out, nulls = blk_get_int(v, 4)
+    assert nulls == set(c4_nulls)
+    assert all(int(x) == 1000 for x in out[:base])
+    from test_cs_block import _get_str as blk_get_str
+    srows, snulls = blk_get_str(v, 1)
+    assert srows == c1 and snulls == set()
+    srows, snulls = blk_get_str(v, 3)
+    assert snulls == set(c3_nulls)
+    assert srows[base:] == c3[base:]
