@@ -368,3 +368,57 @@ def test_cs_encoder_dict_const_ref_cases():
     srows, snulls = blk_get_str(v, 3)
     assert snulls == set(c3_nulls)
     assert srows[base:] == c3[base:]
+
+
+def test_cs_encoder_string_cases():
+    """TestCSEncoder::test_string_encoder cases <1>-<5>: the
+    fixed/var/zero-len-null/bitmap decision matrix the reference
+    asserts per shape."""
+    from test_cs_block import (
+        _str_col as blk_str_col, _get_str as blk_get_str,
+        CA_IS_FIXED, STR_ZERO_LEN_NULL, STR_FIXED_LEN,
+    )
+    # <1> 100 fixed-64 strings + 1 null: IS_FIXED + null bitmap, not
+    # zero-len-null (padding+bitmap beats an offset stream)
+    strs = [bytes([65 + i % 26]) * 64 for i in range(100)] + [b""]
+    v = blk_dec(blk_enc(101, [blk_str_col(strs, null_rows=[100])]))
+    assert v.col[0].attrs & CA_IS_FIXED
+    assert v.col[0].attrs & CA_HAS_NULL_BITMAP
+    assert v.col[0].sm_attr & STR_FIXED_LEN
+    assert v.col[0].sm_fixed_str_len == 64
+    assert not (v.col[0].sm_attr & STR_ZERO_LEN_NULL)
+    out, nulls = blk_get_str(v, 0)
+    assert nulls == {100} and out[:100] == strs[:100]
+    # <2> var lens 1..100 + null, no zero-length datum: var with
+    # zero-len-as-null, NO bitmap
+    strs = [b"\x0f" * (i + 1) for i in range(100)] + [b""]
+    v = blk_dec(blk_enc(101, [blk_str_col(strs, null_rows=[100])]))
+    assert not (v.col[0].attrs & CA_IS_FIXED)
+    assert not (v.col[0].attrs & CA_HAS_NULL_BITMAP)
+    assert v.col[0].sm_attr & STR_ZERO_LEN_NULL
+    out, nulls = blk_get_str(v, 0)
+    assert nulls == {100} and out[:100] == strs[:100]
+    # <3> var lens 0..N + null, WITH a zero-length datum: bitmap, var,
+    # not zero-len-null
+    strs = [b"\x0f" * i for i in range(60)] + [b""]
+    v = blk_dec(blk_enc(61, [blk_str_col(strs, null_rows=[60])]))
+    assert not (v.col[0].attrs & CA_IS_FIXED)
+    assert v.col[0].attrs & CA_HAS_NULL_BITMAP
+    assert not (v.col[0].sm_attr & STR_ZERO_LEN_NULL)
+    out, nulls = blk_get_str(v, 0)
+    assert nulls == {60}
+    assert out[0] == b"" and 0 not in nulls  # real empty vs null
+    # <4> all null: var + zero-len-as-null, no bitmap
+    v = blk_dec(blk_enc(100, [blk_str_col([b""] * 100,
+                                          null_rows=list(range(100)))]))
+    assert not (v.col[0].attrs & CA_IS_FIXED)
+    assert not (v.col[0].attrs & CA_HAS_NULL_BITMAP)
+    assert v.col[0].sm_attr & STR_ZERO_LEN_NULL
+    out, nulls = blk_get_str(v, 0)
+    assert nulls == set(range(100))
+    # <5> all zero-length REAL datums: fixed with fixed_len 0
+    v = blk_dec(blk_enc(100, [blk_str_col([b""] * 100)]))
+    assert v.col[0].attrs & CA_IS_FIXED
+    assert v.col[0].sm_fixed_str_len == 0
+    out, nulls = blk_get_str(v, 0)
+    assert nulls == set() and all(s == b"" for s in out)
