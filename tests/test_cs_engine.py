@@ -109,3 +109,34 @@ def test_cs_load_gpu_scan_parity():
     exp = _expected(flag, qty, ship, null_rows, cutoff)
     assert res_cpu.rows_passed == sum(v[0] for v in exp.values())
     eng.close()
+
+
+def test_cs_load_preserves_nullness_for_nu_nn():
+    """IS NULL / IS NOT NULL white filters over CS-loaded columns: the
+    transcode must preserve nullness exactly for every CS null
+    representation (replace-value, bitmap, dict null ref, zero-len
+    string)."""
+    rows = 3000
+    rng = np.random.default_rng(101)
+    vals_pos = rng.integers(10, 10**6, rows).astype(np.int64)   # replace=min-1
+    vals_full = rng.integers(-2**63, 2**63 - 1, rows).astype(np.int64)
+    vals_full[0], vals_full[1] = -2**63, 2**63 - 1              # bitmap
+    flag = rng.integers(0, 4, rows).astype(np.int64)            # dict null ref
+    tags = [bytes("N%d" % (f % 4), "ascii") for f in flag]
+    nulls = [sorted(int(x) for x in rng.choice(rows, k, replace=False))
+             for k in (100, 150, 200, 250)]
+    blocks = [cs_enc(rows, [
+        cs_int_col(list(vals_pos), null_rows=nulls[0]),
+        cs_int_col(list(vals_full), null_rows=nulls[1]),
+        cs_int_col(list(flag), null_rows=nulls[2], dict_=True),
+        cs_str_col(tags, null_rows=nulls[3], dict_=True),
+    ])]
+    schema, pax = cs.to_pax_blocks(blocks)
+    bs = _pax_blockset(schema, pax)
+    for c in range(4):
+        for op, expect in ((abi.OP_NU, len(nulls[c])),
+                           (abi.OP_NN, rows - len(nulls[c]))):
+            filt = abi.make_filter([dict(col=c, op=op)])
+            agg = abi.make_agg([], [dict(kind=abi.AGG_COUNT)])
+            res = oracle.scan_filter_agg(bs, filt, agg)
+            assert res.rows_passed == expect, (c, op)
