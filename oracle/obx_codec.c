@@ -503,30 +503,135 @@ static int encode_column(enc_buf *mb, obx_col_header *ch,
       break;
     }
     case OBX_ENC_CONST: {
-      /* ObConstEncoder (ob_const_encoder.cpp:100-330). Round 1: only the
-         no-exception, non-null path (exceptions -> dict path is a next row) */
-      if (null_cnt) return OBX_NOT_SUPPORTED;
-      for (uint32_t r = 1; r < rows; r++) {
+      /* ObConstEncoder (ob_const_encoder.cpp:52-350): the most frequent
+         value (null counts as a candidate) becomes the const; every
+         other row is an exception [ref + row_id] ahead of an embedded
+         dict of the distinct values. Reference AUTO suitability bounds
+         are MAX_EXCEPTION_SIZE=32 / MAX_EXCEPTION_PCT=10
+         (ob_const_encoder.h:48-49); a FORCED const accepts anything the
+         format itself can hold (count <= 255, refs <= 255). */
+      int pure_const = (null_cnt == 0);
+      for (uint32_t r = 1; pure_const && r < rows; r++) {
         if (memcmp(data, data + (size_t)r * len, (size_t)len) != 0)
-          return OBX_NOT_SUPPORTED;
+          pure_const = 0;
       }
+      int all_null = (null_cnt == rows);
+      if (pure_const || all_null) {
+        /* store_meta_without_dict (ob_const_encoder.cpp:147-200):
+           const_ref 1 marks the null-const, 0 a value-const */
+        obx_const_meta cm;
+        memset(&cm, 0, sizeof(cm));
+        cm.version = 0;
+        cm.count = 0;
+        cm.const_ref = all_null ? 1 : 0;
+        cm.offset = (uint16_t)sizeof(cm);
+        int64_t cell = all_null ? 0 : ((sc == OBX_SC_INT) ? tss : len);
+        if (mb->len + (int64_t)sizeof(cm) + cell > mb->cap)
+          return OBX_BUF_NOT_ENOUGH;
+        memcpy(mb->p + mb->len, &cm, sizeof(cm));
+        /* store_value (ob_const_encoder.cpp:200-240): ObIntSC stores
+           type_store_size bytes, others datum bytes */
+        if (cell) memcpy(mb->p + mb->len + sizeof(cm), data, (size_t)cell);
+        mb->len += (int64_t)sizeof(cm) + cell;
+        ch->type = OBX_COL_CONST;
+        ch->attr |= OBX_COL_ATTR_FIX_LENGTH;
+        ch->length = (uint32_t)(sizeof(cm) + cell);
+        break;
+      }
+      /* exception path: dict + [exc refs][exc row_ids] */
+      dict_builder db;
+      int rc = dict_build(&db, data, nulls, rows, len);
+      if (rc) { dict_free(&db); return rc; }
+      dict_sort(&db, len, sc);
+      uint32_t count = db.count;
+      /* choose the const ref: most frequent value; null (ref == count)
+         competes too (ob_const_encoder.cpp:72-94) */
+      uint32_t *freq = (uint32_t *)calloc((size_t)count + 1, 4);
+      if (!freq) { dict_free(&db); return OBX_BUF_NOT_ENOUGH; }
+      for (uint32_t r = 0; r < rows; r++) freq[dict_ref(&db, r)]++;
+      uint32_t const_ref = 0, max_cnt = 0;
+      for (uint32_t i = 0; i <= count; i++)
+        if (freq[i] > max_cnt) { max_cnt = freq[i]; const_ref = i; }
+      free(freq);
+      uint32_t exc = rows - max_cnt;
+      uint64_t max_ref = null_cnt ? count : (count ? count - 1 : 0);
+      uint64_t max_row_id = 0;
+      for (uint32_t r = rows; r-- > 0;) {
+        if (dict_ref(&db, r) != const_ref) { max_row_id = r; break; }
+      }
+      if (exc > 255 || max_ref > 255) {
+        dict_free(&db);
+        return OBX_NOT_SUPPORTED; /* format bound (uint8 fields) */
+      }
+      int64_t row_id_byte = obx_byte_packed_int_size(max_row_id);
       obx_const_meta cm;
       memset(&cm, 0, sizeof(cm));
       cm.version = 0;
-      cm.count = 0;
-      cm.const_ref = 0;
-      cm.offset = (uint16_t)sizeof(cm);
-      int64_t cell = (sc == OBX_SC_INT) ? tss : len;
-      if (mb->len + (int64_t)sizeof(cm) + cell > mb->cap)
+      cm.count = (uint8_t)exc;
+      cm.const_ref = (uint8_t)const_ref;
+      cm.attr = (uint8_t)(row_id_byte & 7);
+      cm.offset = (uint16_t)(sizeof(cm) + exc * (row_id_byte + 1));
+      if (mb->len + (int64_t)cm.offset > mb->cap) {
+        dict_free(&db);
         return OBX_BUF_NOT_ENOUGH;
+      }
       memcpy(mb->p + mb->len, &cm, sizeof(cm));
-      /* store_value (ob_const_encoder.cpp:200-240): ObIntSC stores
-         type_store_size bytes, others datum bytes */
-      memcpy(mb->p + mb->len + sizeof(cm), data, (size_t)cell);
-      mb->len += (int64_t)sizeof(cm) + cell;
+      /* dict_ref_gen before row_id_gen (ob_const_encoder.cpp:321-345) */
+      uint8_t *eref = mb->p + mb->len + sizeof(cm);
+      uint8_t *erid = eref + exc;
+      uint32_t j = 0;
+      for (uint32_t r = 0; r < rows && j < exc; r++) {
+        uint32_t x = dict_ref(&db, r);
+        if (x != const_ref) {
+          eref[j] = (uint8_t)x;
+          uint64_t rr = r;
+          memcpy(erid + (size_t)j * row_id_byte, &rr, (size_t)row_id_byte);
+          j++;
+        }
+      }
+      mb->len += (int64_t)cm.offset;
+      /* embedded dict meta at +offset, like RLE's */
+      int64_t entry_len = len;
+      if (sc == OBX_SC_INT) {
+        uint64_t mx = 0;
+        for (uint32_t i = 0; i < count; i++) {
+          uint64_t v = 0;
+          memcpy(&v, db.entries + (size_t)i * len, (size_t)len);
+          v &= mask;
+          if (v > mx) mx = v;
+        }
+        entry_len = obx_int_size(mx);
+      }
+      obx_dict_meta dm;
+      memset(&dm, 0, sizeof(dm));
+      dm.version = 0;
+      dm.row_ref_size = 0; /* refs live in the exception list */
+      dm.count = count;
+      dm.data_size = (uint16_t)entry_len;
+      dm.attr = OBX_DICT_ATTR_FIX_LENGTH | OBX_DICT_ATTR_IS_SORTED;
+      int64_t dsize = (int64_t)sizeof(dm) + (int64_t)count * entry_len;
+      if (mb->len + dsize > mb->cap) {
+        dict_free(&db);
+        return OBX_BUF_NOT_ENOUGH;
+      }
+      memcpy(mb->p + mb->len, &dm, sizeof(dm));
+      uint8_t *pay = mb->p + mb->len + sizeof(dm);
+      for (uint32_t i = 0; i < count; i++) {
+        if (sc == OBX_SC_INT) {
+          uint64_t v = 0;
+          memcpy(&v, db.entries + (size_t)i * len, (size_t)len);
+          v &= mask;
+          memcpy(pay + (size_t)i * entry_len, &v, (size_t)entry_len);
+        } else {
+          memcpy(pay + (size_t)i * entry_len, db.entries + (size_t)i * len,
+                 (size_t)len);
+        }
+      }
+      mb->len += dsize;
+      dict_free(&db);
       ch->type = OBX_COL_CONST;
       ch->attr |= OBX_COL_ATTR_FIX_LENGTH;
-      ch->length = (uint32_t)(sizeof(cm) + cell);
+      ch->length = (uint32_t)(cm.offset + dsize);
       break;
     }
     case OBX_ENC_INTEGER_BASE_DIFF: {
