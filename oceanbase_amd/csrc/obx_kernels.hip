@@ -856,8 +856,11 @@ extern "C" __global__ void k_lower_leaves(
   blk_leaf o;
   o.mask = 0; o.lo = 0; o.hi = 0; o.mode = OBX_LEAF_VALUE; o.invert = 0;
 
-  const bool is_dict = (c.enc == OBX_D_DICT || c.enc == OBX_D_RLE ||
-                        (c.enc == OBX_D_CONST && c.runs > 0));
+  /* NOT const-with-exceptions: REF_MASK row evaluation reads a per-row
+     ref (packed stream or RLE runs), which CONST does not have — its
+     refs live in the exception list, so it takes the VALUE slow path
+     (col_value2 walks the exceptions). */
+  const bool is_dict = (c.enc == OBX_D_DICT || c.enc == OBX_D_RLE);
   if (is_dict && c.count < 64) {
     if (lf.op == 8) {
       o.mask = 1ull << c.count;
