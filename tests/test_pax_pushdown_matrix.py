@@ -1,0 +1,153 @@
+"""The reference's shared pushdown-filter matrix
+(TestColumnDecoder::basic_filter_pushdown_* in
+unittest/storage/blocksstable/encoding/test_column_decoder.h:696-1050,
+instantiated per decoder in test_general_column_decoder.cpp): ROW_CNT=64
+rows split 24/10/10/10 across four ascending seeds plus 10 trailing
+nulls, every white op with the popcounts the reference asserts — here
+run across EVERY obx PAX encoding of the column (the reference
+instantiates the same matrix for DICT/RLE/INTDIFF/HEX/STRING_DIFF/
+STRING_PREFIX/COLUMN_EQUAL/COLUMN_SUBSTR)."""
+import os
+import sys
+
+import numpy as np
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from oceanbase_amd import abi, oracle  # noqa: E402
+
+from test_cs_pipeline_equiv import _pax_blockset  # noqa: E402
+
+ROW_CNT = 64
+SEG = [24, 10, 10, 10, 10]  # seed0 x24, seed1/2/3 x10, null x10
+
+
+def _counts(op, lo_i, hi_i=None, in_i=None):
+    """expected popcount over the seed segments (segment k holds seed k;
+    the last segment is null and fails every value op)."""
+    n = 0
+    for k, cnt in enumerate(SEG[:4]):
+        if op == abi.OP_EQ:
+            ok = k == lo_i
+        elif op == abi.OP_NE:
+            ok = k != lo_i
+        elif op == abi.OP_LT:
+            ok = k < lo_i
+        elif op == abi.OP_LE:
+            ok = k <= lo_i
+        elif op == abi.OP_GT:
+            ok = k > lo_i
+        elif op == abi.OP_GE:
+            ok = k >= lo_i
+        elif op == abi.OP_BT:
+            ok = lo_i <= k <= hi_i
+        elif op == abi.OP_IN:
+            ok = k in in_i
+        else:
+            ok = False
+        if ok:
+            n += cnt
+    if op == abi.OP_NU:
+        n = SEG[4]
+    if op == abi.OP_NN:
+        n = ROW_CNT - SEG[4]
+    return n
+
+
+def _nulls_tail():
+    nb = np.zeros((ROW_CNT + 7) // 8, dtype=np.uint8)
+    for r in range(ROW_CNT - 10, ROW_CNT):
+        nb[r >> 3] |= 1 << (r & 7)
+    return nb
+
+
+def _int_bs(enc, seeds):
+    vals = np.zeros(ROW_CNT, dtype=np.int64)
+    p = 0
+    for k, cnt in enumerate(SEG[:4]):
+        vals[p:p + cnt] = seeds[k]
+        p += cnt
+    schema = oracle.make_schema([(abi.T_INT, 0, 19, 8)])
+    block = oracle.encode_block(schema, [vals], [enc], [_nulls_tail()])
+    return _pax_blockset(schema, [block])
+
+
+def _char_bs(enc, seeds, ln=4, extra_coleq=False):
+    data = np.zeros(ROW_CNT * ln, dtype=np.uint8)
+    p = 0
+    for k, cnt in enumerate(SEG[:4]):
+        s = seeds[k]
+        for r in range(p, p + cnt):
+            data[r * ln:(r + 1) * ln] = np.frombuffer(s, dtype=np.uint8)
+        p += cnt
+    if not extra_coleq:
+        schema = oracle.make_schema([(abi.T_CHAR, 0, 0, ln)])
+        block = oracle.encode_block(schema, [data], [enc], [_nulls_tail()])
+        return _pax_blockset(schema, [block]), 0
+    # span encodings need a reference column: col0 RAW carries the same
+    # values, col1 is the span-encoded copy the matrix filters on
+    schema = oracle.make_schema([(abi.T_CHAR, 0, 0, ln)] * 2)
+    block = oracle.encode_block(schema, [data.copy(), data.copy()],
+                                [abi.ENC_RAW, enc],
+                                [_nulls_tail(), _nulls_tail()])
+    return _pax_blockset(schema, [block]), 1
+
+
+def _run_matrix(bs, col, keys):
+    """keys: the four seed values as leaf operands (int64)."""
+    def count(op, lo=0, hi=0, il=None):
+        leaf = dict(col=col, op=op, lo=lo, hi=hi)
+        if il is not None:
+            leaf["in_list"] = il
+        res = oracle.scan_filter_agg(
+            bs, abi.make_filter([leaf]),
+            abi.make_agg([], [dict(kind=abi.AGG_COUNT)]))
+        return res.rows_passed
+
+    s5 = keys[3] + 12345  # a value not present (seed5 analogue)
+    assert count(abi.OP_EQ, keys[0]) == _counts(abi.OP_EQ, 0)
+    assert count(abi.OP_EQ, keys[1]) == _counts(abi.OP_EQ, 1)
+    assert count(abi.OP_NE, keys[0]) == _counts(abi.OP_NE, 0)
+    assert count(abi.OP_NU) == _counts(abi.OP_NU, 0)
+    assert count(abi.OP_NN) == _counts(abi.OP_NN, 0)
+    assert count(abi.OP_LT, keys[1]) == _counts(abi.OP_LT, 1)
+    assert count(abi.OP_LE, keys[1]) == _counts(abi.OP_LE, 1)
+    assert count(abi.OP_GT, keys[1]) == _counts(abi.OP_GT, 1)
+    assert count(abi.OP_GE, keys[1]) == _counts(abi.OP_GE, 1)
+    assert count(abi.OP_BT, keys[1], keys[2]) == _counts(abi.OP_BT, 1, 2)
+    assert count(abi.OP_BT, keys[0], keys[3]) == _counts(abi.OP_BT, 0, 3)
+    assert count(abi.OP_IN, il=[keys[1], keys[2], s5]) == \
+        _counts(abi.OP_IN, 0, in_i={1, 2})
+    assert count(abi.OP_IN, il=[s5]) == 0
+
+
+INT_SEEDS = [10000, 10001, 10002, 10003]
+
+
+@pytest.mark.parametrize("enc", [abi.ENC_RAW, abi.ENC_DICT, abi.ENC_RLE,
+                                 abi.ENC_INT_DIFF, abi.ENC_CONST])
+def test_int_matrix(enc):
+    bs = _int_bs(enc, INT_SEEDS)
+    _run_matrix(bs, 0, INT_SEEDS)
+
+
+CHAR_SEEDS = [b"s100", b"s101", b"s102", b"s103"]
+CHAR_KEYS = [int.from_bytes(s, "little") for s in CHAR_SEEDS]
+
+
+@pytest.mark.parametrize("enc", [abi.ENC_RAW, abi.ENC_DICT, abi.ENC_RLE,
+                                 abi.ENC_HEX, abi.ENC_SDIFF,
+                                 abi.ENC_STRING_PREFIX])
+def test_char_matrix(enc):
+    bs, col = _char_bs(enc, CHAR_SEEDS)
+    _run_matrix(bs, col, CHAR_KEYS)
+
+
+@pytest.mark.parametrize("enc", [abi.ENC_COLUMN_EQUAL,
+                                 abi.ENC_COLUMN_SUBSTR])
+def test_span_matrix(enc):
+    """TestColumnEqualMicroDecoder / TestInterColumnSubstringDecoder
+    instantiations: the matrix on a span-encoded column."""
+    bs, col = _char_bs(enc, CHAR_SEEDS, extra_coleq=True)
+    _run_matrix(bs, col, CHAR_KEYS)
