@@ -151,3 +151,57 @@ def test_span_matrix(enc):
     instantiations: the matrix on a span-encoded column."""
     bs, col = _char_bs(enc, CHAR_SEEDS, extra_coleq=True)
     _run_matrix(bs, col, CHAR_KEYS)
+
+
+# ---- the same matrix over CS-format columns (recreating the shapes of
+# unittest/storage/blocksstable/cs_encoding/test_integer_pd_filter.cpp,
+# test_int_dict_pd_filter.cpp, test_str_dict_pd_filter.cpp,
+# test_string_pd_filter.cpp through the transformer-role load path) ----
+
+from oceanbase_amd import cs  # noqa: E402
+from test_cs_block import (  # noqa: E402
+    _enc as cs_enc, _int_col as cs_int_col, _str_col as cs_str_col,
+)
+
+
+def _cs_int_bs(seeds, dict_=False, enc=0):
+    vals, p = [0] * ROW_CNT, 0
+    for k, cnt in enumerate(SEG[:4]):
+        for r in range(p, p + cnt):
+            vals[r] = seeds[k]
+        p += cnt
+    nulls = list(range(ROW_CNT - 10, ROW_CNT))
+    block = cs_enc(ROW_CNT, [cs_int_col(vals, null_rows=nulls,
+                                        dict_=dict_, enc=enc)])
+    schema, pax = cs.to_pax_blocks([block])
+    return _pax_blockset(schema, pax)
+
+
+def _cs_str_bs(seeds, dict_=False):
+    strs, p = [b""] * ROW_CNT, 0
+    for k, cnt in enumerate(SEG[:4]):
+        for r in range(p, p + cnt):
+            strs[r] = seeds[k]
+        p += cnt
+    nulls = list(range(ROW_CNT - 10, ROW_CNT))
+    for r in nulls:
+        strs[r] = seeds[0]  # placeholder; bytes skipped for null rows
+    block = cs_enc(ROW_CNT, [cs_str_col(strs, null_rows=nulls,
+                                        dict_=dict_)])
+    schema, pax = cs.to_pax_blocks([block])
+    return _pax_blockset(schema, pax)
+
+
+@pytest.mark.parametrize("dict_,enc", [(False, 0), (False, 6), (True, 0)])
+def test_cs_integer_matrix(dict_, enc):
+    """test_integer_pd_filter / test_int_dict_pd_filter shapes."""
+    bs = _cs_int_bs(INT_SEEDS, dict_=dict_, enc=enc)
+    _run_matrix(bs, 0, INT_SEEDS)
+
+
+@pytest.mark.parametrize("dict_", [False, True])
+def test_cs_string_matrix(dict_):
+    """test_string_pd_filter / test_str_dict_pd_filter shapes (fixed-len
+    strings; the load path maps them to char(N) PAX columns)."""
+    bs = _cs_str_bs(CHAR_SEEDS, dict_=dict_)
+    _run_matrix(bs, 0, CHAR_KEYS)
