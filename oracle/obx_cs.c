@@ -121,6 +121,16 @@ int64_t obx_cs_int_stream_enc3(const int64_t *vals, const uint8_t *nulls,
                                uint32_t rows, uint8_t enc_type,
                                int use_null_replace, int64_t null_replaced,
                                uint8_t *buf, size_t cap) {
+  return obx_cs_int_stream_enc4(vals, nulls, rows, enc_type,
+                                use_null_replace, null_replaced, 0, buf,
+                                cap);
+}
+
+int64_t obx_cs_int_stream_enc4(const int64_t *vals, const uint8_t *nulls,
+                               uint32_t rows, uint8_t enc_type,
+                               int use_null_replace, int64_t null_replaced,
+                               uint32_t precision_width_size,
+                               uint8_t *buf, size_t cap) {
   if (!vals || !rows || !buf) return -1;
   /* build_signed_stream_meta (ob_stream_encoding_struct.cpp:118-166):
      base ONLY when min < 0 (range = max - min); for min >= 0 the width
@@ -159,6 +169,13 @@ int64_t obx_cs_int_stream_enc3(const int64_t *vals, const uint8_t *nulls,
   if (use_null_replace) {
     m.attr |= OBX_CS_REPLACE_NULL;
     m.null_replaced = (uint64_t)null_replaced;
+  }
+  if (precision_width_size) {
+    /* set_precision_width_size: tag of the decimal datum width */
+    m.attr |= OBX_CS_DECIMAL_INT;
+    m.precision_width_tag = precision_width_size <= 1 ? 0
+                            : precision_width_size <= 2 ? 1
+                            : precision_width_size <= 4 ? 2 : 3;
   }
   m.width_tag = range <= 0xFF ? 0 : range <= 0xFFFF ? 1
                 : range <= 0xFFFFFFFFull ? 2 : 3;

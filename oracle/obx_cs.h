@@ -92,6 +92,16 @@ int64_t obx_cs_int_stream_enc3(const int64_t *vals, const uint8_t *nulls,
                                int use_null_replace, int64_t null_replaced,
                                uint8_t *buf, size_t cap);
 
+/* enc3 plus the DECIMAL_INT attribute: precision_width_size in bytes
+ * (4 or 8, 0 = not decimal) serialized as the precision width tag
+ * (ObIntegerStreamMeta::set_precision_width_size,
+ * ob_stream_encoding_struct.h) */
+int64_t obx_cs_int_stream_enc4(const int64_t *vals, const uint8_t *nulls,
+                               uint32_t rows, uint8_t enc_type,
+                               int use_null_replace, int64_t null_replaced,
+                               uint32_t precision_width_size,
+                               uint8_t *buf, size_t cap);
+
 /* Decode [meta][stream] back to int64 values (base re-applied). null
  * positions decode to the stored replace value. Returns consumed bytes
  * or -1. */

@@ -596,3 +596,30 @@ def test_golden_stream_bytes():
             buf, n, rows, out.ctypes.data_as(C.POINTER(C.c_int64)),
             None) == n
         assert (out == v).all()
+
+
+def test_decimal_int_attr_meta():
+    """DECIMAL_INT attribute (ObIntegerColumnEncoder for ObDecimalIntSC:
+    precision_width_size 4 or 8 serialized as a width tag between the
+    conditional vi64 fields and the pfor byte)."""
+    _lib.obx_cs_int_stream_enc4.restype = C.c_int64
+    _lib.obx_cs_int_stream_enc4.argtypes = [
+        C.POINTER(C.c_int64), C.POINTER(C.c_uint8), C.c_uint32, C.c_uint8,
+        C.c_int, C.c_int64, C.c_uint32, C.POINTER(C.c_uint8), C.c_size_t]
+    vals = np.array([1550, 2025, 99], dtype=np.int64)  # cents, prec<=9
+    cap = 256
+    buf = (C.c_uint8 * cap)()
+    n = _lib.obx_cs_int_stream_enc4(
+        vals.ctypes.data_as(C.POINTER(C.c_int64)), None, 3, 1, 0, 0, 8,
+        buf, cap)
+    assert n > 0
+    blob = bytes(buf[:n])
+    # [version=1][attr=DECIMAL_INT=4][type=RAW=1][width_tag=1 (max 2025)]
+    # [precision_width_tag=3 (8 bytes)][pfor=0] + 3 x 2-byte values
+    assert blob[:4] == bytes([1, 4, 1, 1])
+    assert blob[4] == 3 and blob[5] == 0
+    assert len(blob) == 6 + 6
+    out = np.zeros(3, dtype=np.int64)
+    m = _lib.obx_cs_int_stream_dec(
+        buf, n, 3, out.ctypes.data_as(C.POINTER(C.c_int64)), None)
+    assert m == n and list(out) == [1550, 2025, 99]
