@@ -281,8 +281,28 @@ static int encode_column(enc_buf *mb, obx_col_header *ch,
       uint32_t x = db.ref_of_row[r];
       if (x != prev) { runs++; prev = x; }
     }
+    /* const suitability (ObConstEncoder::traverse, ob_const_encoder.h:
+       48-49): exceptions vs the dominant value (null competes) bounded
+       by MAX_EXCEPTION_SIZE=32 and MAX_EXCEPTION_PCT=10 */
+    uint32_t cmax = null_cnt;
+    {
+      uint32_t *freq = (uint32_t *)calloc((size_t)nd + 1, 4);
+      if (freq) {
+        for (uint32_t r = 0; r < rows; r++) {
+          uint32_t x = db.ref_of_row[r]; /* UINT32_MAX sentinel = null */
+          if (x <= nd) freq[x]++;
+        }
+        for (uint32_t i = 0; i <= nd; i++)
+          if (freq[i] > cmax) cmax = freq[i];
+        free(freq);
+      }
+    }
+    uint32_t cexc = rows - cmax;
+    uint32_t cpct = rows / 10 > 1 ? rows / 10 : 1;
     dict_free(&db);
     if (nd == 1 && null_cnt == 0) enc = OBX_ENC_CONST;
+    else if (cexc > 0 && cexc <= 32 && cexc < cpct && nd <= 255)
+      enc = OBX_ENC_CONST;
     else if (runs * 8 <= rows) enc = OBX_ENC_RLE;
     else if (nd <= 64 && nd * 4 <= rows) enc = OBX_ENC_DICT;
     else if (sc == OBX_SC_INT && null_cnt < rows) enc = OBX_ENC_INTEGER_BASE_DIFF;
