@@ -205,3 +205,62 @@ def test_cs_string_matrix(dict_):
     strings; the load path maps them to char(N) PAX columns)."""
     bs = _cs_str_bs(CHAR_SEEDS, dict_=dict_)
     _run_matrix(bs, 0, CHAR_KEYS)
+
+
+def test_exceed_range_compare_filter():
+    """TestIntegerPdFilter::test_exceed_range_compare_filter: operands
+    outside the column's stored domain must degenerate to all/none, not
+    wrap (the packed-domain lowering biases signed domains)."""
+    vals = np.array([100, 200, 300, 150, 250] * 12 + [175, 225, 275, 125],
+                    dtype=np.int64)
+    rows = len(vals)
+    schema = oracle.make_schema([(abi.T_INT, 0, 19, 8)])
+    for enc in (abi.ENC_RAW, abi.ENC_DICT, abi.ENC_INT_DIFF):
+        bs = _pax_blockset(schema, [oracle.encode_block(
+            schema, [vals.copy()], [enc], None)])
+        def cnt(op, lo=0, hi=0):
+            return oracle.scan_filter_agg(
+                bs, abi.make_filter([dict(col=0, op=op, lo=lo, hi=hi)]),
+                abi.make_agg([], [dict(kind=abi.AGG_COUNT)])).rows_passed
+        assert cnt(abi.OP_GT, 2**63 - 1) == 0
+        assert cnt(abi.OP_LT, -2**63) == 0
+        assert cnt(abi.OP_GE, -2**63) == rows
+        assert cnt(abi.OP_LE, 2**63 - 1) == rows
+        assert cnt(abi.OP_BT, -2**63, 2**63 - 1) == rows
+        assert cnt(abi.OP_BT, 10**15, 10**16) == 0
+        assert cnt(abi.OP_EQ, 10**12) == 0
+        assert cnt(abi.OP_NE, 10**12) == rows
+
+
+def test_all_null_integer_through_cs_load():
+    """TestIntegerPdFilter::test_all_null_integer_decoder through the
+    CS load path: an all-null CS integer column must stay all-null
+    after the transform and fail every value filter."""
+    rows = 200
+    block = cs_enc(rows, [cs_int_col([0] * rows,
+                                     null_rows=list(range(rows)))])
+    schema, pax = cs.to_pax_blocks([block])
+    bs = _pax_blockset(schema, pax)
+    def cnt(op, lo=0):
+        return oracle.scan_filter_agg(
+            bs, abi.make_filter([dict(col=0, op=op, lo=lo)]),
+            abi.make_agg([], [dict(kind=abi.AGG_COUNT)])).rows_passed
+    assert cnt(abi.OP_NU) == rows
+    assert cnt(abi.OP_NN) == 0
+    assert cnt(abi.OP_EQ, 0) == 0
+    assert cnt(abi.OP_GE, -2**62) == 0
+
+
+def test_integer_in_at_capacity():
+    """TestIntegerPdFilter::test_integer_many_in_values analogue at the
+    engine's OBX_MAX_IN_LIST=8 bound (larger lists are the reference
+    adapter's fallback; include/obx.h:149)."""
+    vals = np.arange(64, dtype=np.int64) % 16
+    schema = oracle.make_schema([(abi.T_INT, 0, 19, 8)])
+    bs = _pax_blockset(schema, [oracle.encode_block(
+        schema, [vals], [abi.ENC_RAW], None)])
+    il = [0, 2, 4, 6, 8, 10, 12, 14]  # exactly 8 entries
+    res = oracle.scan_filter_agg(
+        bs, abi.make_filter([dict(col=0, op=abi.OP_IN, in_list=il)]),
+        abi.make_agg([], [dict(kind=abi.AGG_COUNT)]))
+    assert res.rows_passed == sum(1 for v in vals if int(v) in set(il))
