@@ -698,6 +698,13 @@ static void dzp_unpack(const uint8_t *in, uint32_t n, uint32_t b,
   }
 }
 
+/* bounds guard for truncated/hostile streams: room for n b-bit values */
+static inline int dzp_room(const uint8_t *ip, const uint8_t *end,
+                           uint32_t n, uint32_t b) {
+  return b <= 64 && ip <= end &&
+         (size_t)(end - ip) >= ((size_t)n * b + 7) / 8;
+}
+
 /* find_most_fit_bx (ob_simd_fixed_pfor.h:30-87) */
 static void dzp_fit(const uint64_t *v, uint32_t n, uint32_t wbits,
                     uint32_t *b_out, uint32_t *bx_out) {
@@ -849,6 +856,7 @@ static const uint8_t *dzp_frame_dec(const uint8_t *ip, const uint8_t *end,
     bx = *ip++;
   }
   if (bx == 0) {
+    if (!dzp_room(ip, end, DZP_BLOCK, b)) return NULL;
     dzp_unpack(ip, DZP_BLOCK, b, zz);
     ip += ((size_t)DZP_BLOCK * b + 7) / 8;
   } else {
@@ -858,8 +866,10 @@ static const uint8_t *dzp_frame_dec(const uint8_t *ip, const uint8_t *end,
     ip += 16;
     uint32_t xn = (uint32_t)(__builtin_popcountll(xmap[0]) +
                              __builtin_popcountll(xmap[1]));
+    if (!dzp_room(ip, end, xn, bx)) return NULL;
     dzp_unpack(ip, xn, bx, exc);
     ip += ((size_t)xn * bx + 7) / 8;
+    if (!dzp_room(ip, end, DZP_BLOCK, b)) return NULL;
     dzp_unpack(ip, DZP_BLOCK, b, zz);
     ip += ((size_t)DZP_BLOCK * b + 7) / 8;
     uint32_t xi = 0;
@@ -964,6 +974,7 @@ int64_t obx_cs_xpfor_dec(const uint8_t *in, size_t in_len, uint32_t count,
     if (ip + 2 > end) return -1;
     uint32_t sh = *ip++;
     uint32_t b = *ip++;
+    if (!dzp_room(ip, end, rem, b)) return -1;
     dzp_unpack(ip, rem, b, v);
     ip += ((size_t)rem * b + 7) / 8;
     for (uint32_t i = 0; i < rem; i++) {
@@ -1006,6 +1017,7 @@ static int64_t dzp_dec_core(const uint8_t *in, size_t in_len,
       bx = *ip++;
     }
     if (bx == 0) {
+      if (!dzp_room(ip, end, DZP_BLOCK, b)) return -1;
       dzp_unpack(ip, DZP_BLOCK, b, zz);
       ip += ((size_t)DZP_BLOCK * b + 7) / 8;
     } else {
@@ -1015,8 +1027,10 @@ static int64_t dzp_dec_core(const uint8_t *in, size_t in_len,
       ip += 16;
       uint32_t xn = (uint32_t)(__builtin_popcountll(xmap[0]) +
                                __builtin_popcountll(xmap[1]));
+      if (!dzp_room(ip, end, xn, bx)) return -1;
       dzp_unpack(ip, xn, bx, exc);
       ip += ((size_t)xn * bx + 7) / 8;
+      if (!dzp_room(ip, end, DZP_BLOCK, b)) return -1;
       dzp_unpack(ip, DZP_BLOCK, b, zz);
       ip += ((size_t)DZP_BLOCK * b + 7) / 8;
       uint32_t xi = 0;
@@ -1040,6 +1054,7 @@ static int64_t dzp_dec_core(const uint8_t *in, size_t in_len,
     uint32_t rem = count - done;
     if (ip >= end) return -1;
     uint32_t b = *ip++;
+    if (!dzp_room(ip, end, rem, b)) return -1;
     dzp_unpack(ip, rem, b, zz);
     ip += ((size_t)rem * b + 7) / 8;
     for (uint32_t i = 0; i < rem; i++) {
