@@ -567,3 +567,42 @@ def test_mixed_block_with_dict_columns():
                if r not in snulls)
     pout, pn = _get_int(v, 2)
     assert list(pout) == plain and pn == set()
+
+
+def test_truncated_block_never_crashes():
+    """every truncated prefix of a valid block must be rejected or
+    decoded within bounds (the PFoR decoders carry explicit room guards
+    for hostile streams; regression for the ASAN truncation fuzz)."""
+    rows = 300
+    rng = np.random.default_rng(31)
+    blob = _enc(rows, [
+        _int_col([int(x) for x in rng.integers(-10**6, 10**6, rows)],
+                 enc=5),
+        _int_col([int(x) for x in np.sort(rng.integers(0, 10**9, rows))],
+                 dict_=True),
+        _str_col([bytes("w%02d" % (i % 40), "ascii") for i in range(rows)],
+                 dict_=True),
+    ])
+    out = np.zeros(rows, dtype=np.int64)
+    nb = np.zeros((rows + 7) // 8, dtype=np.uint8)
+    bout = (C.c_uint8 * (1 << 20))()
+    lens = np.zeros(rows, dtype=np.uint32)
+    step = 1
+    for cut in range(0, len(blob) + 1, step):
+        if cut > 128:
+            step = 13
+        buf = (C.c_uint8 * max(cut, 1)).from_buffer_copy(
+            blob[:cut] or b"\x00")
+        v = BlockView()
+        if _lib.obx_cs_block_dec(buf, cut, C.byref(v)) == 0:
+            _lib.obx_cs_block_get_int(
+                C.byref(v), 0, out.ctypes.data_as(C.POINTER(C.c_int64)),
+                nb.ctypes.data_as(C.POINTER(C.c_uint8)))
+            _lib.obx_cs_block_get_str(
+                C.byref(v), 2, bout, 1 << 20,
+                lens.ctypes.data_as(C.POINTER(C.c_uint32)),
+                nb.ctypes.data_as(C.POINTER(C.c_uint8)))
+    # the full blob still decodes correctly
+    buf = (C.c_uint8 * len(blob)).from_buffer_copy(blob)
+    v = BlockView()
+    assert _lib.obx_cs_block_dec(buf, len(blob), C.byref(v)) == 0
