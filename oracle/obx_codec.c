@@ -1593,10 +1593,21 @@ int obx_decode_block(const obx_col_schema *cols, uint16_t n_cols,
                      const uint16_t *proj_cols, uint16_t n_proj,
                      uint8_t *const *out_cols, uint8_t *const *out_nulls,
                      uint32_t *row_count) {
-  (void)block_len;
+  /* length/shape guard: header + column headers must fit. The column
+     PAYLOADS are trusted after this point — in the reference the block
+     checksum (ObMicroBlockHeader::data_checksum_, verified before any
+     decode) guarantees interior integrity, and this oracle decodes only
+     blocks its own writer produced; full hostile-payload bounds checking
+     is deliberately out of scope (the CS stream layer, which the column
+     layer cross-checks by stream offsets, does carry room guards). */
+  if (block_len < (int64_t)sizeof(obx_micro_header))
+    return OBX_INVALID_ARGUMENT;
   const obx_micro_header *h = (const obx_micro_header *)block;
   if (h->magic != OBX_MICRO_BLOCK_MAGIC) return OBX_INVALID_ARGUMENT;
   if (h->column_count != n_cols) return OBX_INVALID_ARGUMENT;
+  if ((int64_t)h->header_size +
+          (int64_t)n_cols * (int64_t)sizeof(obx_col_header) > block_len)
+    return OBX_INVALID_ARGUMENT;
   const uint8_t *meta_region = block + h->header_size +
                                (int64_t)n_cols * sizeof(obx_col_header);
   const obx_col_header *chp =
