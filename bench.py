@@ -85,11 +85,19 @@ def result_to_bytes(res):
     return bytes(res)  # ctypes struct -> raw bytes
 
 
-def merge_results(blobs, n_aggs):
+_SUM_FAMILY = {abi.AGG_COUNT, abi.AGG_SUM, abi.AGG_SUM_PROD2,
+               abi.AGG_SUM_PROD3, abi.AGG_SUM_MUL}
+
+
+def merge_results(blobs, n_aggs, kinds=None):
     """Merge serialized AggResults (Python ints: exact 256-bit adds).
 
-    Sum-family aggregates only (COUNT/SUM/PROD sums — every bench
-    workload's shape); MIN/MAX cells would need min/max folds here."""
+    Cells are added for sum-family aggregates and min/max-folded for
+    MIN/MAX; `kinds` (list of abi.AGG_* per aggregate) must be passed
+    when any non-sum aggregate is present — without it, a MIN/MAX cell
+    reaching this merge fails loudly instead of being summed wrong."""
+    if kinds is not None:
+        assert len(kinds) == n_aggs
     groups = {}
     scanned = passed = 0
     for blob in blobs:
@@ -97,11 +105,84 @@ def merge_results(blobs, n_aggs):
         scanned += r.rows_scanned
         passed += r.rows_passed
         for key, cnt, cells in abi.result_rows(r, n_aggs):
-            g = groups.setdefault(key, [0] + [0] * n_aggs)
+            g = groups.setdefault(key, [0] + [None] * n_aggs)
             g[0] += cnt
             for a in range(n_aggs):
-                g[a + 1] += cells[a]
+                kind = kinds[a] if kinds is not None else abi.AGG_SUM
+                if kind in _SUM_FAMILY:
+                    g[a + 1] = (g[a + 1] or 0) + cells[a]
+                elif kind == abi.AGG_MIN:
+                    g[a + 1] = cells[a] if g[a + 1] is None \
+                        else min(g[a + 1], cells[a])
+                elif kind == abi.AGG_MAX:
+                    g[a + 1] = cells[a] if g[a + 1] is None \
+                        else max(g[a + 1], cells[a])
+                else:
+                    raise AssertionError(
+                        f"aggregate kind {kind} has no distributed merge")
     return dict(sorted(groups.items())), scanned, passed
+
+
+def measure_traffic(args):
+    """HBM traffic of the dominant kernel, per launch, from TCC fabric
+    counters: re-runs this bench (3 launches) under `rocprofv3 --pmc`
+    in a subprocess and parses the per-dispatch counter CSV. Read bytes
+    = RDREQ x 64 B x 2 (the gfx950 half-count of wide coalesced reads,
+    MI355X_MICROARCH.md §HBM, calibrated on filter-int64 in round 1);
+    write bytes = WRREQ x 64 B (uncalibrated). Returns a dict or None."""
+    import csv
+    import glob
+    import shutil
+    import subprocess
+    import tempfile
+    rocprof = shutil.which("rocprofv3")
+    if rocprof is None:
+        return None
+    me = os.path.abspath(__file__)
+    with tempfile.TemporaryDirectory(prefix="obx_pmc_") as td:
+        cmd = [rocprof, "--pmc", "TCC_EA0_RDREQ", "TCC_EA0_WRREQ",
+               "-d", td, "--", sys.executable, me,
+               "--workload", args.workload, "--rows", str(args.rows or 0),
+               "--block-bytes", str(args.block_bytes),
+               "--seed", str(args.seed), "--steps", "2", "--warmup", "1",
+               "--no-cpu-baseline", "--no-traffic"]
+        env = dict(os.environ, TMPDIR=td)
+        try:
+            subprocess.run(cmd, cwd=td, env=env, check=True,
+                           stdout=subprocess.DEVNULL,
+                           stderr=subprocess.DEVNULL, timeout=600)
+        except Exception:
+            return None
+        per_kernel = {}  # name -> {counter: total, "n": dispatches}
+        for f in glob.glob(os.path.join(td, "**", "*counter_collection.csv"),
+                           recursive=True):
+            with open(f, newline="") as fh:
+                for row in csv.DictReader(fh):
+                    name = row.get("Kernel_Name", "")
+                    cname = row.get("Counter_Name", "")
+                    try:
+                        val = float(row.get("Counter_Value", "0"))
+                    except ValueError:
+                        continue
+                    k = per_kernel.setdefault(name, {})
+                    k[cname] = k.get(cname, 0.0) + val
+                    if cname == "TCC_EA0_RDREQ":
+                        k["n"] = k.get("n", 0) + 1
+        main_k = None
+        for name, k in per_kernel.items():
+            if not name.startswith("k_"):
+                continue
+            if main_k is None or k.get("TCC_EA0_RDREQ", 0) > \
+                    per_kernel[main_k].get("TCC_EA0_RDREQ", 0):
+                main_k = name
+        if main_k is None or per_kernel[main_k].get("n", 0) == 0:
+            return None
+        k = per_kernel[main_k]
+        n = k["n"]
+        return dict(kernel=main_k.split("(")[0],
+                    read_bytes=k.get("TCC_EA0_RDREQ", 0) * 64 * 2 / n,
+                    write_bytes=k.get("TCC_EA0_WRREQ", 0) * 64 / n,
+                    launches=n)
 
 
 def main():
@@ -115,6 +196,8 @@ def main():
     ap.add_argument("--block-bytes", type=int, default=16384)
     ap.add_argument("--seed", type=int, default=42)
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--no-traffic", action="store_true",
+                    help="skip the rocprofv3 TCC-traffic leg")
     ap.add_argument("--trace", action="store_true",
                     help="per-step device-time table (prep/main kernel) — "
                          "the per-operator monitor rebuild (SURVEY §5)")
@@ -164,8 +247,10 @@ def main():
                     .cuda(local_rank)
                 outs = [torch.empty_like(t) for _ in range(world)]
                 dist.all_gather(outs, t)
+                kinds = [agg.aggs[i].kind for i in range(n_aggs)]
                 merged = merge_results(
-                    [bytes(o.cpu().numpy().tobytes()) for o in outs], n_aggs)
+                    [bytes(o.cpu().numpy().tobytes()) for o in outs], n_aggs,
+                    kinds)
                 return merged
             return res
         else:
@@ -181,6 +266,7 @@ def main():
     prep_ms = []
     for _ in range(args.warmup):
         one_step()
+    jit = eng.last_jit() if agg is not None else False
     barrier_sync()
     t_start = time.time()
     for _ in range(args.steps):
@@ -221,32 +307,44 @@ def main():
     roofline["achieved"] = achieved / 1e9
     roofline["peak"] = HBM_PEAK_BYTES / 1e9
 
-    # --- CPU baseline (oracle on host cores; bounded sample; rank0 N=1) ----
+    # --- measured HBM traffic (TCC fabric counters, separate rocprofv3
+    #     child run of the same workload; per launch of the main kernel) ----
+    if rank == 0 and world <= 1 and not args.no_traffic \
+            and os.environ.get("OBX_BENCH_TRAFFIC", "1") != "0":
+        tr = measure_traffic(args)
+        if tr is not None:
+            roofline["traffic"] = tr["read_bytes"] + tr["write_bytes"]
+            roofline["traffic_detail"] = dict(
+                kernel=tr["kernel"],
+                read_bytes=tr["read_bytes"], write_bytes=tr["write_bytes"],
+                note="RDREQ*64B*2 (gfx950 wide-read half-count correction, "
+                     "calibrated r01) + WRREQ*64B, per launch")
+
+    # --- CPU baseline (oracle on host cores; fixed sample, 3 reps; rank0
+    #     N=1). The oracle is -O3 -march=native auto-vectorized C, not the
+    #     reference's hand-written AVX512 — stated in `sample`. ------------
     cpu_baseline = None
     if rank == 0 and world <= 1 and not args.no_cpu_baseline:
         cores = os.cpu_count() or 1
-        # sample sized for ~10-30 s of CPU work: start with ~4M rows probe
         rpb = max(li.total_rows // li.n_blocks, 1)
-        probe_blocks = max(int(4_000_000 // rpb), 1)
-        probe_blocks = min(probe_blocks, li.n_blocks)
-        sub = li.subset(probe_blocks)
-        t0 = time.time()
-        r = oracle.scan_filter_agg(sub, filt, agg, nthreads=0)
-        dt = time.time() - t0
-        probe_rows = r.rows_scanned
-        rate = probe_rows / dt
-        target_rows = rate * 15.0  # ~15 s sample
-        nblk = min(int(target_rows // rpb) + 1, li.n_blocks)
-        if nblk > probe_blocks * 2:
-            sub = li.subset(nblk)
+        nblk = min(max(int(150_000_000 // rpb), 1), li.n_blocks)
+        sub = li.subset(nblk)
+        rates = []
+        srows = 0
+        for _ in range(3):
             t0 = time.time()
             r = oracle.scan_filter_agg(sub, filt, agg, nthreads=0)
             dt = time.time() - t0
+            srows = r.rows_scanned
+            rates.append(srows / dt)
+        rates.sort()
+        spread = (rates[-1] - rates[0]) / rates[1] if rates[1] else 0.0
         cpu_baseline = dict(
-            value=r.rows_scanned / dt, unit="rows/s", cores=cores,
+            value=rates[1], unit="rows/s", cores=cores,
             kind="port",
-            sample=f"{r.rows_scanned} rows ({dt:.1f}s) of the same workload "
-                   f"via the oracle's threaded CPU path")
+            sample=f"fixed {srows}-row sample, median of 3 reps "
+                   f"(spread {spread:.1%}); oracle = auto-vectorized C "
+                   f"(-O3 -march=native), threaded")
 
     if rank == 0:
         line = dict(
@@ -263,6 +361,7 @@ def main():
             vs_baseline=None,  # no published per-path number
             dtype="int64",
             data="synthetic",
+            jit=jit,
             config=dict(workload=wl["name"],
                         rows_per_gpu=rows_per_gpu,
                         total_rows=total_rows,

@@ -57,6 +57,7 @@ enum {
   OBX_BUF_NOT_ENOUGH = -4009,
   OBX_NO_GPU = -7001,            /* product path refused: HIP device required */
   OBX_INTERNAL_ERROR = -4016,
+  OBX_PHYSIC_CHECKSUM_ERROR = -4108, /* OB_PHYSIC_CHECKSUM_ERROR */
 };
 
 /* ---- column encodings (ObColumnHeader::Type,
@@ -234,6 +235,12 @@ int64_t obx_encode_block(const obx_col_schema *cols, uint16_t n_cols,
                          const uint8_t *const *null_bitmaps,
                          uint32_t row_count, const uint8_t *enc_request,
                          uint8_t *out, int64_t out_cap);
+
+/* Payload checksum restating ob_crc64_sse42 semantics (CRC-32C in a u64
+ * accumulator — deps/oblib/src/lib/checksum/ob_crc64.cpp:448): the value of
+ * ObMicroBlockHeader::data_checksum_ over the bytes after the micro header.
+ * Exported by BOTH libraries (independent implementations). */
+uint64_t obx_crc32c(const uint8_t *buf, int64_t len);
 
 /* --- microblock reader (ObMicroBlockDecoder restatement) ----------------- */
 /* Decode whole block: for each requested column, write row_count datums of

@@ -137,34 +137,51 @@ def decode_block(block):
                 vals.append(data[pos:pos + int(lens[r])])
                 pos += int(lens[r])
             nulls = np.unpackbits(nb, bitorder="little")[:rows].astype(bool)
+            # fixed length comes from the stream metadata when declared
+            # (ObStringStreamMeta fixed_str_len / STR_FIXED_LEN attr) so
+            # all-null or all-empty blocks and per-block value skew cannot
+            # change the inferred schema; data inference is the fallback
+            # for var-stored columns that happen to be uniform.
             fl = None
-            nn = [len(s) for r, s in enumerate(vals) if not nulls[r]]
-            if nn and all(l == nn[0] for l in nn):
-                fl = nn[0]
+            if v.col[c].sm_attr & STR_FIXED_LEN and \
+                    v.col[c].sm_fixed_str_len > 0:
+                fl = int(v.col[c].sm_fixed_str_len)
+            else:
+                nn = [len(s) for r, s in enumerate(vals) if not nulls[r]]
+                if nn and all(l == nn[0] for l in nn):
+                    fl = nn[0]
             out.append(dict(kind="str", values=vals, nulls=nulls,
                             fixed_len=fl))
     return rows, out
 
 
-def to_pax_blocks(cs_blocks):
+def to_pax_blocks(cs_blocks, declared_specs=None):
     """Transcode CS blocks (one logical table, identical schemas) into
     (schema, [pax_block_bytes]) ready for the engine's blockset loader.
     The transform is the load-time step; the scan itself runs on the
-    engine's native PAX kernels."""
+    engine's native PAX kernels.
+
+    declared_specs: optional schema-level declaration [(obj_type, scale,
+    precision, len), ...] used for columns whose block carries no usable
+    length (e.g. an all-null string block with no STR_FIXED_LEN attr)."""
     specs = None
     pax = []
     for block in cs_blocks:
         rows, cols = decode_block(block)
         bspecs = []
-        for col in cols:
+        for i, col in enumerate(cols):
             if col["kind"] == "int":
                 bspecs.append((abi.T_INT, 0, 19, 8))
             else:
-                if col["fixed_len"] is None:
+                fl = col["fixed_len"]
+                if fl is None and declared_specs is not None:
+                    fl = declared_specs[i][3]
+                if fl is None:
                     raise ValueError(
                         "var-length string column has no fixed-cell PAX "
-                        "equivalent (round-1 engine scope)")
-                bspecs.append((abi.T_CHAR, 0, 0, col["fixed_len"]))
+                        "equivalent (round-1 engine scope); pass "
+                        "declared_specs for all-null blocks")
+                bspecs.append((abi.T_CHAR, 0, 0, fl))
         if specs is None:
             specs = bspecs
         elif specs != bspecs:

@@ -158,6 +158,24 @@ extern "C" int obx_gpu_close(obx_gpu_ctx *ctx) {
   return OBX_SUCCESS;
 }
 
+/* Load-time payload checksum (the reference verifies data_checksum_ when a
+ * block enters the block cache, before any decode — check_payload_checksum,
+ * ob_micro_block_header.cpp:257-271). ob_crc64_sse42 semantics = CRC-32C in
+ * a u64 accumulator; the host has the crc32 instruction, same as the
+ * reference's hardware path. Independent of the oracle's implementation. */
+__attribute__((target("sse4.2")))
+static uint64_t host_crc32c(const uint8_t *buf, int64_t len) {
+  uint64_t crc = 0;
+  int64_t i = 0;
+  for (; i + 8 <= len; i += 8) {
+    uint64_t w;
+    memcpy(&w, buf + i, 8);
+    crc = __builtin_ia32_crc32di(crc, w);
+  }
+  for (; i < len; i++) crc = __builtin_ia32_crc32qi((uint32_t)crc, buf[i]);
+  return crc;
+}
+
 /* ---- host block parsing (mirrors ObMicroBlockDecoder pointer math,
  * ob_micro_block_decoder.cpp:360-380, and the per-decoder init functions
  * cited in oracle/obx_codec.c) ------------------------------------------- */
@@ -168,6 +186,13 @@ static int parse_block(const obx_col_schema *cols, uint16_t n_cols,
   const obx_micro_header *h = (const obx_micro_header *)block;
   if (h->magic != OBX_MICRO_BLOCK_MAGIC) return OBX_INVALID_ARGUMENT;
   if (h->column_count != n_cols) return OBX_INVALID_ARGUMENT;
+  if (h->data_zlength > (int64_t)block_len ||
+      h->data_zlength < OBX_MICRO_HEADER_SIZE)
+    return OBX_INVALID_ARGUMENT;
+  if ((int64_t)host_crc32c(block + OBX_MICRO_HEADER_SIZE,
+                           (int64_t)h->data_zlength - OBX_MICRO_HEADER_SIZE) !=
+      h->data_checksum)
+    return OBX_PHYSIC_CHECKSUM_ERROR;
   memset(out, 0, sizeof(*out));
   out->row_start_lo = (uint32_t)row_start;
   out->row_start_hi = (uint32_t)(row_start >> 32);
@@ -923,6 +948,10 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
     }
   }
   return OBX_SUCCESS;
+}
+
+extern "C" uint64_t obx_crc32c(const uint8_t *buf, int64_t len) {
+  return host_crc32c(buf, len);
 }
 
 extern "C" double obx_gpu_last_kernel_ms(obx_gpu_ctx *ctx) {

@@ -35,6 +35,32 @@
 /* helpers                                                               */
 /* ===================================================================== */
 
+/* Block payload checksum, restating ob_crc64_sse42 semantics
+ * (deps/oblib/src/lib/checksum/ob_crc64.cpp:448-520 and the crc32q
+ * hardware path): CRC-32C (Castagnoli, reflected poly 0x82f63b78,
+ * init 0, no final xor) accumulated in a 64-bit register — i.e. the
+ * stored int64 data_checksum is the CRC-32C of the payload,
+ * zero-extended. The writer sets it over everything after the micro
+ * header (ob_imicro_block_writer.cpp:193); the decoder verifies it
+ * before trusting interior payloads (check_payload_checksum,
+ * ob_micro_block_header.cpp:257-271). */
+static uint32_t crc32c_tab[256];
+static void crc32c_init(void) {
+  if (crc32c_tab[1]) return;
+  for (uint32_t n = 0; n < 256; n++) {
+    uint32_t c = n;
+    for (int k = 0; k < 8; k++) c = (c & 1) ? 0x82f63b78u ^ (c >> 1) : c >> 1;
+    crc32c_tab[n] = c;
+  }
+}
+uint64_t obx_crc32c(const uint8_t *buf, int64_t len) {
+  crc32c_init();
+  uint64_t crc = 0;
+  for (int64_t i = 0; i < len; i++)
+    crc = crc32c_tab[(crc ^ buf[i]) & 0xff] ^ (crc >> 8);
+  return crc;
+}
+
 static inline int64_t datum_int(const uint8_t *p, int len, int sc) {
   /* read a datum payload as int64 (sign rules per store class) */
   uint64_t v = 0;
@@ -301,7 +327,9 @@ static int encode_column(enc_buf *mb, obx_col_header *ch,
     uint32_t cpct = rows / 10 > 1 ? rows / 10 : 1;
     dict_free(&db);
     if (nd == 1 && null_cnt == 0) enc = OBX_ENC_CONST;
-    else if (cexc > 0 && cexc <= 32 && cexc < cpct && nd <= 255)
+    else if (cexc > 0 && cexc <= 32 && cexc <= cpct && nd <= 255)
+      /* accept-on-equal boundary: the reference rejects only when
+         count_ > MAX(rows*10/100, 1) (ob_const_encoder.cpp:103-104) */
       enc = OBX_ENC_CONST;
     else if (runs * 8 <= rows) enc = OBX_ENC_RLE;
     else if (nd <= 64 && nd * 4 <= rows) enc = OBX_ENC_DICT;
@@ -1106,7 +1134,9 @@ int64_t obx_encode_block(const obx_col_schema *cols, uint16_t n_cols,
   h->original_length = (int32_t)total;
   h->data_length = (int32_t)total;
   h->data_zlength = (int32_t)total;
-  h->data_checksum = 0;
+  h->data_checksum =
+      (int64_t)obx_crc32c(out + OBX_MICRO_HEADER_SIZE,
+                          total - OBX_MICRO_HEADER_SIZE);
   /* set_header_checksum (ob_micro_block_header.cpp:20-50):
      xor of 16-bit words of the listed fields */
   {
@@ -1594,12 +1624,9 @@ int obx_decode_block(const obx_col_schema *cols, uint16_t n_cols,
                      uint8_t *const *out_cols, uint8_t *const *out_nulls,
                      uint32_t *row_count) {
   /* length/shape guard: header + column headers must fit. The column
-     PAYLOADS are trusted after this point — in the reference the block
-     checksum (ObMicroBlockHeader::data_checksum_, verified before any
-     decode) guarantees interior integrity, and this oracle decodes only
-     blocks its own writer produced; full hostile-payload bounds checking
-     is deliberately out of scope (the CS stream layer, which the column
-     layer cross-checks by stream offsets, does carry room guards). */
+     PAYLOADS are trusted only after the data checksum below verifies —
+     the reference's checksum-before-decode model (check_payload_checksum,
+     ob_micro_block_header.cpp:257-271). */
   if (block_len < (int64_t)sizeof(obx_micro_header))
     return OBX_INVALID_ARGUMENT;
   const obx_micro_header *h = (const obx_micro_header *)block;
@@ -1608,6 +1635,12 @@ int obx_decode_block(const obx_col_schema *cols, uint16_t n_cols,
   if ((int64_t)h->header_size +
           (int64_t)n_cols * (int64_t)sizeof(obx_col_header) > block_len)
     return OBX_INVALID_ARGUMENT;
+  if (h->data_zlength > block_len || h->data_zlength < OBX_MICRO_HEADER_SIZE)
+    return OBX_INVALID_ARGUMENT;
+  if ((int64_t)obx_crc32c(block + OBX_MICRO_HEADER_SIZE,
+                          (int64_t)h->data_zlength - OBX_MICRO_HEADER_SIZE) !=
+      h->data_checksum)
+    return OBX_PHYSIC_CHECKSUM_ERROR;
   const uint8_t *meta_region = block + h->header_size +
                                (int64_t)n_cols * sizeof(obx_col_header);
   const obx_col_header *chp =

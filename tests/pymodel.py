@@ -33,6 +33,18 @@ def type_store_size(obj_type):
     return {4: 4, 5: 8, 19: 4, 23: -1, 50: -1}[obj_type]
 
 
+def crc32c(buf):
+    """CRC-32C (Castagnoli, reflected 0x82f63b78, init 0, no final xor) —
+    the value ob_crc64_sse42 stores in data_checksum (zero-extended to i64);
+    independent table-free implementation from the polynomial alone."""
+    crc = 0
+    for b in buf:
+        crc ^= b
+        for _ in range(8):
+            crc = (0x82F63B78 ^ (crc >> 1)) if crc & 1 else crc >> 1
+    return crc
+
+
 def bs_get(buf, pos, length):
     """bit read, little-endian bit order (ob_bit_stream.h semantics)."""
     v = 0
@@ -65,6 +77,10 @@ class Block:
         assert self.magic == 1005, self.magic
         assert self.version == 3
         assert self.column_count == len(schema)
+        (self.data_zlength, self.data_checksum) = struct.unpack_from(
+            "<iq", data, 44)  # after original_length(4)+mmtv(8)+data_length(4)
+        assert self.data_checksum == crc32c(data[64:self.data_zlength]), \
+            "data_checksum mismatch (payload CRC-32C)"
         self.row_index_byte = self.opt & 7
         self.extend_value_bit = (self.opt >> 3) & 7
         self.col_headers = []

@@ -282,3 +282,24 @@ def test_invalid_combine_program_rejected():
                           prog=[0, abi.TOK_AND])  # arity underflow
     with pytest.raises(RuntimeError):
         oracle.filter_block(schema, 1, blk, bad)
+
+
+def test_data_checksum_rejects_corruption():
+    """The decoder verifies data_checksum (CRC-32C of the payload after the
+    64-B header, ob_crc64_sse42 semantics / check_payload_checksum,
+    ob_micro_block_header.cpp:257-271) before trusting interior payloads."""
+    import ctypes as C
+    vals = RNG.integers(-2**62, 2**62, 300, dtype=np.int64)
+    schema = oracle.make_schema([(abi.T_INT, 0, 19, 8)])
+    blk = oracle.encode_block(schema, [_mk_int64(vals)], [abi.ENC_RAW])
+    good = np.frombuffer(bytes(blk), dtype=np.uint8).copy()
+    # clean block decodes
+    oracle.decode_block(schema, 1, good, [0])
+    # flip one payload byte -> OBX_PHYSIC_CHECKSUM_ERROR (-4108)
+    bad = good.copy()
+    bad[200] ^= 0x40
+    try:
+        oracle.decode_block(schema, 1, bad, [0])
+        raise AssertionError("corrupted payload decoded without error")
+    except RuntimeError as e:
+        assert "-4108" in str(e), e
