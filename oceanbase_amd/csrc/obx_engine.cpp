@@ -70,7 +70,9 @@ extern "C" __global__ void k_agg_pass(
     uint32_t, const uint64_t *, const uint8_t *, gslot *);
 extern "C" __global__ void k_lower_leaves(
     const uint8_t *, const dev_block *, uint32_t, const dev_leaf *, uint32_t,
-    blk_leaf *);
+    uint32_t, const int64_t *, blk_leaf *);
+extern "C" __global__ void k_col_minmax(
+    const uint8_t *, const dev_block *, uint32_t, uint32_t, int64_t *);
 
 #define HIP_TRY(x)                                        \
   do {                                                    \
@@ -111,6 +113,19 @@ struct obx_handle {
      1 dict, 2 intdiff, 3 const, 4 slow), bit4 = string; col_cnt = dict
      count capped at 255 */
   std::vector<uint8_t> col_class, col_cnt;
+  /* stored skip index: per-(block,col) [min,max] of non-null decoded
+     values, captured by k_col_minmax at load (device copy feeds
+     k_lower_leaves pruning; host summary feeds the JIT's i64-accumulator
+     range eligibility, obx_jit.inc) */
+  int64_t *d_minmax = nullptr;
+  int64_t col_min[OBX_DEV_MAX_COLS] = {};   /* global over blocks */
+  int64_t col_max[OBX_DEV_MAX_COLS] = {};
+  bool col_known[OBX_DEV_MAX_COLS] = {};    /* bounds known in EVERY block */
+  bool col_dict_every[OBX_DEV_MAX_COLS] = {}; /* class==1 and count<=63
+                                                 in every block */
+  bool col_ext_any[OBX_DEV_MAX_COLS] = {};  /* HAS_EXT in any block */
+  uint32_t col_maxcnt[OBX_DEV_MAX_COLS] = {}; /* max dict count */
+  uint32_t max_block_rows = 0;
 };
 
 struct obx_gpu_ctx {
@@ -148,7 +163,7 @@ extern "C" int obx_gpu_close(obx_gpu_ctx *ctx) {
     (void)hipFree(h.d_buf); (void)hipFree(h.d_blocks); (void)hipFree(h.d_bleaves);
     (void)hipFree(h.d_pleaves); (void)hipFree(h.d_gtable); (void)hipFree(h.d_counters);
     (void)hipFree(h.d_bitmap); (void)hipFree(h.d_row_ids); (void)hipFree(h.d_blk_counts);
-    (void)hipFree(h.d_row_slot);
+    (void)hipFree(h.d_row_slot); (void)hipFree(h.d_minmax);
     for (auto *p : h.d_decode_out) (void)hipFree(p);
   }
   (void)hipStreamDestroy(ctx->stream);
@@ -174,6 +189,11 @@ static uint64_t host_crc32c(const uint8_t *buf, int64_t len) {
   }
   for (; i < len; i++) crc = __builtin_ia32_crc32qi((uint32_t)crc, buf[i]);
   return crc;
+}
+
+static uint32_t grid_for(uint32_t n_blocks) {
+  uint32_t g = n_blocks < 4096u ? n_blocks : 4096u;
+  return g ? g : 1;
 }
 
 /* ---- host block parsing (mirrors ObMicroBlockDecoder pointer math,
@@ -441,7 +461,17 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
       if (dc.flags & OBX_DF_STRING) cls |= 0x10;
       h.col_class.push_back(cls);
       h.col_cnt.push_back((uint8_t)(dc.count > 254 ? 255 : dc.count));
+      if (b == 0) {
+        h.col_dict_every[c] = true;
+        h.col_maxcnt[c] = 0;
+      }
+      if ((cls & 0xF) != 1 || (cls & 0x10) || dc.count > 63)
+        h.col_dict_every[c] = false;
+      if (dc.count > h.col_maxcnt[c]) h.col_maxcnt[c] = dc.count;
+      if (dc.flags & OBX_DF_HAS_EXT) h.col_ext_any[c] = true;
     }
+    if (blocks[b].row_count > h.max_block_rows)
+      h.max_block_rows = blocks[b].row_count;
     row_start += blocks[b].row_count;
   }
   h.total_rows = row_start;
@@ -456,6 +486,36 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
   HIP_TRY(hipMalloc(&h.d_counters, 16 * sizeof(unsigned long long)));
   HIP_TRY(hipMalloc(&h.d_gtable, sizeof(gslot) * OBX_GTABLE_SLOTS));
   HIP_TRY(hipMalloc(&h.d_pleaves, sizeof(dev_leaf) * OBX_DEV_MAX_LEAVES));
+
+  /* stored skip index: capture per-(block,col) min/max once at load
+     (k_col_minmax; outside every timed region) and summarize per column
+     for the JIT's range eligibility */
+  {
+    uint64_t nmm = (uint64_t)bs->n_blocks * bs->n_cols * 2;
+    HIP_TRY(hipMalloc(&h.d_minmax, nmm * sizeof(int64_t)));
+    hipLaunchKernelGGL(k_col_minmax, dim3(grid_for(bs->n_blocks)),
+                       dim3(OBX_WG_HOST), 0, nullptr, h.d_buf, h.d_blocks,
+                       bs->n_blocks, (uint32_t)bs->n_cols, h.d_minmax);
+    std::vector<int64_t> mm(nmm);
+    HIP_TRY(hipMemcpy(mm.data(), h.d_minmax, nmm * sizeof(int64_t),
+                      hipMemcpyDeviceToHost));
+    for (uint16_t c = 0; c < bs->n_cols; c++) {
+      int64_t gmin = INT64_MAX, gmax = INT64_MIN;
+      bool known = true;
+      for (uint32_t b = 0; b < bs->n_blocks; b++) {
+        int64_t mn = mm[2 * ((uint64_t)b * bs->n_cols + c)];
+        int64_t mx = mm[2 * ((uint64_t)b * bs->n_cols + c) + 1];
+        if (mn == INT64_MIN && mx == INT64_MAX) { known = false; break; }
+        if (mn <= mx) { /* empty (all-null) blocks don't widen */
+          if (mn < gmin) gmin = mn;
+          if (mx > gmax) gmax = mx;
+        }
+      }
+      h.col_known[c] = known;
+      h.col_min[c] = gmin;
+      h.col_max[c] = gmax;
+    }
+  }
   h.in_use = true;
   ctx->handles.push_back(h);
   return (int)ctx->handles.size() - 1;
@@ -470,7 +530,7 @@ extern "C" int obx_gpu_free_blocks(obx_gpu_ctx *ctx, int handle) {
   (void)hipFree(h.d_buf); (void)hipFree(h.d_blocks); (void)hipFree(h.d_bleaves);
   (void)hipFree(h.d_pleaves); (void)hipFree(h.d_gtable); (void)hipFree(h.d_counters);
   (void)hipFree(h.d_bitmap); (void)hipFree(h.d_row_ids); (void)hipFree(h.d_blk_counts);
-  (void)hipFree(h.d_row_slot);
+  (void)hipFree(h.d_row_slot); (void)hipFree(h.d_minmax);
   for (auto *&p : h.d_decode_out) { (void)hipFree(p); p = nullptr; }
   h = obx_handle();
   return OBX_SUCCESS;
@@ -667,15 +727,11 @@ static int prep_query(obx_gpu_ctx *ctx, obx_handle &h,
     uint32_t grid = (total + OBX_WG_HOST - 1) / OBX_WG_HOST;
     hipLaunchKernelGGL(k_lower_leaves, dim3(grid), dim3(OBX_WG_HOST), 0, ctx->stream,
                        h.d_buf, h.d_blocks, h.n_blocks, h.d_pleaves, nl,
+                       (uint32_t)h.n_cols, (const int64_t *)h.d_minmax,
                        h.d_bleaves);
   }
   HIP_TRY(hipEventRecord(ctx->ev_p1, ctx->stream));
   return OBX_SUCCESS;
-}
-
-static uint32_t grid_for(uint32_t n_blocks) {
-  uint32_t g = n_blocks < 4096u ? n_blocks : 4096u;
-  return g ? g : 1;
 }
 
 #include "obx_jit.inc"

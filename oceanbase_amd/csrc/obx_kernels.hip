@@ -844,6 +844,7 @@ __device__ __forceinline__ void lower_range(__int128 D, __int128 D2,
 extern "C" __global__ void k_lower_leaves(
     const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
     uint32_t n_blocks, const dev_leaf *__restrict__ pl, uint32_t n_leaves,
+    uint32_t n_cols, const int64_t *__restrict__ minmax,
     blk_leaf *__restrict__ out) {
   uint64_t idx = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
   uint64_t total = (uint64_t)n_blocks * n_leaves;
@@ -919,7 +920,128 @@ extern "C" __global__ void k_lower_leaves(
   } else {
     o.mode = OBX_LEAF_VALUE;
   }
+
+  /* Stored min/max skip-index refinement (the rebuild of the reference's
+   * ObSSTableIndexFilter micro-block pruning,
+   * /root/reference/src/storage/access/ob_sstable_index_filter.cpp: skip
+   * a micro block whose [min,max] cannot satisfy the predicate, or mark
+   * it all-pass). The per-(block,col) bounds are captured once at load
+   * by k_col_minmax (the analogue of the skip index written at encode
+   * time); strings are excluded (VALUE-mode compares map char through
+   * char_key order, which the raw-LE bounds do not follow). */
+  if (minmax && lf.op <= 6 && !(c.flags & OBX_DF_STRING) &&
+      (o.mode == OBX_LEAF_RANGE || o.mode == OBX_LEAF_VALUE)) {
+    int64_t mn = minmax[2 * ((uint64_t)b * n_cols + lf.col)];
+    int64_t mx = minmax[2 * ((uint64_t)b * n_cols + lf.col) + 1];
+    if (!(mn == INT64_MIN && mx == INT64_MAX)) { /* known bounds */
+      bool none = false, all = false;
+      if (mn > mx) { /* no non-null values in the block */
+        none = true;
+      } else {
+        const int64_t lo = lf.vlo, hi = lf.vhi;
+        switch (lf.op) {
+          case 0: none = lo < mn || lo > mx; all = (mn == mx && mn == lo);
+                  break;
+          case 1: none = mn > lo; all = mx <= lo; break; /* LE */
+          case 2: none = mn >= lo; all = mx < lo; break; /* LT */
+          case 3: none = mx < lo; all = mn >= lo; break; /* GE */
+          case 4: none = mx <= lo; all = mn > lo; break; /* GT */
+          case 5: none = (mn == mx && mn == lo); all = lo < mn || lo > mx;
+                  break;
+          case 6: none = mx < lo || mn > hi; all = mn >= lo && mx <= hi;
+                  break;
+        }
+      }
+      if (none) o.mode = OBX_LEAF_NONE;
+      else if (all) o.mode = OBX_LEAF_ALL; /* non-null rows all pass; ext
+                                              nulls still excluded by the
+                                              kernels' ALL handling */
+    }
+  }
   out[idx] = o;
+}
+
+/* Per-(block,column) min/max of the NON-NULL decoded values, captured once
+ * at load time — the engine's stored skip index (the reference writes
+ * ObSkipIndex min/max rows at encode time, storage/blocksstable/index_block;
+ * our container has no index block, so the bounds are derived from the
+ * encoded data at load, outside every timed region). One wave per
+ * (block,col). Output: out[2*(b*n_cols+c)] = {min, max};
+ * {INT64_MIN, INT64_MAX} = unknown (strings/span encodings);
+ * min > max = no non-null values. */
+extern "C" __global__ void k_col_minmax(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, uint32_t n_cols, int64_t *__restrict__ out) {
+  const uint32_t lane = threadIdx.x & 63;
+  const uint32_t wv = threadIdx.x >> 6;
+  uint64_t total = (uint64_t)n_blocks * n_cols;
+  for (uint64_t wid = (uint64_t)blockIdx.x * WAVES + wv; wid < total;
+       wid += (uint64_t)gridDim.x * WAVES) {
+    uint32_t b = (uint32_t)(wid / n_cols), c = (uint32_t)(wid % n_cols);
+    const dev_block &blk = blocks[b];
+    const dev_col &dc = blk.cols[c];
+    blk_view bv;
+    bv.base = buf;
+    bv.bit_bias = 0;
+    bv.rbase_bit = 0;
+    int64_t mn = INT64_MAX, mx = INT64_MIN;
+    bool known = true;
+    if (dc.flags & OBX_DF_STRING) {
+      known = false;
+    } else {
+      switch (dc.enc) {
+        case OBX_D_DICT:
+        case OBX_D_RLE:
+          for (uint32_t e = lane; e < dc.count; e += 64) {
+            int64_t v = dict_entry(bv, dc, e);
+            if (v < mn) mn = v;
+            if (v > mx) mx = v;
+          }
+          break;
+        case OBX_D_CONST:
+          if (dc.runs == 0) {
+            if (lane == 0 && dc.count) mn = mx = dc.base;
+          } else {
+            for (uint32_t e = lane; e < dc.count; e += 64) {
+              int64_t v = dict_entry(bv, dc, e);
+              if (v < mn) mn = v;
+              if (v > mx) mx = v;
+            }
+          }
+          break;
+        case OBX_D_RAW:
+        case OBX_D_INTDIFF:
+          for (uint32_t r = lane; r < blk.row_count; r += 64) {
+            bool isn;
+            int64_t v = col_value(bv, dc, r, isn);
+            if (!isn) {
+              if (v < mn) mn = v;
+              if (v > mx) mx = v;
+            }
+          }
+          break;
+        default:
+          known = false;
+          break;
+      }
+    }
+    if (!known) {
+      mn = INT64_MIN;
+      mx = INT64_MAX;
+    }
+    for (int off = 32; off > 0; off >>= 1) {
+      int64_t omn = (int64_t)shflxor64((uint64_t)mn, off);
+      int64_t omx = (int64_t)shflxor64((uint64_t)mx, off);
+      if (known) { /* unknown is wave-uniform (depends only on dc) */
+        if (omn < mn) mn = omn;
+        if (omx > mx) mx = omx;
+      }
+    }
+    if (lane == 0) {
+      out[2 * wid] = mn;
+      out[2 * wid + 1] = mx;
+    }
+  }
 }
 
 /* ======================================================================
