@@ -124,6 +124,11 @@ struct obx_handle {
   bool col_dict_every[OBX_DEV_MAX_COLS] = {}; /* class==1 and count<=63
                                                  in every block */
   bool col_ext_any[OBX_DEV_MAX_COLS] = {};  /* HAS_EXT in any block */
+  bool col_raw8_every[OBX_DEV_MAX_COLS] = {}; /* 8-B RAW, 64-bit-aligned,
+                                                 no ext, in every block */
+  bool col_rangefam_every[OBX_DEV_MAX_COLS] = {}; /* RAW numeric / INTDIFF
+                                                     (packed-range lowerable)
+                                                     in every block */
   uint32_t col_maxcnt[OBX_DEV_MAX_COLS] = {}; /* max dict count */
   uint32_t max_block_rows = 0;
 };
@@ -463,10 +468,19 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
       h.col_cnt.push_back((uint8_t)(dc.count > 254 ? 255 : dc.count));
       if (b == 0) {
         h.col_dict_every[c] = true;
+        h.col_raw8_every[c] = true;
+        h.col_rangefam_every[c] = true;
         h.col_maxcnt[c] = 0;
       }
       if ((cls & 0xF) != 1 || (cls & 0x10) || dc.count > 63)
         h.col_dict_every[c] = false;
+      if (!(dc.enc == OBX_D_RAW && !(dc.flags & OBX_DF_BITPACK) &&
+            !(dc.flags & OBX_DF_STRING) && !(dc.flags & OBX_DF_HAS_EXT) &&
+            dc.width == 8 && (dc.data_bit & 63) == 0))
+        h.col_raw8_every[c] = false;
+      if (!((dc.enc == OBX_D_RAW || dc.enc == OBX_D_INTDIFF) &&
+            !(dc.flags & OBX_DF_STRING) && !(dc.flags & OBX_DF_HAS_EXT)))
+        h.col_rangefam_every[c] = false;
       if (dc.count > h.col_maxcnt[c]) h.col_maxcnt[c] = dc.count;
       if (dc.flags & OBX_DF_HAS_EXT) h.col_ext_any[c] = true;
     }
@@ -536,13 +550,12 @@ extern "C" int obx_gpu_free_blocks(obx_gpu_ctx *ctx, int handle) {
   return OBX_SUCCESS;
 }
 
-/* build plan header + device leaves; lower per-block leaves on device */
-static int prep_query(obx_gpu_ctx *ctx, obx_handle &h,
-                      const obx_filter_desc *filter, const obx_agg_desc *agg,
-                      dev_plan_hdr &ph) {
+/* build plan header + device leaves (pure: no HIP calls) */
+static int build_plan(const obx_handle &h, const obx_filter_desc *filter,
+                      const obx_agg_desc *agg, dev_plan_hdr &ph,
+                      dev_leaf pl[OBX_DEV_MAX_LEAVES]) {
   memset(&ph, 0, sizeof(ph));
-  dev_leaf pl[OBX_DEV_MAX_LEAVES];
-  memset(pl, 0, sizeof(pl));
+  memset(pl, 0, sizeof(dev_leaf) * OBX_DEV_MAX_LEAVES);
   uint16_t nl = filter ? filter->n_leaves : 0;
   if (nl > OBX_DEV_MAX_LEAVES) return OBX_INVALID_ARGUMENT;
   ph.n_leaves = nl;
@@ -712,6 +725,18 @@ static int prep_query(obx_gpu_ctx *ctx, obx_handle &h,
     }
   }
 
+  return OBX_SUCCESS;
+}
+
+/* build the plan + upload leaves + lower per-block tests on device */
+static int prep_query(obx_gpu_ctx *ctx, obx_handle &h,
+                      const obx_filter_desc *filter, const obx_agg_desc *agg,
+                      dev_plan_hdr &ph, dev_leaf *pl_out = nullptr) {
+  dev_leaf pl[OBX_DEV_MAX_LEAVES];
+  int rc = build_plan(h, filter, agg, ph, pl);
+  if (pl_out) memcpy(pl_out, pl, sizeof(pl));
+  if (rc != OBX_SUCCESS) return rc;
+  uint16_t nl = ph.n_leaves;
   /* upload plan leaves + lower per-block tests on device */
   HIP_TRY(hipEventRecord(ctx->ev_p0, ctx->stream));
   HIP_TRY(hipMemcpyAsync(h.d_pleaves, pl, sizeof(pl), hipMemcpyHostToDevice,
@@ -863,7 +888,8 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
   HIP_TRY(hipSetDevice(ctx->device));
   obx_handle &h = ctx->handles[handle];
   dev_plan_hdr ph;
-  int rc = prep_query(ctx, h, filter, agg, ph);
+  dev_leaf plv[OBX_DEV_MAX_LEAVES];
+  int rc = prep_query(ctx, h, filter, agg, ph, plv);
   if (rc != OBX_SUCCESS) return rc;
   ctx->last_jit = 0;
 
@@ -924,7 +950,7 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
   } else {
     /* plan-specialized JIT kernel when eligible (compiled once per plan
        signature, before the timed region; obx_jit.inc) */
-    jit_entry *je = jit_prepare(h, ph);
+    jit_entry *je = jit_prepare(h, ph, plv);
     ctx->last_jit = je != nullptr;
     HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
     if (je) {
@@ -1008,6 +1034,58 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
 
 extern "C" uint64_t obx_crc32c(const uint8_t *buf, int64_t len) {
   return host_crc32c(buf, len);
+}
+
+/* Debug/test-only: generate the hipRTC source the JIT would compile for a
+ * plan against synthetic load-time column summaries, without a GPU (the
+ * container has no device; tests hipcc-compile the dump offline).
+ * col_flags bits: 1 = dict-everywhere, 2 = ext-any, 4 = bounds known.
+ * Returns the source length (>= 0) or a negative status; 0 = the plan
+ * would not take the JIT path. */
+extern "C" int64_t obx_jit_dump_src(
+    const obx_filter_desc *filter, const obx_agg_desc *agg,
+    const obx_col_schema *cols, uint16_t n_cols, const uint8_t *col_flags,
+    const int64_t *col_min, const int64_t *col_max,
+    const uint32_t *col_maxcnt, uint32_t max_block_rows, int force_v1,
+    char *out, int64_t cap) {
+  if (!cols || n_cols > OBX_DEV_MAX_COLS) return OBX_INVALID_ARGUMENT;
+  obx_handle h;
+  h.n_cols = n_cols;
+  h.n_blocks = 1;
+  h.lds_ok = true;
+  h.max_block_rows = max_block_rows ? max_block_rows : 4096;
+  memcpy(h.cols, cols, sizeof(obx_col_schema) * n_cols);
+  for (uint16_t c = 0; c < n_cols; c++) {
+    h.col_dict_every[c] = (col_flags[c] & 1) != 0;
+    h.col_ext_any[c] = (col_flags[c] & 2) != 0;
+    h.col_known[c] = (col_flags[c] & 4) != 0;
+    h.col_raw8_every[c] = (col_flags[c] & 8) != 0;
+    h.col_rangefam_every[c] = (col_flags[c] & 16) != 0;
+    h.col_min[c] = col_min[c];
+    h.col_max[c] = col_max[c];
+    h.col_maxcnt[c] = col_maxcnt[c];
+    uint8_t cls = h.col_dict_every[c] ? 1 : 0;
+    if (obx_store_class(cols[c].obj_type) == OBX_SC_STRING) cls |= 0x10;
+    h.col_class.push_back(cls);
+    h.col_cnt.push_back(
+        (uint8_t)(h.col_maxcnt[c] > 254 ? 255 : h.col_maxcnt[c]));
+  }
+  dev_plan_hdr ph;
+  dev_leaf pl[OBX_DEV_MAX_LEAVES];
+  int rc = build_plan(h, filter, agg, ph, pl);
+  if (rc != OBX_SUCCESS) return rc;
+  jit_shape js;
+  if (!jit_plan_shape(ph, js)) return 0;
+  if (!jit_blocks_ok(h, ph, js)) return 0;
+  jit_strategy st;
+  if (!force_v1) jit_build_strategy(h, ph, js, pl, st);
+  std::string src = st.ok ? jit_gen_source_v2(ph, js, st)
+                          : jit_gen_source(ph, js);
+  if (out && cap > (int64_t)src.size()) {
+    memcpy(out, src.data(), src.size());
+    out[src.size()] = 0;
+  }
+  return (int64_t)src.size();
 }
 
 extern "C" double obx_gpu_last_kernel_ms(obx_gpu_ctx *ctx) {
