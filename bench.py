@@ -141,7 +141,7 @@ def measure_traffic(args):
     me = os.path.abspath(__file__)
     with tempfile.TemporaryDirectory(prefix="obx_pmc_") as td:
         cmd = [rocprof, "--pmc", "TCC_EA0_RDREQ", "TCC_EA0_WRREQ",
-               "-d", td, "--", sys.executable, me,
+               "--output-format", "csv", "-d", td, "--", sys.executable, me,
                "--workload", args.workload, "--rows", str(args.rows or 0),
                "--block-bytes", str(args.block_bytes),
                "--seed", str(args.seed), "--steps", "2", "--warmup", "1",

@@ -130,6 +130,9 @@ struct obx_handle {
                                                      (packed-range lowerable)
                                                      in every block */
   uint32_t col_maxcnt[OBX_DEV_MAX_COLS] = {}; /* max dict count */
+  uint32_t col_maxw[OBX_DEV_MAX_COLS] = {};   /* max packed width (bits)
+                                                 across blocks; 255 = no
+                                                 packed stream (slow enc) */
   uint32_t max_block_rows = 0;
 };
 
@@ -483,6 +486,13 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
         h.col_rangefam_every[c] = false;
       if (dc.count > h.col_maxcnt[c]) h.col_maxcnt[c] = dc.count;
       if (dc.flags & OBX_DF_HAS_EXT) h.col_ext_any[c] = true;
+      {
+        uint32_t w = 255; /* no packed stream: blocks multi-row reads */
+        if (dc.enc == OBX_D_RAW || dc.enc == OBX_D_DICT ||
+            dc.enc == OBX_D_INTDIFF)
+          w = (dc.flags & OBX_DF_BITPACK) ? dc.width : (uint32_t)dc.width * 8;
+        if (w > h.col_maxw[c]) h.col_maxw[c] = w;
+      }
     }
     if (blocks[b].row_count > h.max_block_rows)
       h.max_block_rows = blocks[b].row_count;
@@ -1046,7 +1056,8 @@ extern "C" int64_t obx_jit_dump_src(
     const obx_filter_desc *filter, const obx_agg_desc *agg,
     const obx_col_schema *cols, uint16_t n_cols, const uint8_t *col_flags,
     const int64_t *col_min, const int64_t *col_max,
-    const uint32_t *col_maxcnt, uint32_t max_block_rows, int force_v1,
+    const uint32_t *col_maxcnt, const uint32_t *col_maxw,
+    uint32_t max_block_rows, int force_v1,
     char *out, int64_t cap) {
   if (!cols || n_cols > OBX_DEV_MAX_COLS) return OBX_INVALID_ARGUMENT;
   obx_handle h;
@@ -1064,6 +1075,7 @@ extern "C" int64_t obx_jit_dump_src(
     h.col_min[c] = col_min[c];
     h.col_max[c] = col_max[c];
     h.col_maxcnt[c] = col_maxcnt[c];
+    h.col_maxw[c] = col_maxw ? col_maxw[c] : 255;
     uint8_t cls = h.col_dict_every[c] ? 1 : 0;
     if (obx_store_class(cols[c].obj_type) == OBX_SC_STRING) cls |= 0x10;
     h.col_class.push_back(cls);
