@@ -123,6 +123,9 @@ struct obx_handle {
   bool col_known[OBX_DEV_MAX_COLS] = {};    /* bounds known in EVERY block */
   bool col_dict_every[OBX_DEV_MAX_COLS] = {}; /* class==1 and count<=63
                                                  in every block */
+  bool col_dict_stable[OBX_DEV_MAX_COLS] = {}; /* dict_every AND the dict
+                                                  payload is byte-identical
+                                                  in every block */
   bool col_ext_any[OBX_DEV_MAX_COLS] = {};  /* HAS_EXT in any block */
   bool col_raw8_every[OBX_DEV_MAX_COLS] = {}; /* 8-B RAW, 64-bit-aligned,
                                                  no ext, in every block */
@@ -500,6 +503,20 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
   }
   h.total_rows = row_start;
   h.lds_ok = lds_ok;
+  /* dict stability: identical dict payload in every block enables the
+     JIT's persistent (whole-kernel) histograms/value tables */
+  for (uint16_t c = 0; c < bs->n_cols; c++) {
+    h.col_dict_stable[c] = h.col_dict_every[c];
+    if (!h.col_dict_stable[c]) continue;
+    const dev_col &d0 = blocks[0].cols[c];
+    size_t dlen = (size_t)d0.count * d0.entry_len;
+    for (uint32_t b = 1; b < bs->n_blocks && h.col_dict_stable[c]; b++) {
+      const dev_col &db = blocks[b].cols[c];
+      if (db.count != d0.count || db.entry_len != d0.entry_len ||
+          memcmp(bs->data + db.dict_byte, bs->data + d0.dict_byte, dlen))
+        h.col_dict_stable[c] = false;
+    }
+  }
 
   HIP_TRY(hipMalloc(&h.d_buf, h.total_bytes + 64));
   HIP_TRY(hipMemset(h.d_buf + h.total_bytes, 0, 64));
@@ -961,7 +978,7 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
     /* plan-specialized JIT kernel when eligible (compiled once per plan
        signature, before the timed region; obx_jit.inc) */
     jit_entry *je = jit_prepare(h, ph, plv);
-    ctx->last_jit = je != nullptr;
+    ctx->last_jit = je ? g_jit_kind : 0; /* 2 = persistent v2, 1 = v1 */
     HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
     if (je) {
       if (jit_launch(je, h, grid_for(h.n_blocks), ctx->stream) != 0)
@@ -1072,6 +1089,7 @@ extern "C" int64_t obx_jit_dump_src(
     h.col_known[c] = (col_flags[c] & 4) != 0;
     h.col_raw8_every[c] = (col_flags[c] & 8) != 0;
     h.col_rangefam_every[c] = (col_flags[c] & 16) != 0;
+    h.col_dict_stable[c] = (col_flags[c] & 32) != 0 && h.col_dict_every[c];
     h.col_min[c] = col_min[c];
     h.col_max[c] = col_max[c];
     h.col_maxcnt[c] = col_maxcnt[c];
