@@ -123,9 +123,14 @@ struct obx_handle {
   bool col_known[OBX_DEV_MAX_COLS] = {};    /* bounds known in EVERY block */
   bool col_dict_every[OBX_DEV_MAX_COLS] = {}; /* class==1 and count<=63
                                                  in every block */
-  bool col_dict_stable[OBX_DEV_MAX_COLS] = {}; /* dict_every AND the dict
+  bool col_dict_any[OBX_DEV_MAX_COLS] = {};   /* dict (incl. char) with
+                                                 count<=63 in every block */
+  bool col_dict_stable[OBX_DEV_MAX_COLS] = {}; /* dict_any AND the dict
                                                   payload is byte-identical
-                                                  in every block */
+                                                  in every block (group/ref
+                                                  persistence; numeric-only
+                                                  uses additionally require
+                                                  col_dict_every) */
   bool col_ext_any[OBX_DEV_MAX_COLS] = {};  /* HAS_EXT in any block */
   bool col_raw8_every[OBX_DEV_MAX_COLS] = {}; /* 8-B RAW, 64-bit-aligned,
                                                  no ext, in every block */
@@ -474,12 +479,15 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
       h.col_cnt.push_back((uint8_t)(dc.count > 254 ? 255 : dc.count));
       if (b == 0) {
         h.col_dict_every[c] = true;
+        h.col_dict_any[c] = true;
         h.col_raw8_every[c] = true;
         h.col_rangefam_every[c] = true;
         h.col_maxcnt[c] = 0;
       }
       if ((cls & 0xF) != 1 || (cls & 0x10) || dc.count > 63)
         h.col_dict_every[c] = false;
+      if ((cls & 0xF) != 1 || dc.count > 63)
+        h.col_dict_any[c] = false; /* dict (possibly char) everywhere */
       if (!(dc.enc == OBX_D_RAW && !(dc.flags & OBX_DF_BITPACK) &&
             !(dc.flags & OBX_DF_STRING) && !(dc.flags & OBX_DF_HAS_EXT) &&
             dc.width == 8 && (dc.data_bit & 63) == 0))
@@ -506,7 +514,7 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
   /* dict stability: identical dict payload in every block enables the
      JIT's persistent (whole-kernel) histograms/value tables */
   for (uint16_t c = 0; c < bs->n_cols; c++) {
-    h.col_dict_stable[c] = h.col_dict_every[c];
+    h.col_dict_stable[c] = h.col_dict_any[c];
     if (!h.col_dict_stable[c]) continue;
     const dev_col &d0 = blocks[0].cols[c];
     size_t dlen = (size_t)d0.count * d0.entry_len;
@@ -1089,7 +1097,10 @@ extern "C" int64_t obx_jit_dump_src(
     h.col_known[c] = (col_flags[c] & 4) != 0;
     h.col_raw8_every[c] = (col_flags[c] & 8) != 0;
     h.col_rangefam_every[c] = (col_flags[c] & 16) != 0;
-    h.col_dict_stable[c] = (col_flags[c] & 32) != 0 && h.col_dict_every[c];
+    h.col_dict_any[c] = h.col_dict_every[c] ||
+        (obx_store_class(cols[c].obj_type) == OBX_SC_STRING &&
+         (col_flags[c] & 1));
+    h.col_dict_stable[c] = (col_flags[c] & 32) != 0 && h.col_dict_any[c];
     h.col_min[c] = col_min[c];
     h.col_max[c] = col_max[c];
     h.col_maxcnt[c] = col_maxcnt[c];
