@@ -142,6 +142,7 @@ struct obx_handle {
                                                  across blocks; 255 = no
                                                  packed stream (slow enc) */
   uint32_t max_block_rows = 0;
+  uint32_t max_block_len = 0;
 };
 
 struct obx_gpu_ctx {
@@ -507,6 +508,7 @@ extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
     }
     if (blocks[b].row_count > h.max_block_rows)
       h.max_block_rows = blocks[b].row_count;
+    if (blen > h.max_block_len) h.max_block_len = (uint32_t)blen;
     row_start += blocks[b].row_count;
   }
   h.total_rows = row_start;
@@ -1090,6 +1092,7 @@ extern "C" int64_t obx_jit_dump_src(
   h.n_blocks = 1;
   h.lds_ok = true;
   h.max_block_rows = max_block_rows ? max_block_rows : 4096;
+  h.max_block_len = OBX_LDS_STAGE_BYTES - 32;
   memcpy(h.cols, cols, sizeof(obx_col_schema) * n_cols);
   for (uint16_t c = 0; c < n_cols; c++) {
     h.col_dict_every[c] = (col_flags[c] & 1) != 0;
