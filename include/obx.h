@@ -292,6 +292,19 @@ int obx_gpu_close(obx_gpu_ctx *ctx);
 /* Stage a blockset into HBM (one large contiguous device buffer + device
  * offset/descriptor arrays). Returns a handle id >= 0 or negative status. */
 int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs);
+
+/* CS (cs_encoding) blockset: `data`/`block_offsets` hold CS-format micro
+ * blocks. The load-time transform of ObCSMicroBlockTransformer
+ * (cs_encoding/ob_cs_micro_block_transformer.cpp) runs GPU-native: the
+ * host parses stream metadata, device kernels decode every stream (RAW
+ * widths, the PFoR/RLE codec families, dict refs, null recovery) into an
+ * HBM arena in the engine's scan layout. Same handle semantics as
+ * obx_gpu_load_blocks; scans/filters/aggregates then run unchanged. */
+int obx_gpu_load_cs_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs);
+
+/* host-side CS block header probe (rows/cols); also exported for tests */
+int obx_cs_host_parse(const uint8_t *buf, int64_t len, uint32_t *rows_out,
+                      uint32_t *ncols_out);
 int obx_gpu_free_blocks(obx_gpu_ctx *ctx, int handle);
 
 /* filter_pushdown_filter equivalent: result bitmap + per-block pass counts +

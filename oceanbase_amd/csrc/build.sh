@@ -13,7 +13,7 @@ for src, out in (("obx_dev.h", "obx_embed_dev_h.inc"),
     open(out, "w").write('R"OBXRAW(' + txt + ')OBXRAW"\n')
 PYEOF
 $HIPCC --offload-arch=gfx950 -O3 -std=c++17 -fPIC -shared \
-    obx_engine.cpp obx_kernels.hip \
+    obx_engine.cpp obx_kernels.hip obx_cs_load.cpp obx_cs_kernels.hip \
     -L/opt/rocm/lib -lhiprtc \
     -o ../libobx.so
 echo "built oceanbase_amd/libobx.so"

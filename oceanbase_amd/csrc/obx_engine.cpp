@@ -1073,6 +1073,121 @@ extern "C" uint64_t obx_crc32c(const uint8_t *buf, int64_t len) {
   return host_crc32c(buf, len);
 }
 
+/* ---- CS load glue (obx_cs_load.cpp drives; these own the handle) ------- */
+int obx_cs_alloc_handle(obx_gpu_ctx *ctx, obx_handle **h_out) {
+  if (!ctx) return OBX_INVALID_ARGUMENT;
+  HIP_TRY(hipSetDevice(ctx->device));
+  *h_out = new obx_handle();
+  return OBX_SUCCESS;
+}
+
+int obx_cs_finish_load(obx_gpu_ctx *ctx, obx_handle *h,
+                       const obx_col_schema *cols, uint16_t n_cols,
+                       std::vector<dev_block> &blocks, uint8_t *d_buf,
+                       uint64_t arena_bytes, uint64_t total_rows) {
+  h->d_buf = d_buf;
+  h->n_blocks = (uint32_t)blocks.size();
+  h->n_cols = n_cols;
+  h->total_rows = total_rows;
+  h->total_bytes = arena_bytes;
+  memcpy(h->cols, cols, sizeof(obx_col_schema) * n_cols);
+  bool lds_ok = true;
+  for (uint32_t b = 0; b < h->n_blocks; b++) {
+    const dev_block &db = blocks[b];
+    if ((db.block_byte & 15) || db.block_len + 24 > OBX_LDS_STAGE_BYTES ||
+        db.row_count > 4096)
+      lds_ok = false;
+    for (uint16_t c = 0; c < n_cols; c++) {
+      const dev_col &dc = db.cols[c];
+      uint8_t cls;
+      switch (dc.enc) {
+        case OBX_D_RAW: cls = 0; break;
+        case OBX_D_DICT: cls = 1; break;
+        case OBX_D_INTDIFF: cls = 2; break;
+        case OBX_D_CONST: cls = dc.runs == 0 ? 3 : 4; break;
+        default: cls = 4; break;
+      }
+      if (dc.flags & OBX_DF_STRING) cls |= 0x10;
+      h->col_class.push_back(cls);
+      h->col_cnt.push_back((uint8_t)(dc.count > 254 ? 255 : dc.count));
+      if (b == 0) {
+        h->col_dict_every[c] = true;
+        h->col_dict_any[c] = true;
+        h->col_raw8_every[c] = true;
+        h->col_rangefam_every[c] = true;
+        h->col_maxcnt[c] = 0;
+      }
+      if ((cls & 0xF) != 1 || (cls & 0x10) || dc.count > 63)
+        h->col_dict_every[c] = false;
+      if ((cls & 0xF) != 1 || dc.count > 63) h->col_dict_any[c] = false;
+      if (!(dc.enc == OBX_D_RAW && !(dc.flags & OBX_DF_BITPACK) &&
+            !(dc.flags & OBX_DF_STRING) && !(dc.flags & OBX_DF_HAS_EXT) &&
+            dc.width == 8 && (dc.data_bit & 63) == 0))
+        h->col_raw8_every[c] = false;
+      if (!((dc.enc == OBX_D_RAW || dc.enc == OBX_D_INTDIFF) &&
+            !(dc.flags & OBX_DF_STRING) && !(dc.flags & OBX_DF_HAS_EXT)))
+        h->col_rangefam_every[c] = false;
+      {
+        uint32_t w = 255;
+        if (dc.enc == OBX_D_RAW || dc.enc == OBX_D_DICT ||
+            dc.enc == OBX_D_INTDIFF)
+          w = (dc.flags & OBX_DF_BITPACK) ? dc.width
+                                          : (uint32_t)dc.width * 8;
+        if (w > h->col_maxw[c]) h->col_maxw[c] = w;
+      }
+      if (dc.flags & OBX_DF_HAS_EXT) h->col_ext_any[c] = true;
+    }
+    if (db.row_count > h->max_block_rows) h->max_block_rows = db.row_count;
+    if (db.block_len > h->max_block_len) h->max_block_len = db.block_len;
+  }
+  h->lds_ok = lds_ok;
+  /* dict payloads live only on the device for CS handles; the persistent
+     JIT path requires the host-verified byte-identical dicts, so it is
+     simply not taken (col_dict_stable stays false -> v1 JIT / generic
+     kernels, which are parity-equal) */
+  HIP_TRY(hipMalloc(&h->d_blocks, sizeof(dev_block) * blocks.size()));
+  HIP_TRY(hipMemcpy(h->d_blocks, blocks.data(),
+                    sizeof(dev_block) * blocks.size(),
+                    hipMemcpyHostToDevice));
+  HIP_TRY(hipMalloc(&h->d_counters, 16 * sizeof(unsigned long long)));
+  HIP_TRY(hipMalloc(&h->d_gtable, sizeof(gslot) * OBX_GTABLE_SLOTS));
+  HIP_TRY(hipMalloc(&h->d_pleaves, sizeof(dev_leaf) * OBX_DEV_MAX_LEAVES));
+  {
+    uint64_t nmm = (uint64_t)h->n_blocks * n_cols * 2;
+    HIP_TRY(hipMalloc(&h->d_minmax, nmm * sizeof(int64_t)));
+    hipLaunchKernelGGL(k_col_minmax, dim3(grid_for(h->n_blocks)),
+                       dim3(OBX_WG_HOST), 0, nullptr, h->d_buf, h->d_blocks,
+                       h->n_blocks, (uint32_t)n_cols, h->d_minmax);
+    std::vector<int64_t> mm(nmm);
+    HIP_TRY(hipMemcpy(mm.data(), h->d_minmax, nmm * sizeof(int64_t),
+                      hipMemcpyDeviceToHost));
+    for (uint16_t c = 0; c < n_cols; c++) {
+      int64_t gmin = INT64_MAX, gmax = INT64_MIN;
+      bool known = true;
+      for (uint32_t b = 0; b < h->n_blocks; b++) {
+        int64_t mn = mm[2 * ((uint64_t)b * n_cols + c)];
+        int64_t mx = mm[2 * ((uint64_t)b * n_cols + c) + 1];
+        if (mn == INT64_MIN && mx == INT64_MAX) { known = false; break; }
+        if (mn <= mx) {
+          if (mn < gmin) gmin = mn;
+          if (mx > gmax) gmax = mx;
+        }
+      }
+      h->col_known[c] = known;
+      h->col_min[c] = gmin;
+      h->col_max[c] = gmax;
+    }
+  }
+  return OBX_SUCCESS;
+}
+
+int obx_cs_publish_handle(obx_gpu_ctx *ctx, obx_handle *h) {
+  h->in_use = true;
+  ctx->handles.push_back(*h);
+  delete h;
+  return (int)ctx->handles.size() - 1;
+}
+
 /* Debug/test-only: generate the hipRTC source the JIT would compile for a
  * plan against synthetic load-time column summaries, without a GPU (the
  * container has no device; tests hipcc-compile the dump offline).

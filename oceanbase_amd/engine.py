@@ -31,6 +31,9 @@ def _load():
     lib.obx_gpu_close.argtypes = [C.c_void_p]
     lib.obx_gpu_load_blocks.restype = C.c_int
     lib.obx_gpu_load_blocks.argtypes = [C.c_void_p, C.POINTER(abi.BlockSet)]
+    lib.obx_gpu_load_cs_blocks.restype = C.c_int
+    lib.obx_gpu_load_cs_blocks.argtypes = [C.c_void_p,
+                                           C.POINTER(abi.BlockSet)]
     lib.obx_gpu_free_blocks.restype = C.c_int
     lib.obx_gpu_free_blocks.argtypes = [C.c_void_p, C.c_int]
     lib.obx_gpu_filter.restype = C.c_int
@@ -105,6 +108,16 @@ class GpuEngine:
         h = self._lib.obx_gpu_load_blocks(self._ctx, C.byref(bs))
         if h < 0:
             raise RuntimeError(f"obx_gpu_load_blocks failed: {h}")
+        return h
+
+    def load_cs(self, cs_blocks, schema):
+        """GPU-native CS load: device kernels decode the CS streams into
+        the engine's scan layout (ObCSMicroBlockTransformer position)."""
+        from . import cs as _cs
+        bs, keep = _cs.make_blockset(cs_blocks, schema)
+        h = self._lib.obx_gpu_load_cs_blocks(self._ctx, C.byref(bs))
+        if h < 0:
+            raise RuntimeError(f"obx_gpu_load_cs_blocks failed: {h}")
         return h
 
     def free(self, handle):

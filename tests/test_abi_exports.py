@@ -34,14 +34,24 @@ def test_engine_exports_every_declared_symbol():
         assert getattr(lib, n, None) is not None, f"missing export: {n}"
 
 
+# product-only host helpers (exported by libobx.so, not the oracle)
+PRODUCT_ONLY = {"obx_cs_host_parse"}
+
+
 def test_oracle_exports_cpu_symbols():
     from oceanbase_amd import oracle  # builds liboracle.so if needed
     lib = C.CDLL(os.path.join(REPO, "oracle", "liboracle.so"))
     for n in _declared_functions():
-        if n.startswith("obx_gpu"):
+        if n.startswith("obx_gpu") or n in PRODUCT_ONLY:
             continue
         assert getattr(lib, n, None) is not None, f"missing export: {n}"
     assert oracle is not None
+
+
+def test_product_exports_cs_host_parse():
+    lib = C.CDLL(os.path.join(REPO, "oceanbase_amd", "libobx.so"))
+    for n in PRODUCT_ONLY:
+        assert getattr(lib, n, None) is not None, f"missing export: {n}" 
 
 
 def test_product_refuses_without_gpu():

@@ -10,7 +10,8 @@ import pytest
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
-from oceanbase_amd import abi, cs, oracle  # noqa: E402
+from oceanbase_amd import abi, oracle  # noqa: E402
+import cs_oracle_util as cs  # noqa: E402
 
 from test_cs_block import (  # noqa: E402
     _enc as cs_enc, _int_col as cs_int_col, _str_col as cs_str_col,

@@ -158,7 +158,7 @@ def test_span_matrix(enc):
 # test_int_dict_pd_filter.cpp, test_str_dict_pd_filter.cpp,
 # test_string_pd_filter.cpp through the transformer-role load path) ----
 
-from oceanbase_amd import cs  # noqa: E402
+import cs_oracle_util as cs  # noqa: E402
 from test_cs_block import (  # noqa: E402
     _enc as cs_enc, _int_col as cs_int_col, _str_col as cs_str_col,
 )
