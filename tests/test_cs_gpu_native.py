@@ -207,8 +207,10 @@ def test_cs_q1_style_end_to_end(eng):
         ]))
         for k, vco in enumerate((qty, price, disc, flag, date)):
             all_cols[k].append(vco)
-    specs = [(abi.T_INT, 0, 19, 8), (abi.T_DECIMAL_INT, 2, 15, 8),
-             (abi.T_INT, 2, 15, 8), (abi.T_CHAR, 0, 0, 1),
+    # scales 0 on both sides: the oracle leg re-encodes through
+    # to_pax_blocks, whose schema is scale-0 T_INT — one_b must agree
+    specs = [(abi.T_INT, 0, 19, 8), (abi.T_INT, 0, 19, 8),
+             (abi.T_INT, 0, 19, 8), (abi.T_CHAR, 0, 0, 1),
              (abi.T_INT, 0, 19, 8)]
     h, _schema = _load(eng, blocks, specs)
     filt = abi.make_filter([dict(col=4, op=abi.OP_LE, lo=10471)])
