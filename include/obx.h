@@ -107,7 +107,26 @@ enum obx_white_op {
   OBX_OP_IN = 7,   /* in-list (operands in obx_filter_leaf.in_list) */
   OBX_OP_NU = 8,   /* is null */
   OBX_OP_NN = 9,   /* is not null */
+  OBX_OP_BLACK = 10, /* black (generic-expression) filter: the leaf carries
+                        a postfix program over column values and constants
+                        (ObPhysicalFilterExecutor::filter_batch,
+                        ob_pushdown_filter.cpp:2066 — decode the referenced
+                        columns, evaluate the expression, keep is-true
+                        rows). SQL three-valued logic: NULL operands
+                        propagate; a NULL or false result drops the row. */
 };
+
+/* black-filter bytecode (one byte per op):
+ *   0x00|i  push column value i (obx_filter_leaf.bcols[i])
+ *   0x40|i  push constant i     (obx_filter_leaf.bconst[i])
+ *   0x50 ADD  0x51 SUB  0x52 MUL  0x53 DIV (int; x/0 -> NULL)  0x54 NEG
+ *   0x60 LT  0x61 LE  0x62 GT  0x63 GE  0x64 EQ  0x65 NE (-> bool)
+ *   0x70 AND  0x71 OR  0x72 NOT (three-valued)
+ * Stack depth <= 8. The program leaves one value; the row passes iff it
+ * is non-NULL and != 0. */
+#define OBX_BX_MAX_PROG 24
+#define OBX_BX_MAX_CONST 4
+#define OBX_BX_MAX_COLS 4
 
 /* ---- aggregate kinds (subset the hot path needs; AVG is rewritten to
  *      SUM+COUNT before the engine, ob_expand_aggregate_utils.cpp:1674) */
@@ -149,12 +168,19 @@ typedef struct obx_blockset {
 /* ---- filter descriptor (white-only AND tree for round 1) ---------------- */
 #define OBX_MAX_IN_LIST 8
 typedef struct obx_filter_leaf {
-  uint16_t col;      /* column index */
+  uint16_t col;      /* column index (BLACK: bcols[0], for pruning) */
   uint8_t op;        /* enum obx_white_op */
   uint8_t n_in;      /* operand count for IN */
   int64_t lo;        /* operand (EQ/LT/...), or BT lower bound */
   int64_t hi;        /* BT upper bound */
   int64_t in_list[OBX_MAX_IN_LIST];
+  /* BLACK leaves only (op == OBX_OP_BLACK) */
+  int64_t bconst[OBX_BX_MAX_CONST];
+  uint16_t bcols[OBX_BX_MAX_COLS];
+  uint8_t n_bprog;
+  uint8_t n_bcols;
+  uint8_t bprog[OBX_BX_MAX_PROG];
+  uint8_t bpad[6];
 } obx_filter_leaf;
 
 /* Filter combine program: the POD restatement of the executor tree's

@@ -11,6 +11,7 @@ ENC_RAW, ENC_DICT, ENC_RLE, ENC_CONST, ENC_INT_DIFF, ENC_SDIFF, ENC_HEX, ENC_STR
 T_INT, T_INT32, T_DATE, T_CHAR, T_DECIMAL_INT = 5, 4, 19, 23, 50
 
 OP_EQ, OP_LE, OP_LT, OP_GE, OP_GT, OP_NE, OP_BT, OP_IN, OP_NU, OP_NN = range(10)
+OP_BLACK = 10
 
 AGG_COUNT, AGG_SUM, AGG_MIN, AGG_MAX, AGG_SUM_PROD2, AGG_SUM_PROD3, AGG_SUM_MUL = range(7)
 
@@ -35,7 +36,20 @@ class BlockSet(C.Structure):
 class FilterLeaf(C.Structure):
     _fields_ = [("col", C.c_uint16), ("op", C.c_uint8), ("n_in", C.c_uint8),
                 ("lo", C.c_int64), ("hi", C.c_int64),
-                ("in_list", C.c_int64 * 8)]
+                ("in_list", C.c_int64 * 8),
+                # black (generic-expression) leaves, op == OP_BLACK
+                ("bconst", C.c_int64 * 4), ("bcols", C.c_uint16 * 4),
+                ("n_bprog", C.c_uint8), ("n_bcols", C.c_uint8),
+                ("bprog", C.c_uint8 * 24), ("bpad", C.c_uint8 * 6)]
+
+
+# black-filter bytecode (include/obx.h OBX_BX_*): postfix program over
+# column values and constants; three-valued logic, wrap-mod-2^64 arith
+BX_COL = 0x00   # | slot index (into the leaf's bcols)
+BX_CONST = 0x40  # | const index
+BX_ADD, BX_SUB, BX_MUL, BX_DIV, BX_NEG = 0x50, 0x51, 0x52, 0x53, 0x54
+BX_LT, BX_LE, BX_GT, BX_GE, BX_EQ, BX_NE = 0x60, 0x61, 0x62, 0x63, 0x64, 0x65
+BX_AND, BX_OR, BX_NOT = 0x70, 0x71, 0x72
 
 
 TOK_AND, TOK_OR = 128, 129
@@ -90,6 +104,21 @@ def make_filter(leaves, prog=None):
         for i, t in enumerate(prog):
             f.prog[i] = t
     for i, lf in enumerate(leaves):
+        if lf.get("op") == OP_BLACK or "bprog" in lf:
+            bcols = lf["bcols"]
+            prog = lf["bprog"]
+            consts = lf.get("bconst", [])
+            f.leaves[i].op = OP_BLACK
+            f.leaves[i].col = bcols[0]
+            f.leaves[i].n_bcols = len(bcols)
+            for j, c in enumerate(bcols):
+                f.leaves[i].bcols[j] = c
+            f.leaves[i].n_bprog = len(prog)
+            for j, b in enumerate(prog):
+                f.leaves[i].bprog[j] = b
+            for j, k in enumerate(consts):
+                f.leaves[i].bconst[j] = k
+            continue
         f.leaves[i].col = lf["col"]
         f.leaves[i].op = lf["op"]
         f.leaves[i].lo = lf.get("lo", 0)

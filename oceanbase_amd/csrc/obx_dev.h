@@ -124,12 +124,20 @@ enum {
 typedef struct dev_leaf {
   int64_t vlo, vhi;     /* VALUE-mode operands (order-mapped for char) */
   uint16_t col;
-  uint8_t op;           /* obx_white_op for VALUE mode */
+  uint8_t op;           /* obx_white_op for VALUE mode (10 = BLACK) */
   uint8_t n_in;
   uint8_t char_len;     /* >0: char column — order-map values before compare
                            (byte-lexicographic, like the oracle's char_key) */
   uint8_t pad[3];
   int64_t in_list[8];   /* VALUE-mode IN */
+  /* BLACK (op 10): postfix program over bcols values + bconst
+     (ob_pushdown_filter.cpp:2066 semantics; obx.h OBX_BX_* bytecode) */
+  int64_t bconst[4];
+  uint16_t bcols[4];
+  uint8_t n_bprog;
+  uint8_t n_bcols;
+  uint8_t bprog[24];
+  uint8_t bpad[6];
 } dev_leaf;
 
 /* per-(block, leaf) lowered test (32 B, n_blocks * n_leaves of them) */

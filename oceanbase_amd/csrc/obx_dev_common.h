@@ -255,10 +255,11 @@ __device__ __forceinline__ leaf_ctx make_leaf_ctx(const dev_block &blk,
   x.cc = make_col_ctx(c, blk_bit);
   x.mask = blf.mask; x.lo = blf.lo; x.hi = blf.hi;
   x.mode = blf.mode; x.invert = blf.invert; x.op = plf.op;
-  /* slow fallback: RLE/exception decode, or IN lists (operand array stays
-     in the global leaf) */
+  /* slow fallback: RLE/exception decode, IN lists, or black programs
+     (operand arrays stay in the global leaf) */
   x.slow = (x.cc.kind == 4 ||
-            (blf.mode == OBX_LEAF_VALUE && plf.op == 7)) ? 1 : 0;
+            (blf.mode == OBX_LEAF_VALUE &&
+             (plf.op == 7 || plf.op == 10))) ? 1 : 0;
   return x;
 }
 
@@ -512,6 +513,87 @@ __device__ __noinline__ int64_t col_value2(const blk_view bv,
   return col_value(bv, c, r, null_out);
 }
 
+/* Black (generic-expression) filter: device evaluation of the postfix
+ * OBX_BX_* program with three-valued logic — independent of the oracle's
+ * C implementation, compared bit-exactly by the parity tests.
+ * __noinline__: cold generic path (register budget). */
+__device__ __noinline__ bool black_eval_vals(const dev_leaf &plf,
+                                             const int64_t *vals,
+                                             const bool *nus) {
+  int64_t st[8];
+  bool nu[8];
+  int sp = 0;
+#pragma clang loop unroll(disable)
+  for (int p = 0; p < plf.n_bprog; p++) {
+    uint8_t op = plf.bprog[p];
+    if (op < 0x40) {
+      if (op >= plf.n_bcols || sp >= 8) return false;
+      st[sp] = vals[op];
+      nu[sp++] = nus[op];
+    } else if (op < 0x50) {
+      int k = op & 0x0F;
+      if (k >= 4 || sp >= 8) return false;
+      st[sp] = plf.bconst[k];
+      nu[sp++] = false;
+    } else if (op == 0x54) {
+      if (sp < 1) return false;
+      st[sp - 1] = (int64_t)(0 - (uint64_t)st[sp - 1]);
+    } else if (op == 0x72) {
+      if (sp < 1) return false;
+      if (!nu[sp - 1]) st[sp - 1] = !st[sp - 1];
+    } else {
+      if (sp < 2) return false;
+      sp--;
+      int64_t a = st[sp - 1], b2 = st[sp];
+      bool na = nu[sp - 1], nb = nu[sp];
+      bool rn = na || nb;
+      int64_t r2 = 0;
+      switch (op) {
+        case 0x50: r2 = (int64_t)((uint64_t)a + (uint64_t)b2); break;
+        case 0x51: r2 = (int64_t)((uint64_t)a - (uint64_t)b2); break;
+        case 0x52: r2 = (int64_t)((uint64_t)a * (uint64_t)b2); break;
+        case 0x53:
+          if (b2 == 0) rn = true;
+          else if (a == INT64_MIN && b2 == -1) r2 = a;
+          else r2 = a / b2;
+          break;
+        case 0x60: r2 = a < b2; break;
+        case 0x61: r2 = a <= b2; break;
+        case 0x62: r2 = a > b2; break;
+        case 0x63: r2 = a >= b2; break;
+        case 0x64: r2 = a == b2; break;
+        case 0x65: r2 = a != b2; break;
+        case 0x70:
+          if ((!na && !a) || (!nb && !b2)) { r2 = 0; rn = false; }
+          else if (rn) r2 = 0;
+          else r2 = 1;
+          break;
+        case 0x71:
+          if ((!na && a) || (!nb && b2)) { r2 = 1; rn = false; }
+          else if (rn) r2 = 0;
+          else r2 = 0;
+          break;
+        default: return false;
+      }
+      st[sp - 1] = r2;
+      nu[sp - 1] = rn;
+    }
+  }
+  if (sp != 1) return false;
+  return !nu[0] && st[0] != 0;
+}
+
+__device__ __noinline__ bool black_match(const blk_view bv,
+                                         const dev_block &blk,
+                                         const dev_leaf &plf, uint32_t r) {
+  int64_t vals[4];
+  bool nus[4];
+#pragma clang loop unroll(disable)
+  for (int j = 0; j < plf.n_bcols; j++)
+    vals[j] = col_value2(bv, blk, blk.cols[plf.bcols[j]], r, nus[j]);
+  return black_eval_vals(plf, vals, nus);
+}
+
 __device__ __forceinline__ bool leaf_value_match(const dev_leaf &lf, int64_t v,
                                                  bool isn) {
   if (lf.op == 8) return isn;
@@ -563,6 +645,7 @@ __device__ __noinline__ bool leaf_match(const blk_view bv,
       return isn != (bool)blf.invert;
     }
     default: {
+      if (plf.op == 10) return black_match(bv, blk, plf, r);
       bool isn;
       int64_t v = col_value2(bv, blk, c, r, isn);
       return leaf_value_match(plf, v, isn);

@@ -619,6 +619,45 @@ static int build_plan(const obx_handle &h, const obx_filter_desc *filter,
     pl[i].vhi = lf->hi;
     pl[i].n_in = lf->n_in;
     for (int j = 0; j < lf->n_in && j < 8; j++) pl[i].in_list[j] = lf->in_list[j];
+    if (lf->op == OBX_OP_BLACK) {
+      /* stack-discipline validation mirroring the oracle's
+         obx__bprog_valid */
+      if (lf->n_bprog == 0 || lf->n_bprog > OBX_BX_MAX_PROG ||
+          lf->n_bcols == 0 || lf->n_bcols > OBX_BX_MAX_COLS)
+        return OBX_INVALID_ARGUMENT;
+      int sp = 0;
+      for (int p = 0; p < lf->n_bprog; p++) {
+        uint8_t op2 = lf->bprog[p];
+        if (op2 < 0x40) {
+          if (op2 >= lf->n_bcols) return OBX_INVALID_ARGUMENT;
+          sp++;
+        } else if (op2 < 0x50) {
+          if ((op2 & 0x0F) >= OBX_BX_MAX_CONST) return OBX_INVALID_ARGUMENT;
+          sp++;
+        } else if (op2 == 0x54 || op2 == 0x72) {
+          if (sp < 1) return OBX_INVALID_ARGUMENT;
+        } else if ((op2 >= 0x50 && op2 <= 0x53) ||
+                   (op2 >= 0x60 && op2 <= 0x65) || op2 == 0x70 ||
+                   op2 == 0x71) {
+          if (sp < 2) return OBX_INVALID_ARGUMENT;
+          sp--;
+        } else {
+          return OBX_INVALID_ARGUMENT;
+        }
+        if (sp > 8) return OBX_INVALID_ARGUMENT;
+      }
+      if (sp != 1) return OBX_INVALID_ARGUMENT;
+      for (int j = 0; j < lf->n_bcols; j++) {
+        if (lf->bcols[j] >= h.n_cols) return OBX_INVALID_ARGUMENT;
+        pl[i].bcols[j] = lf->bcols[j];
+      }
+      pl[i].n_bcols = lf->n_bcols;
+      pl[i].n_bprog = lf->n_bprog;
+      memcpy(pl[i].bprog, lf->bprog, lf->n_bprog);
+      memcpy(pl[i].bconst, lf->bconst, sizeof(pl[i].bconst));
+      pl[i].col = lf->bcols[0]; /* pruning/lowering anchor */
+      continue;
+    }
     /* char columns compare byte-lexicographically: order-map the operands
        exactly like the oracle's char_key (low len LE bytes -> BE int) */
     if (obx_store_class(h.cols[lf->col].obj_type) == OBX_SC_STRING) {

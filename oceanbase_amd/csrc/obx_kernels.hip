@@ -857,6 +857,32 @@ extern "C" __global__ void k_lower_leaves(
   blk_leaf o;
   o.mask = 0; o.lo = 0; o.hi = 0; o.mode = OBX_LEAF_VALUE; o.invert = 0;
 
+  /* black (generic-expression) leaves: the filter-on-dict trick — a
+     single-column program over a small dict evaluates ONCE PER ENTRY
+     into a ref mask (ob_dict_decoder.cpp:1481-1561); a const column
+     folds to ALL/NONE; anything else evaluates per row (VALUE mode,
+     black_match). */
+  if (lf.op == 10) {
+    const dev_col &bc = blocks[b].cols[lf.bcols[0]];
+    if (lf.n_bcols == 1 &&
+        (bc.enc == OBX_D_DICT || bc.enc == OBX_D_RLE) && bc.count < 64) {
+      for (uint32_t e = 0; e < bc.count; e++) {
+        int64_t v = dict_entry(bv, bc, e);
+        bool nu0 = false;
+        if (black_eval_vals(lf, &v, &nu0)) o.mask |= 1ull << e;
+      }
+      o.mode = (o.mask == 0) ? OBX_LEAF_NONE : OBX_LEAF_REF_MASK;
+    } else if (lf.n_bcols == 1 && bc.enc == OBX_D_CONST && bc.runs == 0) {
+      int64_t v = bc.base;
+      bool nu0 = (bc.count == 0);
+      o.mode = black_eval_vals(lf, &v, &nu0) ? OBX_LEAF_ALL : OBX_LEAF_NONE;
+    } else {
+      o.mode = OBX_LEAF_VALUE;
+    }
+    out[idx] = o;
+    return;
+  }
+
   /* NOT const-with-exceptions: REF_MASK row evaluation reads a per-row
      ref (packed stream or RLE runs), which CONST does not have — its
      refs live in the exception list, so it takes the VALUE slow path

@@ -40,6 +40,9 @@ int obx__col_dec_init(void *d, const void *h, const void *ch,
                       const uint8_t *meta_region);
 int obx__col_dec_row(const void *d, const void *h, const obx_col_schema *cs,
                      uint32_t r, int64_t *out, int *is_null);
+int obx__black_eval(const obx_filter_leaf *lf, const int64_t *vals,
+                    const int *nulls);
+int obx__bprog_valid(const obx_filter_leaf *lf);
 int obx__leaf_match(const obx_filter_leaf *lf, int64_t v, int is_null, int sc,
                     int len);
 int obx__combine_leaves(const obx_filter_desc *f, const int *leaf_res);
@@ -164,7 +167,14 @@ static void *worker(void *arg) {
   for (int i = 0; i < 64; i++) dec_of_col[i] = -1;
 #define NEED(c) do { uint16_t cc = (c); \
   if (cc < 64 && dec_of_col[cc] < 0) { dec_of_col[cc] = n_need; need[n_need++] = cc; } } while (0)
-  if (filter) for (int i = 0; i < filter->n_leaves; i++) NEED(filter->leaves[i].col);
+  if (filter) for (int i = 0; i < filter->n_leaves; i++) {
+    const obx_filter_leaf *lf0 = &filter->leaves[i];
+    if (lf0->op == OBX_OP_BLACK) {
+      for (int j = 0; j < lf0->n_bcols; j++) NEED(lf0->bcols[j]);
+    } else {
+      NEED(lf0->col);
+    }
+  }
   if (agg) {
     for (int i = 0; i < agg->n_group_cols; i++) NEED(agg->group_cols[i]);
     for (int i = 0; i < agg->n_aggs; i++) {
@@ -200,6 +210,15 @@ static void *worker(void *arg) {
       if (filter && filter->n_prog == 0) {
         for (int i = 0; i < filter->n_leaves && pass; i++) {
           const obx_filter_leaf *lf = &filter->leaves[i];
+          if (lf->op == OBX_OP_BLACK) {
+            int64_t bvv[OBX_BX_MAX_COLS]; int bnn[OBX_BX_MAX_COLS];
+            for (int j = 0; j < lf->n_bcols; j++)
+              obx__col_dec_row(decbuf + dsz * dec_of_col[lf->bcols[j]], h,
+                               &bs->cols[lf->bcols[j]], r, &bvv[j],
+                               &bnn[j]);
+            pass = obx__black_eval(lf, bvv, bnn);
+            continue;
+          }
           int di = dec_of_col[lf->col];
           int64_t v; int isn;
           obx__col_dec_row(decbuf + dsz * di, h, &bs->cols[lf->col], r, &v,
@@ -212,6 +231,15 @@ static void *worker(void *arg) {
         int leaf_res[8];
         for (int i = 0; i < filter->n_leaves; i++) {
           const obx_filter_leaf *lf = &filter->leaves[i];
+          if (lf->op == OBX_OP_BLACK) {
+            int64_t bvv[OBX_BX_MAX_COLS]; int bnn[OBX_BX_MAX_COLS];
+            for (int j = 0; j < lf->n_bcols; j++)
+              obx__col_dec_row(decbuf + dsz * dec_of_col[lf->bcols[j]], h,
+                               &bs->cols[lf->bcols[j]], r, &bvv[j],
+                               &bnn[j]);
+            leaf_res[i] = obx__black_eval(lf, bvv, bnn);
+            continue;
+          }
           int di = dec_of_col[lf->col];
           int64_t v; int isn;
           obx__col_dec_row(decbuf + dsz * di, h, &bs->cols[lf->col], r, &v,
@@ -316,6 +344,11 @@ int obx_cpu_scan_filter_agg(const obx_blockset *bs,
                             obx_agg_result *out) {
   if (!bs || !out) return OBX_INVALID_ARGUMENT;
   if (filter && !obx__prog_valid(filter)) return OBX_INVALID_ARGUMENT;
+  if (filter)
+    for (int i = 0; i < filter->n_leaves; i++)
+      if (filter->leaves[i].op == OBX_OP_BLACK &&
+          !obx__bprog_valid(&filter->leaves[i]))
+        return OBX_INVALID_ARGUMENT;
   if (nthreads <= 0) nthreads = (int)sysconf(_SC_NPROCESSORS_ONLN);
   if (nthreads < 1) nthreads = 1;
   if (nthreads > 256) nthreads = 256;

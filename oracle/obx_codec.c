@@ -1686,6 +1686,106 @@ static inline int64_t char_key(int64_t raw_le, int len) {
   return (int64_t)be;
 }
 
+/* Black (generic-expression) filter evaluation: the postfix program of
+ * obx.h's OBX_BX_* bytecode over the leaf's referenced column values,
+ * with SQL three-valued logic — the restatement of
+ * ObPhysicalFilterExecutor::filter_batch's decode-then-eval-expr loop
+ * (ob_pushdown_filter.cpp:2066) for the engine's bounded expression
+ * shapes. Arithmetic wraps mod 2^64 (documented container semantics);
+ * x/0 and NULL operands yield NULL; the row passes iff the result is
+ * non-NULL and nonzero. Returns 0 on malformed programs. */
+int obx__black_eval(const obx_filter_leaf *lf, const int64_t *vals,
+                    const int *nulls) {
+  int64_t st[8];
+  int nu[8];
+  int sp = 0;
+  for (int p = 0; p < lf->n_bprog; p++) {
+    uint8_t op = lf->bprog[p];
+    if (op < 0x40) {
+      if (op >= lf->n_bcols || sp >= 8) return 0;
+      st[sp] = vals[op];
+      nu[sp++] = nulls[op];
+    } else if (op < 0x50) {
+      int k = op & 0x0F;
+      if (k >= OBX_BX_MAX_CONST || sp >= 8) return 0;
+      st[sp] = lf->bconst[k];
+      nu[sp++] = 0;
+    } else if (op == 0x54) { /* NEG */
+      if (sp < 1) return 0;
+      st[sp - 1] = (int64_t)(0 - (uint64_t)st[sp - 1]);
+    } else if (op == 0x72) { /* NOT (three-valued) */
+      if (sp < 1) return 0;
+      if (!nu[sp - 1]) st[sp - 1] = !st[sp - 1];
+    } else {
+      if (sp < 2) return 0;
+      sp--;
+      int64_t a = st[sp - 1], b = st[sp];
+      int na = nu[sp - 1], nb = nu[sp];
+      int rn = na || nb;
+      int64_t r = 0;
+      switch (op) {
+        case 0x50: r = (int64_t)((uint64_t)a + (uint64_t)b); break;
+        case 0x51: r = (int64_t)((uint64_t)a - (uint64_t)b); break;
+        case 0x52: r = (int64_t)((uint64_t)a * (uint64_t)b); break;
+        case 0x53:
+          if (b == 0) rn = 1;
+          else if (a == INT64_MIN && b == -1) r = a;
+          else r = a / b;
+          break;
+        case 0x60: r = a < b; break;
+        case 0x61: r = a <= b; break;
+        case 0x62: r = a > b; break;
+        case 0x63: r = a >= b; break;
+        case 0x64: r = a == b; break;
+        case 0x65: r = a != b; break;
+        case 0x70: /* AND: false dominates NULL */
+          if ((!na && !a) || (!nb && !b)) { r = 0; rn = 0; }
+          else if (rn) r = 0;
+          else r = 1;
+          break;
+        case 0x71: /* OR: true dominates NULL */
+          if ((!na && a) || (!nb && b)) { r = 1; rn = 0; }
+          else if (rn) r = 0;
+          else r = 0;
+          break;
+        default: return 0;
+      }
+      st[sp - 1] = r;
+      nu[sp - 1] = rn;
+    }
+  }
+  if (sp != 1) return 0;
+  return !nu[0] && st[0] != 0;
+}
+
+/* stack-discipline validation of a black program */
+int obx__bprog_valid(const obx_filter_leaf *lf) {
+  if (lf->n_bprog == 0 || lf->n_bprog > OBX_BX_MAX_PROG ||
+      lf->n_bcols > OBX_BX_MAX_COLS)
+    return 0;
+  int sp = 0;
+  for (int p = 0; p < lf->n_bprog; p++) {
+    uint8_t op = lf->bprog[p];
+    if (op < 0x40) {
+      if (op >= lf->n_bcols) return 0;
+      sp++;
+    } else if (op < 0x50) {
+      if ((op & 0x0F) >= OBX_BX_MAX_CONST) return 0;
+      sp++;
+    } else if (op == 0x54 || op == 0x72) {
+      if (sp < 1) return 0;
+    } else if ((op >= 0x50 && op <= 0x53) || (op >= 0x60 && op <= 0x65) ||
+               op == 0x70 || op == 0x71) {
+      if (sp < 2) return 0;
+      sp--;
+    } else {
+      return 0;
+    }
+    if (sp > 8) return 0;
+  }
+  return sp == 1;
+}
+
 static inline int leaf_match(const obx_filter_leaf *lf, int64_t v, int is_null,
                              int sc, int len) {
   if (lf->op == OBX_OP_NU) return is_null;
@@ -1779,11 +1879,25 @@ int obx_cpu_filter_block(const obx_col_schema *cols, uint16_t n_cols,
   memset(result_bits, 0, (rows + 7) / 8);
 
   col_dec dec[8];
+  col_dec bdec[8][OBX_BX_MAX_COLS];
   int sc_of[8], len_of[8];
   uint16_t nl = filter ? filter->n_leaves : 0;
   if (nl > 8) return OBX_INVALID_ARGUMENT;
   for (uint16_t i = 0; i < nl; i++) {
-    uint16_t c = filter->leaves[i].col;
+    const obx_filter_leaf *lf = &filter->leaves[i];
+    if (lf->op == OBX_OP_BLACK) {
+      if (!obx__bprog_valid(lf)) return OBX_INVALID_ARGUMENT;
+      for (int j = 0; j < lf->n_bcols; j++) {
+        uint16_t c = lf->bcols[j];
+        if (c >= n_cols) return OBX_INVALID_ARGUMENT;
+        int rc = col_dec_init(&bdec[i][j], h, &chp[c], meta_region);
+        if (rc) return rc;
+      }
+      sc_of[i] = 0;
+      len_of[i] = 8;
+      continue;
+    }
+    uint16_t c = lf->col;
     if (c >= n_cols) return OBX_INVALID_ARGUMENT;
     int rc = col_dec_init(&dec[i], h, &chp[c], meta_region);
     if (rc) return rc;
@@ -1796,21 +1910,38 @@ int obx_cpu_filter_block(const obx_col_schema *cols, uint16_t n_cols,
     int pass = 1;
     if (filter && filter->n_prog == 0) {
       for (uint16_t i = 0; i < nl && pass; i++) {
+        const obx_filter_leaf *lf = &filter->leaves[i];
+        if (lf->op == OBX_OP_BLACK) {
+          int64_t bvv[OBX_BX_MAX_COLS]; int bnn[OBX_BX_MAX_COLS];
+          for (int j = 0; j < lf->n_bcols; j++)
+            if (col_dec_row(&bdec[i][j], h, &cols[lf->bcols[j]], r,
+                            &bvv[j], &bnn[j]))
+              return OBX_INTERNAL_ERROR;
+          pass = obx__black_eval(lf, bvv, bnn);
+          continue;
+        }
         int64_t v; int isn;
-        if (col_dec_row(&dec[i], h, &cols[filter->leaves[i].col], r, &v,
-                        &isn))
+        if (col_dec_row(&dec[i], h, &cols[lf->col], r, &v, &isn))
           return OBX_INTERNAL_ERROR;
-        pass = leaf_match(&filter->leaves[i], v, isn, sc_of[i], len_of[i]);
+        pass = leaf_match(lf, v, isn, sc_of[i], len_of[i]);
       }
     } else if (filter && nl > 0) {
       int leaf_res[8];
       for (uint16_t i = 0; i < nl; i++) {
+        const obx_filter_leaf *lf = &filter->leaves[i];
+        if (lf->op == OBX_OP_BLACK) {
+          int64_t bvv[OBX_BX_MAX_COLS]; int bnn[OBX_BX_MAX_COLS];
+          for (int j = 0; j < lf->n_bcols; j++)
+            if (col_dec_row(&bdec[i][j], h, &cols[lf->bcols[j]], r,
+                            &bvv[j], &bnn[j]))
+              return OBX_INTERNAL_ERROR;
+          leaf_res[i] = obx__black_eval(lf, bvv, bnn);
+          continue;
+        }
         int64_t v; int isn;
-        if (col_dec_row(&dec[i], h, &cols[filter->leaves[i].col], r, &v,
-                        &isn))
+        if (col_dec_row(&dec[i], h, &cols[lf->col], r, &v, &isn))
           return OBX_INTERNAL_ERROR;
-        leaf_res[i] = leaf_match(&filter->leaves[i], v, isn, sc_of[i],
-                                 len_of[i]);
+        leaf_res[i] = leaf_match(lf, v, isn, sc_of[i], len_of[i]);
       }
       pass = combine_leaves(filter, leaf_res);
     }
