@@ -213,10 +213,16 @@ def main():
 
     import torch
     dist = None
+    # RCCL (backend "nccl") for real multi-GPU; OBX_DIST_BACKEND=gloo lets a
+    # 1-GPU box smoke the full multi-rank path (exchange on CPU)
+    backend = os.environ.get("OBX_DIST_BACKEND", "nccl")
+    n_dev = torch.cuda.device_count() if torch.cuda.is_available() else 1
+    device = local_rank % max(n_dev, 1)
     if world > 1:
         import torch.distributed as tdist
-        tdist.init_process_group(backend="nccl")
-        torch.cuda.set_device(local_rank)
+        tdist.init_process_group(backend=backend)
+        if backend == "nccl":
+            torch.cuda.set_device(device)
         dist = tdist
 
     # --- generate this rank's shard (deterministic by global row id) -------
@@ -231,7 +237,7 @@ def main():
     gen_s = time.time() - t0
 
     from oceanbase_amd.engine import GpuEngine
-    eng = GpuEngine(local_rank)
+    eng = GpuEngine(device)
     t0 = time.time()
     h = eng.load(li.bs)
     load_s = time.time() - t0
@@ -243,8 +249,9 @@ def main():
             res = eng.scan_filter_agg(h, filt, agg)
             if dist is not None:
                 blob = result_to_bytes(res)
-                t = torch.frombuffer(bytearray(blob), dtype=torch.uint8) \
-                    .cuda(local_rank)
+                t = torch.frombuffer(bytearray(blob), dtype=torch.uint8)
+                if backend == "nccl":
+                    t = t.cuda(device)
                 outs = [torch.empty_like(t) for _ in range(world)]
                 dist.all_gather(outs, t)
                 kinds = [agg.aggs[i].kind for i in range(n_aggs)]
@@ -284,7 +291,9 @@ def main():
 
     # MAX over ranks
     if dist is not None:
-        t = torch.tensor([elapsed], dtype=torch.float64).cuda(local_rank)
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if backend == "nccl":
+            t = t.cuda(device)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.cpu().item())
 

@@ -401,3 +401,37 @@ def test_pipeline_agg_path_parity(eng):
     assert base.rows_passed == piped.rows_passed
     assert sorted(abi.result_rows(base, 6)) == sorted(
         abi.result_rows(piped, 6))
+
+
+@pytest.mark.gpu
+def test_stored_minmax_skip_index_parity():
+    """Stored min/max skip index (ObSSTableIndexFilter equivalent): RAW
+    fixed columns cluster by block here, so selective range predicates
+    prune most blocks via the bounds captured at load (k_col_minmax ->
+    k_lower_leaves NONE verdicts); results must equal the oracle and the
+    all-pass/none refinements must not change survivors."""
+    from oceanbase_amd.engine import GpuEngine
+    rng = np.random.default_rng(3)
+    rows_pb, nblocks = 2000, 24
+    blks, av = [], []
+    schema = oracle.make_schema([(abi.T_INT, 0, 19, 8),
+                                 (abi.T_INT, 0, 19, 8)])
+    for b in range(nblocks):
+        # values cluster per block: block b holds [b*1000, b*1000+999]
+        a = rng.integers(b * 1000, b * 1000 + 1000, rows_pb).astype(np.int64)
+        x = rng.integers(-10**6, 10**6, rows_pb).astype(np.int64)
+        blks.append(oracle.encode_block(
+            schema, [a.view(np.uint8), x.view(np.uint8)],
+            [abi.ENC_RAW, abi.ENC_RAW]))
+        av.append(a)
+    bs = _manual_blockset(schema, blks)
+    bs.total_rows = rows_pb * nblocks
+    eng = GpuEngine(0)
+    h = eng.load(bs)
+    for lo, hi in ((5000, 6999), (0, 999), (23_000, 23_999), (40_000, 41_000),
+                   (2500, 2503)):
+        filt = abi.make_filter([dict(col=0, op=abi.OP_BT, lo=lo, hi=hi)])
+        got = eng.filter(h, filt)
+        exp = sum(int(((a >= lo) & (a <= hi)).sum()) for a in av)
+        assert got == exp, (lo, hi)
+    eng.close()
