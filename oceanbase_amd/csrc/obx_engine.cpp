@@ -846,8 +846,10 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
   HIP_TRY(hipSetDevice(ctx->device));
   obx_handle &h = ctx->handles[handle];
   dev_plan_hdr ph;
-  int rc = prep_query(ctx, h, filter, nullptr, ph);
+  dev_leaf plv[OBX_DEV_MAX_LEAVES];
+  int rc = prep_query(ctx, h, filter, nullptr, ph, plv);
   if (rc != OBX_SUCCESS) return rc;
+  ctx->last_jit = 0;
   uint64_t bm_words = (h.total_rows + 63) / 64 + 1;
   if (!h.d_bitmap) HIP_TRY(hipMalloc(&h.d_bitmap, bm_words * 8));
   HIP_TRY(hipMemsetAsync(h.d_bitmap, 0, bm_words * 8, ctx->stream));
@@ -859,8 +861,19 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
       HIP_TRY(hipMalloc(&h.d_blk_counts, h.n_blocks * sizeof(uint32_t)));
   }
   HIP_TRY(hipMemsetAsync(h.d_counters, 0, 16 * 8, ctx->stream));
+  /* bitmap-only filters take the specialized wave-per-block kernel
+     (direct global reads of just the leaf streams; obx_jit_v2.inc) */
+  jit_entry *fje = want_row_ids ? nullptr : jit_prepare_filter(h, ph, plv);
+  ctx->last_jit = fje ? 2 : 0;
   HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
-  {
+  if (fje) {
+    void *args[7] = {&h.d_buf, &h.d_blocks, &h.n_blocks, &h.d_pleaves,
+                     &h.d_bleaves, &h.d_bitmap, &h.d_counters};
+    if (hipModuleLaunchKernel(fje->fn, grid_for(h.n_blocks), 1, 1,
+                              OBX_WG_HOST, 1, 1, 0, ctx->stream, args,
+                              nullptr) != hipSuccess)
+      return OBX_INTERNAL_ERROR;
+  } else {
     auto kfn = h.lds_ok ? (ph.n_prog ? k_filter_prog_lds : k_filter_lds)
                         : (ph.n_prog ? k_filter_prog : k_filter);
     hipLaunchKernelGGL(kfn, dim3(grid_for(h.n_blocks)), dim3(OBX_WG_HOST), 0,
