@@ -273,7 +273,7 @@ def main():
     prep_ms = []
     for _ in range(args.warmup):
         one_step()
-    jit = eng.last_jit() if agg is not None else False
+    jit = eng.last_jit()
     barrier_sync()
     t_start = time.time()
     for _ in range(args.steps):
