@@ -1285,6 +1285,16 @@ extern "C" int64_t obx_jit_dump_src(
   dev_leaf pl[OBX_DEV_MAX_LEAVES];
   int rc = build_plan(h, filter, agg, ph, pl);
   if (rc != OBX_SUCCESS) return rc;
+  if (!agg) { /* bitmap-filter path: dump the wave-per-block filter JIT */
+    jit_strategy fst;
+    if (!jit_build_fstrategy(h, ph, pl, fst)) return 0;
+    std::string fsrc = jit_gen_source_filter(ph, fst);
+    if (out && cap > (int64_t)fsrc.size()) {
+      memcpy(out, fsrc.data(), fsrc.size());
+      out[fsrc.size()] = 0;
+    }
+    return (int64_t)fsrc.size();
+  }
   jit_shape js;
   if (!jit_plan_shape(ph, js)) return 0;
   if (!jit_blocks_ok(h, ph, js)) return 0;

@@ -1,0 +1,41 @@
+"""CPU checks of the hipRTC query-codegen source (no GPU needed:
+codegen and strategy selection are host code)."""
+import ctypes as C
+import os
+
+from oceanbase_amd import abi
+
+
+def test_filter_jit_source_strip_mined():
+    """The bitmap-filter JIT strip-mines R rows/lane with one packed read
+    (R*W<=64) and assembles the bitmap via wave-synchronous LDS ORs."""
+    import ctypes as C
+    from oceanbase_amd import abi
+    lib = C.CDLL(os.path.join(os.path.dirname(abi.__file__), "libobx.so"))
+    filt = abi.make_filter([dict(col=0, op=abi.OP_LE, lo=10561)])
+    n = 4
+    cols = (abi.ColSchema * n)()
+    for c in range(n):
+        cols[c].obj_type = abi.T_INT
+    flags = (C.c_uint8 * n)(*[16] * n)      # rangefam
+    cmin = (C.c_int64 * n)()
+    cmax = (C.c_int64 * n)(*[30000] * n)
+    maxcnt = (C.c_uint32 * n)()
+    maxw = (C.c_uint32 * n)(*[13] * n)       # 13-bit packed -> R=4
+    buf = C.create_string_buffer(1 << 20)
+    lib.obx_jit_dump_src.restype = C.c_int64
+    sz = lib.obx_jit_dump_src(C.byref(filt), None, cols, n, flags, cmin,
+                              cmax, maxcnt, maxw, 5460, 0, buf, len(buf))
+    assert sz > 0
+    src = buf.raw[:sz].decode()
+    assert "#define FR 4" in src
+    assert "fm[wv][lane >> 4]" in src       # LDS mask assembly
+    assert "FR * l0W" in src                # one packed read per R rows
+    # wide column (64-bit raw): no strip mining, ballot path
+    maxw2 = (C.c_uint32 * n)(*[64] * n)
+    sz = lib.obx_jit_dump_src(C.byref(filt), None, cols, n, flags, cmin,
+                              cmax, maxcnt, maxw2, 2048, 0, buf, len(buf))
+    assert sz > 0
+    src = buf.raw[:sz].decode()
+    assert "#define FR 1" in src
+    assert "__ballot" in src
