@@ -867,9 +867,14 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
   ctx->last_jit = fje ? 2 : 0;
   HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
   if (fje) {
+    uint32_t fgrid = grid_for(h.n_blocks);
+    if (const char *fg = getenv("OBX_JIT_FGRID")) {
+      int g = atoi(fg);
+      if (g > 0 && g <= 65535) fgrid = (uint32_t)g;
+    }
     void *args[7] = {&h.d_buf, &h.d_blocks, &h.n_blocks, &h.d_pleaves,
                      &h.d_bleaves, &h.d_bitmap, &h.d_counters};
-    if (hipModuleLaunchKernel(fje->fn, grid_for(h.n_blocks), 1, 1,
+    if (hipModuleLaunchKernel(fje->fn, fgrid, 1, 1,
                               OBX_WG_HOST, 1, 1, 0, ctx->stream, args,
                               nullptr) != hipSuccess)
       return OBX_INTERNAL_ERROR;

@@ -28,14 +28,26 @@ def test_filter_jit_source_strip_mined():
                               cmax, maxcnt, maxw, 5460, 0, buf, len(buf))
     assert sz > 0
     src = buf.raw[:sz].decode()
-    assert "#define FR 4" in src
-    assert "fm[wv][lane >> 4]" in src       # LDS mask assembly
-    assert "FR * l0W" in src                # one packed read per R rows
+    # narrow stream, fits LDS budget -> staged mode (stream DMA'd to LDS)
+    assert "__shared__ uint8_t seg[" in src
+    assert "global_load_lds" in src
+    # forced direct mode -> strip-mined loads + LDS mask assembly
+    os.environ["OBX_JIT_FSTAGE"] = "0"
+    try:
+        sz = lib.obx_jit_dump_src(C.byref(filt), None, cols, n, flags,
+                                  cmin, cmax, maxcnt, maxw, 5460, 0, buf,
+                                  len(buf))
+        assert sz > 0
+        src = buf.raw[:sz].decode()
+        assert "#define FR 4" in src
+        assert "fm[wv][lane >> 4]" in src   # LDS mask assembly
+        assert "FR * l0W" in src            # one packed read per R rows
+    finally:
+        del os.environ["OBX_JIT_FSTAGE"]
     # wide column (64-bit raw): no strip mining, ballot path
     maxw2 = (C.c_uint32 * n)(*[64] * n)
     sz = lib.obx_jit_dump_src(C.byref(filt), None, cols, n, flags, cmin,
                               cmax, maxcnt, maxw2, 2048, 0, buf, len(buf))
     assert sz > 0
     src = buf.raw[:sz].decode()
-    assert "#define FR 1" in src
     assert "__ballot" in src
