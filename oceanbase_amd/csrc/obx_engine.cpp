@@ -867,7 +867,12 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
   ctx->last_jit = fje ? 2 : 0;
   HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
   if (fje) {
-    uint32_t fgrid = grid_for(h.n_blocks);
+    /* ~1-2k WGs measured optimal: larger grids thrash concurrent DMA
+       streams, and the striped-counter tail scales with grid anyway */
+    uint32_t fgrid = h.n_blocks / 8;
+    if (fgrid < 512) fgrid = h.n_blocks < 512 ? (h.n_blocks ? h.n_blocks : 1)
+                                              : 512;
+    if (fgrid > 2048) fgrid = 2048;
     if (const char *fg = getenv("OBX_JIT_FGRID")) {
       int g = atoi(fg);
       if (g > 0 && g <= 65535) fgrid = (uint32_t)g;
@@ -895,9 +900,11 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
   ctx->last_ms = ms;
   HIP_TRY(hipEventElapsedTime(&ms, ctx->ev_p0, ctx->ev_p1));
   ctx->last_prep_ms = ms;
-  unsigned long long cnt[2];
-  HIP_TRY(hipMemcpy(cnt, h.d_counters, 16, hipMemcpyDeviceToHost));
+  unsigned long long cnt[16];
+  HIP_TRY(hipMemcpy(cnt, h.d_counters, 16 * 8, hipMemcpyDeviceToHost));
+  /* slot 0: legacy kernels; slots 8..15: JIT filter's striped WG sums */
   h.last_survivors = cnt[0];
+  for (int i2 = 8; i2 < 16; i2++) h.last_survivors += cnt[i2];
   return OBX_SUCCESS;
 }
 
