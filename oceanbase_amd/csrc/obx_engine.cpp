@@ -1077,11 +1077,13 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
   ctx->last_prep_ms = ms;
 
   std::vector<gslot> gt(OBX_GTABLE_SLOTS);
-  unsigned long long cnt[2];
+  unsigned long long cnt[16];
   HIP_TRY(hipMemcpy(gt.data(), h.d_gtable, sizeof(gslot) * OBX_GTABLE_SLOTS,
                     hipMemcpyDeviceToHost));
-  HIP_TRY(hipMemcpy(cnt, h.d_counters, 16, hipMemcpyDeviceToHost));
+  HIP_TRY(hipMemcpy(cnt, h.d_counters, 16 * 8, hipMemcpyDeviceToHost));
   if (cnt[1] != 0) return OBX_BUF_NOT_ENOUGH; /* LDS table overflow */
+  /* slots 8..15: v2 kernel's striped per-wave survivor sums */
+  for (int i2 = 8; i2 < 16; i2++) cnt[0] += cnt[i2];
 
   memset(out, 0, sizeof(*out));
   out->rows_scanned = h.total_rows;
