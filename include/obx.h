@@ -368,6 +368,23 @@ uint64_t obx_gpu_total_rows(obx_gpu_ctx *ctx, int handle);
 uint64_t obx_gpu_total_bytes(obx_gpu_ctx *ctx, int handle);
 uint64_t obx_gpu_last_survivors(obx_gpu_ctx *ctx, int handle);
 
+/* Paged access to the last scan's full sorted group rows — the growth
+ * path past the OBX_MAX_GROUPS inline result. The reference's hash
+ * group-by grows its table unboundedly
+ * (ob_exec_hash_struct_vec.h:1718); here a scan whose group count
+ * exceeds OBX_MAX_GROUPS returns OBX_BUF_NOT_ENOUGH with
+ * obx_agg_result.n_groups set to the true total, and the rows (sorted
+ * by key bytes) are fetched in pages. Device capacity: OBX_GTABLE_BIG
+ * (4096) distinct groups; the engine transparently reruns an
+ * LDS-table-overflowed generic scan through a direct-global
+ * high-cardinality kernel (AND-combined filters only). */
+int obx_gpu_agg_fetch(obx_gpu_ctx *ctx, int handle, uint32_t start,
+                      uint32_t count, obx_group_row *out, uint32_t *n_out,
+                      uint64_t *n_total);
+/* CPU-oracle mirror (test infrastructure) */
+int obx_cpu_agg_fetch(uint32_t start, uint32_t count, obx_group_row *out,
+                      uint32_t *n_out, uint64_t *n_total);
+
 #ifdef __cplusplus
 }
 #endif

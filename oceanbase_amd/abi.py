@@ -16,6 +16,7 @@ OP_BLACK = 10
 AGG_COUNT, AGG_SUM, AGG_MIN, AGG_MAX, AGG_SUM_PROD2, AGG_SUM_PROD3, AGG_SUM_MUL = range(7)
 
 OBX_SUCCESS = 0
+OBX_BUF_NOT_ENOUGH = -4009
 
 MAX_GROUPS = 64
 MAX_KEY_BYTES = 16
@@ -143,6 +144,15 @@ def make_agg(group_cols, aggs):
         a.aggs[i].col_b = e.get("col_b", 0)
         a.aggs[i].col_c = e.get("col_c", 0)
     return a
+
+
+def group_row_tuples(rows, n_aggs):
+    """[GroupRow] -> list of (key_bytes, row_count, [cell ints])."""
+    out = []
+    for g in rows:
+        out.append((bytes(g.key[: g.key_len]), g.row_count,
+                    [g.cells[a].as_int() for a in range(n_aggs)]))
+    return out
 
 
 def result_rows(res, n_aggs):

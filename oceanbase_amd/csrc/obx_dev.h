@@ -38,7 +38,15 @@ typedef unsigned long size_t;
 #define OBX_DEV_MAX_COLS 8
 #define OBX_DEV_MAX_LEAVES 8
 #define OBX_DEV_MAX_AGGS 8
-#define OBX_GTABLE_SLOTS 256   /* global group table (open addressing) */
+#define OBX_GTABLE_SLOTS 256   /* global group table (open addressing);
+                                  LDS-mirrored by the pass kernels */
+#define OBX_GTABLE_SHIFT 8     /* log2(OBX_GTABLE_SLOTS) */
+#define OBX_GTABLE_BIG 4096    /* the high-cardinality direct kernel's
+                                  table: growth past the 64-group inline
+                                  result (ob_exec_hash_struct_vec.h:1718
+                                  grows unboundedly — callers page with
+                                  obx_gpu_agg_fetch) */
+#define OBX_GTABLE_BIG_SHIFT 12
 #define OBX_LTABLE_SLOTS 16    /* per-workgroup LDS group table */
 
 /* column encodings (== ObColumnHeader::Type subset) */

@@ -28,6 +28,9 @@
 #include "../../oracle/obx_format.h"
 
 /* kernels (obx_kernels.hip) */
+extern "C" __global__ void k_scan_agg_direct(
+    const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
+    const blk_leaf *, const dev_plan_hdr, gslot *, unsigned long long *);
 extern "C" __global__ void k_scan_filter_agg(
     const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
     const blk_leaf *, const dev_plan_hdr, gslot *, unsigned long long *);
@@ -106,6 +109,9 @@ struct obx_handle {
   uint8_t *d_row_slot = nullptr;
   uint8_t *d_decode_out[OBX_DEV_MAX_COLS] = {};
   uint64_t last_survivors = 0;
+  gslot *d_gtable_big = nullptr; /* high-cardinality direct kernel table */
+  std::vector<obx_group_row> last_rows; /* full sorted group rows of the
+                                           last scan (obx_gpu_agg_fetch) */
   bool lds_ok = false;   /* all blocks 16-B aligned and <= LDS stage size */
   bool in_use = false;
   /* per-(block,col) encoding summary captured at load time for the JIT
@@ -178,7 +184,7 @@ extern "C" int obx_gpu_close(obx_gpu_ctx *ctx) {
   for (auto &h : ctx->handles) {
     if (!h.in_use) continue;
     (void)hipFree(h.d_buf); (void)hipFree(h.d_blocks); (void)hipFree(h.d_bleaves);
-    (void)hipFree(h.d_pleaves); (void)hipFree(h.d_gtable); (void)hipFree(h.d_counters);
+    (void)hipFree(h.d_pleaves); (void)hipFree(h.d_gtable); (void)hipFree(h.d_counters); (void)hipFree(h.d_gtable_big);
     (void)hipFree(h.d_bitmap); (void)hipFree(h.d_row_ids); (void)hipFree(h.d_blk_counts);
     (void)hipFree(h.d_row_slot); (void)hipFree(h.d_minmax);
     for (auto *p : h.d_decode_out) (void)hipFree(p);
@@ -579,7 +585,7 @@ extern "C" int obx_gpu_free_blocks(obx_gpu_ctx *ctx, int handle) {
   obx_handle &h = ctx->handles[handle];
   if (!h.in_use) return OBX_INVALID_ARGUMENT;
   (void)hipFree(h.d_buf); (void)hipFree(h.d_blocks); (void)hipFree(h.d_bleaves);
-  (void)hipFree(h.d_pleaves); (void)hipFree(h.d_gtable); (void)hipFree(h.d_counters);
+  (void)hipFree(h.d_pleaves); (void)hipFree(h.d_gtable); (void)hipFree(h.d_counters); (void)hipFree(h.d_gtable_big);
   (void)hipFree(h.d_bitmap); (void)hipFree(h.d_row_ids); (void)hipFree(h.d_blk_counts);
   (void)hipFree(h.d_row_slot); (void)hipFree(h.d_minmax);
   for (auto *&p : h.d_decode_out) { (void)hipFree(p); p = nullptr; }
@@ -1081,8 +1087,33 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
   HIP_TRY(hipMemcpy(gt.data(), h.d_gtable, sizeof(gslot) * OBX_GTABLE_SLOTS,
                     hipMemcpyDeviceToHost));
   HIP_TRY(hipMemcpy(cnt, h.d_counters, 16 * 8, hipMemcpyDeviceToHost));
-  if (cnt[1] != 0) return OBX_BUF_NOT_ENOUGH; /* LDS table overflow */
-  /* slots 8..15: v2 kernel's striped per-wave survivor sums */
+  if (cnt[1] != 0) {
+    /* a workgroup shard overflowed its LDS group table: rerun with the
+       direct-global high-cardinality kernel (growth path; AND-only
+       filters — OR-programs keep the capacity error) */
+    if (ph.n_prog != 0) return OBX_BUF_NOT_ENOUGH;
+    if (!h.d_gtable_big)
+      HIP_TRY(hipMalloc(&h.d_gtable_big, sizeof(gslot) * OBX_GTABLE_BIG));
+    std::vector<gslot> init(OBX_GTABLE_BIG);
+    memset(init.data(), 0, sizeof(gslot) * OBX_GTABLE_BIG);
+    for (auto &s2 : init) s2.key = OBX_KEY_EMPTY;
+    HIP_TRY(hipMemcpyAsync(h.d_gtable_big, init.data(),
+                           sizeof(gslot) * OBX_GTABLE_BIG,
+                           hipMemcpyHostToDevice, ctx->stream));
+    HIP_TRY(hipMemsetAsync(h.d_counters, 0, 16 * 8, ctx->stream));
+    hipLaunchKernelGGL(k_scan_agg_direct, dim3(grid_for(h.n_blocks)),
+                       dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf,
+                       h.d_blocks, h.n_blocks, h.d_pleaves, h.d_bleaves,
+                       ph, h.d_gtable_big, h.d_counters);
+    HIP_TRY(hipStreamSynchronize(ctx->stream));
+    gt.resize(OBX_GTABLE_BIG);
+    HIP_TRY(hipMemcpy(gt.data(), h.d_gtable_big,
+                      sizeof(gslot) * OBX_GTABLE_BIG,
+                      hipMemcpyDeviceToHost));
+    HIP_TRY(hipMemcpy(cnt, h.d_counters, 16 * 8, hipMemcpyDeviceToHost));
+    if (cnt[2] != 0) return OBX_BUF_NOT_ENOUGH; /* >4096 distinct groups */
+  }
+  /* slots 8..15: striped per-wave survivor sums */
   for (int i2 = 8; i2 < 16; i2++) cnt[0] += cnt[i2];
 
   memset(out, 0, sizeof(*out));
@@ -1102,10 +1133,11 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
     memcpy(kb, &b->key, 8);
     return memcmp(ka, kb, 8) < 0;
   });
-  if (live.size() > OBX_MAX_GROUPS) return OBX_BUF_NOT_ENOUGH;
-  out->n_groups = (uint32_t)live.size();
+  /* materialize ALL rows (the growth path's pagination source), then
+     surface the inline result when it fits */
+  h.last_rows.assign(live.size(), obx_group_row());
   for (size_t i = 0; i < live.size(); i++) {
-    obx_group_row *g = &out->groups[i];
+    obx_group_row *g = &h.last_rows[i];
     memcpy(g->key, &live[i]->key, 8);
     g->key_len = key_len;
     g->row_count = live[i]->count;
@@ -1132,6 +1164,15 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
       }
     }
   }
+  if (h.last_rows.size() > OBX_MAX_GROUPS) {
+    /* more groups than the inline result holds: rows stay paged behind
+       obx_gpu_agg_fetch (the ObHashGroupByOp growth path) */
+    out->n_groups = (uint32_t)h.last_rows.size();
+    return OBX_BUF_NOT_ENOUGH;
+  }
+  out->n_groups = (uint32_t)h.last_rows.size();
+  for (size_t i = 0; i < h.last_rows.size(); i++)
+    out->groups[i] = h.last_rows[i];
   return OBX_SUCCESS;
 }
 
@@ -1349,4 +1390,23 @@ extern "C" uint64_t obx_gpu_total_bytes(obx_gpu_ctx *ctx, int handle) {
 extern "C" uint64_t obx_gpu_last_survivors(obx_gpu_ctx *ctx, int handle) {
   if (!ctx || handle < 0 || handle >= (int)ctx->handles.size()) return 0;
   return ctx->handles[handle].last_survivors;
+}
+
+/* paged access to the last scan's full (sorted) group rows — the result
+ * surface for > OBX_MAX_GROUPS groups (the reference's hash group-by
+ * grows unboundedly, ob_exec_hash_struct_vec.h:1718; our device table
+ * holds OBX_GTABLE_SLOTS groups and the host pages them out). */
+extern "C" int obx_gpu_agg_fetch(obx_gpu_ctx *ctx, int handle,
+                                 uint32_t start, uint32_t count,
+                                 obx_group_row *out, uint32_t *n_out,
+                                 uint64_t *n_total) {
+  if (!ctx || handle < 0 || handle >= (int)ctx->handles.size() || !out)
+    return OBX_INVALID_ARGUMENT;
+  obx_handle &h = ctx->handles[handle];
+  uint64_t total = h.last_rows.size();
+  if (n_total) *n_total = total;
+  uint32_t n = 0;
+  for (; n < count && start + n < total; n++) out[n] = h.last_rows[start + n];
+  if (n_out) *n_out = n;
+  return OBX_SUCCESS;
 }

@@ -472,7 +472,7 @@ __device__ void scan_filter_agg_body(
   for (uint32_t s = tid; s < OBX_LTABLE_SLOTS; s += WG) {
     uint64_t key = tab.key[s];
     if (key == OBX_KEY_EMPTY) continue;
-    uint32_t idx = (uint32_t)((key * 0x9E3779B97F4A7C15ull) >> 56) &
+    uint32_t idx = (uint32_t)((key * 0x9E3779B97F4A7C15ull) >> (64 - OBX_GTABLE_SHIFT)) &
                    (OBX_GTABLE_SLOTS - 1);
     for (int probe = 0; probe < OBX_GTABLE_SLOTS; probe++) {
       unsigned long long cur_k = atomicCAS(&gtable[idx].key, OBX_KEY_EMPTY,
@@ -1091,7 +1091,7 @@ extern "C" __global__ void k_col_minmax(
 
 /* find-or-insert a key in the global group table; returns slot or -1 */
 __device__ __forceinline__ int g_slot(gslot *gtable, uint64_t key) {
-  uint32_t idx = (uint32_t)((key * 0x9E3779B97F4A7C15ull) >> 56) &
+  uint32_t idx = (uint32_t)((key * 0x9E3779B97F4A7C15ull) >> (64 - OBX_GTABLE_SHIFT)) &
                  (OBX_GTABLE_SLOTS - 1);
   for (int probe = 0; probe < OBX_GTABLE_SLOTS; probe++) {
     unsigned long long k = __hip_atomic_load(
@@ -1345,4 +1345,128 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_agg_pass(
       }
     }
   }
+}
+
+/* ---------------- high-cardinality direct-global scan --------------------
+ * Growth path past the per-workgroup LDS group table (the reference's
+ * ObHashGroupByOp hash table grows unboundedly,
+ * ob_exec_hash_struct_vec.h:1718): when the generic kernel reports an LDS
+ * table overflow, the host reruns the scan with this kernel. No LDS
+ * table: every surviving row probes the global group table (open
+ * addressing over OBX_GTABLE_SLOTS) and accumulates with global atomics.
+ * Correctness-first slow path — plans that fit the LDS table never take
+ * it. AND-combined leaves only (the host keeps OBX_BUF_NOT_ENOUGH for
+ * OR-programs). */
+__device__ __forceinline__ int g_slot_probe(gslot *gt, uint64_t key) {
+  uint32_t idx = (uint32_t)((key * 0x9E3779B97F4A7C15ull) >>
+                            (64 - OBX_GTABLE_BIG_SHIFT)) &
+                 (OBX_GTABLE_BIG - 1);
+  for (int p = 0; p < OBX_GTABLE_BIG; p++) {
+    unsigned long long k = gt[idx].key;
+    if (k == key) return (int)idx;
+    if (k == OBX_KEY_EMPTY) {
+      unsigned long long c = atomicCAS(&gt[idx].key, OBX_KEY_EMPTY,
+                                       (unsigned long long)key);
+      if (c == OBX_KEY_EMPTY || c == key) return (int)idx;
+    }
+    idx = (idx + 1) & (OBX_GTABLE_BIG - 1);
+  }
+  return -1;
+}
+
+extern "C" __global__ __launch_bounds__(WG, 2) void k_scan_agg_direct(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, const dev_leaf *__restrict__ plan_leaves,
+    const blk_leaf *__restrict__ bleaves, const dev_plan_hdr ph,
+    gslot *__restrict__ gtable, unsigned long long *__restrict__ counters) {
+  const uint32_t tid = threadIdx.x;
+  const uint32_t lane = tid & 63, wv = tid >> 6;
+  unsigned long long lane_cnt = 0;
+  for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
+    const dev_block &cur = blocks[b];
+    const blk_leaf *bl = bleaves + (size_t)b * ph.n_leaves;
+    bool skip = false;
+    for (uint32_t i = 0; i < ph.n_leaves; i++)
+      if (leaf_class(cur, plan_leaves[i], bl[i]) == 0) { skip = true; break; }
+    if (skip) continue;
+    blk_view bv;
+    bv.base = buf;
+    bv.bit_bias = 0;
+    bv.rbase_bit = 0;
+    const dev_col *gd0 = nullptr, *gd1 = nullptr;
+    if (ph.n_group_cols > 0) gd0 = &cur.cols[ph.need_cols[ph.group_idx[0]]];
+    if (ph.n_group_cols > 1) gd1 = &cur.cols[ph.need_cols[ph.group_idx[1]]];
+    const uint32_t kl0 = ph.group_len[0], kl1 = ph.group_len[1];
+    const uint32_t rows = cur.row_count;
+    for (uint32_t r = tid; r < rows; r += WG) {
+      bool live = true;
+      for (uint32_t i = 0; live && i < ph.n_leaves; i++)
+        live = leaf_match(bv, cur, plan_leaves[i], bl[i], r);
+      if (!live) continue;
+      lane_cnt++;
+      uint64_t key = build_group_key(bv, ph.n_group_cols, gd0, gd1, kl0,
+                                     kl1, r);
+      int gi = g_slot_probe(gtable, key);
+      if (gi < 0) { /* global table full: surfaced as counters[2] */
+        atomicAdd(&counters[2], 1ull);
+        continue;
+      }
+      atomicAdd(&gtable[gi].count, 1ull);
+      for (uint32_t a = 0; a < ph.n_aggs; a++) {
+        const dev_agg ag = ph.aggs[a];
+        if (ag.kind == 0 && ag.ia == 0xFF) continue; /* COUNT(*): count */
+        bool na = false, nb = false, nc = false;
+        int64_t va = 0, vb = 0, vc = 0;
+        if (ag.ia != 0xFF)
+          va = col_value2(bv, cur, cur.cols[ph.need_cols[ag.ia]], r, na);
+        if (ag.kind >= 4 && ag.ib != 0xFF)
+          vb = col_value2(bv, cur, cur.cols[ph.need_cols[ag.ib]], r, nb);
+        if (ag.kind == 5 && ag.ic != 0xFF)
+          vc = col_value2(bv, cur, cur.cols[ph.need_cols[ag.ic]], r, nc);
+        switch (ag.kind) {
+          case 0: /* COUNT(col) */
+            if (!na) g_acc_i128(gtable[gi].cells[a], 1ull, 0ull);
+            break;
+          case 1: /* SUM */
+            if (!na) {
+              i128v v = i128_from_i64(va);
+              g_acc_i128(gtable[gi].cells[a], v.lo, (uint64_t)v.hi);
+            }
+            break;
+          case 2:
+          case 3: /* MIN / MAX */
+            if (!na) {
+              cas_minmax(&gtable[gi].cells[a][0], va, ag.kind == 2);
+              gtable[gi].cells[a][1] = 1;
+            }
+            break;
+          case 4: /* SUM_PROD2 */
+            if (!na && !nb) {
+              i128v p2 = i128_mul_i64(va, ag.one_b - vb);
+              g_acc_i128(gtable[gi].cells[a], p2.lo, (uint64_t)p2.hi);
+            }
+            break;
+          case 5: /* SUM_PROD3 */
+            if (!na && !nb && !nc) {
+              i128v p3 = i128_mul_pos_i64(i128_mul_i64(va, ag.one_b - vb),
+                                          ag.one_c + vc);
+              g_acc_i128(gtable[gi].cells[a], p3.lo, (uint64_t)p3.hi);
+            }
+            break;
+          case 6: /* SUM_MUL */
+            if (!na && !nb) {
+              i128v p = i128_mul_i64(va, vb);
+              g_acc_i128(gtable[gi].cells[a], p.lo, (uint64_t)p.hi);
+            }
+            break;
+          default:
+            break;
+        }
+      }
+    }
+  }
+  for (int off = 32; off > 0; off >>= 1)
+    lane_cnt += (unsigned long long)__shfl_xor((long long)lane_cnt, off, 64);
+  if (lane == 0 && lane_cnt)
+    atomicAdd(&counters[8 + ((blockIdx.x * WAVES + wv) & 7)], lane_cnt);
 }

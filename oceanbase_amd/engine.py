@@ -191,6 +191,38 @@ class GpuEngine:
             raise RuntimeError(f"obx_gpu_scan_filter_agg failed: {rc}")
         return res
 
+    def scan_filter_agg_paged(self, handle, filter_desc, agg_desc):
+        """Scan allowing > OBX_MAX_GROUPS groups: returns
+        (AggResult, [GroupRow...]) with the FULL sorted row list fetched
+        through the pagination surface (obx_gpu_agg_fetch)."""
+        res = abi.AggResult()
+        rc = self._lib.obx_gpu_scan_filter_agg(
+            self._ctx, handle,
+            C.byref(filter_desc) if filter_desc is not None else None,
+            C.byref(agg_desc) if agg_desc is not None else None,
+            C.byref(res))
+        if rc not in (0, abi.OBX_BUF_NOT_ENOUGH):
+            raise RuntimeError(f"obx_gpu_scan_filter_agg failed: {rc}")
+        return res, self.agg_fetch_all(handle)
+
+    def agg_fetch_all(self, handle, page=256):
+        rows, start = [], 0
+        buf = (abi.GroupRow * page)()
+        n_out = C.c_uint32()
+        total = C.c_uint64()
+        while True:
+            rc = self._lib.obx_gpu_agg_fetch(self._ctx, handle, start, page,
+                                             buf, C.byref(n_out),
+                                             C.byref(total))
+            if rc != 0:
+                raise RuntimeError(f"obx_gpu_agg_fetch failed: {rc}")
+            for i in range(n_out.value):
+                rows.append(abi.GroupRow.from_buffer_copy(buf[i]))
+            start += n_out.value
+            if start >= total.value or n_out.value == 0:
+                break
+        return rows
+
     def last_kernel_ms(self):
         return float(self._lib.obx_gpu_last_kernel_ms(self._ctx))
 

@@ -214,3 +214,29 @@ def scan_filter_agg(bs, filter_desc, agg_desc, nthreads=0):
     if st != abi.OBX_SUCCESS:
         raise RuntimeError(f"obx_cpu_scan_filter_agg failed: {st}")
     return res
+
+
+def scan_filter_agg_paged(bs, filter_desc, agg_desc, nthreads=0):
+    """Scan allowing > OBX_MAX_GROUPS groups; returns (AggResult, rows)."""
+    res = abi.AggResult()
+    st = _lib.obx_cpu_scan_filter_agg(
+        C.byref(bs), C.byref(filter_desc) if filter_desc is not None else None,
+        C.byref(agg_desc) if agg_desc is not None else None, nthreads,
+        C.byref(res))
+    if st not in (abi.OBX_SUCCESS, abi.OBX_BUF_NOT_ENOUGH):
+        raise RuntimeError(f"obx_cpu_scan_filter_agg failed: {st}")
+    rows, start, page = [], 0, 256
+    buf = (abi.GroupRow * page)()
+    n_out = C.c_uint32()
+    total = C.c_uint64()
+    while True:
+        rc = _lib.obx_cpu_agg_fetch(start, page, buf, C.byref(n_out),
+                                    C.byref(total))
+        if rc != 0:
+            raise RuntimeError(f"obx_cpu_agg_fetch failed: {rc}")
+        for i in range(n_out.value):
+            rows.append(abi.GroupRow.from_buffer_copy(buf[i]))
+        start += n_out.value
+        if start >= total.value or n_out.value == 0:
+            break
+    return res, rows

@@ -123,13 +123,17 @@ typedef struct grp {
   int has_val[8]; /* for MIN/MAX init */
 } grp;
 
+#define OBX_CPU_MAX_GROUPS 65536 /* growth cap (the reference's table
+                                     grows unboundedly; this bounds test
+                                     memory) */
 typedef struct work_ctx {
   const obx_blockset *bs;
   const obx_filter_desc *filter;
   const obx_agg_desc *agg;
   _Atomic uint64_t *next_block;
-  grp groups[OBX_MAX_GROUPS];
-  uint32_t n_groups;
+  grp *groups;          /* grown on demand (ob_exec_hash_struct_vec.h:1718
+                           grows unboundedly) */
+  uint32_t n_groups, gcap;
   uint64_t rows_scanned, rows_passed;
   int rc;
 } work_ctx;
@@ -143,7 +147,14 @@ static grp *find_group(work_ctx *w, const uint8_t *key, uint8_t key_len) {
     if (g->key_len == key_len && memcmp(g->key, key, (size_t)key_len + 1) == 0)
       return g;
   }
-  if (w->n_groups >= OBX_MAX_GROUPS) return NULL;
+  if (w->n_groups >= w->gcap) {
+    uint32_t nc = w->gcap ? w->gcap * 2 : OBX_MAX_GROUPS;
+    if (nc > OBX_CPU_MAX_GROUPS) return NULL;
+    grp *ng = (grp *)realloc(w->groups, (size_t)nc * sizeof(grp));
+    if (!ng) return NULL;
+    w->groups = ng;
+    w->gcap = nc;
+  }
   grp *g = &w->groups[w->n_groups++];
   memset(g, 0, sizeof(*g));
   memcpy(g->key, key, (size_t)key_len + 1);
@@ -338,6 +349,9 @@ static void *worker(void *arg) {
   return NULL;
 }
 
+static obx_group_row *g_last_rows = NULL;
+static uint64_t g_last_n = 0;
+
 int obx_cpu_scan_filter_agg(const obx_blockset *bs,
                             const obx_filter_desc *filter,
                             const obx_agg_desc *agg, int nthreads,
@@ -414,16 +428,49 @@ int obx_cpu_scan_filter_agg(const obx_blockset *bs,
     }
     merged.groups[j] = tmp;
   }
-  out->n_groups = merged.n_groups;
-  for (uint32_t i = 0; i < merged.n_groups; i++) {
-    grp *s = &merged.groups[i];
-    obx_group_row *o = &out->groups[i];
-    memcpy(o->key, s->key, OBX_MAX_KEY_BYTES);
-    o->key_len = s->key_len;
-    o->row_count = s->row_count;
-    for (int a = 0; a < 8; a++)
-      memcpy(&o->cells[a], &s->cell[a], sizeof(obx_agg_cell));
+  /* cache ALL rows for paged access (obx_cpu_agg_fetch), surface the
+     inline result when it fits */
+  free(g_last_rows);
+  g_last_rows = NULL;
+  g_last_n = 0;
+  if (merged.n_groups) {
+    g_last_rows =
+        (obx_group_row *)calloc(merged.n_groups, sizeof(obx_group_row));
+    if (g_last_rows) {
+      for (uint32_t i = 0; i < merged.n_groups; i++) {
+        grp *s = &merged.groups[i];
+        obx_group_row *o = &g_last_rows[i];
+        memcpy(o->key, s->key, OBX_MAX_KEY_BYTES);
+        o->key_len = s->key_len;
+        o->row_count = s->row_count;
+        for (int a = 0; a < 8; a++)
+          memcpy(&o->cells[a], &s->cell[a], sizeof(obx_agg_cell));
+      }
+      g_last_n = merged.n_groups;
+    }
   }
+  out->n_groups = merged.n_groups;
+  if (merged.n_groups <= OBX_MAX_GROUPS) {
+    for (uint32_t i = 0; i < merged.n_groups; i++)
+      out->groups[i] = g_last_rows[i];
+  } else if (rc == OBX_SUCCESS) {
+    rc = OBX_BUF_NOT_ENOUGH; /* rows stay paged behind obx_cpu_agg_fetch */
+  }
+  for (int t = 0; t < nthreads; t++) free(ws[t].groups);
+  free(merged.groups);
   free(ws); free(tids);
   return rc;
+}
+
+/* paged access to the last scan's sorted group rows (single-threaded test
+ * infrastructure; mirrors obx_gpu_agg_fetch) */
+int obx_cpu_agg_fetch(uint32_t start, uint32_t count, obx_group_row *out,
+                      uint32_t *n_out, uint64_t *n_total) {
+  if (!out) return OBX_INVALID_ARGUMENT;
+  if (n_total) *n_total = g_last_n;
+  uint32_t n = 0;
+  for (; n < count && start + n < g_last_n; n++)
+    out[n] = g_last_rows[start + n];
+  if (n_out) *n_out = n;
+  return OBX_SUCCESS;
 }

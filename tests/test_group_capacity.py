@@ -1,0 +1,142 @@
+"""Group-by capacity growth past OBX_MAX_GROUPS (=64 inline result rows).
+
+The reference's hash group-by grows its table unboundedly
+(ob_exec_hash_struct_vec.h:1718). Here: a scan with more groups returns
+OBX_BUF_NOT_ENOUGH with n_groups = the true total and the full sorted
+rows are paged out through obx_{cpu,gpu}_agg_fetch; on device, the
+generic kernel's LDS-table overflow triggers a rerun through the
+direct-global high-cardinality kernel (k_scan_agg_direct, up to
+OBX_GTABLE_BIG=4096 distinct groups).
+"""
+import ctypes as C
+
+import numpy as np
+import pytest
+
+from oceanbase_amd import abi, oracle
+
+
+def _blockset(schema, blocks_bytes, total_rows):
+    aligned, offs = [], [0]
+    for b in blocks_bytes:
+        body = b[:-16]
+        pad = (-len(body)) % 16
+        aligned.append(body + b"\x00" * pad)
+        offs.append(offs[-1] + len(body) + pad)
+    data = np.frombuffer(b"".join(aligned) + b"\x00" * 16, dtype=np.uint8)
+    offarr = np.array(offs, dtype=np.uint64)
+    bs = abi.BlockSet()
+    bs.data = data.ctypes.data_as(C.POINTER(C.c_uint8))
+    bs.block_offsets = offarr.ctypes.data_as(C.POINTER(C.c_uint64))
+    bs.n_blocks = len(blocks_bytes)
+    bs.n_cols = len(schema)
+    bs.cols = C.cast(schema, C.POINTER(abi.ColSchema))
+    bs.total_rows = total_rows
+    bs._keep = (data, offarr)
+    return bs
+
+
+def _make_many_groups(seed=7, rows_per_block=1024, n_blocks=12,
+                      d0=18, d1=17):
+    rng = np.random.default_rng(seed)
+    schema = oracle.make_schema([(abi.T_INT, 0, 0, 8),
+                                 (abi.T_CHAR, 0, 0, 1),
+                                 (abi.T_CHAR, 0, 0, 1)])
+    blocks, v_all, g0_all, g1_all = [], [], [], []
+    for _ in range(n_blocks):
+        vals = rng.integers(-10000, 10000, rows_per_block).astype(np.int64)
+        g0 = (65 + rng.integers(0, d0, rows_per_block)).astype(np.uint8)
+        g1 = (97 + rng.integers(0, d1, rows_per_block)).astype(np.uint8)
+        blocks.append(oracle.encode_block(
+            schema, [vals.view(np.uint8), g0, g1],
+            [abi.ENC_RAW, abi.ENC_DICT, abi.ENC_DICT]))
+        v_all.append(vals)
+        g0_all.append(g0)
+        g1_all.append(g1)
+    bs = _blockset(schema, blocks, rows_per_block * n_blocks)
+    return (bs, np.concatenate(v_all), np.concatenate(g0_all),
+            np.concatenate(g1_all))
+
+
+def _expected(vals, g0, g1, lo):
+    """Python recompute: filter val < lo, group by (g0,g1)."""
+    sel = vals < lo
+    groups = {}
+    for v, a, b in zip(vals[sel], g0[sel], g1[sel]):
+        key = bytes([a, b])
+        c = groups.setdefault(key, [0, 0, None])
+        c[0] += 1
+        c[1] += int(v)
+        c[2] = int(v) if c[2] is None else min(c[2], int(v))
+    return dict(sorted(groups.items()))
+
+
+def _descs():
+    filt = abi.make_filter([dict(col=0, op=abi.OP_LT, lo=5000)])
+    agg = abi.make_agg([1, 2], [dict(kind=abi.AGG_COUNT),
+                                dict(kind=abi.AGG_SUM, col_a=0),
+                                dict(kind=abi.AGG_MIN, col_a=0)])
+    return filt, agg
+
+
+def test_oracle_group_growth_paged():
+    bs, vals, g0, g1 = _make_many_groups()
+    filt, agg = _descs()
+    # inline result refuses (>64 groups) ...
+    with pytest.raises(RuntimeError):
+        oracle.scan_filter_agg(bs, filt, agg)
+    # ... the paged surface returns everything
+    res, rows = oracle.scan_filter_agg_paged(bs, filt, agg)
+    exp = _expected(vals, g0, g1, 5000)
+    assert res.n_groups == len(exp) > 64
+    got = {k: (cnt, cells[1], cells[2])
+           for k, cnt, cells in abi.group_row_tuples(rows, 3)}
+    assert set(got.keys()) == set(exp.keys())
+    for k, (cnt, s, mn) in exp.items():
+        assert got[k] == (cnt, s, mn), k
+    # pagination windows agree with the full fetch
+    buf = (abi.GroupRow * 10)()
+    n_out = C.c_uint32()
+    total = C.c_uint64()
+    lib = oracle._lib
+    assert lib.obx_cpu_agg_fetch(5, 10, buf, C.byref(n_out),
+                                 C.byref(total)) == 0
+    assert total.value == len(exp) and n_out.value == 10
+    assert bytes(buf[0].key[:buf[0].key_len]) == sorted(exp)[5]
+
+
+def test_oracle_small_result_still_inline():
+    bs, vals, g0, g1 = _make_many_groups(d0=4, d1=3)
+    filt, agg = _descs()
+    res = oracle.scan_filter_agg(bs, filt, agg)  # <=64 groups: no error
+    exp = _expected(vals, g0, g1, 5000)
+    assert res.n_groups == len(exp) <= 64
+
+
+@pytest.mark.gpu
+def test_gpu_group_growth_matches_oracle():
+    from oceanbase_amd.engine import GpuEngine
+    eng = GpuEngine()
+    bs, vals, g0, g1 = _make_many_groups()
+    filt, agg = _descs()
+    h = eng.load(bs)
+    with pytest.raises(RuntimeError):
+        eng.scan_filter_agg(h, filt, agg)  # inline refuses
+    res, rows = eng.scan_filter_agg_paged(h, filt, agg)
+    _, crows = oracle.scan_filter_agg_paged(bs, filt, agg)
+    assert res.n_groups == len(crows) > 64
+    assert abi.group_row_tuples(rows, 3) == abi.group_row_tuples(crows, 3)
+    eng.free(h)
+
+
+@pytest.mark.gpu
+def test_gpu_fetch_small_result():
+    from oceanbase_amd.engine import GpuEngine
+    eng = GpuEngine()
+    bs, vals, g0, g1 = _make_many_groups(d0=4, d1=3)
+    filt, agg = _descs()
+    h = eng.load(bs)
+    res = eng.scan_filter_agg(h, filt, agg)
+    rows = eng.agg_fetch_all(h)
+    assert abi.group_row_tuples(rows, 3) == abi.result_rows(res, 3)
+    eng.free(h)
