@@ -1096,7 +1096,15 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
       HIP_TRY(hipMalloc(&h.d_gtable_big, sizeof(gslot) * OBX_GTABLE_BIG));
     std::vector<gslot> init(OBX_GTABLE_BIG);
     memset(init.data(), 0, sizeof(gslot) * OBX_GTABLE_BIG);
-    for (auto &s2 : init) s2.key = OBX_KEY_EMPTY;
+    for (auto &s2 : init) {
+      s2.key = OBX_KEY_EMPTY;
+      for (uint32_t a = 0; a < ph.n_aggs; a++) {
+        if (ph.aggs[a].kind == OBX_AGG_MIN)
+          s2.cells[a][0] = (unsigned long long)INT64_MAX;
+        else if (ph.aggs[a].kind == OBX_AGG_MAX)
+          s2.cells[a][0] = (unsigned long long)INT64_MIN;
+      }
+    }
     HIP_TRY(hipMemcpyAsync(h.d_gtable_big, init.data(),
                            sizeof(gslot) * OBX_GTABLE_BIG,
                            hipMemcpyHostToDevice, ctx->stream));
@@ -1355,6 +1363,7 @@ extern "C" int64_t obx_jit_dump_src(
   if (!jit_blocks_ok(h, ph, js)) return 0;
   jit_strategy st;
   if (!force_v1) jit_build_strategy(h, ph, js, pl, st);
+  if (!st.ok && js.has_mm) return 0; /* v1 cannot do MIN/MAX */
   std::string src = st.ok ? jit_gen_source_v2(ph, js, st)
                           : jit_gen_source(ph, js);
   if (out && cap > (int64_t)src.size()) {

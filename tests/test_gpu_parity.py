@@ -435,3 +435,25 @@ def test_stored_minmax_skip_index_parity():
         exp = sum(int(((a >= lo) & (a <= hi)).sum()) for a in av)
         assert got == exp, (lo, hi)
     eng.close()
+
+
+def test_q1_with_minmax_takes_v2_jit(eng):
+    """MIN/MAX aggregates no longer force the generic fallback: CTX
+    columns accumulate in LDS CAS cells, dict columns read their min/max
+    off the histogram at flush — all bit-exact vs the oracle."""
+    li, h = _load_lineitem(eng, 4, 150000)
+    filt = abi.make_filter(
+        [dict(col=6, op=abi.OP_LE, lo=oracle.date_days(1998, 9, 2))])
+    aggs = [dict(kind=abi.AGG_COUNT),
+            dict(kind=abi.AGG_SUM, col_a=1),
+            dict(kind=abi.AGG_MIN, col_a=1),   # CTX raw8 -> wmm
+            dict(kind=abi.AGG_MAX, col_a=1),
+            dict(kind=abi.AGG_MIN, col_a=0),   # dict col -> hist flush
+            dict(kind=abi.AGG_MAX, col_a=0)]
+    agg = abi.make_agg([4, 5], aggs)
+    res_gpu = eng.scan_filter_agg(h, filt, agg)
+    import ctypes as Ct
+    assert eng._lib.obx_gpu_last_jit(eng._ctx) == 2  # persistent v2 ran
+    res_cpu = oracle.scan_filter_agg(li.bs, filt, agg)
+    assert abi.result_rows(res_gpu, 6) == abi.result_rows(res_cpu, 6)
+    eng.free(h)

@@ -153,7 +153,9 @@ def _make_null_groups(seed=13, rows_per_block=1024, n_blocks=10,
                                  (abi.T_CHAR, 0, 0, 1)])
     blocks = []
     for _ in range(n_blocks):
-        v0 = rng.integers(-5000, 5000, rows_per_block).astype(np.int64)
+        # all-positive values: a zero-initialized (sentinel-less) min/max
+        # table would return 0 instead of the true min (regression guard)
+        v0 = rng.integers(1000, 5000, rows_per_block).astype(np.int64)
         v1 = rng.integers(0, 50, rows_per_block).astype(np.int64)
         v2 = rng.integers(0, 9, rows_per_block).astype(np.int64)
         nulls = np.zeros((rows_per_block + 7) // 8, dtype=np.uint8)
@@ -178,6 +180,7 @@ def _null_descs():
         [dict(kind=abi.AGG_COUNT, col_a=0),      # COUNT(col) with NULLs
          dict(kind=abi.AGG_SUM, col_a=0),
          dict(kind=abi.AGG_MAX, col_a=0),
+         dict(kind=abi.AGG_MIN, col_a=0),
          dict(kind=abi.AGG_SUM_PROD2, col_a=0, col_b=2),
          dict(kind=abi.AGG_SUM_PROD3, col_a=0, col_b=2, col_c=1)])
     return filt, agg
@@ -204,7 +207,7 @@ def test_gpu_growth_all_agg_kinds_matches_oracle():
     cres, crows = oracle.scan_filter_agg_paged(bs, filt, agg)
     assert res.n_groups == cres.n_groups > 64
     assert res.rows_passed == cres.rows_passed
-    assert abi.group_row_tuples(rows, 5) == abi.group_row_tuples(crows, 5)
+    assert abi.group_row_tuples(rows, 6) == abi.group_row_tuples(crows, 6)
     eng.free(h)
 
 

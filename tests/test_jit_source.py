@@ -51,3 +51,39 @@ def test_filter_jit_source_strip_mined():
     assert sz > 0
     src = buf.raw[:sz].decode()
     assert "__ballot" in src
+
+
+def test_v2_scan_source_min_max():
+    """MIN/MAX aggregates are v2-eligible: CTX columns via LDS CAS cells
+    (wmm), dict columns read off the histogram at flush."""
+    from oceanbase_amd import oracle
+    lib = C.CDLL(os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        "oceanbase_amd", "libobx.so"))
+    filt = abi.make_filter([dict(col=6, op=abi.OP_LE,
+                                 lo=oracle.date_days(1998, 9, 2))])
+    aggs = [dict(kind=abi.AGG_COUNT),
+            dict(kind=abi.AGG_SUM, col_a=1),
+            dict(kind=abi.AGG_MIN, col_a=1),    # CTX (raw8) -> wmm
+            dict(kind=abi.AGG_MAX, col_a=0)]    # dict col -> hist flush
+    agg = abi.make_agg([4, 5], aggs)
+    n = 7
+    cols = (abi.ColSchema * n)()
+    for c in range(n):
+        cols[c].obj_type = abi.T_INT
+    flags = (C.c_uint8 * n)(*[37, 12, 37, 37, 37, 37, 20])
+    cmin = (C.c_int64 * n)(*[0, 90000, 0, 0, 0, 0, 8000])
+    cmax = (C.c_int64 * n)(*[50, 10500000, 10, 8, 2, 1, 11000])
+    maxcnt = (C.c_uint32 * n)(*[50, 0, 11, 9, 3, 2, 0])
+    maxw = (C.c_uint32 * n)(*[6, 64, 4, 4, 2, 1, 13])
+    buf = C.create_string_buffer(1 << 20)
+    lib.obx_jit_dump_src.restype = C.c_int64
+    sz = lib.obx_jit_dump_src(C.byref(filt), C.byref(agg), cols, n, flags,
+                              cmin, cmax, maxcnt, maxw, 1365, 0, buf,
+                              len(buf))
+    assert sz > 0
+    src = buf.raw[:sz].decode()
+    assert "k_jit_scan" in src
+    assert "wmm[" in src                    # CAS min cells (CTX MIN)
+    assert "cas_minmax" in src
+    assert "INT64_MAX" in src               # MIN sentinel init
