@@ -1,0 +1,147 @@
+"""Deeper seeded fuzzing (CPU-only surfaces).
+
+1. Black-filter programs: 60 random valid postfix programs over random
+   multi-column data — oracle bitmap vs the independent python evaluator.
+2. CS integer codecs vs the reference binaries (when oracle/_ref is
+   built): 150 random (codec, width, count, distribution) shapes,
+   byte-identical encode + reference-decodes-our-bytes.
+"""
+import ctypes as C
+import os
+import random
+
+import numpy as np
+import pytest
+
+from oceanbase_amd import abi, oracle
+from test_black_filter import py_black_eval
+
+M64 = (1 << 64) - 1
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _rand_prog(rng, n_cols, n_consts):
+    """Random postfix program ending with a comparison/logic root."""
+    arith = [abi.BX_ADD, abi.BX_SUB, abi.BX_MUL, abi.BX_DIV]
+    cmps = [abi.BX_LT, abi.BX_LE, abi.BX_GT, abi.BX_GE, abi.BX_EQ,
+            abi.BX_NE]
+    prog = []
+    depth = 0
+    # build 1-3 comparison terms, then AND/OR them
+    n_terms = rng.randint(1, 3)
+    for _ in range(n_terms):
+        # each side: col/const with optional arith
+        for _side in range(2):
+            prog.append(rng.choice(
+                [abi.BX_COL | rng.randrange(n_cols),
+                 abi.BX_CONST | rng.randrange(n_consts)]))
+            if rng.random() < 0.4:
+                prog.append(rng.choice(
+                    [abi.BX_COL | rng.randrange(n_cols),
+                     abi.BX_CONST | rng.randrange(n_consts)]))
+                prog.append(rng.choice(arith))
+            if rng.random() < 0.15:
+                prog.append(abi.BX_NEG)
+        prog.append(rng.choice(cmps))
+        depth += 1
+    while depth > 1:
+        prog.append(rng.choice([abi.BX_AND, abi.BX_OR]))
+        depth -= 1
+    if rng.random() < 0.2:
+        prog.append(abi.BX_NOT)
+    return prog
+
+
+def test_black_program_fuzz_oracle_vs_pymodel():
+    rng = random.Random(424242)
+    nrng = np.random.default_rng(11)
+    rows = 800
+    n_cols, n_consts = 3, 3
+    data = [nrng.integers(-1000, 1000, rows).astype(np.int64)
+            for _ in range(n_cols)]
+    null_rows = set(range(0, rows, 17))
+    nulls = np.zeros((rows + 7) // 8, dtype=np.uint8)
+    for r in null_rows:
+        nulls[r >> 3] |= 1 << (r & 7)
+    schema = oracle.make_schema([(abi.T_INT, 0, 0, 8)] * n_cols)
+    blk = oracle.encode_block(
+        schema, [d.view(np.uint8) for d in data],
+        [abi.ENC_RAW] * n_cols, [nulls, None, None])
+    n_checked = 0
+    for i in range(60):
+        prog = _rand_prog(rng, n_cols, n_consts)
+        if len(prog) > abi.__dict__.get("BX_MAX_PROG", 24) and len(prog) > 24:
+            continue
+        consts = [rng.randint(-500, 500) for _ in range(n_consts)]
+        filt = abi.make_filter([dict(op=abi.OP_BLACK,
+                                     bcols=list(range(n_cols)),
+                                     bconst=consts, bprog=prog)])
+        bits, pc = oracle.filter_block(schema, n_cols, blk, filt)
+        exp = 0
+        for r in range(rows):
+            vals = [int(d[r]) for d in data]
+            nl = [r in null_rows, False, False]
+            want = py_black_eval(prog, consts, vals, nl)
+            got = bool(bits[r >> 3] & (1 << (r & 7)))
+            assert got == want, (i, r, prog)
+            exp += want
+        assert pc == exp
+        n_checked += 1
+    assert n_checked >= 50
+
+
+_ref_so = os.path.join(REPO, "oracle", "_ref", "librefcodec.so")
+needs_ref = pytest.mark.skipif(not os.path.exists(_ref_so),
+                               reason="oracle/_ref not built here")
+
+
+@needs_ref
+def test_cs_codec_fuzz_vs_reference():
+    from test_ref_parity import _codec_libs, CODECS
+    ref, ours = _codec_libs()
+    rng = np.random.default_rng(20260915)
+    FN = dict((e, f) for f, e in CODECS)
+    n = 0
+    for _ in range(150):
+        enc_type = int(rng.choice(list(FN.keys())))
+        wb = int(rng.choice([1, 2, 4, 8]))
+        cnt = int(rng.integers(1, 700))
+        if enc_type == 6:
+            # the reference SIMD codec pads to whole 128-value frames;
+            # stream encoders only select it for full frames
+            cnt = max(128, (cnt // 128) * 128)
+        mask = (1 << (8 * wb)) - 1
+        kind = rng.integers(0, 4)
+        if kind == 0:      # random
+            vals = rng.integers(0, 1 << min(8 * wb, 63), cnt,
+                                dtype=np.uint64) & mask
+        elif kind == 1:    # ramp with jitter
+            vals = (np.arange(cnt, dtype=np.uint64) * 3 +
+                    rng.integers(0, 5, cnt, dtype=np.uint64)) & mask
+        elif kind == 2:    # runs
+            vals = np.repeat(rng.integers(0, 100, max(cnt // 7, 1),
+                                          dtype=np.uint64),
+                             7)[:cnt].copy() & mask
+            if len(vals) < cnt:
+                vals = np.pad(vals, (0, cnt - len(vals)))
+        else:              # small range
+            vals = rng.integers(0, 17, cnt, dtype=np.uint64) & mask
+        raw = vals.astype(f"<u{wb}").tobytes()
+        f = getattr(ours, FN[enc_type])
+        f.restype = C.c_int64
+        f.argtypes = [C.c_char_p, C.c_uint32, C.c_uint32, C.c_char_p,
+                      C.c_size_t]
+        cap = len(raw) * 3 + 8192
+        b1 = C.create_string_buffer(cap)
+        b2 = C.create_string_buffer(cap)
+        n1 = f(raw, cnt, wb, b1, cap)
+        n2 = ref.ref_codec_encode(enc_type, raw, cnt, wb, b2, cap)
+        assert n1 == n2 and b1.raw[:n1] == b2.raw[:n2], (enc_type, wb, cnt,
+                                                         int(kind))
+        if not (enc_type == 6 and cnt % 128):
+            dec = C.create_string_buffer(len(raw))
+            m = ref.ref_codec_decode(enc_type, b1.raw[:n1], n1, cnt, wb,
+                                     dec, len(raw))
+            assert m == n1 and dec.raw == raw, (enc_type, wb, cnt)
+        n += 1
+    assert n == 150
