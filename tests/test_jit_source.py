@@ -87,3 +87,34 @@ def test_v2_scan_source_min_max():
     assert "wmm[" in src                    # CAS min cells (CTX MIN)
     assert "cas_minmax" in src
     assert "INT64_MAX" in src               # MIN sentinel init
+
+
+def test_filter_jit_folds_or_program():
+    """AND/OR combine programs compile to one closed-form expression in
+    the staged filter JIT (block verdict in three-valued logic, row eval
+    as booleans)."""
+    lib = C.CDLL(os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        "oceanbase_amd", "libobx.so"))
+    filt = abi.make_filter(
+        [dict(col=0, op=abi.OP_LE, lo=10561),
+         dict(col=1, op=abi.OP_GT, lo=5),
+         dict(col=2, op=abi.OP_EQ, lo=3)],
+        prog=[0, 1, abi.TOK_AND, 2, abi.TOK_OR])
+    n = 4
+    cols = (abi.ColSchema * n)()
+    for c in range(n):
+        cols[c].obj_type = abi.T_INT
+    flags = (C.c_uint8 * n)(*[16] * n)
+    cmin = (C.c_int64 * n)()
+    cmax = (C.c_int64 * n)(*[30000] * n)
+    maxcnt = (C.c_uint32 * n)()
+    maxw = (C.c_uint32 * n)(*[13, 8, 6, 20])
+    buf = C.create_string_buffer(1 << 20)
+    lib.obx_jit_dump_src.restype = C.c_int64
+    sz = lib.obx_jit_dump_src(C.byref(filt), None, cols, n, flags, cmin,
+                              cmax, maxcnt, maxw, 4000, 0, buf, len(buf))
+    assert sz > 0
+    src = buf.raw[:sz].decode()
+    assert "f3or(f3and(c0, c1), c2)" in src      # block verdict fold
+    assert "(lf0 && lf1) || lf2" in src          # row fold

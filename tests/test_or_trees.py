@@ -102,6 +102,9 @@ def test_gpu_or_tree_parity(case_seed):
     h = eng.load(li.bs)
     survivors = eng.filter(h, fd)
     assert survivors == res_cpu.rows_passed, (case_seed, prog)
+    # the bitmap path compiles the combine program into the staged
+    # filter JIT (codegen-folded expression)
+    assert eng._lib.obx_gpu_last_jit(eng._ctx) == 2, (case_seed, prog)
     # and through the fused agg path (scalar count)
     agg = abi.make_agg([], [dict(kind=abi.AGG_COUNT)])
     res_gpu = eng.scan_filter_agg(h, fd, agg)
