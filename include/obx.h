@@ -120,6 +120,7 @@ enum obx_white_op {
  *   0x00|i  push column value i (obx_filter_leaf.bcols[i])
  *   0x40|i  push constant i     (obx_filter_leaf.bconst[i])
  *   0x50 ADD  0x51 SUB  0x52 MUL  0x53 DIV (int; x/0 -> NULL)  0x54 NEG
+ *   0x55 MOD (x%0 -> NULL; INT64_MIN%-1 == 0)
  *   0x60 LT  0x61 LE  0x62 GT  0x63 GE  0x64 EQ  0x65 NE (-> bool)
  *   0x70 AND  0x71 OR  0x72 NOT (three-valued)
  * Stack depth <= 8. The program leaves one value; the row passes iff it

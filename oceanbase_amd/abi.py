@@ -49,6 +49,7 @@ class FilterLeaf(C.Structure):
 BX_COL = 0x00   # | slot index (into the leaf's bcols)
 BX_CONST = 0x40  # | const index
 BX_ADD, BX_SUB, BX_MUL, BX_DIV, BX_NEG = 0x50, 0x51, 0x52, 0x53, 0x54
+BX_MOD = 0x55
 BX_LT, BX_LE, BX_GT, BX_GE, BX_EQ, BX_NE = 0x60, 0x61, 0x62, 0x63, 0x64, 0x65
 BX_AND, BX_OR, BX_NOT = 0x70, 0x71, 0x72
 

@@ -557,6 +557,11 @@ __device__ __noinline__ bool black_eval_vals(const dev_leaf &plf,
           else if (a == INT64_MIN && b2 == -1) r2 = a;
           else r2 = a / b2;
           break;
+        case 0x55: /* MOD: x % 0 -> NULL; INT64_MIN % -1 == 0 */
+          if (b2 == 0) rn = true;
+          else if (a == INT64_MIN && b2 == -1) r2 = 0;
+          else r2 = a % b2;
+          break;
         case 0x60: r2 = a < b2; break;
         case 0x61: r2 = a <= b2; break;
         case 0x62: r2 = a > b2; break;

@@ -1732,6 +1732,11 @@ int obx__black_eval(const obx_filter_leaf *lf, const int64_t *vals,
           else if (a == INT64_MIN && b == -1) r = a;
           else r = a / b;
           break;
+        case 0x55: /* MOD: x % 0 -> NULL; INT64_MIN % -1 == 0 */
+          if (b == 0) rn = 1;
+          else if (a == INT64_MIN && b == -1) r = 0;
+          else r = a % b;
+          break;
         case 0x60: r = a < b; break;
         case 0x61: r = a <= b; break;
         case 0x62: r = a > b; break;
@@ -1774,8 +1779,8 @@ int obx__bprog_valid(const obx_filter_leaf *lf) {
       sp++;
     } else if (op == 0x54 || op == 0x72) {
       if (sp < 1) return 0;
-    } else if ((op >= 0x50 && op <= 0x53) || (op >= 0x60 && op <= 0x65) ||
-               op == 0x70 || op == 0x71) {
+    } else if ((op >= 0x50 && op <= 0x53) || op == 0x55 ||
+               (op >= 0x60 && op <= 0x65) || op == 0x70 || op == 0x71) {
       if (sp < 2) return 0;
       sp--;
     } else {

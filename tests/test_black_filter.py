@@ -60,6 +60,14 @@ def py_black_eval(prog, consts, vals, nulls):
                     r = a
                 else:
                     r = int(a / b) if (a < 0) != (b < 0) and a % b else a // b
+            elif op == 0x55:  # MOD, C truncated-division remainder
+                if b == 0:
+                    rn = True
+                elif a == -(1 << 63) and b == -1:
+                    r = 0
+                else:
+                    q = int(a / b) if (a < 0) != (b < 0) and a % b else a // b
+                    r = s64(a - q * b)
             elif 0x60 <= op <= 0x65:
                 r = int({0x60: a < b, 0x61: a <= b, 0x62: a > b,
                          0x63: a >= b, 0x64: a == b, 0x65: a != b}[op])

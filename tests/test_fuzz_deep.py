@@ -22,7 +22,8 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 def _rand_prog(rng, n_cols, n_consts):
     """Random postfix program ending with a comparison/logic root."""
-    arith = [abi.BX_ADD, abi.BX_SUB, abi.BX_MUL, abi.BX_DIV]
+    arith = [abi.BX_ADD, abi.BX_SUB, abi.BX_MUL, abi.BX_DIV,
+             abi.BX_MOD]
     cmps = [abi.BX_LT, abi.BX_LE, abi.BX_GT, abi.BX_GE, abi.BX_EQ,
             abi.BX_NE]
     prog = []
