@@ -231,3 +231,38 @@ def test_gpu_growth_with_black_filter():
     assert res.n_groups == cres.n_groups > 64
     assert abi.group_row_tuples(rows, 2) == abi.group_row_tuples(crows, 2)
     eng.free(h)
+
+
+def test_cpu_fetch_edges():
+    bs, vals, g0, g1 = _make_many_groups(d0=4, d1=3)
+    filt, agg = _descs()
+    oracle.scan_filter_agg(bs, filt, agg)
+    lib = oracle._lib
+    buf = (abi.GroupRow * 8)()
+    n_out = C.c_uint32(99)
+    total = C.c_uint64()
+    # start beyond total -> 0 rows, total still reported
+    assert lib.obx_cpu_agg_fetch(10_000, 8, buf, C.byref(n_out),
+                                 C.byref(total)) == 0
+    assert n_out.value == 0 and 0 < total.value <= 64
+    # count 0 -> 0 rows
+    assert lib.obx_cpu_agg_fetch(0, 0, buf, C.byref(n_out),
+                                 C.byref(total)) == 0
+    assert n_out.value == 0
+
+
+@pytest.mark.gpu
+def test_gpu_fetch_edges():
+    from oceanbase_amd.engine import GpuEngine
+    eng = GpuEngine()
+    bs, vals, g0, g1 = _make_many_groups(d0=4, d1=3)
+    filt, agg = _descs()
+    h = eng.load(bs)
+    eng.scan_filter_agg(h, filt, agg)
+    buf = (abi.GroupRow * 8)()
+    n_out = C.c_uint32(99)
+    total = C.c_uint64()
+    assert eng._lib.obx_gpu_agg_fetch(eng._ctx, h, 10_000, 8, buf,
+                                      C.byref(n_out), C.byref(total)) == 0
+    assert n_out.value == 0 and 0 < total.value <= 64
+    eng.free(h)
