@@ -362,6 +362,10 @@ int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
 /* Timing helpers: milliseconds of device time of the last
  * filter/scan_filter_agg call (HIP events on the context stream). */
 double obx_gpu_last_kernel_ms(obx_gpu_ctx *ctx);
+/* device time of the last query's prep phase (plan upload +
+ * k_lower_leaves) — with last_kernel_ms, the per-operator stats the
+ * reference surfaces through sql_plan_monitor. */
+double obx_gpu_last_prep_ms(obx_gpu_ctx *ctx);
 /* 1 if the last scan_filter_agg ran the hipRTC plan-specialized kernel
  * (query codegen), 0 if the precompiled generic kernels ran. */
 int obx_gpu_last_jit(obx_gpu_ctx *ctx);
@@ -390,3 +394,16 @@ int obx_cpu_agg_fetch(uint32_t start, uint32_t count, obx_group_row *out,
 }
 #endif
 #endif /* OBX_H_ */
+
+/* TEST INFRASTRUCTURE: render the hipRTC source the JIT would generate
+ * for a plan over a synthetic one-block handle (no GPU needed — codegen
+ * and strategy selection are host code). agg == NULL renders the
+ * bitmap-filter kernel. Returns source bytes (0 = plan not
+ * JIT-eligible), negative on invalid arguments. */
+int64_t obx_jit_dump_src(const obx_filter_desc *filter,
+                         const obx_agg_desc *agg,
+                         const obx_col_schema *cols, uint16_t n_cols,
+                         const uint8_t *col_flags, const int64_t *col_min,
+                         const int64_t *col_max, const uint32_t *col_maxcnt,
+                         const uint32_t *col_maxw, uint32_t max_block_rows,
+                         int force_v1, char *out, int64_t cap);
