@@ -35,7 +35,7 @@ def test_engine_exports_every_declared_symbol():
 
 
 # product-only host helpers (exported by libobx.so, not the oracle)
-PRODUCT_ONLY = {"obx_cs_host_parse"}
+PRODUCT_ONLY = {"obx_cs_host_parse", "obx_jit_dump_src"}
 
 
 def test_oracle_exports_cpu_symbols():
