@@ -960,6 +960,7 @@ extern "C" int obx_gpu_decode(obx_gpu_ctx *ctx, int handle,
     return OBX_INVALID_ARGUMENT;
   HIP_TRY(hipSetDevice(ctx->device));
   obx_handle &h = ctx->handles[handle];
+  HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
   for (uint16_t i = 0; i < n_proj; i++) {
     uint16_t c = proj_cols[i];
     if (c >= h.n_cols) return OBX_INVALID_ARGUMENT;
@@ -970,7 +971,11 @@ extern "C" int obx_gpu_decode(obx_gpu_ctx *ctx, int handle,
                        (uint32_t)c, (uint32_t)h.cols[c].len,
                        h.d_decode_out[c], (uint8_t *)nullptr);
   }
+  HIP_TRY(hipEventRecord(ctx->ev_stop, ctx->stream));
   HIP_TRY(hipStreamSynchronize(ctx->stream));
+  float ms = 0;
+  HIP_TRY(hipEventElapsedTime(&ms, ctx->ev_start, ctx->ev_stop));
+  ctx->last_ms = ms; /* per-operator monitoring, as the scan verbs */
   return OBX_SUCCESS;
 }
 
