@@ -390,11 +390,6 @@ int obx_gpu_agg_fetch(obx_gpu_ctx *ctx, int handle, uint32_t start,
 int obx_cpu_agg_fetch(uint32_t start, uint32_t count, obx_group_row *out,
                       uint32_t *n_out, uint64_t *n_total);
 
-#ifdef __cplusplus
-}
-#endif
-#endif /* OBX_H_ */
-
 /* TEST INFRASTRUCTURE: render the hipRTC source the JIT would generate
  * for a plan over a synthetic one-block handle (no GPU needed — codegen
  * and strategy selection are host code). agg == NULL renders the
@@ -407,3 +402,8 @@ int64_t obx_jit_dump_src(const obx_filter_desc *filter,
                          const int64_t *col_max, const uint32_t *col_maxcnt,
                          const uint32_t *col_maxw, uint32_t max_block_rows,
                          int force_v1, char *out, int64_t cap);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* OBX_H_ */
