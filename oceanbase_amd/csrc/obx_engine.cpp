@@ -28,6 +28,9 @@
 #include "../../oracle/obx_format.h"
 
 /* kernels (obx_kernels.hip) */
+extern "C" __global__ void k_decode_lds(
+    const uint8_t *, const dev_block *, uint32_t, uint32_t, uint32_t,
+    uint8_t *, uint8_t *);
 extern "C" __global__ void k_scan_agg_direct(
     const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
     const blk_leaf *, const dev_plan_hdr, gslot *, unsigned long long *);
@@ -966,7 +969,8 @@ extern "C" int obx_gpu_decode(obx_gpu_ctx *ctx, int handle,
     if (c >= h.n_cols) return OBX_INVALID_ARGUMENT;
     if (!h.d_decode_out[c])
       HIP_TRY(hipMalloc(&h.d_decode_out[c], h.total_rows * h.cols[c].len));
-    hipLaunchKernelGGL(k_decode, dim3(grid_for(h.n_blocks)), dim3(OBX_WG_HOST), 0,
+    auto kfn = h.lds_ok ? k_decode_lds : k_decode;
+    hipLaunchKernelGGL(kfn, dim3(grid_for(h.n_blocks)), dim3(OBX_WG_HOST), 0,
                        ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
                        (uint32_t)c, (uint32_t)h.cols[c].len,
                        h.d_decode_out[c], (uint8_t *)nullptr);
