@@ -780,8 +780,13 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_decode(
       int64_t v = col_value2(bv, cur, cur.cols[col], r, isn);
       uint64_t uv = isn ? 0 : (uint64_t)v;
       uint8_t *dst = out + (row_start + r) * datum_len;
-      for (uint32_t i = 0; i < datum_len; i++)
-        dst[i] = (uint8_t)(uv >> (i * 8));
+      if (datum_len == 8)       /* aligned: hipMalloc base + 8-B stride */
+        *(uint64_t *)dst = uv;
+      else if (datum_len == 4)
+        *(uint32_t *)dst = (uint32_t)uv;
+      else
+        for (uint32_t i = 0; i < datum_len; i++)
+          dst[i] = (uint8_t)(uv >> (i * 8));
       if (out_null) out_null[row_start + r] = isn ? 1 : 0;
     }
   }
@@ -812,8 +817,13 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_decode_lds(
       int64_t v = col_value2(bv, cur, cur.cols[col], r, isn);
       uint64_t uv = isn ? 0 : (uint64_t)v;
       uint8_t *dst = out + (row_start + r) * datum_len;
-      for (uint32_t i = 0; i < datum_len; i++)
-        dst[i] = (uint8_t)(uv >> (i * 8));
+      if (datum_len == 8)       /* aligned: hipMalloc base + 8-B stride */
+        *(uint64_t *)dst = uv;
+      else if (datum_len == 4)
+        *(uint32_t *)dst = (uint32_t)uv;
+      else
+        for (uint32_t i = 0; i < datum_len; i++)
+          dst[i] = (uint8_t)(uv >> (i * 8));
       if (out_null) out_null[row_start + r] = isn ? 1 : 0;
     }
   }
