@@ -872,7 +872,7 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
   HIP_TRY(hipMemsetAsync(h.d_counters, 0, 16 * 8, ctx->stream));
   /* bitmap-only filters take the specialized wave-per-block kernel
      (direct global reads of just the leaf streams; obx_jit_v2.inc) */
-  jit_entry *fje = want_row_ids ? nullptr : jit_prepare_filter(h, ph, plv);
+  jit_entry *fje = jit_prepare_filter(h, ph, plv, want_row_ids);
   ctx->last_jit = fje ? 2 : 0;
   HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
   if (fje) {
@@ -886,8 +886,10 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
       int g = atoi(fg);
       if (g > 0 && g <= 65535) fgrid = (uint32_t)g;
     }
-    void *args[7] = {&h.d_buf, &h.d_blocks, &h.n_blocks, &h.d_pleaves,
-                     &h.d_bleaves, &h.d_bitmap, &h.d_counters};
+    void *rid = want_row_ids ? (void *)h.d_row_ids : nullptr;
+    void *bcn = want_row_ids ? (void *)h.d_blk_counts : nullptr;
+    void *args[9] = {&h.d_buf, &h.d_blocks, &h.n_blocks, &h.d_pleaves,
+                     &h.d_bleaves, &h.d_bitmap, &rid, &bcn, &h.d_counters};
     if (hipModuleLaunchKernel(fje->fn, fgrid, 1, 1,
                               OBX_WG_HOST, 1, 1, 0, ctx->stream, args,
                               nullptr) != hipSuccess)
@@ -1360,6 +1362,10 @@ extern "C" int64_t obx_jit_dump_src(
   if (!agg) { /* bitmap-filter path: dump the wave-per-block filter JIT */
     jit_strategy fst;
     if (!jit_build_fstrategy(h, ph, pl, fst)) return 0;
+    if (getenv("OBX_DUMP_RID") && fst.fstage) { /* test hook */
+      fst.frid = 1;
+      fst.fnch = (uint16_t)((h.max_block_rows + 63) / 64);
+    }
     std::string fsrc = jit_gen_source_filter(ph, fst);
     if (out && cap > (int64_t)fsrc.size()) {
       memcpy(out, fsrc.data(), fsrc.size());

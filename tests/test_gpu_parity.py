@@ -92,6 +92,33 @@ def test_filter_parity_config3(eng):
     assert survivors == res.rows_passed
 
 
+def test_row_ids_through_filter_jit(eng):
+    """Selection vectors (get_row_ids shape) through the staged filter
+    JIT's mask-capture path: ordered block-local row ids + per-block
+    counts, bit-exact vs the oracle, with the JIT engaged."""
+    li, h = _load_lineitem(eng, 3, 120000)
+    cutoff = oracle.date_days(1997, 6, 1)
+    fd = abi.make_filter([dict(col=0, op=abi.OP_LE, lo=cutoff)])
+    survivors = eng.filter(h, fd, want_row_ids=True)
+    assert eng._lib.obx_gpu_last_jit(eng._ctx) == 2  # RID JIT engaged
+    row_ids, _n = eng.fetch_row_ids(h)
+    blk_counts = eng.fetch_blk_counts(h, li.n_blocks)
+    off = 0
+    total = 0
+    for b in range(li.n_blocks):
+        blk = li.block(b)
+        bits, pc = oracle.filter_block(li.schema, li.n_cols, blk, fd)
+        total += pc
+        assert blk_counts[b] == pc, b
+        rows_b = int(np.frombuffer(blk[16:20], dtype=np.uint32)[0])
+        want_ids = [r for r in range(rows_b)
+                    if (bits[r >> 3] >> (r & 7)) & 1]
+        got_ids = row_ids[off:off + pc].tolist()
+        assert got_ids == want_ids, b
+        off += rows_b
+    assert survivors == total
+
+
 @pytest.mark.parametrize("op,lo,hi", [
     (abi.OP_EQ, 24, 0), (abi.OP_NE, 24, 0), (abi.OP_LT, 1, 0),
     (abi.OP_LE, 50, 0), (abi.OP_GT, 49, 0), (abi.OP_GE, 51, 0),
