@@ -146,3 +146,44 @@ def test_cs_codec_fuzz_vs_reference():
             assert m == n1 and dec.raw == raw, (enc_type, wb, cnt)
         n += 1
     assert n == 150
+
+
+@pytest.mark.gpu
+def test_black_program_fuzz_gpu_vs_oracle():
+    """The device black VM across 30 random programs (multi-block set,
+    NULLs included): survivor counts + bitmaps vs the oracle."""
+    from oceanbase_amd.engine import GpuEngine
+    rng = random.Random(777)
+    nrng = np.random.default_rng(23)
+    rows_pb, nblocks = 2000, 5
+    n_cols, n_consts = 3, 3
+    schema = oracle.make_schema([(abi.T_INT, 0, 0, 8)] * n_cols)
+    blocks = []
+    for _ in range(nblocks):
+        data = [nrng.integers(-1000, 1000, rows_pb).astype(np.int64)
+                for _ in range(n_cols)]
+        nulls = np.zeros((rows_pb + 7) // 8, dtype=np.uint8)
+        for r in range(0, rows_pb, 13):
+            nulls[r >> 3] |= 1 << (r & 7)
+        blocks.append(oracle.encode_block(
+            schema, [d.view(np.uint8) for d in data],
+            [abi.ENC_RAW] * n_cols, [nulls, None, None]))
+    import test_group_capacity as tgc
+    bs = tgc._blockset(schema, blocks, rows_pb * nblocks)
+    eng = GpuEngine()
+    h = eng.load(bs)
+    checked = 0
+    for i in range(30):
+        prog = _rand_prog(rng, n_cols, n_consts)
+        if len(prog) > 24:
+            continue
+        consts = [rng.randint(-500, 500) for _ in range(n_consts)]
+        filt = abi.make_filter([dict(op=abi.OP_BLACK,
+                                     bcols=list(range(n_cols)),
+                                     bconst=consts, bprog=prog)])
+        res_cpu = oracle.scan_filter_agg(bs, filt, None)
+        survivors = eng.filter(h, filt)
+        assert survivors == res_cpu.rows_passed, (i, prog)
+        checked += 1
+    assert checked >= 25
+    eng.free(h)
