@@ -645,7 +645,7 @@ static int build_plan(const obx_handle &h, const obx_filter_desc *filter,
           sp++;
         } else if (op2 == 0x54 || op2 == 0x72) {
           if (sp < 1) return OBX_INVALID_ARGUMENT;
-        } else if ((op2 >= 0x50 && op2 <= 0x53) ||
+        } else if ((op2 >= 0x50 && op2 <= 0x53) || op2 == 0x55 ||
                    (op2 >= 0x60 && op2 <= 0x65) || op2 == 0x70 ||
                    op2 == 0x71) {
           if (sp < 2) return OBX_INVALID_ARGUMENT;
