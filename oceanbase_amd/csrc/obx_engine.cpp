@@ -31,6 +31,9 @@
 extern "C" __global__ void k_decode_lds(
     const uint8_t *, const dev_block *, uint32_t, uint32_t, uint32_t,
     uint8_t *, uint8_t *);
+extern "C" __global__ void k_decode_multi(
+    const uint8_t *, const dev_block *, uint32_t, const uint16_t *,
+    const uint8_t *, uint32_t, uint8_t *const *, uint8_t *const *);
 extern "C" __global__ void k_scan_agg_direct(
     const uint8_t *, const dev_block *, uint32_t, const dev_leaf *,
     const blk_leaf *, const dev_plan_hdr, gslot *, unsigned long long *);
@@ -113,6 +116,7 @@ struct obx_handle {
   uint8_t *d_decode_out[OBX_DEV_MAX_COLS] = {};
   uint64_t last_survivors = 0;
   gslot *d_gtable_big = nullptr; /* high-cardinality direct kernel table */
+  void *d_proj_scratch = nullptr; /* k_decode_multi proj/len/out arrays */
   std::vector<obx_group_row> last_rows; /* full sorted group rows of the
                                            last scan (obx_gpu_agg_fetch) */
   bool lds_ok = false;   /* all blocks 16-B aligned and <= LDS stage size */
@@ -187,7 +191,7 @@ extern "C" int obx_gpu_close(obx_gpu_ctx *ctx) {
   for (auto &h : ctx->handles) {
     if (!h.in_use) continue;
     (void)hipFree(h.d_buf); (void)hipFree(h.d_blocks); (void)hipFree(h.d_bleaves);
-    (void)hipFree(h.d_pleaves); (void)hipFree(h.d_gtable); (void)hipFree(h.d_counters); (void)hipFree(h.d_gtable_big);
+    (void)hipFree(h.d_pleaves); (void)hipFree(h.d_gtable); (void)hipFree(h.d_counters); (void)hipFree(h.d_gtable_big); (void)hipFree(h.d_proj_scratch);
     (void)hipFree(h.d_bitmap); (void)hipFree(h.d_row_ids); (void)hipFree(h.d_blk_counts);
     (void)hipFree(h.d_row_slot); (void)hipFree(h.d_minmax);
     for (auto *p : h.d_decode_out) (void)hipFree(p);
@@ -588,7 +592,7 @@ extern "C" int obx_gpu_free_blocks(obx_gpu_ctx *ctx, int handle) {
   obx_handle &h = ctx->handles[handle];
   if (!h.in_use) return OBX_INVALID_ARGUMENT;
   (void)hipFree(h.d_buf); (void)hipFree(h.d_blocks); (void)hipFree(h.d_bleaves);
-  (void)hipFree(h.d_pleaves); (void)hipFree(h.d_gtable); (void)hipFree(h.d_counters); (void)hipFree(h.d_gtable_big);
+  (void)hipFree(h.d_pleaves); (void)hipFree(h.d_gtable); (void)hipFree(h.d_counters); (void)hipFree(h.d_gtable_big); (void)hipFree(h.d_proj_scratch);
   (void)hipFree(h.d_bitmap); (void)hipFree(h.d_row_ids); (void)hipFree(h.d_blk_counts);
   (void)hipFree(h.d_row_slot); (void)hipFree(h.d_minmax);
   for (auto *&p : h.d_decode_out) { (void)hipFree(p); p = nullptr; }
@@ -966,6 +970,40 @@ extern "C" int obx_gpu_decode(obx_gpu_ctx *ctx, int handle,
   HIP_TRY(hipSetDevice(ctx->device));
   obx_handle &h = ctx->handles[handle];
   HIP_TRY(hipEventRecord(ctx->ev_start, ctx->stream));
+  if (h.lds_ok && n_proj > 1 && n_proj <= OBX_DEV_MAX_COLS) {
+    /* one block stage serves every projected column */
+    uint16_t pc[OBX_DEV_MAX_COLS];
+    uint8_t ln[OBX_DEV_MAX_COLS];
+    uint8_t *po[OBX_DEV_MAX_COLS];
+    for (uint16_t i = 0; i < n_proj; i++) {
+      uint16_t c = proj_cols[i];
+      if (c >= h.n_cols) return OBX_INVALID_ARGUMENT;
+      if (!h.d_decode_out[c])
+        HIP_TRY(hipMalloc(&h.d_decode_out[c], h.total_rows * h.cols[c].len));
+      pc[i] = c;
+      ln[i] = (uint8_t)h.cols[c].len;
+      po[i] = h.d_decode_out[c];
+    }
+    if (!h.d_proj_scratch)
+      HIP_TRY(hipMalloc(&h.d_proj_scratch,
+                        OBX_DEV_MAX_COLS * (8 + 2 + 1) + 16));
+    /* pointers first (8-aligned base), then u16 cols, then u8 lens */
+    uint8_t *base = (uint8_t *)h.d_proj_scratch;
+    uint8_t *outp = base;
+    uint8_t *pcp = base + OBX_DEV_MAX_COLS * 8;
+    uint8_t *lnp = pcp + OBX_DEV_MAX_COLS * 2;
+    /* synchronous copies: the sources are stack arrays */
+    HIP_TRY(hipMemcpy(outp, po, sizeof(void *) * n_proj,
+                      hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(pcp, pc, sizeof(uint16_t) * n_proj,
+                      hipMemcpyHostToDevice));
+    HIP_TRY(hipMemcpy(lnp, ln, n_proj, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_decode_multi, dim3(grid_for(h.n_blocks)),
+                       dim3(OBX_WG_HOST), 0, ctx->stream, h.d_buf,
+                       h.d_blocks, h.n_blocks, (const uint16_t *)pcp,
+                       lnp, (uint32_t)n_proj, (uint8_t *const *)outp,
+                       (uint8_t *const *)nullptr);
+  } else {
   for (uint16_t i = 0; i < n_proj; i++) {
     uint16_t c = proj_cols[i];
     if (c >= h.n_cols) return OBX_INVALID_ARGUMENT;
@@ -976,6 +1014,7 @@ extern "C" int obx_gpu_decode(obx_gpu_ctx *ctx, int handle,
                        ctx->stream, h.d_buf, h.d_blocks, h.n_blocks,
                        (uint32_t)c, (uint32_t)h.cols[c].len,
                        h.d_decode_out[c], (uint8_t *)nullptr);
+  }
   }
   HIP_TRY(hipEventRecord(ctx->ev_stop, ctx->stream));
   HIP_TRY(hipStreamSynchronize(ctx->stream));

@@ -829,6 +829,49 @@ extern "C" __global__ __launch_bounds__(WG, 2) void k_decode_lds(
   }
 }
 
+extern "C" __global__ __launch_bounds__(WG, 2) void k_decode_multi(
+    const uint8_t *__restrict__ buf, const dev_block *__restrict__ blocks,
+    uint32_t n_blocks, const uint16_t *__restrict__ proj,
+    const uint8_t *__restrict__ lens, uint32_t n_proj,
+    uint8_t *const *__restrict__ outs, uint8_t *const *__restrict__ onulls) {
+  /* one stage per block, ALL projected columns decoded from LDS — the
+     per-column variant re-reads every block once per column */
+  __shared__ uint8_t lds_blk[OBX_LDS_STAGE_BYTES + 32];
+  const uint32_t tid = threadIdx.x;
+  for (uint32_t b = blockIdx.x; b < n_blocks; b += gridDim.x) {
+    const dev_block &cur = blocks[b];
+    __syncthreads(); /* drain previous block's LDS reads */
+    stage_issue(buf, cur, lds_blk);
+    stage_wait();
+    blk_view bv;
+    bv.base = lds_blk;
+    bv.bit_bias = cur.block_byte * 8;
+    bv.rbase_bit = 0;
+    const uint32_t rows = cur.row_count;
+    const uint64_t row_start = dev_block_row_start(&cur);
+    for (uint32_t p = 0; p < n_proj; p++) {
+      const uint32_t col = proj[p];
+      const uint32_t datum_len = lens[p];
+      uint8_t *out = outs[p];
+      uint8_t *out_null = onulls ? onulls[p] : nullptr;
+      for (uint32_t r = tid; r < rows; r += WG) {
+        bool isn;
+        int64_t v = col_value2(bv, cur, cur.cols[col], r, isn);
+        uint64_t uv = isn ? 0 : (uint64_t)v;
+        uint8_t *dst = out + (row_start + r) * datum_len;
+        if (datum_len == 8)
+          *(uint64_t *)dst = uv;
+        else if (datum_len == 4)
+          *(uint32_t *)dst = (uint32_t)uv;
+        else
+          for (uint32_t i = 0; i < datum_len; i++)
+            dst[i] = (uint8_t)(uv >> (i * 8));
+        if (out_null) out_null[row_start + r] = isn ? 1 : 0;
+      }
+    }
+  }
+}
+
 /* ---------------- device-side filter lowering ---------------------------
  * One thread per (block, leaf): translate the leaf into the block's packed
  * domain (the reference evaluates dict-domain filters once per dict entry,
