@@ -266,3 +266,41 @@ def test_gpu_fetch_edges():
                                       C.byref(n_out), C.byref(total)) == 0
     assert n_out.value == 0 and 0 < total.value <= 64
     eng.free(h)
+
+
+@pytest.mark.gpu
+def test_jws_with_null_factor_axis():
+    """Weighted-joint-histogram fold when the factor axis column HAS
+    NULLs: the null bucket must contribute zero to the product sums but
+    still count into the plain SUM (vt[null] = 0 at flush)."""
+    from oceanbase_amd.engine import GpuEngine
+    rng = np.random.default_rng(31)
+    rows_pb, nblocks = 2000, 6
+    schema = oracle.make_schema([(abi.T_INT, 0, 19, 8),   # price (CTX)
+                                 (abi.T_INT, 0, 19, 8),   # disc (dict)
+                                 (abi.T_CHAR, 0, 0, 1)])  # group
+    blocks = []
+    # identical dict payloads across blocks (JWS needs dict stability):
+    # draw disc from the same value set in every block
+    for _ in range(nblocks):
+        price = rng.integers(1000, 100000, rows_pb).astype(np.int64)
+        disc = rng.integers(0, 10, rows_pb).astype(np.int64)
+        grp = (65 + rng.integers(0, 3, rows_pb)).astype(np.uint8)
+        nulls = np.zeros((rows_pb + 7) // 8, dtype=np.uint8)
+        for r in range(0, rows_pb, 9):
+            nulls[r >> 3] |= 1 << (r & 7)   # disc NULL every 9th row
+        blocks.append(oracle.encode_block(
+            schema, [price.view(np.uint8), disc.view(np.uint8), grp],
+            [abi.ENC_RAW, abi.ENC_DICT, abi.ENC_DICT],
+            [None, nulls, None]))
+    bs = _blockset(schema, blocks, rows_pb * nblocks)
+    agg = abi.make_agg([2], [dict(kind=abi.AGG_COUNT),
+                             dict(kind=abi.AGG_SUM, col_a=0),
+                             dict(kind=abi.AGG_SUM_PROD2, col_a=0,
+                                  col_b=1)])
+    eng = GpuEngine()
+    h = eng.load(bs)
+    res_gpu = eng.scan_filter_agg(h, None, agg)
+    res_cpu = oracle.scan_filter_agg(bs, None, agg)
+    assert abi.result_rows(res_gpu, 3) == abi.result_rows(res_cpu, 3)
+    eng.free(h)
