@@ -346,3 +346,51 @@ def test_v2_jit_dict_null_aggregates():
     res_cpu = oracle.scan_filter_agg(bs, None, agg)
     assert abi.result_rows(res_gpu, 6) == abi.result_rows(res_cpu, 6)
     eng.free(h)
+
+
+@pytest.mark.gpu
+def test_v2_jit_null_weight_and_prod3():
+    """NULLs in the product WEIGHT column (CTX, ext bits) and in both
+    factor dict columns, PROD2+PROD3 fused, through the v2 JIT."""
+    from oceanbase_amd.engine import GpuEngine
+    rng = np.random.default_rng(43)
+    rows_pb, nblocks = 1500, 6
+    schema = oracle.make_schema([(abi.T_INT, 0, 19, 8),   # price + nulls
+                                 (abi.T_INT, 0, 19, 8),   # disc dict+nulls
+                                 (abi.T_INT, 0, 19, 8),   # tax dict+nulls
+                                 (abi.T_CHAR, 0, 0, 1)])
+    blocks = []
+    for _ in range(nblocks):
+        price = rng.integers(1000, 100000, rows_pb).astype(np.int64)
+        disc = rng.integers(0, 10, rows_pb).astype(np.int64)
+        tax = rng.integers(0, 8, rows_pb).astype(np.int64)
+        grp = (65 + rng.integers(0, 4, rows_pb)).astype(np.uint8)
+        nb_ = (rows_pb + 7) // 8
+        n0 = np.zeros(nb_, dtype=np.uint8)
+        n1 = np.zeros(nb_, dtype=np.uint8)
+        n2 = np.zeros(nb_, dtype=np.uint8)
+        for r in range(0, rows_pb, 11):
+            n0[r >> 3] |= 1 << (r & 7)
+        for r in range(3, rows_pb, 13):
+            n1[r >> 3] |= 1 << (r & 7)
+        for r in range(5, rows_pb, 17):
+            n2[r >> 3] |= 1 << (r & 7)
+        blocks.append(oracle.encode_block(
+            schema, [price.view(np.uint8), disc.view(np.uint8),
+                     tax.view(np.uint8), grp],
+            [abi.ENC_RAW, abi.ENC_DICT, abi.ENC_DICT, abi.ENC_DICT],
+            [n0, n1, n2, None]))
+    bs = _blockset(schema, blocks, rows_pb * nblocks)
+    agg = abi.make_agg([3], [
+        dict(kind=abi.AGG_COUNT),
+        dict(kind=abi.AGG_SUM, col_a=0),
+        dict(kind=abi.AGG_SUM_PROD2, col_a=0, col_b=1),
+        dict(kind=abi.AGG_SUM_PROD3, col_a=0, col_b=1, col_c=2)])
+    eng = GpuEngine()
+    h = eng.load(bs)
+    res_gpu = eng.scan_filter_agg(h, None, agg)
+    jit_kind = eng._lib.obx_gpu_last_jit(eng._ctx)
+    res_cpu = oracle.scan_filter_agg(bs, None, agg)
+    assert abi.result_rows(res_gpu, 4) == abi.result_rows(res_cpu, 4), \
+        f"jit={jit_kind}"
+    eng.free(h)
