@@ -394,3 +394,38 @@ def test_v2_jit_null_weight_and_prod3():
     assert abi.result_rows(res_gpu, 4) == abi.result_rows(res_cpu, 4), \
         f"jit={jit_kind}"
     eng.free(h)
+
+
+@pytest.mark.gpu
+def test_filter_jit_null_ops_and_rid_edges():
+    """NU/NN/IN leaves on a dict column with NULL refs through the
+    bitmap JIT, and selection-vector edges (zero survivors, all
+    survivors) through the RID capture path."""
+    from oceanbase_amd.engine import GpuEngine
+    rng = np.random.default_rng(47)
+    rows_pb, nblocks = 1500, 5
+    schema = oracle.make_schema([(abi.T_INT, 0, 19, 8)])
+    blocks = []
+    for _ in range(nblocks):
+        v = rng.integers(0, 12, rows_pb).astype(np.int64)
+        nulls = np.zeros((rows_pb + 7) // 8, dtype=np.uint8)
+        for r in range(0, rows_pb, 7):
+            nulls[r >> 3] |= 1 << (r & 7)
+        blocks.append(oracle.encode_block(
+            schema, [v.view(np.uint8)], [abi.ENC_DICT], [nulls]))
+    bs = _blockset(schema, blocks, rows_pb * nblocks)
+    eng = GpuEngine()
+    h = eng.load(bs)
+    for fd in (abi.make_filter([dict(col=0, op=abi.OP_NU)]),
+               abi.make_filter([dict(col=0, op=abi.OP_NN)]),
+               abi.make_filter([dict(col=0, op=abi.OP_IN,
+                                     in_list=[2, 5, 11])]),
+               abi.make_filter([dict(col=0, op=abi.OP_LT, lo=-5)]),   # none
+               abi.make_filter([dict(col=0, op=abi.OP_LE, lo=99)])):  # all nn
+        res_cpu = oracle.scan_filter_agg(bs, fd, None)
+        got = eng.filter(h, fd, want_row_ids=True)
+        assert got == res_cpu.rows_passed
+        ids, _ = eng.fetch_row_ids(h)
+        counts = eng.fetch_blk_counts(h, nblocks)
+        assert int(counts.sum()) == res_cpu.rows_passed
+    eng.free(h)
