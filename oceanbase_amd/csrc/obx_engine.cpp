@@ -872,6 +872,11 @@ extern "C" int obx_gpu_filter(obx_gpu_ctx *ctx, int handle,
       HIP_TRY(hipMalloc(&h.d_row_ids, h.total_rows * sizeof(int32_t)));
     if (!h.d_blk_counts)
       HIP_TRY(hipMalloc(&h.d_blk_counts, h.n_blocks * sizeof(uint32_t)));
+    /* skipped blocks (NONE verdict) write nothing: stale counts from a
+       previous filter would survive (caught by the zero-survivor edge
+       test) */
+    HIP_TRY(hipMemsetAsync(h.d_blk_counts, 0,
+                           h.n_blocks * sizeof(uint32_t), ctx->stream));
   }
   HIP_TRY(hipMemsetAsync(h.d_counters, 0, 16 * 8, ctx->stream));
   /* bitmap-only filters take the specialized wave-per-block kernel
