@@ -181,19 +181,57 @@ static void *worker(void *arg) {
   if (filter) for (int i = 0; i < filter->n_leaves; i++) {
     const obx_filter_leaf *lf0 = &filter->leaves[i];
     if (lf0->op == OBX_OP_BLACK) {
-      for (int j = 0; j < lf0->n_bcols; j++) NEED(lf0->bcols[j]);
+      for (int j = 0; j < lf0->n_bcols; j++) {
+        if (lf0->bcols[j] >= bs->n_cols) {
+          w->rc = OBX_INVALID_ARGUMENT;
+          free(decbuf);
+          return NULL;
+        }
+        NEED(lf0->bcols[j]);
+      }
     } else {
+      if (lf0->col >= bs->n_cols) {
+        w->rc = OBX_INVALID_ARGUMENT;
+        free(decbuf);
+        return NULL;
+      }
       NEED(lf0->col);
     }
   }
   if (agg) {
-    for (int i = 0; i < agg->n_group_cols; i++) NEED(agg->group_cols[i]);
+    for (int i = 0; i < agg->n_group_cols; i++) {
+      if (agg->group_cols[i] >= bs->n_cols) {
+        w->rc = OBX_INVALID_ARGUMENT;
+        free(decbuf);
+        return NULL;
+      }
+      NEED(agg->group_cols[i]);
+    }
     for (int i = 0; i < agg->n_aggs; i++) {
       const obx_agg_expr *e = &agg->aggs[i];
+      if (e->col_a != UINT16_MAX && e->col_a >= bs->n_cols) {
+        w->rc = OBX_INVALID_ARGUMENT;
+        free(decbuf);
+        return NULL;
+      }
       if (e->col_a != UINT16_MAX) NEED(e->col_a);
       if (e->kind == OBX_AGG_SUM_PROD2 || e->kind == OBX_AGG_SUM_PROD3 ||
-          e->kind == OBX_AGG_SUM_MUL) NEED(e->col_b);
-      if (e->kind == OBX_AGG_SUM_PROD3) NEED(e->col_c);
+          e->kind == OBX_AGG_SUM_MUL) {
+        if (e->col_b >= bs->n_cols) {
+          w->rc = OBX_INVALID_ARGUMENT;
+          free(decbuf);
+          return NULL;
+        }
+        NEED(e->col_b);
+      }
+      if (e->kind == OBX_AGG_SUM_PROD3) {
+        if (e->col_c >= bs->n_cols) {
+          w->rc = OBX_INVALID_ARGUMENT;
+          free(decbuf);
+          return NULL;
+        }
+        NEED(e->col_c);
+      }
     }
   }
 #undef NEED
