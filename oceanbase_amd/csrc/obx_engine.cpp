@@ -729,8 +729,11 @@ static int build_plan(const obx_handle &h, const obx_filter_desc *filter,
     return (uint8_t)slot_of_col[c];
   };
   if (agg) {
+    if (agg->n_group_cols > 2 || agg->n_aggs > OBX_DEV_MAX_AGGS)
+      return OBX_INVALID_ARGUMENT;
     ph.n_group_cols = agg->n_group_cols;
     for (int g = 0; g < agg->n_group_cols; g++) {
+      if (agg->group_cols[g] >= h.n_cols) return OBX_INVALID_ARGUMENT;
       ph.group_idx[g] = need(agg->group_cols[g]);
       ph.group_len[g] = h.cols[agg->group_cols[g]].len;
     }
@@ -740,9 +743,12 @@ static int build_plan(const obx_handle &h, const obx_filter_desc *filter,
       dev_agg *da = &ph.aggs[a];
       da->kind = e->kind;
       da->ia = da->ib = da->ic = 0xFF;
+      if (e->col_a != UINT16_MAX && e->col_a >= h.n_cols)
+        return OBX_INVALID_ARGUMENT;
       if (e->col_a != UINT16_MAX) da->ia = need(e->col_a);
       if (e->kind == OBX_AGG_SUM_PROD2 || e->kind == OBX_AGG_SUM_PROD3 ||
           e->kind == OBX_AGG_SUM_MUL) {
+        if (e->col_b >= h.n_cols) return OBX_INVALID_ARGUMENT;
         da->ib = need(e->col_b);
         static const int64_t P10[19] = {1ll,10ll,100ll,1000ll,10000ll,
           100000ll,1000000ll,10000000ll,100000000ll,1000000000ll,
@@ -751,6 +757,7 @@ static int build_plan(const obx_handle &h, const obx_filter_desc *filter,
           100000000000000000ll,1000000000000000000ll};
         da->one_b = P10[h.cols[e->col_b].scale];
         if (e->kind == OBX_AGG_SUM_PROD3) {
+          if (e->col_c >= h.n_cols) return OBX_INVALID_ARGUMENT;
           da->ic = need(e->col_c);
           da->one_c = P10[h.cols[e->col_c].scale];
         }
