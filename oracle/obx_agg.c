@@ -241,6 +241,24 @@ static void *worker(void *arg) {
     if (b >= bs->n_blocks) break;
     const uint8_t *block = bs->data + bs->block_offsets[b];
     const obx_micro_header *h = (const obx_micro_header *)block;
+    /* read == verify (negligible: HW CRC-32C at ~20 GB/s/core) */
+    int64_t blen = (int64_t)(bs->block_offsets[b + 1] - bs->block_offsets[b]);
+    if (h->magic != OBX_MICRO_BLOCK_MAGIC ||
+        (int64_t)h->header_size +
+                (int64_t)bs->n_cols * (int64_t)sizeof(obx_col_header) >
+            blen ||
+        h->data_zlength > blen || h->data_zlength < OBX_MICRO_HEADER_SIZE) {
+      w->rc = OBX_INVALID_ARGUMENT;
+      free(decbuf);
+      return NULL;
+    }
+    if ((int64_t)obx_crc32c(block + OBX_MICRO_HEADER_SIZE,
+                            (int64_t)h->data_zlength -
+                                OBX_MICRO_HEADER_SIZE) != h->data_checksum) {
+      w->rc = OBX_PHYSIC_CHECKSUM_ERROR;
+      free(decbuf);
+      return NULL;
+    }
     const uint8_t *meta_region = block + h->header_size +
         (int64_t)bs->n_cols * sizeof(obx_col_header);
     const obx_col_header *chp =

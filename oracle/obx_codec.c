@@ -1872,9 +1872,19 @@ int obx_cpu_filter_block(const obx_col_schema *cols, uint16_t n_cols,
                          const obx_filter_desc *filter,
                          uint8_t *result_bits, uint32_t *row_count,
                          uint32_t *popcnt) {
-  (void)block_len;
   const obx_micro_header *h = (const obx_micro_header *)block;
   if (h->magic != OBX_MICRO_BLOCK_MAGIC) return OBX_INVALID_ARGUMENT;
+  /* read == verify, as decode (the reference checks the block checksum
+     before any use, ObMicroBlockReader init / block cache) */
+  if ((int64_t)h->header_size +
+          (int64_t)n_cols * (int64_t)sizeof(obx_col_header) > block_len)
+    return OBX_INVALID_ARGUMENT;
+  if (h->data_zlength > block_len || h->data_zlength < OBX_MICRO_HEADER_SIZE)
+    return OBX_INVALID_ARGUMENT;
+  if ((int64_t)obx_crc32c(block + OBX_MICRO_HEADER_SIZE,
+                          (int64_t)h->data_zlength - OBX_MICRO_HEADER_SIZE) !=
+      h->data_checksum)
+    return OBX_PHYSIC_CHECKSUM_ERROR;
   const uint8_t *meta_region = block + h->header_size +
                                (int64_t)n_cols * sizeof(obx_col_header);
   const obx_col_header *chp =
