@@ -1056,6 +1056,7 @@ extern "C" int obx_gpu_scan_filter_agg(obx_gpu_ctx *ctx, int handle,
     return OBX_INVALID_ARGUMENT;
   HIP_TRY(hipSetDevice(ctx->device));
   obx_handle &h = ctx->handles[handle];
+  h.last_rows.clear(); /* no stale pagination after a failed scan */
   dev_plan_hdr ph;
   dev_leaf plv[OBX_DEV_MAX_LEAVES];
   int rc = prep_query(ctx, h, filter, agg, ph, plv);

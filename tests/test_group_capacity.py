@@ -472,6 +472,8 @@ def test_gpu_gtable_overflow_clean_error():
     with pytest.raises(RuntimeError) as ei:
         eng.scan_filter_agg(h, None, agg)
     assert "-4009" in str(ei.value)  # OBX_BUF_NOT_ENOUGH
+    # pagination surface is empty after the failed scan (no stale rows)
+    assert eng.agg_fetch_all(h) == []
     # engine still healthy: a small plan on the same handle succeeds
     small = abi.make_agg([], [dict(kind=abi.AGG_COUNT)])
     res = eng.scan_filter_agg(h, None, small)
