@@ -458,6 +458,12 @@ static int parse_block(const obx_col_schema *cols, uint16_t n_cols,
 extern "C" int obx_gpu_load_blocks(obx_gpu_ctx *ctx, const obx_blockset *bs) {
   if (!ctx || !bs) return OBX_INVALID_ARGUMENT;
   if (bs->n_cols > OBX_DEV_MAX_COLS) return OBX_NOT_SUPPORTED;
+  for (uint16_t c = 0; c < bs->n_cols; c++) {
+    /* decimal scale bounds the P10 tables (reference: decimal-int
+       precision <= 18 digits on this path) */
+    if (bs->cols[c].scale < 0 || bs->cols[c].scale > 18)
+      return OBX_NOT_SUPPORTED;
+  }
   HIP_TRY(hipSetDevice(ctx->device));
   obx_handle h;
   h.n_blocks = bs->n_blocks;

@@ -419,6 +419,10 @@ int obx_cpu_scan_filter_agg(const obx_blockset *bs,
       if (filter->leaves[i].op == OBX_OP_BLACK &&
           !obx__bprog_valid(&filter->leaves[i]))
         return OBX_INVALID_ARGUMENT;
+  for (uint16_t c = 0; c < bs->n_cols; c++) {
+    if (bs->cols[c].scale < 0 || bs->cols[c].scale > 18)
+      return OBX_NOT_SUPPORTED; /* POW10 bound (decimal-int <= 18 digits) */
+  }
   if (nthreads <= 0) nthreads = (int)sysconf(_SC_NPROCESSORS_ONLN);
   if (nthreads < 1) nthreads = 1;
   if (nthreads > 256) nthreads = 256;

@@ -171,3 +171,19 @@ def test_out_of_range_columns_rejected_gpu():
         eng.scan_filter_agg(h, None, abi.make_agg(
             [9], [dict(kind=abi.AGG_COUNT)]))
     eng.free(h)
+
+
+def test_bad_scale_rejected():
+    """Out-of-range decimal scale in the ABI schema indexes the P10
+    tables: both sides must reject it."""
+    import numpy as np
+    from test_group_capacity import _blockset
+    rng = np.random.default_rng(79)
+    schema = oracle.make_schema([(abi.T_DECIMAL_INT, 2, 15, 8)])
+    v = rng.integers(0, 100, 500).astype(np.int64)
+    blk = oracle.encode_block(schema, [v.view(np.uint8)], [abi.ENC_RAW])
+    bs = _blockset(schema, [blk], 500)
+    schema[0].scale = 77  # hostile schema byte
+    agg = abi.make_agg([], [dict(kind=abi.AGG_SUM, col_a=0)])
+    with pytest.raises(RuntimeError):
+        oracle.scan_filter_agg(bs, None, agg)
