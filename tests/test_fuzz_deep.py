@@ -256,3 +256,64 @@ def test_cs_corruption_fuzz_never_crashes():
     # no crash across all 200 corruptions is the property; both outcomes
     # are legal depending on where the flip landed
     assert outcomes["ok"] + outcomes["err"] == 200, outcomes
+
+
+def test_auto_encoding_roundtrip_fuzz():
+    """120 random column shapes through OBX_ENC_AUTO: whatever encoding
+    the writer picks, decode + filter popcount must round-trip, and the
+    independent python model must agree with the C decoder."""
+    import pymodel
+    rng = np.random.default_rng(83)
+    checked = 0
+    for i in range(120):
+        rows = int(rng.integers(1, 3000))
+        kind = int(rng.integers(0, 6))
+        if kind == 0:    # few distinct -> dict/const
+            vals = rng.choice(rng.integers(-50, 50, 5), rows)
+        elif kind == 1:  # runs -> rle
+            vals = np.repeat(rng.integers(0, 9, max(rows // 9, 1)),
+                             9)[:rows].copy()
+            if len(vals) < rows:
+                vals = np.pad(vals, (0, rows - len(vals)))
+        elif kind == 2:  # narrow range -> intdiff/bitpack
+            base = int(rng.integers(-10**9, 10**9))
+            vals = base + rng.integers(0, 100, rows)
+        elif kind == 3:  # constant
+            vals = np.full(rows, int(rng.integers(-10**6, 10**6)))
+        elif kind == 4:  # wide random -> raw
+            vals = rng.integers(-10**12, 10**12, rows)
+        else:            # constant + few exceptions -> const/exc
+            vals = np.full(rows, 7)
+            for r in range(0, rows, 97):
+                vals[r] = int(rng.integers(-100, 100))
+        vals = vals.astype(np.int64)
+        nulls = None
+        if rng.random() < 0.4:
+            nulls = np.zeros((rows + 7) // 8, dtype=np.uint8)
+            for r in range(0, rows, int(rng.integers(5, 40))):
+                nulls[r >> 3] |= 1 << (r & 7)
+        schema = oracle.make_schema([(abi.T_INT, 0, 19, 8)])
+        blk = oracle.encode_block(schema, [vals.view(np.uint8)],
+                                  [abi.ENC_AUTO],
+                                  [nulls] if nulls is not None else None)
+        # python model agrees with the C encoder's bytes
+        pb = pymodel.Block(blk, [(abi.T_INT, 0, 19, 8)])
+        got = pb.decode_col(0)
+        null_set = set()
+        if nulls is not None:
+            null_set = {r for r in range(rows)
+                        if nulls[r >> 3] & (1 << (r & 7))}
+        for r in range(rows):
+            if r in null_set:
+                continue
+            gv = got[r][0] if isinstance(got[r], tuple) else got[r]
+            assert gv == int(vals[r]), (i, r, kind)
+        # filter popcount cross-check
+        lo = int(np.median(vals))
+        fd = abi.make_filter([dict(col=0, op=abi.OP_LE, lo=lo)])
+        _bits, pc = oracle.filter_block(schema, 1, blk, fd)
+        expect = sum(1 for r in range(rows)
+                     if r not in null_set and int(vals[r]) <= lo)
+        assert pc == expect, (i, kind)
+        checked += 1
+    assert checked == 120
